@@ -1,0 +1,103 @@
+// Edge-feature gather for SetConv (capability of reference
+// model/flot/gconv.py:60-68).  Forward emits the (B, C+3, K, N) edge-conv
+// input directly:
+//   out[b, c,   j, n] = feats[b, idx[b,n,j], c] - feats[b, n, c]   (c < C)
+//   out[b, C+c, j, n] = xyz  [b, idx[b,n,j], c] - xyz  [b, n, c]   (c < 3)
+// Thread layout: one thread per (b, j, n) with n fastest, so every store
+// for a fixed (c, j) is wave-coalesced over n; the neighbour-row reads are
+// inherently scattered (kNN gather) and ride L2.
+//
+// Backward (w.r.t. feats only; graph positions carry no gradient in
+// PV-RAFT):
+//   g[b, n, c] = -sum_j gout[b, c, j, n] + sum_{(n',j): idx[n',j]==n} gout[b, c, j, n']
+// Kernel A writes the centre term (unique writes), kernel B scatters the
+// neighbour term with fp32 global atomics.
+#include <hip/hip_runtime.h>
+#include "common.h"
+
+__global__ void gather_edge_fwd_kernel(
+    const float *__restrict__ feats,  // (B, N, C)
+    const int *__restrict__ idx,      // (B, N, K)
+    const float *__restrict__ xyz,    // (B, N, 3)
+    float *__restrict__ out,          // (B, C+3, K, N)
+    int B, int N, int K, int C) {
+  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)B * K * N;
+  if (gid >= total) return;
+  const int n = (int)(gid % N);
+  const int j = (int)((gid / N) % K);
+  const int b = (int)(gid / ((long)N * K));
+
+  const int nb = idx[((long)b * N + n) * K + j];
+  const float *f_nb = feats + ((long)b * N + nb) * C;
+  const float *f_ct = feats + ((long)b * N + n) * C;
+  float *dst = out + (((long)b * (C + 3)) * K + j) * N + n;
+  const long cstride = (long)K * N;
+  for (int c = 0; c < C; ++c) dst[c * cstride] = f_nb[c] - f_ct[c];
+
+  const float *x_nb = xyz + ((long)b * N + nb) * 3;
+  const float *x_ct = xyz + ((long)b * N + n) * 3;
+  dst += (long)C * cstride;
+  for (int c = 0; c < 3; ++c) dst[c * cstride] = x_nb[c] - x_ct[c];
+}
+
+// centre term: g[b,n,c] = -sum_j gout[b,c,j,n]; one thread per (b,c,n)
+__global__ void gather_edge_bwd_center_kernel(
+    const float *__restrict__ gout,  // (B, C+3, K, N)
+    float *__restrict__ gfeats,      // (B, N, C)
+    int B, int N, int K, int C) {
+  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)B * C * N;
+  if (gid >= total) return;
+  const int n = (int)(gid % N);
+  const int c = (int)((gid / N) % C);
+  const int b = (int)(gid / ((long)N * C));
+
+  const float *src = gout + (((long)b * (C + 3) + c) * K) * N + n;
+  float acc = 0.f;
+  for (int j = 0; j < K; ++j) acc += src[(long)j * N];
+  gfeats[((long)b * N + n) * C + c] = -acc;
+}
+
+// neighbour term: scatter-add, one thread per (b,j,n) looping channels
+__global__ void gather_edge_bwd_scatter_kernel(
+    const float *__restrict__ gout,  // (B, C+3, K, N)
+    const int *__restrict__ idx,     // (B, N, K)
+    float *__restrict__ gfeats,      // (B, N, C)
+    int B, int N, int K, int C) {
+  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)B * K * N;
+  if (gid >= total) return;
+  const int n = (int)(gid % N);
+  const int j = (int)((gid / N) % K);
+  const int b = (int)(gid / ((long)N * K));
+
+  const int nb = idx[((long)b * N + n) * K + j];
+  const float *src = gout + (((long)b * (C + 3)) * K + j) * N + n;
+  float *dst = gfeats + ((long)b * N + nb) * C;
+  const long cstride = (long)K * N;
+  for (int c = 0; c < C; ++c) atomicAdd(dst + c, src[c * cstride]);
+}
+
+void launch_gather_edge_fwd(const float *feats, const int *idx, const float *xyz,
+                            float *out, int B, int N, int K, int C,
+                            hipStream_t stream) {
+  const long total = (long)B * K * N;
+  const int threads = 256;
+  hipLaunchKernelGGL(gather_edge_fwd_kernel,
+                     dim3((total + threads - 1) / threads), dim3(threads), 0,
+                     stream, feats, idx, xyz, out, B, N, K, C);
+}
+
+void launch_gather_edge_bwd(const float *gout, const int *idx, float *gfeats,
+                            int B, int N, int K, int C, hipStream_t stream) {
+  const int threads = 256;
+  const long t1 = (long)B * C * N;
+  hipLaunchKernelGGL(gather_edge_bwd_center_kernel,
+                     dim3((t1 + threads - 1) / threads), dim3(threads), 0,
+                     stream, gout, gfeats, B, N, K, C);
+  const long t2 = (long)B * K * N;
+  hipLaunchKernelGGL(gather_edge_bwd_scatter_kernel,
+                     dim3((t2 + threads - 1) / threads), dim3(threads), 0,
+                     stream, gout, idx, gfeats, B, N, K, C);
+}

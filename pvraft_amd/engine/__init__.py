@@ -1,0 +1,2 @@
+from .trainer import Trainer, VAL_ITERS
+from .refine_trainer import RefineTrainer

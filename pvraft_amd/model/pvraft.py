@@ -1,0 +1,154 @@
+"""PV-RAFT scene-flow models.
+
+``PVRaft`` has the capabilities of reference model/RAFTSceneFlow.py (RSF):
+two weight-independent encoders (feature + context), the truncated
+correlation field, and an iterative GRU update loop producing one flow
+estimate per iteration.  ``PVRaftRefine`` (reference
+model/RAFTSceneFlowRefine.py: RSF_refine) runs the same backbone frozen
+(no_grad) and applies a trainable refinement head to the final flow using
+pc1's graph, returning a single refined flow.
+
+State-dict layout matches the reference (feature_extractor.*,
+context_extractor.*, corr_block.*, update_block.*, refine_block.*) so
+checkpoints are interchangeable both ways.
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+import torch.nn as nn
+from torch import Tensor
+
+from .corr import CorrBlock
+from .encoder import PointEncoder
+from .refine import RefineHead
+from .update import UpdateBlock
+
+
+class PVRaft(nn.Module):
+    def __init__(
+        self,
+        corr_levels: int = 3,
+        base_scales: float = 0.25,
+        truncate_k: int = 512,
+        resolution: int = 3,
+        hidden_dim: int = 64,
+        context_dim: int = 64,
+    ):
+        super().__init__()
+        self.hidden_dim = hidden_dim
+        self.context_dim = context_dim
+        self.feature_extractor = PointEncoder()
+        self.context_extractor = PointEncoder()
+        self.corr_block = CorrBlock(
+            num_levels=corr_levels,
+            base_scale=base_scales,
+            resolution=resolution,
+            truncate_k=truncate_k,
+        )
+        self.update_block = UpdateBlock(hidden_dim=hidden_dim)
+
+    @classmethod
+    def from_args(cls, args) -> "PVRaft":
+        return cls(
+            corr_levels=args.corr_levels,
+            base_scales=args.base_scales,
+            truncate_k=args.truncate_k,
+        )
+
+    def forward(self, p, num_iters: int = 12) -> List[Tensor]:
+        xyz1, xyz2 = p
+        fmap1, graph1 = self.feature_extractor(xyz1)
+        fmap2, _ = self.feature_extractor(xyz2)
+
+        field = self.corr_block.build(fmap1, fmap2, xyz2)
+
+        fct1, graph_context = self.context_extractor(xyz1)
+        net, inp = torch.split(fct1, [self.hidden_dim, self.context_dim], dim=1)
+        net = torch.tanh(net)
+        inp = torch.relu(inp)
+
+        coords1, coords2 = xyz1, xyz1
+        flow_predictions = []
+        for _ in range(num_iters):
+            coords2 = coords2.detach()
+            corr = self.corr_block(field, coords2)
+            flow = coords2 - coords1
+            net, delta_flow = self.update_block(net, inp, corr, flow, graph_context)
+            coords2 = coords2 + delta_flow
+            flow_predictions.append(coords2 - coords1)
+        return flow_predictions
+
+
+class PVRaftRefine(nn.Module):
+    def __init__(
+        self,
+        corr_levels: int = 3,
+        base_scales: float = 0.25,
+        truncate_k: int = 512,
+        resolution: int = 3,
+        hidden_dim: int = 64,
+        context_dim: int = 64,
+    ):
+        super().__init__()
+        self.hidden_dim = hidden_dim
+        self.context_dim = context_dim
+        self.feature_extractor = PointEncoder()
+        self.context_extractor = PointEncoder()
+        self.corr_block = CorrBlock(
+            num_levels=corr_levels,
+            base_scale=base_scales,
+            resolution=resolution,
+            truncate_k=truncate_k,
+        )
+        self.update_block = UpdateBlock(hidden_dim=hidden_dim)
+        self.refine_block = RefineHead()
+
+    @classmethod
+    def from_args(cls, args) -> "PVRaftRefine":
+        return cls(
+            corr_levels=args.corr_levels,
+            base_scales=args.base_scales,
+            truncate_k=args.truncate_k,
+        )
+
+    def freeze_backbone(self) -> None:
+        """Freeze everything except the refinement head.
+
+        The reference intends this at engine_refine.py:51-54 but sets
+        requires_grad on Modules (a no-op for params); actual freezing there
+        comes from the no_grad forward.  Here both are done explicitly.
+        """
+        for m in (self.feature_extractor, self.context_extractor, self.corr_block, self.update_block):
+            for pmt in m.parameters():
+                pmt.requires_grad_(False)
+
+    def forward(self, p, num_iters: int = 32) -> Tensor:
+        with torch.no_grad():
+            xyz1, xyz2 = p
+            fmap1, graph1 = self.feature_extractor(xyz1)
+            fmap2, _ = self.feature_extractor(xyz2)
+            field = self.corr_block.build(fmap1, fmap2, xyz2)
+
+            fct1, graph_context = self.context_extractor(xyz1)
+            net, inp = torch.split(fct1, [self.hidden_dim, self.context_dim], dim=1)
+            net = torch.tanh(net)
+            inp = torch.relu(inp)
+
+            coords1, coords2 = xyz1, xyz1
+            for _ in range(num_iters):
+                coords2 = coords2.detach()
+                corr = self.corr_block(field, coords2)
+                flow = coords2 - coords1
+                net, delta_flow = self.update_block(net, inp, corr, flow, graph_context)
+                coords2 = coords2 + delta_flow
+        return self.refine_block(coords2 - coords1, graph1)
+
+
+def build_model(args):
+    """CLI helper: pick the model family from the --refine flag."""
+    if getattr(args, "refine", False):
+        return PVRaftRefine.from_args(args)
+    return PVRaft.from_args(args)

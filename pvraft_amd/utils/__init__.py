@@ -1,0 +1,4 @@
+from .loss import compute_loss, sequence_loss
+from .metrics import compute_epe, compute_epe_train
+from .checkpoint import save_checkpoint, load_checkpoint, checkpoint_dir
+from .logging import setup_logger, ScalarLogger

@@ -1,0 +1,105 @@
+"""Model-level tests: shapes, iteration behaviour, reference state-dict
+parity, refine head, gradients end-to-end (CPU, small N)."""
+
+import torch
+
+from pvraft_amd.model import PVRaft, PVRaftRefine
+
+
+def tiny_model(refine=False, truncate_k=16):
+    cls = PVRaftRefine if refine else PVRaft
+    return cls(corr_levels=3, base_scales=0.25, truncate_k=truncate_k)
+
+
+def clouds(B=2, N=48):
+    xyz1 = torch.randn(B, N, 3)
+    xyz2 = xyz1 + 0.1 * torch.randn(B, N, 3)
+    return [xyz1, xyz2]
+
+
+def test_forward_returns_per_iteration_flows():
+    model = tiny_model()
+    p = clouds()
+    flows = model(p, num_iters=3)
+    assert len(flows) == 3
+    for f in flows:
+        assert f.shape == (2, 48, 3)
+        assert torch.isfinite(f).all()
+
+
+def test_backward_reaches_all_parameters():
+    model = tiny_model()
+    flows = model(clouds(), num_iters=2)
+    loss = sum(f.abs().mean() for f in flows)
+    loss.backward()
+    missing = [n for n, p in model.named_parameters() if p.grad is None]
+    assert missing == [], f"params without grad: {missing}"
+
+
+def test_state_dict_matches_reference_layout():
+    """Key names/shapes follow the reference (checkpoint interchange)."""
+    model = tiny_model(refine=True)
+    sd = model.state_dict()
+    expected = {
+        "feature_extractor.feat_conv1.fc1.weight": (16, 6, 1, 1),
+        "feature_extractor.feat_conv2.fc2.weight": (64, 48, 1, 1)[:2] + (1,),
+        "context_extractor.feat_conv3.gn3.weight": (128,),
+        "corr_block.out_conv.0.weight": (128, 81, 1),
+        "corr_block.out_conv.2.weight": (1,),  # PReLU
+        "corr_block.out_conv.3.weight": (64, 128, 1),
+        "corr_block.knn_conv.0.weight": (64, 4, 1, 1),
+        "corr_block.knn_out.weight": (64, 64, 1),
+        "update_block.motion_encoder.conv.weight": (61, 128, 1),
+        "update_block.gru.convz.weight": (64, 192, 1),
+        "update_block.flow_head.conv1.weight": (64, 64, 1),
+        "update_block.flow_head.setconv.fc1.weight": (64, 67, 1, 1),
+        "update_block.flow_head.out_conv.2.weight": (3, 64, 1),
+        "refine_block.ref_conv1.fc1.weight": (16, 6, 1, 1),
+        "refine_block.fc.weight": (3, 128),
+    }
+    for key, shape in expected.items():
+        assert key in sd, f"missing {key}"
+        assert tuple(sd[key].shape) == tuple(shape), (key, sd[key].shape, shape)
+
+
+def test_setconv_mid_width_rule():
+    """gconv.py:21-24: mid = out//2 if in odd else (in+out)//2."""
+    from pvraft_amd.model import SetConv
+
+    assert SetConv(3, 32).fc1.out_channels == 16
+    assert SetConv(32, 64).fc1.out_channels == 48
+    assert SetConv(64, 128).fc1.out_channels == 96
+    assert SetConv(64, 64).fc1.out_channels == 64
+
+
+def test_refine_model_returns_single_flow_and_only_refine_grads():
+    model = tiny_model(refine=True)
+    model.freeze_backbone()
+    flow = model(clouds(), num_iters=2)
+    assert flow.shape == (2, 48, 3)
+    flow.abs().mean().backward()
+    for n, p in model.named_parameters():
+        if n.startswith("refine_block"):
+            assert p.grad is not None, n
+        else:
+            assert p.grad is None, n
+
+
+def test_stage1_checkpoint_loads_into_refine_model():
+    m1 = tiny_model(refine=False)
+    m2 = tiny_model(refine=True)
+    missing, unexpected = m2.load_state_dict(m1.state_dict(), strict=False)
+    assert unexpected == []
+    assert all(k.startswith("refine_block") for k in missing)
+
+
+def test_iteration_zero_of_longer_run_matches_shorter_run():
+    """The GRU loop is causal: first k flows do not depend on later iters."""
+    model = tiny_model()
+    model.eval()
+    p = clouds(B=1, N=40)
+    with torch.no_grad():
+        f2 = model(p, num_iters=2)
+        f4 = model(p, num_iters=4)
+    assert torch.allclose(f2[0], f4[0], atol=1e-5)
+    assert torch.allclose(f2[1], f4[1], atol=1e-5)
