@@ -23,9 +23,16 @@ class PointEncoder(nn.Module):
         self.feat_conv2 = SetConv(width, 2 * width)
         self.feat_conv3 = SetConv(2 * width, 4 * width)
 
-    def forward(self, pc: Tensor):
-        """pc (B, N, 3) -> ((B, 4*width, N), Graph)."""
-        graph = Graph.build(pc, self.num_neighbors)
+    def forward(self, pc: Tensor, graph: Graph = None):
+        """pc (B, N, 3) -> ((B, 4*width, N), Graph).
+
+        ``graph`` may be passed in when the caller already built the kNN
+        graph of this cloud (the model shares pc1's graph between the
+        feature and context encoders -- identical numerics, one build
+        fewer than the reference, which rebuilds it per encoder).
+        """
+        if graph is None:
+            graph = Graph.build(pc, self.num_neighbors)
         x = self.feat_conv1(pc, graph)
         x = self.feat_conv2(x, graph)
         x = self.feat_conv3(x, graph)

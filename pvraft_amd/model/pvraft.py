@@ -23,6 +23,7 @@ from torch import Tensor
 
 from .corr import CorrBlock
 from .encoder import PointEncoder
+from .graph import Graph
 from .refine import RefineHead
 from .update import UpdateBlock
 
@@ -60,12 +61,14 @@ class PVRaft(nn.Module):
 
     def forward(self, p, num_iters: int = 12) -> List[Tensor]:
         xyz1, xyz2 = p
-        fmap1, graph1 = self.feature_extractor(xyz1)
+        graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
+        fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
         fmap2, _ = self.feature_extractor(xyz2)
 
         field = self.corr_block.build(fmap1, fmap2, xyz2)
 
-        fct1, graph_context = self.context_extractor(xyz1)
+        # pc1's graph is shared with the context encoder (same cloud)
+        fct1, graph_context = self.context_extractor(xyz1, graph=graph1)
         net, inp = torch.split(fct1, [self.hidden_dim, self.context_dim], dim=1)
         net = torch.tanh(net)
         inp = torch.relu(inp)
@@ -128,11 +131,12 @@ class PVRaftRefine(nn.Module):
     def forward(self, p, num_iters: int = 32) -> Tensor:
         with torch.no_grad():
             xyz1, xyz2 = p
-            fmap1, graph1 = self.feature_extractor(xyz1)
+            graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
+            fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
             fmap2, _ = self.feature_extractor(xyz2)
             field = self.corr_block.build(fmap1, fmap2, xyz2)
 
-            fct1, graph_context = self.context_extractor(xyz1)
+            fct1, graph_context = self.context_extractor(xyz1, graph=graph1)
             net, inp = torch.split(fct1, [self.hidden_dim, self.context_dim], dim=1)
             net = torch.tanh(net)
             inp = torch.relu(inp)
