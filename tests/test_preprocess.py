@@ -104,7 +104,8 @@ def test_ft3d_backprojection_geometry():
     # pixel at cx maps to x = 0
     disp1 = np.full((1, 960), 5.0, dtype=np.float32)
     pc1 = backproject(disp1)
-    assert abs(pc1[0, 479, 0] + (479 - 479.5) / -5.0) < 1e-5
+    # x = -(px - cx) / disp = -(479 - 479.5)/5 = +0.1
+    assert abs(pc1[0, 479, 0] - 0.1) < 1e-5
 
 
 def test_kitti_depth_and_projection():
