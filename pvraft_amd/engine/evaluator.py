@@ -61,6 +61,7 @@ def evaluate(args):
         log.info(f"Loaded weights from {path}")
     model.eval()
 
+    dump = bool(getattr(args, "dump_results", False))
     sums = [0.0] * 5
     n = 0
     for batch in loader:
@@ -75,6 +76,14 @@ def evaluate(args):
         epe3d, accs, accr, outl = compute_epe(final.float(), batch)
         for j, v in enumerate((loss.item(), epe3d, accs, accr, outl)):
             sums[j] += v
+        if dump:
+            import numpy as np
+
+            d = os.path.join(args.root, "result", args.dataset, str(n))
+            os.makedirs(d, exist_ok=True)
+            np.save(os.path.join(d, "pc1.npy"), batch["sequence"][0].cpu().numpy())
+            np.save(os.path.join(d, "pc2.npy"), batch["sequence"][1].cpu().numpy())
+            np.save(os.path.join(d, "flow.npy"), final.cpu().numpy())
         n += 1
     means = [s / max(n, 1) for s in sums]
     log.info(

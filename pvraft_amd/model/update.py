@@ -49,6 +49,19 @@ class ConvGRU(nn.Module):
         return (1 - z) * h + z * q
 
 
+class ConvRNN(nn.Module):
+    """Vanilla tanh RNN alternative to ConvGRU (reference update.py:43-54;
+    defined for parity -- the reference never instantiates it either)."""
+
+    def __init__(self, input_dim: int = 128, hidden_dim: int = 64):
+        super().__init__()
+        self.convx = nn.Conv1d(input_dim, hidden_dim, 1)
+        self.convh = nn.Conv1d(hidden_dim, hidden_dim, 1)
+
+    def forward(self, h: Tensor, x: Tensor) -> Tensor:
+        return torch.tanh(self.convx(x) + self.convh(h))
+
+
 class FlowHead(nn.Module):
     def __init__(self, input_dim: int = 64):
         super().__init__()

@@ -17,6 +17,7 @@ def parse_args():
     parser = argparse.ArgumentParser(description="Testing Argument")
     add_common_args(parser, training=False)
     parser.add_argument("--gamma", help="exponential weights", default=0.8, type=float)
+    parser.add_argument("--dump_results", help="write result/<dataset>/<i>/{pc1,pc2,flow}.npy for visual.py", action="store_true")
     return parser.parse_args()
 
 
