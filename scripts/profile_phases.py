@@ -7,6 +7,10 @@ GRU loop forward, loss+backward, optimizer.  Run on the GPU box:
 """
 
 import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time
 
 import torch
