@@ -134,14 +134,72 @@ def gather_edge_concat(feats: Tensor, idx: Tensor, xyz: Tensor) -> Tensor:
     return reference.gather_edge_concat(feats, idx, xyz)
 
 
+class _CorrTruncate(torch.autograd.Function):
+    """Truncated correlation with gather-based backward.
+
+    Forward (no_grad): chunked rocBLAS GEMM + per-row top-K -- the N x M
+    matrix exists only one row-chunk at a time and is NOT saved for
+    backward (plain autograd through matmul+topk would retain all of it,
+    reference corr.py:34-37 does exactly that).  Backward uses the saved
+    top-K indices:
+        d f1[:, n]  = scale * sum_k g[n, k] * f2[:, idx[n, k]]
+        d f2[:, m] += scale * sum_{n,k: idx=m} g[n, k] * f1[:, n]
+    which is O(N*K*C) instead of O(N*M*C).
+    """
+
+    CHUNK = 2048
+
+    @staticmethod
+    def forward(ctx, fmap1: Tensor, fmap2: Tensor, xyz2: Tensor, truncate_k: int):
+        with torch.no_grad():
+            corr, idx, txyz = reference.corr_truncate(
+                fmap1, fmap2, xyz2, truncate_k, chunk=_CorrTruncate.CHUNK
+            )
+        ctx.save_for_backward(fmap1, fmap2, idx)
+        return corr, idx, txyz
+
+    @staticmethod
+    def backward(ctx, g_corr: Tensor, _g_idx, _g_xyz):
+        fmap1, fmap2, idx = ctx.saved_tensors
+        B, C, N = fmap1.shape
+        M = fmap2.shape[2]
+        K = idx.shape[2]
+        scale = 1.0 / math.sqrt(C)
+        g_corr = g_corr * scale
+        g1 = torch.empty_like(fmap1)
+        g2 = torch.zeros_like(fmap2)
+        f2t = fmap2.transpose(1, 2)  # B, M, C
+        chunk = _CorrTruncate.CHUNK
+        for s in range(0, N, chunk):
+            e = min(s + chunk, N)
+            idx_c = idx[:, s:e]  # B, n, K
+            g_c = g_corr[:, s:e]  # B, n, K
+            flat = idx_c.reshape(B, -1)
+            f2g = f2t.gather(1, flat.unsqueeze(-1).expand(B, flat.shape[1], C)).view(
+                B, e - s, K, C
+            )
+            g1[:, :, s:e] = torch.einsum("bnkc,bnk->bcn", f2g, g_c)
+            # scatter f1 * g into columns of g2
+            vals = fmap1[:, :, s:e].unsqueeze(-1) * g_c.unsqueeze(1)  # B, C, n, K
+            g2.scatter_add_(
+                2,
+                flat.unsqueeze(1).expand(B, C, flat.shape[1]),
+                vals.reshape(B, C, -1),
+            )
+        return g1, g2, None, None
+
+
 def corr_truncate(fmap1: Tensor, fmap2: Tensor, xyz2: Tensor, truncate_k: int):
     """(B,C,N),(B,C,M),(B,M,3) -> corr (B,N,K), idx (B,N,K), xyz (B,N,K,3).
 
-    GEMM-shaped: runs on rocBLAS via chunked bmm + topk on both backends for
-    now (the chunking keeps the N x M matrix tiled); a fused MFMA+top-K HIP
-    kernel replaces the GPU path when profiling justifies it.
+    GEMM runs on rocBLAS (chunked bmm + topk; the N x M matrix is tiled);
+    on GPU the custom-backward path avoids retaining it for autograd.
     """
-    return reference.corr_truncate(fmap1.float(), fmap2.float(), xyz2, truncate_k)
+    fmap1 = fmap1.float()
+    fmap2 = fmap2.float()
+    if fmap1.is_cuda and os.environ.get("PVRAFT_REF_OPS", "0") != "1":
+        return _CorrTruncate.apply(fmap1, fmap2, xyz2, truncate_k)
+    return reference.corr_truncate(fmap1, fmap2, xyz2, truncate_k)
 
 
 def voxel_corr(

@@ -142,3 +142,26 @@ def test_knn_corr_grad_is_gather():
     # each row has exactly k ones scattered at the selected indices
     assert corr.grad.sum().item() == pytest.approx(B * N * k)
     assert ((corr.grad == 0) | (corr.grad == 1)).all()
+
+
+def test_corr_truncate_custom_backward_matches_autograd():
+    """The gather-based backward (_CorrTruncate, GPU path) vs plain autograd."""
+    from pvraft_amd.ops import _CorrTruncate
+
+    B, C, N, M, K = 2, 8, 15, 21, 6
+    f1 = torch.randn(B, C, N, requires_grad=True)
+    f2 = torch.randn(B, C, M, requires_grad=True)
+    xyz2 = torch.randn(B, M, 3)
+    f1r = f1.detach().clone().requires_grad_(True)
+    f2r = f2.detach().clone().requires_grad_(True)
+
+    corr, idx, txyz = _CorrTruncate.apply(f1, f2, xyz2, K)
+    corr_r, idx_r, txyz_r = R.corr_truncate(f1r, f2r, xyz2, K)
+    assert torch.allclose(corr, corr_r, atol=1e-6)
+    assert torch.equal(idx, idx_r)
+
+    g = torch.randn_like(corr)
+    corr.backward(g)
+    corr_r.backward(g)
+    assert torch.allclose(f1.grad, f1r.grad, atol=1e-5)
+    assert torch.allclose(f2.grad, f2r.grad, atol=1e-5)
