@@ -8,6 +8,8 @@ void launch_gather_edge_fwd(const float*, const int*, const float*, float*,
                             int, int, int, int, hipStream_t);
 void launch_gather_edge_bwd(const float*, const int*, float*, int, int, int,
                             int, hipStream_t);
+void launch_gather_edge_bwd_csr(const float*, const int*, const int*, float*,
+                                int, int, int, int, hipStream_t);
 void launch_voxel_corr_fwd(const float*, const float*, const float*, float*,
                            int, int, int, int, float, hipStream_t);
 void launch_voxel_corr_bwd(const float*, const float*, const float*, float*,
@@ -16,6 +18,13 @@ void launch_knn_corr_fwd(const float*, const float*, const float*, float*,
                          int*, int, int, int, int, hipStream_t);
 void launch_knn_corr_bwd(const float*, const int*, float*, int, int, int, int,
                          hipStream_t);
+
+void launch_gn_fwd(const void*, void*, float*, float*, float*, const float*,
+                   const float*, int, long, long, int, int, float, int, float,
+                   bool, hipStream_t);
+void launch_gn_bwd(const void*, const void*, const float*, const float*,
+                   const float*, const float*, float*, float*, void*, int,
+                   long, long, int, int, int, float, bool, hipStream_t);
 
 namespace {
 
@@ -61,6 +70,20 @@ torch::Tensor gather_edge_concat_bwd(torch::Tensor gout, torch::Tensor idx,
   launch_gather_edge_bwd(gout.data_ptr<float>(), idx.data_ptr<int>(),
                          gfeats.data_ptr<float>(), B, N, K, (int)C, stream());
   return gfeats;
+}
+
+torch::Tensor gather_edge_bwd_csr(torch::Tensor gT, torch::Tensor order,
+                                  torch::Tensor offsets, int64_t K) {
+  check_f32(gT, "gT");
+  TORCH_CHECK(order.scalar_type() == torch::kInt32 && order.is_contiguous());
+  TORCH_CHECK(offsets.scalar_type() == torch::kInt32 && offsets.is_contiguous());
+  const int B = gT.size(0), NK = gT.size(1), C = gT.size(2);
+  const int N = NK / (int)K;
+  auto grad = torch::empty({B, N, C}, gT.options());
+  launch_gather_edge_bwd_csr(gT.data_ptr<float>(), order.data_ptr<int>(),
+                             offsets.data_ptr<int>(), grad.data_ptr<float>(),
+                             B, N, (int)K, C, stream());
+  return grad;
 }
 
 torch::Tensor voxel_corr_fwd(torch::Tensor corr, torch::Tensor xyz,
@@ -117,10 +140,74 @@ torch::Tensor knn_corr_bwd(torch::Tensor gout, torch::Tensor idx, int64_t K) {
   return gcorr;
 }
 
+// x (B, C, S) contiguous (S = flattened spatial); weight/bias (C) fp32.
+// act: 0 none, 1 LeakyReLU(slope).  Returns {y, mean (B*G), rstd (B*G)}.
+std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
+                                              torch::Tensor weight,
+                                              torch::Tensor bias, double eps,
+                                              int64_t act, double slope) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous GPU");
+  TORCH_CHECK(x.dim() == 3, "x must be (B, C, S)");
+  check_f32(weight, "weight");
+  check_f32(bias, "bias");
+  const int B = x.size(0), C = x.size(1);
+  const long S = x.size(2);
+  TORCH_CHECK(C % G == 0, "C % G != 0");
+  const int rows = B * G;
+  const long row_len = (C / G) * S;
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto ws = torch::zeros({rows, 2}, fopt);
+  auto mean = torch::empty({rows}, fopt);
+  auto rstd = torch::empty({rows}, fopt);
+  auto y = torch::empty_like(x);
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32,
+              "group_norm_act: dtype must be float32 or bfloat16");
+  launch_gn_fwd(x.data_ptr(), y.data_ptr(), ws.data_ptr<float>(),
+                mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                weight.data_ptr<float>(), bias.data_ptr<float>(), rows,
+                row_len, S, C, (int)G, (float)eps, (int)act, (float)slope,
+                bf16, stream());
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
+                                              torch::Tensor mean,
+                                              torch::Tensor rstd, int64_t G,
+                                              torch::Tensor weight,
+                                              torch::Tensor bias, int64_t act,
+                                              double slope) {
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous(), "dy/x must be contiguous");
+  const int B = x.size(0), C = x.size(1);
+  const long S = x.size(2);
+  const int rows = B * G;
+  const long row_len = (C / G) * S;
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto row_ws = torch::zeros({rows, 2}, fopt);
+  auto chan_ws = torch::zeros({C, 2}, fopt);
+  auto dx = torch::empty_like(x);
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32,
+              "group_norm_act: dtype must be float32 or bfloat16");
+  TORCH_CHECK(dy.scalar_type() == x.scalar_type(), "dy/x dtype mismatch");
+  launch_gn_bwd(dy.data_ptr(), x.data_ptr(), mean.data_ptr<float>(),
+                rstd.data_ptr<float>(), weight.data_ptr<float>(),
+                bias.data_ptr<float>(), row_ws.data_ptr<float>(),
+                chan_ws.data_ptr<float>(), dx.data_ptr(), rows, row_len, S, C,
+                (int)G, (int)act, (float)slope, bf16, stream());
+  // chan_ws[:, 0] = dbias, chan_ws[:, 1] = dweight
+  auto dbias = chan_ws.select(1, 0).contiguous();
+  auto dweight = chan_ws.select(1, 1).contiguous();
+  return {dx, dweight, dbias};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("group_norm_act_fwd", &group_norm_act_fwd);
+  m.def("group_norm_act_bwd", &group_norm_act_bwd);
   m.def("knn_graph", &knn_graph, "fused kNN graph (CDNA4)");
   m.def("gather_edge_concat_fwd", &gather_edge_concat_fwd);
   m.def("gather_edge_concat_bwd", &gather_edge_concat_bwd);
+  m.def("gather_edge_bwd_csr", &gather_edge_bwd_csr);
   m.def("voxel_corr_fwd", &voxel_corr_fwd);
   m.def("voxel_corr_bwd", &voxel_corr_bwd);
   m.def("knn_corr_fwd", &knn_corr_fwd);

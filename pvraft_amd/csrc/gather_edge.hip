@@ -79,6 +79,51 @@ __global__ void gather_edge_bwd_scatter_kernel(
   for (int c = 0; c < C; ++c) atomicAdd(dst + c, src[c * cstride]);
 }
 
+// CSR backward: deterministic, atomic-free.  The caller transposes the
+// first C channels of gout to gT (B, N, K, C) (so each edge's gradient is
+// a contiguous C-vector) and provides the inverse adjacency in CSR form:
+// order (B, N*K) = edge ids sorted by target node, offsets (B, N+1).
+//   grad[b, m, c] = sum_{t in [off[m], off[m+1])} gT[b, order[t], c]
+//                 - sum_j gT[b, m*K + j, c]
+// Thread = (b, m, c) with c fastest: both loops read contiguous C-vectors,
+// wave-coalesced; writes are unique.  Replaces the fp32-atomic scatter
+// (measured 15% of the train step).
+__global__ void gather_edge_bwd_csr_kernel(
+    const float *__restrict__ gT,      // (B, N*K, C)
+    const int *__restrict__ order,     // (B, N*K)
+    const int *__restrict__ offsets,   // (B, N+1)
+    float *__restrict__ grad,          // (B, N, C)
+    int B, int N, int K, int C) {
+  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)B * N * C;
+  if (gid >= total) return;
+  const int c = (int)(gid % C);
+  const long t1 = gid / C;
+  const int m = (int)(t1 % N);
+  const int b = (int)(t1 / N);
+
+  const float *gbase = gT + (long)b * N * K * C;
+  const int *ord = order + (long)b * N * K;
+  const int lo = offsets[(long)b * (N + 1) + m];
+  const int hi = offsets[(long)b * (N + 1) + m + 1];
+  float acc = 0.f;
+  for (int t = lo; t < hi; ++t) acc += gbase[(long)ord[t] * C + c];
+  const float *own = gbase + ((long)m * K) * C + c;
+  float cen = 0.f;
+  for (int j = 0; j < K; ++j) cen += own[(long)j * C];
+  grad[gid] = acc - cen;
+}
+
+void launch_gather_edge_bwd_csr(const float *gT, const int *order,
+                                const int *offsets, float *grad, int B, int N,
+                                int K, int C, hipStream_t stream) {
+  const long total = (long)B * N * C;
+  const int threads = 256;
+  hipLaunchKernelGGL(gather_edge_bwd_csr_kernel,
+                     dim3((total + threads - 1) / threads), dim3(threads), 0,
+                     stream, gT, order, offsets, grad, B, N, K, C);
+}
+
 void launch_gather_edge_fwd(const float *feats, const int *idx, const float *xyz,
                             float *out, int B, int N, int K, int C,
                             hipStream_t stream) {

@@ -1,0 +1,283 @@
+// K9: GroupNorm (+ fused LeakyReLU) forward/backward for CDNA4.
+//
+// Why custom: ATen's GroupNorm forward launches ONE workgroup per
+// (batch, group) row -- B=2, G=8 => 16 workgroups on a 256-CU chip, and it
+// measured 30% of the PV-RAFT train step.  Here both directions use
+// multi-workgroup reductions (fp32 accumulation, block partials combined
+// with a handful of atomics into a small workspace) so the chip is filled
+// regardless of B.
+//
+// Layout: x (B, C, S) contiguous (S = flattened spatial, e.g. K*N), G
+// groups, group g = channels [g*Cg, (g+1)*Cg) -- a group's data is one
+// contiguous block of Cg*S elements, reduced as a flat row.  Each
+// workgroup owns a CONTIGUOUS chunk of one row, so a thread's channel
+// changes only every ~S/256 iterations and per-channel partials flush with
+// O(channels-touched) atomics, not O(elements).
+// dtype: fp32 or bf16 IO (template), fp32 math everywhere.
+//
+// act: 0 = identity, 1 = LeakyReLU(slope) fused into the normalize pass
+// (backward recomputes the pre-activation sign from xhat, gamma, beta).
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include "common.h"
+
+#define GN_THREADS 256
+
+template <typename T>
+DEV_INLINE float ld(const T *p) {
+  return (float)*p;
+}
+
+template <typename T>
+DEV_INLINE void st(T *p, float v) {
+  *p = (T)v;
+}
+
+DEV_INLINE float block_sum(float v) {
+  v = wave_sum(v);
+  __shared__ float sh[GN_THREADS / WAVE];
+  if (lane_id() == 0) sh[wave_id()] = v;
+  __syncthreads();
+  float t = 0.f;
+  if (threadIdx.x < GN_THREADS / WAVE) t = sh[threadIdx.x];
+  __syncthreads();  // sh reused across calls
+  return wave_sum(t);  // valid in wave 0
+}
+
+// chunk bounds for block `chunk` of `blocks_per_row` over a row of row_len
+DEV_INLINE void chunk_range(long row_len, int blocks_per_row, int chunk,
+                            long &lo, long &hi) {
+  const long per = (row_len + blocks_per_row - 1) / blocks_per_row;
+  lo = (long)chunk * per;
+  hi = min(lo + per, row_len);
+}
+
+// ---------------------------------------------------------------- forward
+
+// pass 1: partial sum/sumsq per row chunk -> ws[row] = {sum, sumsq}
+template <typename T>
+__global__ __launch_bounds__(GN_THREADS) void gn_fwd_reduce_kernel(
+    const T *__restrict__ x, float *__restrict__ ws, long row_len, int rows,
+    int blocks_per_row) {
+  const int row = blockIdx.x / blocks_per_row;
+  if (row >= rows) return;
+  long lo, hi;
+  chunk_range(row_len, blocks_per_row, blockIdx.x % blocks_per_row, lo, hi);
+  const T *base = x + (long)row * row_len;
+  float s = 0.f, ss = 0.f;
+  for (long i = lo + threadIdx.x; i < hi; i += GN_THREADS) {
+    const float v = ld(base + i);
+    s += v;
+    ss += v * v;
+  }
+  s = block_sum(s);
+  ss = block_sum(ss);
+  if (threadIdx.x == 0) {
+    atomicAdd(&ws[row * 2 + 0], s);
+    atomicAdd(&ws[row * 2 + 1], ss);
+  }
+}
+
+// pass 2: rows threads -> mean/rstd
+__global__ void gn_fwd_finalize_kernel(const float *__restrict__ ws,
+                                       float *__restrict__ mean,
+                                       float *__restrict__ rstd, long row_len,
+                                       int rows, float eps) {
+  const int row = blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= rows) return;
+  const float m = ws[row * 2 + 0] / (float)row_len;
+  const float var = ws[row * 2 + 1] / (float)row_len - m * m;
+  mean[row] = m;
+  rstd[row] = rsqrtf(fmaxf(var, 0.f) + eps);
+}
+
+// pass 3: y = act((x - mean) * rstd * gamma + beta)
+template <typename T, int ACT>
+__global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
+    const T *__restrict__ x, T *__restrict__ y, const float *__restrict__ mean,
+    const float *__restrict__ rstd, const float *__restrict__ gamma,
+    const float *__restrict__ beta, long S, int C, int G, long total,
+    float slope) {
+  const int Cg = C / G;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long c = (i / S) % C;
+    const long row = i / (S * Cg);  // == b * G + g
+    float v = (ld(x + i) - mean[row]) * rstd[row] * gamma[c] + beta[c];
+    if (ACT == 1) v = v > 0.f ? v : v * slope;
+    st(y + i, v);
+  }
+}
+
+// ---------------------------------------------------------------- backward
+
+// partial sums: per-row {sum dxhat, sum dxhat*xhat}, per-channel
+// {sum dy_norm, sum dy_norm*xhat}; block = contiguous chunk of one row
+template <typename T, int ACT>
+__global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
+    const T *__restrict__ dy, const T *__restrict__ x,
+    const float *__restrict__ mean, const float *__restrict__ rstd,
+    const float *__restrict__ gamma, const float *__restrict__ beta,
+    float *__restrict__ row_ws,   // (rows, 2)
+    float *__restrict__ chan_ws,  // (C, 2)
+    long row_len, long S, int C, int G, int blocks_per_row, int rows,
+    float slope) {
+  const int row = blockIdx.x / blocks_per_row;
+  if (row >= rows) return;
+  long lo, hi;
+  chunk_range(row_len, blocks_per_row, blockIdx.x % blocks_per_row, lo, hi);
+  const int Cg = C / G;
+  const long base_off = (long)row * row_len;
+  const float m = mean[row];
+  const float r = rstd[row];
+
+  float sum_dx = 0.f, sum_dxx = 0.f;
+  long cur_c = -1;
+  float c_dg = 0.f, c_db = 0.f;
+  for (long i = lo + threadIdx.x; i < hi; i += GN_THREADS) {
+    const long gi = base_off + i;
+    const long c = (gi / S) % C;
+    if (c != cur_c) {
+      if (cur_c >= 0) {
+        atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
+        atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
+      }
+      cur_c = c;
+      c_dg = 0.f;
+      c_db = 0.f;
+    }
+    const float xhat = (ld(x + gi) - m) * r;
+    float g = ld(dy + gi);
+    if (ACT == 1) {
+      const float pre = xhat * gamma[c] + beta[c];
+      g = pre > 0.f ? g : g * slope;
+    }
+    c_db += g;
+    c_dg += g * xhat;
+    const float dxhat = g * gamma[c];
+    sum_dx += dxhat;
+    sum_dxx += dxhat * xhat;
+  }
+  if (cur_c >= 0) {
+    atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
+    atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
+  }
+  sum_dx = block_sum(sum_dx);
+  sum_dxx = block_sum(sum_dxx);
+  if (threadIdx.x == 0) {
+    atomicAdd(&row_ws[row * 2 + 0], sum_dx);
+    atomicAdd(&row_ws[row * 2 + 1], sum_dxx);
+  }
+}
+
+template <typename T, int ACT>
+__global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
+    const T *__restrict__ dy, const T *__restrict__ x,
+    const float *__restrict__ mean, const float *__restrict__ rstd,
+    const float *__restrict__ gamma, const float *__restrict__ beta,
+    const float *__restrict__ row_ws, T *__restrict__ dx, long S, int C,
+    int G, long row_len, long total, float slope) {
+  const int Cg = C / G;
+  const float inv_n = 1.0f / (float)row_len;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long c = (i / S) % C;
+    const long row = i / (S * Cg);
+    const float r = rstd[row];
+    const float xhat = (ld(x + i) - mean[row]) * r;
+    float g = ld(dy + i);
+    if (ACT == 1) {
+      const float pre = xhat * gamma[c] + beta[c];
+      g = pre > 0.f ? g : g * slope;
+    }
+    const float dxhat = g * gamma[c];
+    const float v =
+        (dxhat - (row_ws[row * 2 + 0] + xhat * row_ws[row * 2 + 1]) * inv_n) * r;
+    st(dx + i, v);
+  }
+}
+
+// --------------------------------------------------------------- launchers
+
+static int pick_blocks_per_row(long row_len, int rows) {
+  int bpr = 1;
+  while ((long)bpr * rows < 2048 && (long)bpr * GN_THREADS < row_len) bpr *= 2;
+  return bpr;
+}
+
+template <typename T>
+void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
+                 const float *gamma, const float *beta, int rows, long row_len,
+                 long S, int C, int G, float eps, int act, float slope,
+                 hipStream_t stream) {
+  const long total = (long)rows * row_len;
+  const int bpr = pick_blocks_per_row(row_len, rows);
+  hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, dim3(rows * bpr),
+                     dim3(GN_THREADS), 0, stream, x, ws, row_len, rows, bpr);
+  hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
+                     0, stream, ws, mean, rstd, row_len, rows, eps);
+  const int apply_blocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
+  if (act == 1)
+    hipLaunchKernelGGL((gn_fwd_apply_kernel<T, 1>), dim3(apply_blocks),
+                       dim3(GN_THREADS), 0, stream, x, y, mean, rstd, gamma,
+                       beta, S, C, G, total, slope);
+  else
+    hipLaunchKernelGGL((gn_fwd_apply_kernel<T, 0>), dim3(apply_blocks),
+                       dim3(GN_THREADS), 0, stream, x, y, mean, rstd, gamma,
+                       beta, S, C, G, total, slope);
+}
+
+template <typename T>
+void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
+                 const float *gamma, const float *beta, float *row_ws,
+                 float *chan_ws, T *dx, int rows, long row_len, long S, int C,
+                 int G, int act, float slope, hipStream_t stream) {
+  const long total = (long)rows * row_len;
+  const int bpr = pick_blocks_per_row(row_len, rows);
+  const int apply_blocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
+  if (act == 1) {
+    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 1>), dim3(rows * bpr),
+                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
+                       beta, row_ws, chan_ws, row_len, S, C, G, bpr, rows, slope);
+    hipLaunchKernelGGL((gn_bwd_apply_kernel<T, 1>), dim3(apply_blocks),
+                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
+                       beta, row_ws, dx, S, C, G, row_len, total, slope);
+  } else {
+    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 0>), dim3(rows * bpr),
+                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
+                       beta, row_ws, chan_ws, row_len, S, C, G, bpr, rows, slope);
+    hipLaunchKernelGGL((gn_bwd_apply_kernel<T, 0>), dim3(apply_blocks),
+                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
+                       beta, row_ws, dx, S, C, G, row_len, total, slope);
+  }
+}
+
+// type-erased entry points (bindings.cpp is host-compiled, no HIP types)
+void launch_gn_fwd(const void *x, void *y, float *ws, float *mean, float *rstd,
+                   const float *gamma, const float *beta, int rows,
+                   long row_len, long S, int C, int G, float eps, int act,
+                   float slope, bool bf16, hipStream_t stream) {
+  if (bf16)
+    gn_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)x, (__hip_bfloat16 *)y,
+                                ws, mean, rstd, gamma, beta, rows, row_len, S,
+                                C, G, eps, act, slope, stream);
+  else
+    gn_fwd_impl<float>((const float *)x, (float *)y, ws, mean, rstd, gamma,
+                       beta, rows, row_len, S, C, G, eps, act, slope, stream);
+}
+
+void launch_gn_bwd(const void *dy, const void *x, const float *mean,
+                   const float *rstd, const float *gamma, const float *beta,
+                   float *row_ws, float *chan_ws, void *dx, int rows,
+                   long row_len, long S, int C, int G, int act, float slope,
+                   bool bf16, hipStream_t stream) {
+  if (bf16)
+    gn_bwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)dy,
+                                (const __hip_bfloat16 *)x, mean, rstd, gamma,
+                                beta, row_ws, chan_ws, (__hip_bfloat16 *)dx,
+                                rows, row_len, S, C, G, act, slope, stream);
+  else
+    gn_bwd_impl<float>((const float *)dy, (const float *)x, mean, rstd, gamma,
+                       beta, row_ws, chan_ws, (float *)dx, rows, row_len, S, C,
+                       G, act, slope, stream);
+}

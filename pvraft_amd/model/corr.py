@@ -32,6 +32,7 @@ import torch.nn as nn
 from torch import Tensor
 
 from pvraft_amd import ops
+from .norm import FusedGroupNorm
 
 
 @dataclass
@@ -57,13 +58,13 @@ class CorrBlock(nn.Module):
         self.knn = knn
         self.out_conv = nn.Sequential(
             nn.Conv1d((resolution ** 3) * num_levels, 128, 1),
-            nn.GroupNorm(8, 128),
+            FusedGroupNorm(8, 128),
             nn.PReLU(),
             nn.Conv1d(128, 64, 1),
         )
         self.knn_conv = nn.Sequential(
             nn.Conv2d(4, 64, 1),
-            nn.GroupNorm(8, 64),
+            FusedGroupNorm(8, 64),
             nn.PReLU(),
         )
         self.knn_out = nn.Conv1d(64, 64, 1)

@@ -9,7 +9,8 @@ by the fused gather kernel (ops.gather_edge_concat), never stored.
 
 from __future__ import annotations
 
-from dataclasses import dataclass
+from dataclasses import dataclass, field
+from typing import Optional, Tuple
 
 import torch
 from torch import Tensor
@@ -21,6 +22,7 @@ from pvraft_amd import ops
 class Graph:
     idx: Tensor  # (B, N, k) int64 neighbour indices (self included)
     xyz: Tensor  # (B, N, 3) the cloud the graph was built on
+    _csr: Optional[Tuple[Tensor, Tensor]] = field(default=None, repr=False)
 
     @property
     def k(self) -> int:
@@ -29,3 +31,21 @@ class Graph:
     @staticmethod
     def build(xyz: Tensor, k: int) -> "Graph":
         return Graph(idx=ops.knn_graph(xyz, k), xyz=xyz)
+
+    def csr(self) -> Optional[Tuple[Tensor, Tensor]]:
+        """Inverse adjacency in CSR form, for the deterministic SetConv
+        backward: (order (B, N*k) int32 = edge ids sorted by target node,
+        offsets (B, N+1) int32).  Built lazily once per graph (GPU only)
+        and shared by every SetConv/FlowHead call on this graph.
+        """
+        if not self.idx.is_cuda:
+            return None
+        if self._csr is None:
+            B, N, k = self.idx.shape
+            flat = self.idx.reshape(B, N * k)
+            order = flat.argsort(dim=1)
+            targets = flat.gather(1, order)
+            bounds = torch.arange(N + 1, device=flat.device).expand(B, N + 1)
+            offsets = torch.searchsorted(targets, bounds, side="left")
+            self._csr = (order.to(torch.int32).contiguous(), offsets.to(torch.int32).contiguous())
+        return self._csr

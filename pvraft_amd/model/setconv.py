@@ -20,6 +20,7 @@ from torch import Tensor
 
 from pvraft_amd import ops
 from .graph import Graph
+from .norm import FusedGroupNorm
 
 
 class SetConv(nn.Module):
@@ -27,18 +28,18 @@ class SetConv(nn.Module):
         super().__init__()
         mid = out_ch // 2 if in_ch % 2 != 0 else (out_ch + in_ch) // 2
         self.fc1 = nn.Conv2d(in_ch + 3, mid, 1, bias=False)
-        self.gn1 = nn.GroupNorm(8, mid, affine=True)
+        self.gn1 = FusedGroupNorm(8, mid, act="lrelu", slope=0.1)
         self.fc2 = nn.Conv1d(mid, out_ch, 1, bias=False)
-        self.gn2 = nn.GroupNorm(8, out_ch, affine=True)
+        self.gn2 = FusedGroupNorm(8, out_ch, act="lrelu", slope=0.1)
         self.fc3 = nn.Conv1d(out_ch, out_ch, 1, bias=False)
-        self.gn3 = nn.GroupNorm(8, out_ch, affine=True)
-        self.lrelu = nn.LeakyReLU(negative_slope=0.1)
+        self.gn3 = FusedGroupNorm(8, out_ch, act="lrelu", slope=0.1)
 
     def forward(self, feats: Tensor, graph: Graph) -> Tensor:
-        """feats (B, N, C) -> (B, N, out_ch)."""
-        x = ops.gather_edge_concat(feats, graph.idx, graph.xyz)  # B, C+3, K, N
-        x = self.lrelu(self.gn1(self.fc1(x)))
+        """feats (B, N, C) -> (B, N, out_ch).  LeakyReLU(0.1) is fused into
+        each GroupNorm (reference order gconv.py:71-83: conv -> GN -> lrelu)."""
+        x = ops.gather_edge_concat(feats, graph.idx, graph.xyz, csr=graph.csr())  # B, C+3, K, N
+        x = self.gn1(self.fc1(x))
         x = x.max(dim=2)[0]  # max-pool over neighbours -> B, mid, N
-        x = self.lrelu(self.gn2(self.fc2(x)))
-        x = self.lrelu(self.gn3(self.fc3(x)))
+        x = self.gn2(self.fc2(x))
+        x = self.gn3(self.fc3(x))
         return x.transpose(1, 2)

@@ -66,19 +66,31 @@ def hip_available() -> bool:
 
 
 class _GatherEdgeConcat(torch.autograd.Function):
-    """out (B,C+3,K,N) = concat(feat[nbr]-feat[center], xyz[nbr]-xyz[center])."""
+    """out (B,C+3,K,N) = concat(feat[nbr]-feat[center], xyz[nbr]-xyz[center]).
+
+    Backward prefers the deterministic CSR path (inverse adjacency sorted
+    by target node, no atomics); falls back to the fp32-atomic scatter when
+    no CSR is provided (direct op calls outside a Graph).
+    """
 
     @staticmethod
-    def forward(ctx, feats: Tensor, idx: Tensor, xyz: Tensor) -> Tensor:
-        ctx.save_for_backward(idx)
+    def forward(ctx, feats: Tensor, idx: Tensor, xyz: Tensor, order, offsets) -> Tensor:
+        ctx.save_for_backward(idx, order, offsets)
         ctx.C = feats.shape[2]
         return _EXT.gather_edge_concat_fwd(feats, idx, xyz)
 
     @staticmethod
     def backward(ctx, grad_out: Tensor):
-        (idx,) = ctx.saved_tensors
-        g = _EXT.gather_edge_concat_bwd(grad_out.contiguous(), idx, ctx.C)
-        return g, None, None
+        idx, order, offsets = ctx.saved_tensors
+        C = ctx.C
+        grad_out = grad_out.contiguous()
+        if order is not None:
+            B, _, K, N = grad_out.shape
+            gT = grad_out[:, :C].permute(0, 3, 2, 1).reshape(B, N * K, C).contiguous()
+            g = _EXT.gather_edge_bwd_csr(gT, order, offsets, K)
+        else:
+            g = _EXT.gather_edge_concat_bwd(grad_out, idx, C)
+        return g, None, None, None, None
 
 
 class _VoxelCorr(torch.autograd.Function):
@@ -111,6 +123,62 @@ class _KnnCorr(torch.autograd.Function):
         return g, None, None, None
 
 
+class _GroupNormAct(torch.autograd.Function):
+    """GroupNorm with optionally fused LeakyReLU (HIP, fp32/bf16 IO)."""
+
+    @staticmethod
+    def forward(ctx, x, num_groups, weight, bias, eps, act, slope):
+        w = weight.float().contiguous()
+        b = bias.float().contiguous()
+        y, mean, rstd = _EXT.group_norm_act_fwd(x, num_groups, w, b, eps, act, slope)
+        ctx.save_for_backward(x, mean, rstd, w, b)
+        ctx.conf = (num_groups, act, slope, weight.dtype)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, mean, rstd, w, b = ctx.saved_tensors
+        num_groups, act, slope, wdtype = ctx.conf
+        dx, dw, db = _EXT.group_norm_act_bwd(
+            dy.contiguous(), x, mean, rstd, num_groups, w, b, act, slope
+        )
+        return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None
+
+
+def group_norm_act(
+    x: Tensor,
+    num_groups: int,
+    weight: Tensor,
+    bias: Tensor,
+    eps: float = 1e-5,
+    act: str = "none",
+    slope: float = 0.1,
+) -> Tensor:
+    """GroupNorm over (B, C, *spatial) with optional fused LeakyReLU.
+
+    GPU: single HIP pipeline (multi-workgroup reduction; ATen's GroupNorm
+    uses one workgroup per (batch, group) which starves MI355X's 256 CUs).
+    CPU / reference mode: F.group_norm (+ activation).
+    """
+    act_id = {"none": 0, "lrelu": 1}[act]
+    if _use_hip(x):
+        shape = x.shape
+        y = _GroupNormAct.apply(
+            x.reshape(shape[0], shape[1], -1).contiguous(),
+            num_groups,
+            weight,
+            bias,
+            eps,
+            act_id,
+            slope,
+        )
+        return y.view(shape)
+    y = torch.nn.functional.group_norm(x, num_groups, weight, bias, eps)
+    if act_id == 1:
+        y = torch.nn.functional.leaky_relu(y, slope)
+    return y
+
+
 # ---------------------------------------------------------------------------
 # public functional API (model code calls these)
 # ---------------------------------------------------------------------------
@@ -124,12 +192,18 @@ def knn_graph(xyz: Tensor, k: int) -> Tensor:
     return reference.knn_idx(xyz, k)
 
 
-def gather_edge_concat(feats: Tensor, idx: Tensor, xyz: Tensor) -> Tensor:
-    """(B,N,C),(B,N,K),(B,N,3) -> (B,C+3,K,N) edge-conv input."""
+def gather_edge_concat(feats: Tensor, idx: Tensor, xyz: Tensor, csr=None) -> Tensor:
+    """(B,N,C),(B,N,K),(B,N,3) -> (B,C+3,K,N) edge-conv input.
+
+    ``csr`` = (order, offsets) from Graph.csr() selects the deterministic
+    atomic-free backward.
+    """
     feats = feats.contiguous()
     if _use_hip(feats):
+        order, offsets = csr if csr is not None else (None, None)
         return _GatherEdgeConcat.apply(
-            feats.float(), idx.to(torch.int32).contiguous(), xyz.contiguous().float()
+            feats.float(), idx.to(torch.int32).contiguous(), xyz.contiguous().float(),
+            order, offsets,
         )
     return reference.gather_edge_concat(feats, idx, xyz)
 

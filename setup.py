@@ -19,6 +19,7 @@ SOURCES = [
     "pvraft_amd/csrc/gather_edge.hip",
     "pvraft_amd/csrc/voxel_corr.hip",
     "pvraft_amd/csrc/knn_corr.hip",
+    "pvraft_amd/csrc/group_norm.hip",
 ]
 
 setup(
