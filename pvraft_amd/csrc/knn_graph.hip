@@ -1,11 +1,18 @@
 // K1: fused kNN graph construction (capability of reference
 // model/flot/graph.py:53-60, which materialises the full B x N x N distance
 // matrix and argsorts it).  Here: tiled candidate streaming through LDS +
-// per-query top-k maintained in LDS -- the N x N matrix never exists.
+// per-query top-k selection -- the N x N matrix never exists.
 //
-// Geometry: 128 threads (2 waves) per block, one query point per thread.
-// Per-thread top-k rows are padded to K+1 floats so the rescan (all threads
-// reading slot j of their own row) spreads across LDS banks.
+// Two-phase split/merge: B*N is small (~16k queries) next to 256 CUs, so a
+// one-thread-per-query kernel leaves most of the chip idle.  Phase 1 splits
+// the candidate range into SPLITS independent slices (grid x SPLITS blocks,
+// each keeping a per-query top-k of its slice); phase 2 merges the SPLITS
+// partial lists per query.  Partial lists live in a (B, N, SPLITS, k)
+// scratch tensor.
+//
+// Per-thread top-k: an LDS row of k (dist, idx) pairs, padded to k+1 so
+// the replace-worst rescan (all threads touching slot j of their own row)
+// spreads across banks.
 #include <hip/hip_runtime.h>
 #include "common.h"
 
@@ -13,17 +20,54 @@
 #define KNN_TILE 128
 #define KNN_MAXK 48  // model uses 32 (reference extractor.py:10)
 
-__global__ __launch_bounds__(KNN_THREADS) void knn_graph_kernel(
-    const float *__restrict__ xyz,  // (B, N, 3)
-    int *__restrict__ out_idx,      // (B, N, k)
-    int B, int N, int k) {
+struct TopK {
+  float *dist;
+  int *idx;
+  int filled = 0;
+  float worst = -1.f;
+  int worst_slot = 0;
+
+  __device__ void rescan(int k) {
+    worst = dist[0];
+    worst_slot = 0;
+    for (int j = 1; j < k; ++j)
+      if (dist[j] > worst) {
+        worst = dist[j];
+        worst_slot = j;
+      }
+  }
+
+  __device__ void push(float d, int i, int k) {
+    if (filled < k) {
+      dist[filled] = d;
+      idx[filled] = i;
+      if (++filled == k) rescan(k);
+    } else if (d < worst) {
+      dist[worst_slot] = d;
+      idx[worst_slot] = i;
+      rescan(k);
+    }
+  }
+};
+
+// phase 1: per-query top-k of candidate slice [split*len, ...)
+__global__ __launch_bounds__(KNN_THREADS) void knn_graph_partial_kernel(
+    const float *__restrict__ xyz,   // (B, N, 3)
+    float *__restrict__ part_dist,   // (B, N, SPLITS, k)
+    int *__restrict__ part_idx,      // (B, N, SPLITS, k)
+    int B, int N, int k, int splits) {
   __shared__ float s_tile[KNN_TILE * 3];
   __shared__ float s_dist[KNN_THREADS * (KNN_MAXK + 1)];
   __shared__ int s_idx[KNN_THREADS * (KNN_MAXK + 1)];
 
   const int b = blockIdx.y;
+  const int split = blockIdx.z;
   const int q = blockIdx.x * KNN_THREADS + threadIdx.x;
   const bool active = q < N;
+
+  const long slice_len = ((long)N + splits - 1) / splits;
+  const long lo = split * slice_len;
+  const long hi = min(lo + slice_len, (long)N);
 
   float qx = 0.f, qy = 0.f, qz = 0.f;
   if (active) {
@@ -33,17 +77,13 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_graph_kernel(
     qz = p[2];
   }
 
-  float *my_dist = s_dist + threadIdx.x * (KNN_MAXK + 1);
-  int *my_idx = s_idx + threadIdx.x * (KNN_MAXK + 1);
+  TopK top;
+  top.dist = s_dist + threadIdx.x * (KNN_MAXK + 1);
+  top.idx = s_idx + threadIdx.x * (KNN_MAXK + 1);
 
-  int filled = 0;          // slots used so far (< k during warmup)
-  float worst = -1.f;      // current k-th distance
-  int worst_slot = 0;
-
-  for (int tile = 0; tile < N; tile += KNN_TILE) {
-    const int tile_n = min(KNN_TILE, N - tile);
+  for (long tile = lo; tile < hi; tile += KNN_TILE) {
+    const int tile_n = (int)min((long)KNN_TILE, hi - tile);
     __syncthreads();
-    // cooperative stage: thread t loads candidate t of the tile
     if (threadIdx.x < tile_n) {
       const float *p = xyz + ((long)b * N + tile + threadIdx.x) * 3;
       s_tile[threadIdx.x * 3 + 0] = p[0];
@@ -52,48 +92,53 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_graph_kernel(
     }
     __syncthreads();
     if (!active) continue;
-
     for (int c = 0; c < tile_n; ++c) {
       const float dx = s_tile[c * 3 + 0] - qx;
       const float dy = s_tile[c * 3 + 1] - qy;
       const float dz = s_tile[c * 3 + 2] - qz;
-      const float d = dx * dx + dy * dy + dz * dz;
-      if (filled < k) {
-        my_dist[filled] = d;
-        my_idx[filled] = tile + c;
-        ++filled;
-        if (filled == k) {  // initial scan for the worst slot
-          worst = my_dist[0];
-          worst_slot = 0;
-          for (int j = 1; j < k; ++j)
-            if (my_dist[j] > worst) {
-              worst = my_dist[j];
-              worst_slot = j;
-            }
-        }
-      } else if (d < worst) {
-        my_dist[worst_slot] = d;
-        my_idx[worst_slot] = tile + c;
-        worst = my_dist[0];
-        worst_slot = 0;
-        for (int j = 1; j < k; ++j)
-          if (my_dist[j] > worst) {
-            worst = my_dist[j];
-            worst_slot = j;
-          }
-      }
+      top.push(dx * dx + dy * dy + dz * dz, (int)(tile + c), k);
     }
   }
 
   if (active) {
-    int *dst = out_idx + ((long)b * N + q) * k;
-    for (int j = 0; j < k; ++j) dst[j] = my_idx[j];
+    float *dd = part_dist + (((long)b * N + q) * splits + split) * k;
+    int *di = part_idx + (((long)b * N + q) * splits + split) * k;
+    for (int j = 0; j < k; ++j) {
+      dd[j] = j < top.filled ? top.dist[j] : INFINITY;
+      di[j] = j < top.filled ? top.idx[j] : -1;
+    }
   }
 }
 
-void launch_knn_graph(const float *xyz, int *out_idx, int B, int N, int k,
+// phase 2: merge the SPLITS partial lists of each query
+__global__ __launch_bounds__(KNN_THREADS) void knn_graph_merge_kernel(
+    const float *__restrict__ part_dist, const int *__restrict__ part_idx,
+    int *__restrict__ out_idx,  // (B, N, k)
+    long nq, int k, int splits) {
+  __shared__ float s_dist[KNN_THREADS * (KNN_MAXK + 1)];
+  __shared__ int s_idx[KNN_THREADS * (KNN_MAXK + 1)];
+  const long q = (long)blockIdx.x * KNN_THREADS + threadIdx.x;
+  if (q >= nq) return;
+  TopK top;
+  top.dist = s_dist + threadIdx.x * (KNN_MAXK + 1);
+  top.idx = s_idx + threadIdx.x * (KNN_MAXK + 1);
+  const float *dd = part_dist + q * splits * k;
+  const int *di = part_idx + q * splits * k;
+  for (int t = 0; t < splits * k; ++t)
+    if (di[t] >= 0) top.push(dd[t], di[t], k);
+  int *dst = out_idx + q * k;
+  for (int j = 0; j < k; ++j) dst[j] = top.idx[j];
+}
+
+void launch_knn_graph(const float *xyz, float *part_dist, int *part_idx,
+                      int *out_idx, int B, int N, int k, int splits,
                       hipStream_t stream) {
-  dim3 grid((N + KNN_THREADS - 1) / KNN_THREADS, B);
-  hipLaunchKernelGGL(knn_graph_kernel, grid, dim3(KNN_THREADS), 0, stream,
-                     xyz, out_idx, B, N, k);
+  dim3 grid((N + KNN_THREADS - 1) / KNN_THREADS, B, splits);
+  hipLaunchKernelGGL(knn_graph_partial_kernel, grid, dim3(KNN_THREADS), 0,
+                     stream, xyz, part_dist, part_idx, B, N, k, splits);
+  const long nq = (long)B * N;
+  hipLaunchKernelGGL(knn_graph_merge_kernel,
+                     dim3((nq + KNN_THREADS - 1) / KNN_THREADS),
+                     dim3(KNN_THREADS), 0, stream, part_dist, part_idx,
+                     out_idx, nq, k, splits);
 }
