@@ -26,6 +26,13 @@ void launch_gn_fwd(const void*, void*, float*, float*, float*, const float*,
 void launch_gn_bwd(const void*, const void*, const float*, const float*,
                    const float*, const float*, float*, float*, void*, int,
                    long, long, int, int, int, float, bool, hipStream_t);
+void launch_gnmp_fwd(const void*, void*, unsigned char*, float*, float*,
+                     float*, const float*, const float*, int, long, long, int,
+                     int, int, float, int, float, bool, hipStream_t);
+void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
+                     const float*, const float*, const float*, const float*,
+                     float*, float*, void*, int, long, long, int, int, int,
+                     int, float, bool, hipStream_t);
 
 namespace {
 
@@ -210,9 +217,65 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dweight, dbias};
 }
 
+// x (B, C, K, N); GN stats over full (K, N); returns pooled
+// {y (B,C,N), argmax (B,C,N) u8, mean, rstd}
+std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
+    torch::Tensor x, int64_t G, torch::Tensor weight, torch::Tensor bias,
+    double eps, int64_t act, double slope) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4, "x must be contiguous (B,C,K,N)");
+  check_f32(weight, "weight");
+  check_f32(bias, "bias");
+  const int B = x.size(0), C = x.size(1), K = x.size(2);
+  const long N = x.size(3);
+  TORCH_CHECK(K <= 255, "maxpool K must fit u8 argmax");
+  const int rows = B * G;
+  const long row_len = (C / G) * K * N;
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto ws = torch::zeros({rows, 2}, fopt);
+  auto mean = torch::empty({rows}, fopt);
+  auto rstd = torch::empty({rows}, fopt);
+  auto y = torch::empty({B, C, N}, x.options());
+  auto am = torch::empty({B, C, N}, x.options().dtype(torch::kUInt8));
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32, "fp32/bf16 only");
+  launch_gnmp_fwd(x.data_ptr(), y.data_ptr(), am.data_ptr<unsigned char>(),
+                  ws.data_ptr<float>(), mean.data_ptr<float>(),
+                  rstd.data_ptr<float>(), weight.data_ptr<float>(),
+                  bias.data_ptr<float>(), rows, row_len, N, K, C, (int)G,
+                  (float)eps, (int)act, (float)slope, bf16, stream());
+  return {y, am, mean, rstd};
+}
+
+std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
+    torch::Tensor dy, torch::Tensor x, torch::Tensor am, torch::Tensor mean,
+    torch::Tensor rstd, int64_t G, torch::Tensor weight, torch::Tensor bias,
+    int64_t act, double slope) {
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous(), "dy/x must be contiguous");
+  const int B = x.size(0), C = x.size(1), K = x.size(2);
+  const long N = x.size(3);
+  const int rows = B * G;
+  const long row_len = (C / G) * K * N;
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto row_ws = torch::zeros({rows, 2}, fopt);
+  auto chan_ws = torch::zeros({C, 2}, fopt);
+  auto dx = torch::empty_like(x);
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  launch_gnmp_bwd(dy.data_ptr(), x.data_ptr(), am.data_ptr<unsigned char>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  weight.data_ptr<float>(), bias.data_ptr<float>(),
+                  row_ws.data_ptr<float>(), chan_ws.data_ptr<float>(),
+                  dx.data_ptr(), rows, row_len, N, K, C, (int)G, (int)act,
+                  (float)slope, bf16, stream());
+  auto dbias = chan_ws.select(1, 0).contiguous();
+  auto dweight = chan_ws.select(1, 1).contiguous();
+  return {dx, dweight, dbias};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_act_fwd", &group_norm_act_fwd);
   m.def("group_norm_act_bwd", &group_norm_act_bwd);
+  m.def("group_norm_act_maxpool_fwd", &group_norm_act_maxpool_fwd);
+  m.def("group_norm_act_maxpool_bwd", &group_norm_act_maxpool_bwd);
   m.def("knn_graph", &knn_graph, "fused kNN graph (CDNA4)");
   m.def("gather_edge_concat_fwd", &gather_edge_concat_fwd);
   m.def("gather_edge_concat_bwd", &gather_edge_concat_bwd);

@@ -197,6 +197,152 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
   }
 }
 
+// ------------------------------------------------- GN + act + max-pool(K)
+//
+// SetConv stage 1 is GN -> LeakyReLU -> max over the K neighbour axis
+// (reference gconv.py:71-75).  Fusing the pool into the normalize pass
+// avoids materialising the (B, C, K, N) activation (208 MB at the default
+// config) and the separate pool forward/backward kernels.  Statistics are
+// over the FULL (K, N) spatial extent (pre-pool, matching the reference);
+// the fwd stats passes above are reused unchanged.
+//
+// y[b,c,n] = max_k act(xhat[b,c,k,n]*gamma[c]+beta[c]), argmax saved (u8).
+// Backward: dy lives on the pooled (B,C,N) domain; dxhat is nonzero only
+// at argmax positions, but the GN mean/var coupling makes dx dense.
+
+template <typename T, int ACT>
+__global__ __launch_bounds__(GN_THREADS) void gnmp_fwd_apply_kernel(
+    const T *__restrict__ x, T *__restrict__ y, unsigned char *__restrict__ am,
+    const float *__restrict__ mean, const float *__restrict__ rstd,
+    const float *__restrict__ gamma, const float *__restrict__ beta, long N,
+    int K, int C, int G, long total_out, float slope) {
+  const int Cg = C / G;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_out;
+       i += (long)gridDim.x * blockDim.x) {
+    const long n = i % N;
+    const long c = i / N % C;
+    const long b = i / (N * C);
+    const long row = b * G + c / Cg;
+    const float m = mean[row];
+    const float r = rstd[row];
+    const float ga = gamma[c], be = beta[c];
+    const T *base = x + ((b * C + c) * K) * N + n;
+    float best = -INFINITY;
+    int bk = 0;
+    for (int k = 0; k < K; ++k) {
+      float v = (ld(base + (long)k * N) - m) * r * ga + be;
+      if (ACT == 1) v = v > 0.f ? v : v * slope;
+      if (v > best) {
+        best = v;
+        bk = k;
+      }
+    }
+    st(y + i, best);
+    am[i] = (unsigned char)bk;
+  }
+}
+
+// backward pass 1: row/channel sums over the POOLED domain (selected
+// positions only carry dxhat / dgamma / dbeta)
+template <typename T, int ACT>
+__global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_reduce_kernel(
+    const T *__restrict__ dy, const T *__restrict__ x,
+    const unsigned char *__restrict__ am, const float *__restrict__ mean,
+    const float *__restrict__ rstd, const float *__restrict__ gamma,
+    const float *__restrict__ beta, float *__restrict__ row_ws,
+    float *__restrict__ chan_ws, long N, int K, int C, int G, long total_out,
+    float slope) {
+  const int Cg = C / G;
+  // contiguous chunking over the pooled domain so channel runs are long
+  const long per_block = (total_out + gridDim.x - 1) / gridDim.x;
+  const long lo = (long)blockIdx.x * per_block;
+  const long hi = min(lo + per_block, total_out);
+  float sum_dx = 0.f, sum_dxx = 0.f;
+  long cur_c = -1, cur_row = -1;
+  float c_dg = 0.f, c_db = 0.f;
+  for (long i = lo + threadIdx.x; i < hi; i += GN_THREADS) {
+    const long n = i % N;
+    const long c = i / N % C;
+    const long b = i / (N * C);
+    const long row = b * G + c / Cg;
+    if (c != cur_c) {
+      if (cur_c >= 0) {
+        atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
+        atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
+      }
+      cur_c = c;
+      c_dg = 0.f;
+      c_db = 0.f;
+    }
+    if (row != cur_row) {
+      if (cur_row >= 0) {
+        atomicAdd(&row_ws[cur_row * 2 + 0], sum_dx);
+        atomicAdd(&row_ws[cur_row * 2 + 1], sum_dxx);
+      }
+      cur_row = row;
+      sum_dx = 0.f;
+      sum_dxx = 0.f;
+    }
+    const int k = am[i];
+    const float xv = ld(x + ((b * C + c) * K + k) * N + n);
+    const float xhat = (xv - mean[row]) * rstd[row];
+    float g = ld(dy + i);
+    if (ACT == 1) {
+      const float pre = xhat * gamma[c] + beta[c];
+      g = pre > 0.f ? g : g * slope;
+    }
+    c_db += g;
+    c_dg += g * xhat;
+    const float dxhat = g * gamma[c];
+    sum_dx += dxhat;
+    sum_dxx += dxhat * xhat;
+  }
+  if (cur_c >= 0) {
+    atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
+    atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
+  }
+  if (cur_row >= 0) {
+    atomicAdd(&row_ws[cur_row * 2 + 0], sum_dx);
+    atomicAdd(&row_ws[cur_row * 2 + 1], sum_dxx);
+  }
+}
+
+// backward pass 2: dense dx over the full (B, C, K, N) domain
+template <typename T, int ACT>
+__global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
+    const T *__restrict__ dy, const T *__restrict__ x,
+    const unsigned char *__restrict__ am, const float *__restrict__ mean,
+    const float *__restrict__ rstd, const float *__restrict__ gamma,
+    const float *__restrict__ beta, const float *__restrict__ row_ws,
+    T *__restrict__ dx, long N, int K, int C, int G, long row_len,
+    long total, float slope) {
+  const int Cg = C / G;
+  const float inv_n = 1.0f / (float)row_len;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long n = i % N;
+    const long k = i / N % K;
+    const long c = i / (N * K) % C;
+    const long b = i / (N * K * C);
+    const long row = b * G + c / Cg;
+    const float r = rstd[row];
+    const float xhat = (ld(x + i) - mean[row]) * r;
+    const long pooled = (b * C + c) * N + n;
+    float dxhat = 0.f;
+    if ((int)k == (int)am[pooled]) {
+      float g = ld(dy + pooled);
+      if (ACT == 1) {
+        const float pre = xhat * gamma[c] + beta[c];
+        g = pre > 0.f ? g : g * slope;
+      }
+      dxhat = g * gamma[c];
+    }
+    const float v =
+        (dxhat - (row_ws[row * 2 + 0] + xhat * row_ws[row * 2 + 1]) * inv_n) * r;
+    st(dx + i, v);
+  }
+}
+
 // --------------------------------------------------------------- launchers
 
 static int pick_blocks_per_row(long row_len, int rows) {
@@ -252,7 +398,91 @@ void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
   }
 }
 
+template <typename T>
+void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
+                   float *rstd, const float *gamma, const float *beta, int rows,
+                   long row_len, long N, int K, int C, int G, float eps,
+                   int act, float slope, hipStream_t stream) {
+  const int B = rows / G;
+  const long total_out = (long)B * C * N;
+  const int bpr = pick_blocks_per_row(row_len, rows);
+  hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, dim3(rows * bpr),
+                     dim3(GN_THREADS), 0, stream, x, ws, row_len, rows, bpr);
+  hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
+                     0, stream, ws, mean, rstd, row_len, rows, eps);
+  const int blocks = (int)min((total_out + GN_THREADS - 1) / GN_THREADS, (long)65535);
+  if (act == 1)
+    hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, 1>), dim3(blocks),
+                       dim3(GN_THREADS), 0, stream, x, y, am, mean, rstd,
+                       gamma, beta, N, K, C, G, total_out, slope);
+  else
+    hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, 0>), dim3(blocks),
+                       dim3(GN_THREADS), 0, stream, x, y, am, mean, rstd,
+                       gamma, beta, N, K, C, G, total_out, slope);
+}
+
+template <typename T>
+void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
+                   const float *mean, const float *rstd, const float *gamma,
+                   const float *beta, float *row_ws, float *chan_ws, T *dx,
+                   int rows, long row_len, long N, int K, int C, int G,
+                   int act, float slope, hipStream_t stream) {
+  const int B = rows / G;
+  const long total_out = (long)B * C * N;
+  const long total = (long)rows * row_len;
+  const int rblocks = (int)min((total_out + GN_THREADS - 1) / GN_THREADS, (long)4096);
+  const int ablocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
+  if (act == 1) {
+    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 1>), dim3(rblocks),
+                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
+                       gamma, beta, row_ws, chan_ws, N, K, C, G, total_out, slope);
+    hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, 1>), dim3(ablocks),
+                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
+                       gamma, beta, row_ws, dx, N, K, C, G, row_len, total, slope);
+  } else {
+    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 0>), dim3(rblocks),
+                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
+                       gamma, beta, row_ws, chan_ws, N, K, C, G, total_out, slope);
+    hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, 0>), dim3(ablocks),
+                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
+                       gamma, beta, row_ws, dx, N, K, C, G, row_len, total, slope);
+  }
+}
+
 // type-erased entry points (bindings.cpp is host-compiled, no HIP types)
+void launch_gnmp_fwd(const void *x, void *y, unsigned char *am, float *ws,
+                     float *mean, float *rstd, const float *gamma,
+                     const float *beta, int rows, long row_len, long N, int K,
+                     int C, int G, float eps, int act, float slope, bool bf16,
+                     hipStream_t stream) {
+  if (bf16)
+    gnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)x,
+                                  (__hip_bfloat16 *)y, am, ws, mean, rstd,
+                                  gamma, beta, rows, row_len, N, K, C, G, eps,
+                                  act, slope, stream);
+  else
+    gnmp_fwd_impl<float>((const float *)x, (float *)y, am, ws, mean, rstd,
+                         gamma, beta, rows, row_len, N, K, C, G, eps, act,
+                         slope, stream);
+}
+
+void launch_gnmp_bwd(const void *dy, const void *x, const unsigned char *am,
+                     const float *mean, const float *rstd, const float *gamma,
+                     const float *beta, float *row_ws, float *chan_ws,
+                     void *dx, int rows, long row_len, long N, int K, int C,
+                     int G, int act, float slope, bool bf16,
+                     hipStream_t stream) {
+  if (bf16)
+    gnmp_bwd_impl<__hip_bfloat16>(
+        (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, am, mean, rstd,
+        gamma, beta, row_ws, chan_ws, (__hip_bfloat16 *)dx, rows, row_len, N,
+        K, C, G, act, slope, stream);
+  else
+    gnmp_bwd_impl<float>((const float *)dy, (const float *)x, am, mean, rstd,
+                         gamma, beta, row_ws, chan_ws, (float *)dx, rows,
+                         row_len, N, K, C, G, act, slope, stream);
+}
+
 void launch_gn_fwd(const void *x, void *y, float *ws, float *mean, float *rstd,
                    const float *gamma, const float *beta, int rows,
                    long row_len, long S, int C, int G, float eps, int act,

@@ -32,7 +32,8 @@ import torch.nn as nn
 from torch import Tensor
 
 from pvraft_amd import ops
-from .norm import FusedGroupNorm
+from .norm import FusedGroupNorm, PReLUAny
+from .pointwise import PwConv1d, PwConv2d
 
 
 @dataclass
@@ -57,17 +58,17 @@ class CorrBlock(nn.Module):
         self.base_scale = base_scale
         self.knn = knn
         self.out_conv = nn.Sequential(
-            nn.Conv1d((resolution ** 3) * num_levels, 128, 1),
+            PwConv1d((resolution ** 3) * num_levels, 128, 1),
             FusedGroupNorm(8, 128),
-            nn.PReLU(),
-            nn.Conv1d(128, 64, 1),
+            PReLUAny(),
+            PwConv1d(128, 64, 1),
         )
         self.knn_conv = nn.Sequential(
-            nn.Conv2d(4, 64, 1),
+            PwConv2d(4, 64, 1),
             FusedGroupNorm(8, 64),
-            nn.PReLU(),
+            PReLUAny(),
         )
-        self.knn_out = nn.Conv1d(64, 64, 1)
+        self.knn_out = PwConv1d(64, 64, 1)
 
     def build(self, fmap1: Tensor, fmap2: Tensor, xyz2: Tensor) -> CorrField:
         """Compute the truncated correlation field (once per pair)."""

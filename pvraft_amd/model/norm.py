@@ -14,6 +14,20 @@ from torch import Tensor
 from pvraft_amd import ops
 
 
+class PReLUAny(nn.PReLU):
+    """nn.PReLU that follows the input dtype.
+
+    ATen's GroupNorm always upcasts to fp32 under autocast, so the
+    reference never feeds PReLU bf16; FusedGroupNorm keeps bf16, and
+    F.prelu refuses mixed dtypes -- cast the (fp32) slope to the input.
+    """
+
+    def forward(self, x: Tensor) -> Tensor:
+        import torch.nn.functional as F
+
+        return F.prelu(x, self.weight.to(x.dtype))
+
+
 class FusedGroupNorm(nn.GroupNorm):
     def __init__(self, num_groups: int, num_channels: int, act: str = "none",
                  slope: float = 0.1, **kw):

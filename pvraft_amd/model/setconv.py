@@ -21,25 +21,34 @@ from torch import Tensor
 from pvraft_amd import ops
 from .graph import Graph
 from .norm import FusedGroupNorm
+from .pointwise import PwConv1d, PwConv2d
 
 
 class SetConv(nn.Module):
     def __init__(self, in_ch: int, out_ch: int):
         super().__init__()
         mid = out_ch // 2 if in_ch % 2 != 0 else (out_ch + in_ch) // 2
-        self.fc1 = nn.Conv2d(in_ch + 3, mid, 1, bias=False)
+        self.fc1 = PwConv2d(in_ch + 3, mid, 1, bias=False)
         self.gn1 = FusedGroupNorm(8, mid, act="lrelu", slope=0.1)
-        self.fc2 = nn.Conv1d(mid, out_ch, 1, bias=False)
+        self.fc2 = PwConv1d(mid, out_ch, 1, bias=False)
         self.gn2 = FusedGroupNorm(8, out_ch, act="lrelu", slope=0.1)
-        self.fc3 = nn.Conv1d(out_ch, out_ch, 1, bias=False)
+        self.fc3 = PwConv1d(out_ch, out_ch, 1, bias=False)
         self.gn3 = FusedGroupNorm(8, out_ch, act="lrelu", slope=0.1)
 
     def forward(self, feats: Tensor, graph: Graph) -> Tensor:
-        """feats (B, N, C) -> (B, N, out_ch).  LeakyReLU(0.1) is fused into
-        each GroupNorm (reference order gconv.py:71-83: conv -> GN -> lrelu)."""
+        """feats (B, N, C) -> (B, N, out_ch).
+
+        Reference order gconv.py:71-83: conv -> GN -> lrelu (-> max over
+        neighbours after stage 1).  LeakyReLU is fused into each GroupNorm;
+        stage 1 additionally fuses the neighbour max-pool, so the
+        (B, C, K, N) activation never materialises post-conv.
+        """
         x = ops.gather_edge_concat(feats, graph.idx, graph.xyz, csr=graph.csr())  # B, C+3, K, N
-        x = self.gn1(self.fc1(x))
-        x = x.max(dim=2)[0]  # max-pool over neighbours -> B, mid, N
+        x = self.fc1(x)
+        x = ops.group_norm_act_maxpool(
+            x, self.gn1.num_groups, self.gn1.weight, self.gn1.bias,
+            self.gn1.eps, act="lrelu", slope=0.1,
+        )  # B, mid, N
         x = self.gn2(self.fc2(x))
         x = self.gn3(self.fc3(x))
         return x.transpose(1, 2)

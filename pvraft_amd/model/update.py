@@ -15,15 +15,16 @@ import torch.nn.functional as F
 from torch import Tensor
 
 from .graph import Graph
+from .pointwise import PwConv1d
 from .setconv import SetConv
 
 
 class MotionEncoder(nn.Module):
     def __init__(self):
         super().__init__()
-        self.conv_corr = nn.Conv1d(64, 64, 1)
-        self.conv_flow = nn.Conv1d(3, 64, 1)
-        self.conv = nn.Conv1d(64 + 64, 64 - 3, 1)
+        self.conv_corr = PwConv1d(64, 64, 1)
+        self.conv_flow = PwConv1d(3, 64, 1)
+        self.conv = PwConv1d(64 + 64, 64 - 3, 1)
 
     def forward(self, flow: Tensor, corr: Tensor) -> Tensor:
         """flow (B, N, 3), corr (B, 64, N) -> (B, 64, N)."""
@@ -37,9 +38,9 @@ class MotionEncoder(nn.Module):
 class ConvGRU(nn.Module):
     def __init__(self, input_dim: int = 128, hidden_dim: int = 64):
         super().__init__()
-        self.convz = nn.Conv1d(input_dim + hidden_dim, hidden_dim, 1)
-        self.convr = nn.Conv1d(input_dim + hidden_dim, hidden_dim, 1)
-        self.convq = nn.Conv1d(input_dim + hidden_dim, hidden_dim, 1)
+        self.convz = PwConv1d(input_dim + hidden_dim, hidden_dim, 1)
+        self.convr = PwConv1d(input_dim + hidden_dim, hidden_dim, 1)
+        self.convq = PwConv1d(input_dim + hidden_dim, hidden_dim, 1)
 
     def forward(self, h: Tensor, x: Tensor) -> Tensor:
         hx = torch.cat([h, x], dim=1)
@@ -55,8 +56,8 @@ class ConvRNN(nn.Module):
 
     def __init__(self, input_dim: int = 128, hidden_dim: int = 64):
         super().__init__()
-        self.convx = nn.Conv1d(input_dim, hidden_dim, 1)
-        self.convh = nn.Conv1d(hidden_dim, hidden_dim, 1)
+        self.convx = PwConv1d(input_dim, hidden_dim, 1)
+        self.convh = PwConv1d(hidden_dim, hidden_dim, 1)
 
     def forward(self, h: Tensor, x: Tensor) -> Tensor:
         return torch.tanh(self.convx(x) + self.convh(h))
@@ -65,12 +66,12 @@ class ConvRNN(nn.Module):
 class FlowHead(nn.Module):
     def __init__(self, input_dim: int = 64):
         super().__init__()
-        self.conv1 = nn.Conv1d(input_dim, 64, 1)
+        self.conv1 = PwConv1d(input_dim, 64, 1)
         self.setconv = SetConv(64, 64)
         self.out_conv = nn.Sequential(
-            nn.Conv1d(128, 64, 1),
+            PwConv1d(128, 64, 1),
             nn.ReLU(),
-            nn.Conv1d(64, 3, 1),
+            PwConv1d(64, 3, 1),
         )
 
     def forward(self, x: Tensor, graph: Graph) -> Tensor:

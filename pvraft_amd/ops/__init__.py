@@ -179,6 +179,53 @@ def group_norm_act(
     return y
 
 
+class _GroupNormActMaxpool(torch.autograd.Function):
+    """GroupNorm + activation + max-pool over the K axis, fused (HIP).
+
+    Stats over the full (K, N) spatial extent; output is the pooled
+    (B, C, N) tensor -- the (B, C, K, N) activation never hits HBM.
+    """
+
+    @staticmethod
+    def forward(ctx, x, num_groups, weight, bias, eps, act, slope):
+        w = weight.float().contiguous()
+        b = bias.float().contiguous()
+        y, am, mean, rstd = _EXT.group_norm_act_maxpool_fwd(x, num_groups, w, b, eps, act, slope)
+        ctx.save_for_backward(x, am, mean, rstd, w, b)
+        ctx.conf = (num_groups, act, slope, weight.dtype)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, am, mean, rstd, w, b = ctx.saved_tensors
+        num_groups, act, slope, wdtype = ctx.conf
+        dx, dw, db = _EXT.group_norm_act_maxpool_bwd(
+            dy.contiguous(), x, am, mean, rstd, num_groups, w, b, act, slope
+        )
+        return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None
+
+
+def group_norm_act_maxpool(
+    x: Tensor,
+    num_groups: int,
+    weight: Tensor,
+    bias: Tensor,
+    eps: float = 1e-5,
+    act: str = "lrelu",
+    slope: float = 0.1,
+) -> Tensor:
+    """(B, C, K, N) -> (B, C, N): GroupNorm -> activation -> max over K."""
+    act_id = {"none": 0, "lrelu": 1}[act]
+    if _use_hip(x):
+        return _GroupNormActMaxpool.apply(
+            x.contiguous(), num_groups, weight, bias, eps, act_id, slope
+        )
+    y = torch.nn.functional.group_norm(x, num_groups, weight, bias, eps)
+    if act_id == 1:
+        y = torch.nn.functional.leaky_relu(y, slope)
+    return y.max(dim=2)[0]
+
+
 # ---------------------------------------------------------------------------
 # public functional API (model code calls these)
 # ---------------------------------------------------------------------------

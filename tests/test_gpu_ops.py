@@ -173,3 +173,34 @@ def test_gather_edge_csr_backward_matches_atomic():
     out_ref.backward(go)
     assert torch.allclose(feats.grad, feats3.grad, atol=1e-3), (feats.grad - feats3.grad).abs().max()
     assert torch.allclose(feats2.grad, feats3.grad, atol=1e-3)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_group_norm_act_maxpool_fwd_bwd(dtype):
+    import torch.nn.functional as F
+
+    B, C, K, N, G = 2, 96, 32, 555, 8
+    x = torch.randn(B, C, K, N, device=dev(), dtype=dtype, requires_grad=True)
+    w = torch.randn(C, device=dev(), requires_grad=True)
+    b = torch.randn(C, device=dev(), requires_grad=True)
+    xr = x.detach().clone().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+
+    y = ops.group_norm_act_maxpool(x, G, w, b, 1e-5, act="lrelu", slope=0.1)
+    y_ref = F.leaky_relu(F.group_norm(xr.float(), G, wr, br, 1e-5), 0.1).max(dim=2)[0]
+    atol = 1e-4 if dtype == torch.float32 else 5e-2
+    assert y.shape == (B, C, N)
+    assert torch.allclose(y.float(), y_ref.to(y.dtype).float(), atol=atol), (
+        (y.float() - y_ref.to(y.dtype).float()).abs().max()
+    )
+
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dtype))
+    y_ref.backward(g)
+    gatol = 1e-3 if dtype == torch.float32 else 1e-1
+    assert torch.allclose(x.grad.float(), xr.grad.float(), atol=gatol), (
+        (x.grad.float() - xr.grad.float()).abs().max()
+    )
+    assert torch.allclose(w.grad, wr.grad, rtol=2e-2, atol=2e-1)
+    assert torch.allclose(b.grad, br.grad, rtol=2e-2, atol=2e-1)
