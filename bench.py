@@ -41,6 +41,8 @@ def parse_args():
     p.add_argument("--iters", type=int, default=8)
     p.add_argument("--truncate_k", type=int, default=512)
     p.add_argument("--no-amp", dest="amp", action="store_false")
+    p.add_argument("--no-graph", dest="graph", action="store_false",
+                   help="disable hipGraph capture of the train step")
     p.add_argument("--device", type=str, default=None, help="force device (cpu for plumbing tests)")
     return p.parse_args()
 
@@ -74,17 +76,34 @@ def main():
     optimizer = torch.optim.Adam(model.parameters(), lr=1e-3)
     batch = synthetic_batch(args.batch, args.points, device=device, seed=100 + info.rank)
 
-    def step():
-        reducer.zero_grad()
-        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=amp):
-            est_flow = model(batch["sequence"], num_iters=args.iters)
-            loss = sequence_loss(est_flow, batch, gamma=0.8)
-        loss.backward()
-        reducer.finalize()
-        optimizer.step()
-        return loss
-
     model.train()
+    use_graph = args.graph and cuda
+    if use_graph:
+        from pvraft_amd.engine.graphed import build_graphed_step
+
+        reducer.hooks_enabled = False
+        graphed = build_graphed_step(
+            model, batch, num_iters=args.iters, gamma=0.8, reducer=reducer, amp=amp
+        )
+
+        def step():
+            loss = graphed.replay()
+            reducer.reduce_all()
+            optimizer.step()
+            return loss
+
+    else:
+
+        def step():
+            reducer.zero_grad()
+            with torch.autocast("cuda", dtype=torch.bfloat16, enabled=amp):
+                est_flow = model(batch["sequence"], num_iters=args.iters)
+                loss = sequence_loss(est_flow, batch, gamma=0.8)
+            loss.backward()
+            reducer.finalize()
+            optimizer.step()
+            return loss
+
     for _ in range(args.warmup):
         step()
 
@@ -133,6 +152,7 @@ def main():
                         "gru_iters": args.iters,
                         "truncate_k": args.truncate_k,
                         "parallelism": f"dp{info.world_size}",
+                        "hipgraph": use_graph,
                     },
                 }
             )

@@ -168,11 +168,29 @@ class GradReducer:
             b.pending = len(b.params)
             b.work = None
 
+    hooks_enabled: bool = True
+
     def _on_grad(self, p: torch.nn.Parameter) -> None:
+        if not self.hooks_enabled:
+            return
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0 and self.enabled:
             b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM, group=self.pg, async_op=True)
+
+    def reduce_all(self) -> None:
+        """Eager mean-all-reduce of every bucket (hipGraph mode: the
+        backward ran inside a captured graph, hooks were disabled)."""
+        if not self.enabled:
+            return
+        world = dist.get_world_size()
+        works = [
+            dist.all_reduce(b.flat, op=dist.ReduceOp.SUM, group=self.pg, async_op=True)
+            for b in self.buckets
+        ]
+        for w, b in zip(works, self.buckets):
+            w.wait()
+            b.flat /= world
 
     def finalize(self) -> None:
         world = dist.get_world_size() if self.enabled else 1
