@@ -5,12 +5,12 @@
 
 void launch_knn_graph(const float*, float*, int*, int*, int, int, int, int,
                       hipStream_t);
-void launch_gather_edge_fwd(const float*, const int*, const float*, float*,
-                            int, int, int, int, hipStream_t);
-void launch_gather_edge_bwd(const float*, const int*, float*, int, int, int,
-                            int, hipStream_t);
-void launch_gather_edge_bwd_csr(const float*, const int*, const int*, float*,
-                                int, int, int, int, hipStream_t);
+void launch_gather_edge_fwd(const void*, const int*, const float*, void*,
+                            int, int, int, int, bool, hipStream_t);
+void launch_gather_edge_bwd(const void*, const int*, float*, int, int, int,
+                            int, bool, hipStream_t);
+void launch_gather_edge_bwd_csr(const void*, const int*, const int*, void*,
+                                int, int, int, int, bool, hipStream_t);
 void launch_voxel_corr_fwd(const float*, const float*, const float*, float*,
                            int, int, int, int, float, hipStream_t);
 void launch_voxel_corr_bwd(const float*, const float*, const float*, float*,
@@ -66,39 +66,45 @@ torch::Tensor knn_graph(torch::Tensor xyz, int64_t k) {
 
 torch::Tensor gather_edge_concat_fwd(torch::Tensor feats, torch::Tensor idx,
                                      torch::Tensor xyz) {
-  check_f32(feats, "feats");
+  TORCH_CHECK(feats.is_cuda() && feats.is_contiguous(), "feats must be contiguous GPU");
   check_f32(xyz, "xyz");
   TORCH_CHECK(idx.scalar_type() == torch::kInt32 && idx.is_contiguous(), "idx must be contiguous int32");
+  const bool bf16 = feats.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || feats.scalar_type() == torch::kFloat32, "fp32/bf16 only");
   const int B = feats.size(0), N = feats.size(1), C = feats.size(2), K = idx.size(2);
   auto out = torch::empty({B, C + 3, K, N}, feats.options());
-  launch_gather_edge_fwd(feats.data_ptr<float>(), idx.data_ptr<int>(),
-                         xyz.data_ptr<float>(), out.data_ptr<float>(), B, N, K,
-                         C, stream());
+  launch_gather_edge_fwd(feats.data_ptr(), idx.data_ptr<int>(),
+                         xyz.data_ptr<float>(), out.data_ptr(), B, N, K, C,
+                         bf16, stream());
   return out;
 }
 
 torch::Tensor gather_edge_concat_bwd(torch::Tensor gout, torch::Tensor idx,
                                      int64_t C) {
-  check_f32(gout, "gout");
+  TORCH_CHECK(gout.is_cuda() && gout.is_contiguous(), "gout must be contiguous GPU");
+  const bool bf16 = gout.scalar_type() == torch::kBFloat16;
   const int B = gout.size(0), K = gout.size(2), N = gout.size(3);
   TORCH_CHECK(gout.size(1) == C + 3, "gout channel mismatch");
-  auto gfeats = torch::empty({B, N, C}, gout.options());
-  launch_gather_edge_bwd(gout.data_ptr<float>(), idx.data_ptr<int>(),
-                         gfeats.data_ptr<float>(), B, N, K, (int)C, stream());
-  return gfeats;
+  auto gfeats = torch::empty({B, N, C}, gout.options().dtype(torch::kFloat32));
+  launch_gather_edge_bwd(gout.data_ptr(), idx.data_ptr<int>(),
+                         gfeats.data_ptr<float>(), B, N, K, (int)C, bf16,
+                         stream());
+  return bf16 ? gfeats.to(torch::kBFloat16) : gfeats;
 }
 
 torch::Tensor gather_edge_bwd_csr(torch::Tensor gT, torch::Tensor order,
                                   torch::Tensor offsets, int64_t K) {
-  check_f32(gT, "gT");
+  TORCH_CHECK(gT.is_cuda() && gT.is_contiguous(), "gT must be contiguous GPU");
+  const bool bf16 = gT.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || gT.scalar_type() == torch::kFloat32, "fp32/bf16 only");
   TORCH_CHECK(order.scalar_type() == torch::kInt32 && order.is_contiguous());
   TORCH_CHECK(offsets.scalar_type() == torch::kInt32 && offsets.is_contiguous());
   const int B = gT.size(0), NK = gT.size(1), C = gT.size(2);
   const int N = NK / (int)K;
   auto grad = torch::empty({B, N, C}, gT.options());
-  launch_gather_edge_bwd_csr(gT.data_ptr<float>(), order.data_ptr<int>(),
-                             offsets.data_ptr<int>(), grad.data_ptr<float>(),
-                             B, N, (int)K, C, stream());
+  launch_gather_edge_bwd_csr(gT.data_ptr(), order.data_ptr<int>(),
+                             offsets.data_ptr<int>(), grad.data_ptr(), B, N,
+                             (int)K, C, bf16, stream());
   return grad;
 }
 

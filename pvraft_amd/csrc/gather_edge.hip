@@ -5,21 +5,33 @@
 //   out[b, C+c, j, n] = xyz  [b, idx[b,n,j], c] - xyz  [b, n, c]   (c < 3)
 // Thread layout: one thread per (b, j, n) with n fastest, so every store
 // for a fixed (c, j) is wave-coalesced over n; the neighbour-row reads are
-// inherently scattered (kNN gather) and ride L2.
+// inherently scattered (kNN gather) and ride L2.  fp32 or bf16 IO (the
+// dominant tensor of the encoder under bf16 autocast); positions stay fp32.
 //
 // Backward (w.r.t. feats only; graph positions carry no gradient in
-// PV-RAFT):
-//   g[b, n, c] = -sum_j gout[b, c, j, n] + sum_{(n',j): idx[n',j]==n} gout[b, c, j, n']
-// Kernel A writes the centre term (unique writes), kernel B scatters the
-// neighbour term with fp32 global atomics.
+// PV-RAFT) -- two paths:
+//  * CSR (preferred): deterministic, atomic-free; see
+//    gather_edge_bwd_csr_kernel below.
+//  * atomic fallback: centre-term kernel + fp32 global atomics scatter.
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include "common.h"
 
+template <typename T>
+DEV_INLINE float ldf(const T *p) {
+  return (float)*p;
+}
+template <typename T>
+DEV_INLINE void stf(T *p, float v) {
+  *p = (T)v;
+}
+
+template <typename T>
 __global__ void gather_edge_fwd_kernel(
-    const float *__restrict__ feats,  // (B, N, C)
+    const T *__restrict__ feats,      // (B, N, C)
     const int *__restrict__ idx,      // (B, N, K)
     const float *__restrict__ xyz,    // (B, N, 3)
-    float *__restrict__ out,          // (B, C+3, K, N)
+    T *__restrict__ out,              // (B, C+3, K, N)
     int B, int N, int K, int C) {
   const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long total = (long)B * K * N;
@@ -29,70 +41,33 @@ __global__ void gather_edge_fwd_kernel(
   const int b = (int)(gid / ((long)N * K));
 
   const int nb = idx[((long)b * N + n) * K + j];
-  const float *f_nb = feats + ((long)b * N + nb) * C;
-  const float *f_ct = feats + ((long)b * N + n) * C;
-  float *dst = out + (((long)b * (C + 3)) * K + j) * N + n;
+  const T *f_nb = feats + ((long)b * N + nb) * C;
+  const T *f_ct = feats + ((long)b * N + n) * C;
+  T *dst = out + (((long)b * (C + 3)) * K + j) * N + n;
   const long cstride = (long)K * N;
-  for (int c = 0; c < C; ++c) dst[c * cstride] = f_nb[c] - f_ct[c];
+  for (int c = 0; c < C; ++c)
+    stf(dst + c * cstride, ldf(f_nb + c) - ldf(f_ct + c));
 
   const float *x_nb = xyz + ((long)b * N + nb) * 3;
   const float *x_ct = xyz + ((long)b * N + n) * 3;
   dst += (long)C * cstride;
-  for (int c = 0; c < 3; ++c) dst[c * cstride] = x_nb[c] - x_ct[c];
-}
-
-// centre term: g[b,n,c] = -sum_j gout[b,c,j,n]; one thread per (b,c,n)
-__global__ void gather_edge_bwd_center_kernel(
-    const float *__restrict__ gout,  // (B, C+3, K, N)
-    float *__restrict__ gfeats,      // (B, N, C)
-    int B, int N, int K, int C) {
-  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long total = (long)B * C * N;
-  if (gid >= total) return;
-  const int n = (int)(gid % N);
-  const int c = (int)((gid / N) % C);
-  const int b = (int)(gid / ((long)N * C));
-
-  const float *src = gout + (((long)b * (C + 3) + c) * K) * N + n;
-  float acc = 0.f;
-  for (int j = 0; j < K; ++j) acc += src[(long)j * N];
-  gfeats[((long)b * N + n) * C + c] = -acc;
-}
-
-// neighbour term: scatter-add, one thread per (b,j,n) looping channels
-__global__ void gather_edge_bwd_scatter_kernel(
-    const float *__restrict__ gout,  // (B, C+3, K, N)
-    const int *__restrict__ idx,     // (B, N, K)
-    float *__restrict__ gfeats,      // (B, N, C)
-    int B, int N, int K, int C) {
-  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long total = (long)B * K * N;
-  if (gid >= total) return;
-  const int n = (int)(gid % N);
-  const int j = (int)((gid / N) % K);
-  const int b = (int)(gid / ((long)N * K));
-
-  const int nb = idx[((long)b * N + n) * K + j];
-  const float *src = gout + (((long)b * (C + 3)) * K + j) * N + n;
-  float *dst = gfeats + ((long)b * N + nb) * C;
-  const long cstride = (long)K * N;
-  for (int c = 0; c < C; ++c) atomicAdd(dst + c, src[c * cstride]);
+  for (int c = 0; c < 3; ++c) stf(dst + c * cstride, x_nb[c] - x_ct[c]);
 }
 
 // CSR backward: deterministic, atomic-free.  The caller transposes the
-// first C channels of gout to gT (B, N, K, C) (so each edge's gradient is
-// a contiguous C-vector) and provides the inverse adjacency in CSR form:
+// first C channels of gout to gT (B, N*K, C) (each edge's gradient is a
+// contiguous C-vector) and provides the inverse adjacency in CSR form:
 // order (B, N*K) = edge ids sorted by target node, offsets (B, N+1).
 //   grad[b, m, c] = sum_{t in [off[m], off[m+1])} gT[b, order[t], c]
 //                 - sum_j gT[b, m*K + j, c]
 // Thread = (b, m, c) with c fastest: both loops read contiguous C-vectors,
-// wave-coalesced; writes are unique.  Replaces the fp32-atomic scatter
-// (measured 15% of the train step).
+// wave-coalesced; writes are unique.
+template <typename T>
 __global__ void gather_edge_bwd_csr_kernel(
-    const float *__restrict__ gT,      // (B, N*K, C)
+    const T *__restrict__ gT,          // (B, N*K, C)
     const int *__restrict__ order,     // (B, N*K)
     const int *__restrict__ offsets,   // (B, N+1)
-    float *__restrict__ grad,          // (B, N, C)
+    T *__restrict__ grad,              // (B, N, C)
     int B, int N, int K, int C) {
   const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long total = (long)B * N * C;
@@ -102,47 +77,110 @@ __global__ void gather_edge_bwd_csr_kernel(
   const int m = (int)(t1 % N);
   const int b = (int)(t1 / N);
 
-  const float *gbase = gT + (long)b * N * K * C;
+  const T *gbase = gT + (long)b * N * K * C;
   const int *ord = order + (long)b * N * K;
   const int lo = offsets[(long)b * (N + 1) + m];
   const int hi = offsets[(long)b * (N + 1) + m + 1];
   float acc = 0.f;
-  for (int t = lo; t < hi; ++t) acc += gbase[(long)ord[t] * C + c];
-  const float *own = gbase + ((long)m * K) * C + c;
+  for (int t = lo; t < hi; ++t) acc += ldf(gbase + (long)ord[t] * C + c);
+  const T *own = gbase + ((long)m * K) * C + c;
   float cen = 0.f;
-  for (int j = 0; j < K; ++j) cen += own[(long)j * C];
-  grad[gid] = acc - cen;
+  for (int j = 0; j < K; ++j) cen += ldf(own + (long)j * C);
+  stf(grad + gid, acc - cen);
 }
 
-void launch_gather_edge_bwd_csr(const float *gT, const int *order,
-                                const int *offsets, float *grad, int B, int N,
-                                int K, int C, hipStream_t stream) {
-  const long total = (long)B * N * C;
-  const int threads = 256;
-  hipLaunchKernelGGL(gather_edge_bwd_csr_kernel,
-                     dim3((total + threads - 1) / threads), dim3(threads), 0,
-                     stream, gT, order, offsets, grad, B, N, K, C);
+// atomic fallback: centre term (unique writes) ...
+template <typename T>
+__global__ void gather_edge_bwd_center_kernel(
+    const T *__restrict__ gout,  // (B, C+3, K, N)
+    float *__restrict__ gfeats,  // (B, N, C) fp32 accumulator
+    int B, int N, int K, int C) {
+  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)B * C * N;
+  if (gid >= total) return;
+  const int n = (int)(gid % N);
+  const int c = (int)((gid / N) % C);
+  const int b = (int)(gid / ((long)N * C));
+
+  const T *src = gout + (((long)b * (C + 3) + c) * K) * N + n;
+  float acc = 0.f;
+  for (int j = 0; j < K; ++j) acc += ldf(src + (long)j * N);
+  gfeats[((long)b * N + n) * C + c] = -acc;
 }
 
-void launch_gather_edge_fwd(const float *feats, const int *idx, const float *xyz,
-                            float *out, int B, int N, int K, int C,
-                            hipStream_t stream) {
+// ... + neighbour scatter with fp32 atomics
+template <typename T>
+__global__ void gather_edge_bwd_scatter_kernel(
+    const T *__restrict__ gout, const int *__restrict__ idx,
+    float *__restrict__ gfeats, int B, int N, int K, int C) {
+  const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)B * K * N;
+  if (gid >= total) return;
+  const int n = (int)(gid % N);
+  const int j = (int)((gid / N) % K);
+  const int b = (int)(gid / ((long)N * K));
+
+  const int nb = idx[((long)b * N + n) * K + j];
+  const T *src = gout + (((long)b * (C + 3)) * K + j) * N + n;
+  float *dst = gfeats + ((long)b * N + nb) * C;
+  const long cstride = (long)K * N;
+  for (int c = 0; c < C; ++c) atomicAdd(dst + c, ldf(src + c * cstride));
+}
+
+// ------------------------------------------------------------- launchers
+
+void launch_gather_edge_fwd(const void *feats, const int *idx,
+                            const float *xyz, void *out, int B, int N, int K,
+                            int C, bool bf16, hipStream_t stream) {
   const long total = (long)B * K * N;
   const int threads = 256;
-  hipLaunchKernelGGL(gather_edge_fwd_kernel,
-                     dim3((total + threads - 1) / threads), dim3(threads), 0,
-                     stream, feats, idx, xyz, out, B, N, K, C);
+  const dim3 grid((total + threads - 1) / threads);
+  if (bf16)
+    hipLaunchKernelGGL(gather_edge_fwd_kernel<__hip_bfloat16>, grid,
+                       dim3(threads), 0, stream, (const __hip_bfloat16 *)feats,
+                       idx, xyz, (__hip_bfloat16 *)out, B, N, K, C);
+  else
+    hipLaunchKernelGGL(gather_edge_fwd_kernel<float>, grid, dim3(threads), 0,
+                       stream, (const float *)feats, idx, xyz, (float *)out, B,
+                       N, K, C);
 }
 
-void launch_gather_edge_bwd(const float *gout, const int *idx, float *gfeats,
-                            int B, int N, int K, int C, hipStream_t stream) {
+void launch_gather_edge_bwd_csr(const void *gT, const int *order,
+                                const int *offsets, void *grad, int B, int N,
+                                int K, int C, bool bf16, hipStream_t stream) {
+  const long total = (long)B * N * C;
+  const int threads = 256;
+  const dim3 grid((total + threads - 1) / threads);
+  if (bf16)
+    hipLaunchKernelGGL(gather_edge_bwd_csr_kernel<__hip_bfloat16>, grid,
+                       dim3(threads), 0, stream, (const __hip_bfloat16 *)gT,
+                       order, offsets, (__hip_bfloat16 *)grad, B, N, K, C);
+  else
+    hipLaunchKernelGGL(gather_edge_bwd_csr_kernel<float>, grid, dim3(threads),
+                       0, stream, (const float *)gT, order, offsets,
+                       (float *)grad, B, N, K, C);
+}
+
+void launch_gather_edge_bwd(const void *gout, const int *idx, float *gfeats,
+                            int B, int N, int K, int C, bool bf16,
+                            hipStream_t stream) {
   const int threads = 256;
   const long t1 = (long)B * C * N;
-  hipLaunchKernelGGL(gather_edge_bwd_center_kernel,
-                     dim3((t1 + threads - 1) / threads), dim3(threads), 0,
-                     stream, gout, gfeats, B, N, K, C);
   const long t2 = (long)B * K * N;
-  hipLaunchKernelGGL(gather_edge_bwd_scatter_kernel,
-                     dim3((t2 + threads - 1) / threads), dim3(threads), 0,
-                     stream, gout, idx, gfeats, B, N, K, C);
+  if (bf16) {
+    hipLaunchKernelGGL(gather_edge_bwd_center_kernel<__hip_bfloat16>,
+                       dim3((t1 + threads - 1) / threads), dim3(threads), 0,
+                       stream, (const __hip_bfloat16 *)gout, gfeats, B, N, K, C);
+    hipLaunchKernelGGL(gather_edge_bwd_scatter_kernel<__hip_bfloat16>,
+                       dim3((t2 + threads - 1) / threads), dim3(threads), 0,
+                       stream, (const __hip_bfloat16 *)gout, idx, gfeats, B, N,
+                       K, C);
+  } else {
+    hipLaunchKernelGGL(gather_edge_bwd_center_kernel<float>,
+                       dim3((t1 + threads - 1) / threads), dim3(threads), 0,
+                       stream, (const float *)gout, gfeats, B, N, K, C);
+    hipLaunchKernelGGL(gather_edge_bwd_scatter_kernel<float>,
+                       dim3((t2 + threads - 1) / threads), dim3(threads), 0,
+                       stream, (const float *)gout, idx, gfeats, B, N, K, C);
+  }
 }

@@ -247,9 +247,11 @@ def gather_edge_concat(feats: Tensor, idx: Tensor, xyz: Tensor, csr=None) -> Ten
     """
     feats = feats.contiguous()
     if _use_hip(feats):
+        if feats.dtype not in (torch.float32, torch.bfloat16):
+            feats = feats.float()
         order, offsets = csr if csr is not None else (None, None)
         return _GatherEdgeConcat.apply(
-            feats.float(), idx.to(torch.int32).contiguous(), xyz.contiguous().float(),
+            feats, idx.to(torch.int32).contiguous(), xyz.contiguous().float(),
             order, offsets,
         )
     return reference.gather_edge_concat(feats, idx, xyz)

@@ -17,8 +17,13 @@ def compute_loss(est_flow: Tensor, batch) -> Tensor:
     mask = batch["ground_truth"][0][..., 0]
     true_flow = batch["ground_truth"][1]
     error = est_flow - true_flow
-    error = error[mask > 0]
-    return torch.mean(torch.abs(error))
+    # masked mean-L1 (identical to the reference's error[mask > 0].abs().mean(),
+    # loss.py:34-38) written with static shapes so the training step is
+    # hipGraph-capturable (boolean indexing has a data-dependent shape)
+    m = (mask > 0).to(error.dtype)
+    total = (error.abs() * m.unsqueeze(-1)).sum()
+    count = m.sum() * error.shape[-1]
+    return total / count.clamp(min=1)
 
 
 def sequence_loss(est_flow: Sequence[Tensor], batch, gamma: float = 0.8) -> Tensor:
