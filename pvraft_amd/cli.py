@@ -34,6 +34,7 @@ def add_common_args(parser: argparse.ArgumentParser, training: bool) -> None:
     parser.add_argument("--refine", help="refine mode", action="store_true")
     parser.add_argument("--num_workers", help="dataloader workers per rank", default=8, type=int)
     parser.add_argument("--amp", help="bf16 autocast compute", action="store_true")
+    parser.add_argument("--no_hipgraph", dest="hipgraph", help="disable hipGraph train-step capture", action="store_false")
     parser.add_argument("--synth_len", help="synthetic dataset length", default=256, type=int)
     if training:
         parser.add_argument("--gamma", help="exponential weights", default=0.8, type=float)

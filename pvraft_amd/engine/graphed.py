@@ -31,10 +31,11 @@ class GraphedTrainStep:
     in-place before each replay.
     """
 
-    def __init__(self, fn: Callable[[], torch.Tensor], warmup: int = 3):
+    def __init__(self, fn: Callable[[], tuple], warmup: int = 3):
         self.fn = fn
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self.static_loss: Optional[torch.Tensor] = None
+        self.static_final_flow: Optional[torch.Tensor] = None
         self._warmup = warmup
 
     def capture(self) -> None:
@@ -49,7 +50,7 @@ class GraphedTrainStep:
 
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
-            self.static_loss = self.fn()
+            self.static_loss, self.static_final_flow = self.fn()
 
     def replay(self) -> torch.Tensor:
         self.graph.replay()
@@ -57,8 +58,13 @@ class GraphedTrainStep:
 
 
 def build_graphed_step(model, batch, num_iters: int, gamma: float,
-                       reducer: GradReducer, amp: bool, warmup: int = 3) -> GraphedTrainStep:
-    """Standard stage-1 step: zero -> fwd(iters) -> sequence_loss -> bwd."""
+                       reducer: GradReducer, amp: bool, warmup: int = 3,
+                       loss_fn=None) -> GraphedTrainStep:
+    """Standard stage-1 step: zero -> fwd(iters) -> sequence_loss -> bwd.
+
+    ``loss_fn(flows, batch) -> loss`` overrides the default sequence loss
+    (the refine stage uses compute_loss on a single flow).
+    """
     from pvraft_amd.utils import sequence_loss
 
     def fn():
@@ -66,9 +72,13 @@ def build_graphed_step(model, batch, num_iters: int, gamma: float,
             b.flat.zero_()
         with torch.autocast("cuda", dtype=torch.bfloat16, enabled=amp):
             flows = model(batch["sequence"], num_iters=num_iters)
-            loss = sequence_loss(flows, batch, gamma=gamma)
+            if loss_fn is not None:
+                loss = loss_fn(flows, batch)
+            else:
+                loss = sequence_loss(flows, batch, gamma=gamma)
         loss.backward()
-        return loss
+        final = flows[-1] if isinstance(flows, (list, tuple)) else flows
+        return loss, final
 
     step = GraphedTrainStep(fn, warmup=warmup)
     step.capture()

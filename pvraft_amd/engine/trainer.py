@@ -174,7 +174,47 @@ class Trainer:
     def _final_flow(self, est_flow):
         return est_flow[-1] if isinstance(est_flow, (list, tuple)) else est_flow
 
+    # ------------------------------------------------------ hipGraph path
+
+    def _graph_enabled(self) -> bool:
+        return self.device.type == "cuda" and bool(getattr(self.args, "hipgraph", True))
+
+    def _ensure_graph(self, batch: Batch) -> bool:
+        """Capture the train step into a hipGraph on first use (static
+        shapes: fixed max_points + drop_last).  Returns False when the
+        batch shape cannot be graphed (falls back to eager)."""
+        shape = tuple(batch["sequence"][0].shape)
+        if getattr(self, "_graph_step", None) is not None:
+            return self._graph_shape == shape
+        from .graphed import build_graphed_step
+
+        self._static_batch = Batch.__new__(Batch)
+        self._static_batch.data = {
+            key: [t.clone() for t in batch.data[key]] for key in batch.data
+        }
+        self.reducer.hooks_enabled = False
+        self._graph_step = build_graphed_step(
+            self.model,
+            self._static_batch,
+            num_iters=self.args.iters,
+            gamma=getattr(self.args, "gamma", 0.8),
+            reducer=self.reducer,
+            amp=self.amp,
+            loss_fn=None if self.loss_is_sequence else (lambda flows, b: self._loss(flows, b)),
+        )
+        self._graph_shape = shape
+        self.log.info("train step captured into a hipGraph")
+        return True
+
     def train_step(self, batch: Batch):
+        if self._graph_enabled() and self._ensure_graph(batch):
+            for key in batch.data:
+                for dst, src in zip(self._static_batch.data[key], batch.data[key]):
+                    dst.copy_(src, non_blocking=True)
+            loss = self._graph_step.replay()
+            self.reducer.reduce_all()
+            self.optimizer.step()
+            return loss, self._graph_step.static_final_flow
         self.reducer.zero_grad()
         with torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.amp):
             est_flow = self.model(batch["sequence"], num_iters=self.args.iters)
