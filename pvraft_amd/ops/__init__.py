@@ -274,7 +274,10 @@ class _CorrTruncate(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, fmap1: Tensor, fmap2: Tensor, xyz2: Tensor, truncate_k: int):
-        with torch.no_grad():
+        # keep the GEMM + topk in fp32 even under bf16 autocast (selection
+        # quality and value precision of the correlation field follow the
+        # reference, corr.py:95-99); autocast would otherwise hijack the bmm
+        with torch.no_grad(), torch.autocast("cuda", enabled=False):
             corr, idx, txyz = reference.corr_truncate(
                 fmap1, fmap2, xyz2, truncate_k, chunk=_CorrTruncate.CHUNK
             )
