@@ -32,7 +32,7 @@ from typing import Optional
 import torch
 from torch.utils.data import DataLoader, distributed as dist_data
 
-from pvraft_amd.data import FT3D, Batch, Kitti, SyntheticSceneFlow
+from pvraft_amd.data import FT3D, Batch, CudaPrefetcher, Kitti, SyntheticSceneFlow
 from pvraft_amd.model import PVRaft
 from pvraft_amd.parallel import (
     GradReducer,
@@ -230,8 +230,8 @@ class Trainer:
             self.train_sampler.set_epoch(epoch)
         run_loss, run_epe, seen = 0.0, 0.0, 0
         t0 = time.time()
-        for i, batch in enumerate(self.train_loader):
-            batch = batch.to(self.device, non_blocking=True)
+        loader = CudaPrefetcher(self.train_loader, self.device)
+        for i, batch in enumerate(loader):
             loss, final_flow = self.train_step(batch)
             with torch.no_grad():
                 epe = compute_epe_train(final_flow.float(), batch)
