@@ -289,29 +289,14 @@ class _CorrTruncate(torch.autograd.Function):
         fmap1, fmap2, idx = ctx.saved_tensors
         B, C, N = fmap1.shape
         M = fmap2.shape[2]
-        K = idx.shape[2]
         scale = 1.0 / math.sqrt(C)
-        g_corr = g_corr * scale
-        g1 = torch.empty_like(fmap1)
-        g2 = torch.zeros_like(fmap2)
-        f2t = fmap2.transpose(1, 2)  # B, M, C
-        chunk = _CorrTruncate.CHUNK
-        for s in range(0, N, chunk):
-            e = min(s + chunk, N)
-            idx_c = idx[:, s:e]  # B, n, K
-            g_c = g_corr[:, s:e]  # B, n, K
-            flat = idx_c.reshape(B, -1)
-            f2g = f2t.gather(1, flat.unsqueeze(-1).expand(B, flat.shape[1], C)).view(
-                B, e - s, K, C
-            )
-            g1[:, :, s:e] = torch.einsum("bnkc,bnk->bcn", f2g, g_c)
-            # scatter f1 * g into columns of g2
-            vals = fmap1[:, :, s:e].unsqueeze(-1) * g_c.unsqueeze(1)  # B, C, n, K
-            g2.scatter_add_(
-                2,
-                flat.unsqueeze(1).expand(B, C, flat.shape[1]),
-                vals.reshape(B, C, -1),
-            )
+        # top-K indices are unique within each row, so the sparse gradient
+        # expands to a dense (B, N, M) buffer with a CONFLICT-FREE scatter
+        # (no atomics), and both fmap gradients become plain rocBLAS GEMMs.
+        gfull = torch.zeros(B, N, M, dtype=g_corr.dtype, device=g_corr.device)
+        gfull.scatter_(2, idx, g_corr * scale)
+        g1 = torch.bmm(fmap2, gfull.transpose(1, 2))  # (B,C,M)@(B,M,N) -> (B,C,N)
+        g2 = torch.bmm(fmap1, gfull)  # (B,C,N)@(B,N,M) -> (B,C,M)
         return g1, g2, None, None
 
 

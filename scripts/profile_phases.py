@@ -68,6 +68,26 @@ def main():
         corr = model.corr_block(field, coords)
         timeit("update block (1 iter)", lambda: model.update_block(net, inp, corr, coords - xyz1, gctx))
 
+    # component-wise forward+backward (isolates backward regressions)
+    def comp_fwd_bwd(label, make_out):
+        def run():
+            with amp:
+                out = make_out()
+            g = torch.autograd.grad(out.float().sum(), [p for p in model.parameters() if p.requires_grad], allow_unused=True)
+            return g
+        timeit(label, run, reps=3)
+
+    comp_fwd_bwd("encoder fwd+bwd", lambda: model.feature_extractor(xyz1)[0])
+    def corr_build_out():
+        fm1, _ = model.feature_extractor(xyz1)
+        fm2, _ = model.feature_extractor(xyz2)
+        fld = model.corr_block.build(fm1, fm2, xyz2)
+        return fld.corr
+    comp_fwd_bwd("enc x2 + corr build fwd+bwd", corr_build_out)
+    field_d = model.corr_block.build(fmap1.detach().requires_grad_(True), fmap2.detach(), xyz2)
+    comp_fwd_bwd("voxel lookup fwd+bwd(conv params)", lambda: model.corr_block._voxel_feature(field_d, xyz1))
+    comp_fwd_bwd("knn lookup fwd+bwd(conv params)", lambda: model.corr_block._knn_feature(field_d, xyz1))
+
     def full_forward():
         with amp:
             return model(batch["sequence"], num_iters=args.iters)
