@@ -45,7 +45,9 @@ from pvraft_amd.utils import (
     compute_epe,
     compute_epe_train,
     load_checkpoint,
+    load_train_state,
     save_checkpoint,
+    save_train_state,
     sequence_loss,
     setup_logger,
 )
@@ -161,8 +163,14 @@ class Trainer:
             )
         epoch = load_checkpoint(path, self.model, strict=True)
         self.begin_epoch = epoch + 1
-        for _ in range(epoch):
-            self.lr_scheduler.step()
+        # exact resume when the extended train state exists (optimizer
+        # moments + scheduler); else replay the scheduler like the reference
+        state = load_train_state(self.args, self.optimizer, self.lr_scheduler)
+        if state is not None and state.get("epoch") == epoch:
+            self.best_val_epe = float(state.get("best_val_epe", float("inf")))
+        else:
+            for _ in range(epoch):
+                self.lr_scheduler.step()
         broadcast_module(self.model)
         self.log.info(f"Loaded weights from {path} (epoch {epoch})")
 
@@ -250,6 +258,8 @@ class Trainer:
             self.scalars.add_scalar("Train/EPE", run_epe / seen, step)
         self.lr_scheduler.step()  # once per epoch (reference engine.py:168)
         save_checkpoint(self.model, self.args, epoch, mode="train", rank=self.dist.rank)
+        save_train_state(self.args, epoch, self.optimizer, self.lr_scheduler,
+                         self.best_val_epe, rank=self.dist.rank)
 
     # ------------------------------------------------------------- evaluation
 

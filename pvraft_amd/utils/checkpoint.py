@@ -49,6 +49,47 @@ def save_checkpoint(
     return path
 
 
+def save_train_state(
+    args,
+    epoch: int,
+    optimizer,
+    lr_scheduler=None,
+    best_val_epe: float = float("inf"),
+    rank: int = 0,
+) -> Optional[str]:
+    """Extension over the reference format: optimizer/scheduler state for
+    exact resume (the reference drops Adam moments on resume -- SURVEY.md
+    item 5.4).  Written beside the model checkpoints as train_state.pt;
+    the model files themselves keep the reference layout untouched.
+    """
+    if rank != 0:
+        return None
+    ckpt_dir = checkpoint_dir(args.root, args.exp_path)
+    os.makedirs(ckpt_dir, exist_ok=True)
+    path = os.path.join(ckpt_dir, "train_state.pt")
+    torch.save(
+        {
+            "epoch": epoch,
+            "optimizer": optimizer.state_dict(),
+            "lr_scheduler": lr_scheduler.state_dict() if lr_scheduler is not None else None,
+            "best_val_epe": best_val_epe,
+        },
+        path,
+    )
+    return path
+
+
+def load_train_state(args, optimizer, lr_scheduler=None) -> Optional[dict]:
+    path = os.path.join(checkpoint_dir(args.root, args.exp_path), "train_state.pt")
+    if not os.path.isfile(path):
+        return None
+    state = torch.load(path, map_location="cpu", weights_only=True)
+    optimizer.load_state_dict(state["optimizer"])
+    if lr_scheduler is not None and state.get("lr_scheduler") is not None:
+        lr_scheduler.load_state_dict(state["lr_scheduler"])
+    return state
+
+
 def load_checkpoint(path: str, model: nn.Module, strict: bool = True) -> int:
     """Load a reference-format checkpoint; returns the stored epoch.
 

@@ -42,3 +42,30 @@ def test_nonzero_rank_writes_nothing(tmp_path):
     model = PVRaft(truncate_k=8)
     assert save_checkpoint(model, args, epoch=1, mode="train", rank=1) is None
     assert not os.path.exists(os.path.join(str(tmp_path), "experiments", "exp", "checkpoints"))
+
+
+def test_train_state_roundtrip(tmp_path):
+    import torch.nn as nn
+
+    args = make_args(tmp_path)
+    model = nn.Linear(4, 4)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    sched = torch.optim.lr_scheduler.CosineAnnealingLR(opt, T_max=100)
+    model(torch.randn(2, 4)).sum().backward()
+    opt.step()
+    sched.step()
+
+    from pvraft_amd.utils import load_train_state, save_train_state
+
+    path = save_train_state(args, epoch=2, optimizer=opt, lr_scheduler=sched, best_val_epe=0.5)
+    assert path.endswith("train_state.pt")
+
+    opt2 = torch.optim.Adam(model.parameters(), lr=1e-3)
+    sched2 = torch.optim.lr_scheduler.CosineAnnealingLR(opt2, T_max=100)
+    state = load_train_state(args, opt2, sched2)
+    assert state["epoch"] == 2 and state["best_val_epe"] == 0.5
+    assert sched2.last_epoch == sched.last_epoch
+    s1 = opt.state_dict()["state"]
+    s2 = opt2.state_dict()["state"]
+    for k in s1:
+        assert torch.equal(s1[k]["exp_avg"], s2[k]["exp_avg"])
