@@ -112,7 +112,11 @@ __global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
 // ---------------------------------------------------------------- backward
 
 // partial sums: per-row {sum dxhat, sum dxhat*xhat}, per-channel
-// {sum dy_norm, sum dy_norm*xhat}; block = contiguous chunk of one row
+// {sum dy_norm, sum dy_norm*xhat}.  Grid (spatial chunks, C, B): each
+// block covers ONE channel's spatial slice, reduces its four partials
+// block-wide and issues exactly four global atomics -- per-element or
+// per-thread atomic flushes onto the tiny (C,2)/(rows,2) workspaces
+// serialize catastrophically (measured 100-400x slower).
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
@@ -120,53 +124,41 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
     const float *__restrict__ gamma, const float *__restrict__ beta,
     float *__restrict__ row_ws,   // (rows, 2)
     float *__restrict__ chan_ws,  // (C, 2)
-    long row_len, long S, int C, int G, int blocks_per_row, int rows,
-    float slope) {
-  const int row = blockIdx.x / blocks_per_row;
-  if (row >= rows) return;
-  long lo, hi;
-  chunk_range(row_len, blocks_per_row, blockIdx.x % blocks_per_row, lo, hi);
+    long S, int C, int G, float slope) {
+  const int c = blockIdx.y;
+  const int b = blockIdx.z;
   const int Cg = C / G;
-  const long base_off = (long)row * row_len;
+  const int row = b * G + c / Cg;
   const float m = mean[row];
   const float r = rstd[row];
+  const float ga = gamma[c], be = beta[c];
+  const T *xb = x + ((long)b * C + c) * S;
+  const T *dyb = dy + ((long)b * C + c) * S;
 
-  float sum_dx = 0.f, sum_dxx = 0.f;
-  long cur_c = -1;
-  float c_dg = 0.f, c_db = 0.f;
-  for (long i = lo + threadIdx.x; i < hi; i += GN_THREADS) {
-    const long gi = base_off + i;
-    const long c = (gi / S) % C;
-    if (c != cur_c) {
-      if (cur_c >= 0) {
-        atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
-        atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
-      }
-      cur_c = c;
-      c_dg = 0.f;
-      c_db = 0.f;
-    }
-    const float xhat = (ld(x + gi) - m) * r;
-    float g = ld(dy + gi);
+  float sum_dx = 0.f, sum_dxx = 0.f, c_dg = 0.f, c_db = 0.f;
+  for (long i = (long)blockIdx.x * GN_THREADS + threadIdx.x; i < S;
+       i += (long)gridDim.x * GN_THREADS) {
+    const float xhat = (ld(xb + i) - m) * r;
+    float g = ld(dyb + i);
     if (ACT == 1) {
-      const float pre = xhat * gamma[c] + beta[c];
+      const float pre = xhat * ga + be;
       g = pre > 0.f ? g : g * slope;
     }
     c_db += g;
     c_dg += g * xhat;
-    const float dxhat = g * gamma[c];
+    const float dxhat = g * ga;
     sum_dx += dxhat;
     sum_dxx += dxhat * xhat;
   }
-  if (cur_c >= 0) {
-    atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
-    atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
-  }
   sum_dx = block_sum(sum_dx);
   sum_dxx = block_sum(sum_dxx);
+  c_db = block_sum(c_db);
+  c_dg = block_sum(c_dg);
   if (threadIdx.x == 0) {
     atomicAdd(&row_ws[row * 2 + 0], sum_dx);
     atomicAdd(&row_ws[row * 2 + 1], sum_dxx);
+    atomicAdd(&chan_ws[c * 2 + 0], c_db);
+    atomicAdd(&chan_ws[c * 2 + 1], c_dg);
   }
 }
 
@@ -243,67 +235,50 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_fwd_apply_kernel(
 }
 
 // backward pass 1: row/channel sums over the POOLED domain (selected
-// positions only carry dxhat / dgamma / dbeta)
+// positions only carry dxhat / dgamma / dbeta).  Grid (N chunks, C, B):
+// one channel per block, four block-reduced partials, four atomics.
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_reduce_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
     const unsigned char *__restrict__ am, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
     const float *__restrict__ beta, float *__restrict__ row_ws,
-    float *__restrict__ chan_ws, long N, int K, int C, int G, long total_out,
-    float slope) {
+    float *__restrict__ chan_ws, long N, int K, int C, int G, float slope) {
+  const int c = blockIdx.y;
+  const int b = blockIdx.z;
   const int Cg = C / G;
-  // contiguous chunking over the pooled domain so channel runs are long
-  const long per_block = (total_out + gridDim.x - 1) / gridDim.x;
-  const long lo = (long)blockIdx.x * per_block;
-  const long hi = min(lo + per_block, total_out);
-  float sum_dx = 0.f, sum_dxx = 0.f;
-  long cur_c = -1, cur_row = -1;
-  float c_dg = 0.f, c_db = 0.f;
-  for (long i = lo + threadIdx.x; i < hi; i += GN_THREADS) {
-    const long n = i % N;
-    const long c = i / N % C;
-    const long b = i / (N * C);
-    const long row = b * G + c / Cg;
-    if (c != cur_c) {
-      if (cur_c >= 0) {
-        atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
-        atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
-      }
-      cur_c = c;
-      c_dg = 0.f;
-      c_db = 0.f;
-    }
-    if (row != cur_row) {
-      if (cur_row >= 0) {
-        atomicAdd(&row_ws[cur_row * 2 + 0], sum_dx);
-        atomicAdd(&row_ws[cur_row * 2 + 1], sum_dxx);
-      }
-      cur_row = row;
-      sum_dx = 0.f;
-      sum_dxx = 0.f;
-    }
-    const int k = am[i];
-    const float xv = ld(x + ((b * C + c) * K + k) * N + n);
-    const float xhat = (xv - mean[row]) * rstd[row];
-    float g = ld(dy + i);
+  const int row = b * G + c / Cg;
+  const float m = mean[row];
+  const float r = rstd[row];
+  const float ga = gamma[c], be = beta[c];
+  const T *xb = x + ((long)b * C + c) * K * N;
+  const long pooled_base = ((long)b * C + c) * N;
+
+  float sum_dx = 0.f, sum_dxx = 0.f, c_dg = 0.f, c_db = 0.f;
+  for (long n = (long)blockIdx.x * GN_THREADS + threadIdx.x; n < N;
+       n += (long)gridDim.x * GN_THREADS) {
+    const int k = am[pooled_base + n];
+    const float xhat = (ld(xb + (long)k * N + n) - m) * r;
+    float g = ld(dy + pooled_base + n);
     if (ACT == 1) {
-      const float pre = xhat * gamma[c] + beta[c];
+      const float pre = xhat * ga + be;
       g = pre > 0.f ? g : g * slope;
     }
     c_db += g;
     c_dg += g * xhat;
-    const float dxhat = g * gamma[c];
+    const float dxhat = g * ga;
     sum_dx += dxhat;
     sum_dxx += dxhat * xhat;
   }
-  if (cur_c >= 0) {
-    atomicAdd(&chan_ws[cur_c * 2 + 0], c_db);
-    atomicAdd(&chan_ws[cur_c * 2 + 1], c_dg);
-  }
-  if (cur_row >= 0) {
-    atomicAdd(&row_ws[cur_row * 2 + 0], sum_dx);
-    atomicAdd(&row_ws[cur_row * 2 + 1], sum_dxx);
+  sum_dx = block_sum(sum_dx);
+  sum_dxx = block_sum(sum_dxx);
+  c_db = block_sum(c_db);
+  c_dg = block_sum(c_dg);
+  if (threadIdx.x == 0) {
+    atomicAdd(&row_ws[row * 2 + 0], sum_dx);
+    atomicAdd(&row_ws[row * 2 + 1], sum_dxx);
+    atomicAdd(&chan_ws[c * 2 + 0], c_db);
+    atomicAdd(&chan_ws[c * 2 + 1], c_dg);
   }
 }
 
@@ -373,25 +348,36 @@ void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
                        beta, S, C, G, total, slope);
 }
 
+static int pick_chunks(long spatial, long bc) {
+  // enough (chunk, C, B) blocks to fill the chip, but no more than the
+  // spatial extent supports
+  long want = 4096 / (bc > 0 ? bc : 1);
+  long cap = (spatial + GN_THREADS - 1) / GN_THREADS;
+  long chunks = want < 1 ? 1 : want;
+  if (chunks > cap) chunks = cap;
+  return (int)(chunks < 1 ? 1 : chunks);
+}
+
 template <typename T>
 void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
                  const float *gamma, const float *beta, float *row_ws,
                  float *chan_ws, T *dx, int rows, long row_len, long S, int C,
                  int G, int act, float slope, hipStream_t stream) {
   const long total = (long)rows * row_len;
-  const int bpr = pick_blocks_per_row(row_len, rows);
+  const int B = rows / G;
+  const dim3 rgrid(pick_chunks(S, (long)B * C), C, B);
   const int apply_blocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
   if (act == 1) {
-    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 1>), dim3(rows * bpr),
-                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
-                       beta, row_ws, chan_ws, row_len, S, C, G, bpr, rows, slope);
+    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 1>), rgrid, dim3(GN_THREADS),
+                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws,
+                       chan_ws, S, C, G, slope);
     hipLaunchKernelGGL((gn_bwd_apply_kernel<T, 1>), dim3(apply_blocks),
                        dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
                        beta, row_ws, dx, S, C, G, row_len, total, slope);
   } else {
-    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 0>), dim3(rows * bpr),
-                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
-                       beta, row_ws, chan_ws, row_len, S, C, G, bpr, rows, slope);
+    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 0>), rgrid, dim3(GN_THREADS),
+                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws,
+                       chan_ws, S, C, G, slope);
     hipLaunchKernelGGL((gn_bwd_apply_kernel<T, 0>), dim3(apply_blocks),
                        dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
                        beta, row_ws, dx, S, C, G, row_len, total, slope);
@@ -428,21 +414,20 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                    int rows, long row_len, long N, int K, int C, int G,
                    int act, float slope, hipStream_t stream) {
   const int B = rows / G;
-  const long total_out = (long)B * C * N;
   const long total = (long)rows * row_len;
-  const int rblocks = (int)min((total_out + GN_THREADS - 1) / GN_THREADS, (long)4096);
+  const dim3 rgrid(pick_chunks(N, (long)B * C), C, B);
   const int ablocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
   if (act == 1) {
-    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 1>), dim3(rblocks),
-                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
-                       gamma, beta, row_ws, chan_ws, N, K, C, G, total_out, slope);
+    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 1>), rgrid, dim3(GN_THREADS),
+                       0, stream, dy, x, am, mean, rstd, gamma, beta, row_ws,
+                       chan_ws, N, K, C, G, slope);
     hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, 1>), dim3(ablocks),
                        dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
                        gamma, beta, row_ws, dx, N, K, C, G, row_len, total, slope);
   } else {
-    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 0>), dim3(rblocks),
-                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
-                       gamma, beta, row_ws, chan_ws, N, K, C, G, total_out, slope);
+    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 0>), rgrid, dim3(GN_THREADS),
+                       0, stream, dy, x, am, mean, rstd, gamma, beta, row_ws,
+                       chan_ws, N, K, C, G, slope);
     hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, 0>), dim3(ablocks),
                        dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
                        gamma, beta, row_ws, dx, N, K, C, G, row_len, total, slope);
