@@ -20,32 +20,55 @@
 #define KNN_TILE 128
 #define KNN_MAXK 48  // model uses 32 (reference extractor.py:10)
 
+// Per-thread top-k as a MAX-HEAP over an LDS row: the common reject path
+// (d >= root) is one compare; replace-root is a log2(k) sift-down instead
+// of a k-wide rescan.
 struct TopK {
   float *dist;
   int *idx;
   int filled = 0;
-  float worst = -1.f;
-  int worst_slot = 0;
+  float worst = INFINITY;  // heap root once filled == k
 
-  __device__ void rescan(int k) {
-    worst = dist[0];
-    worst_slot = 0;
-    for (int j = 1; j < k; ++j)
-      if (dist[j] > worst) {
-        worst = dist[j];
-        worst_slot = j;
-      }
+  __device__ void sift_down(int k) {
+    int j = 0;
+    for (;;) {
+      const int l = 2 * j + 1, r = l + 1;
+      int m = j;
+      if (l < k && dist[l] > dist[m]) m = l;
+      if (r < k && dist[r] > dist[m]) m = r;
+      if (m == j) break;
+      const float td = dist[j];
+      dist[j] = dist[m];
+      dist[m] = td;
+      const int ti = idx[j];
+      idx[j] = idx[m];
+      idx[m] = ti;
+      j = m;
+    }
   }
 
   __device__ void push(float d, int i, int k) {
     if (filled < k) {
-      dist[filled] = d;
-      idx[filled] = i;
-      if (++filled == k) rescan(k);
+      int j = filled++;
+      dist[j] = d;
+      idx[j] = i;
+      while (j > 0) {  // sift-up
+        const int par = (j - 1) >> 1;
+        if (dist[par] >= dist[j]) break;
+        const float td = dist[j];
+        dist[j] = dist[par];
+        dist[par] = td;
+        const int ti = idx[j];
+        idx[j] = idx[par];
+        idx[par] = ti;
+        j = par;
+      }
+      if (filled == k) worst = dist[0];
     } else if (d < worst) {
-      dist[worst_slot] = d;
-      idx[worst_slot] = i;
-      rescan(k);
+      dist[0] = d;
+      idx[0] = i;
+      sift_down(k);
+      worst = dist[0];
     }
   }
 };
