@@ -1,0 +1,63 @@
+"""Inference wrapper: hipGraph-captured scene-flow prediction.
+
+Evaluation/serving runs the backbone 32 GRU iterations per pair
+(reference test.py:120) with fixed shapes (bs x max_points), so the whole
+no-grad forward is captured once and replayed per sample -- the same
+launch-collapse as the training step, applied to serving.
+
+    pred = Predictor(model, points=8192, batch=1, iters=32)
+    flow = pred(xyz1, xyz2)   # (B, N, 3)
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+
+class Predictor:
+    def __init__(self, model, points: int, batch: int = 1, iters: int = 32,
+                 amp: bool = False, use_graph: bool = True):
+        self.model = model.eval()
+        self.iters = iters
+        self.amp = amp
+        self.device = next(model.parameters()).device
+        self.use_graph = use_graph and self.device.type == "cuda"
+        self._graph: Optional[torch.cuda.CUDAGraph] = None
+        self._in1 = torch.zeros(batch, points, 3, device=self.device)
+        self._in2 = torch.zeros(batch, points, 3, device=self.device)
+        self._out: Optional[torch.Tensor] = None
+
+    def _forward(self) -> torch.Tensor:
+        with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.amp):
+            flows = self.model([self._in1, self._in2], num_iters=self.iters)
+        out = flows[-1] if isinstance(flows, (list, tuple)) else flows
+        return out.float()
+
+    def _capture(self) -> None:
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._forward()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._out = self._forward()
+
+    @torch.no_grad()
+    def __call__(self, xyz1: torch.Tensor, xyz2: torch.Tensor) -> torch.Tensor:
+        if not self.use_graph:
+            self._in1, self._in2 = xyz1.to(self.device), xyz2.to(self.device)
+            return self._forward()
+        if self._graph is None:
+            self._in1.copy_(xyz1)
+            self._in2.copy_(xyz2)
+            self._capture()
+            return self._out.clone()
+        self._in1.copy_(xyz1, non_blocking=True)
+        self._in2.copy_(xyz2, non_blocking=True)
+        self._graph.replay()
+        return self._out.clone()
