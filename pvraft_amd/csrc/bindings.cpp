@@ -22,17 +22,19 @@ void launch_knn_corr_bwd(const float*, const int*, float*, int, int, int, int,
 
 void launch_gn_fwd(const void*, void*, float*, float*, float*, const float*,
                    const float*, int, long, long, int, int, float, int, float,
-                   bool, hipStream_t);
+                   const float*, bool, hipStream_t);
 void launch_gn_bwd(const void*, const void*, const float*, const float*,
-                   const float*, const float*, float*, float*, void*, int,
-                   long, long, int, int, int, float, bool, hipStream_t);
+                   const float*, const float*, float*, float*, float*, void*,
+                   int, long, long, int, int, int, float, const float*, bool,
+                   hipStream_t);
 void launch_gnmp_fwd(const void*, void*, unsigned char*, float*, float*,
                      float*, const float*, const float*, int, long, long, int,
-                     int, int, float, int, float, bool, hipStream_t);
+                     int, int, float, int, float, const float*, bool,
+                     hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
-                     float*, float*, void*, int, long, long, int, int, int,
-                     int, float, bool, hipStream_t);
+                     float*, float*, float*, void*, int, long, long, int, int,
+                     int, int, float, const float*, bool, hipStream_t);
 
 namespace {
 
@@ -145,7 +147,7 @@ std::vector<torch::Tensor> knn_corr_fwd(torch::Tensor corr, torch::Tensor xyz,
   const int B = corr.size(0), N = corr.size(1), K = corr.size(2);
   TORCH_CHECK(K <= 512, "HIP knn_corr supports K <= 512");
   TORCH_CHECK(k >= 1 && k <= K, "need 1 <= k <= K");
-  auto out = torch::empty({B, 4, N, k}, corr.options());
+  auto out = torch::empty({B, 4, k, N}, corr.options());
   auto idx = torch::empty({B, N, k}, corr.options().dtype(torch::kInt32));
   launch_knn_corr_fwd(corr.data_ptr<float>(), xyz.data_ptr<float>(),
                       coords.data_ptr<float>(), out.data_ptr<float>(),
@@ -155,7 +157,7 @@ std::vector<torch::Tensor> knn_corr_fwd(torch::Tensor corr, torch::Tensor xyz,
 
 torch::Tensor knn_corr_bwd(torch::Tensor gout, torch::Tensor idx, int64_t K) {
   check_f32(gout, "gout");
-  const int B = gout.size(0), N = gout.size(2), k = gout.size(3);
+  const int B = gout.size(0), k = gout.size(2), N = gout.size(3);
   auto gcorr = torch::zeros({B, N, K}, gout.options());
   launch_knn_corr_bwd(gout.data_ptr<float>(), idx.data_ptr<int>(),
                       gcorr.data_ptr<float>(), B, N, (int)K, k, stream());
@@ -164,10 +166,17 @@ torch::Tensor knn_corr_bwd(torch::Tensor gout, torch::Tensor idx, int64_t K) {
 
 // x (B, C, S) contiguous (S = flattened spatial); weight/bias (C) fp32.
 // act: 0 none, 1 LeakyReLU(slope).  Returns {y, mean (B*G), rstd (B*G)}.
+// slope_t: scalar fp32 tensor for act==2 (learnable PReLU slope)
 std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
                                               torch::Tensor weight,
                                               torch::Tensor bias, double eps,
-                                              int64_t act, double slope) {
+                                              int64_t act, double slope,
+                                              c10::optional<torch::Tensor> slope_t) {
+  const float* slope_ptr = nullptr;
+  if (act == 2) {
+    TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
+    slope_ptr = slope_t->data_ptr<float>();
+  }
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous GPU");
   TORCH_CHECK(x.dim() == 3, "x must be (B, C, S)");
   check_f32(weight, "weight");
@@ -189,7 +198,7 @@ std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
                 mean.data_ptr<float>(), rstd.data_ptr<float>(),
                 weight.data_ptr<float>(), bias.data_ptr<float>(), rows,
                 row_len, S, C, (int)G, (float)eps, (int)act, (float)slope,
-                bf16, stream());
+                slope_ptr, bf16, stream());
   return {y, mean, rstd};
 }
 
@@ -198,7 +207,13 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
                                               torch::Tensor rstd, int64_t G,
                                               torch::Tensor weight,
                                               torch::Tensor bias, int64_t act,
-                                              double slope) {
+                                              double slope,
+                                              c10::optional<torch::Tensor> slope_t) {
+  const float* slope_ptr = nullptr;
+  if (act == 2) {
+    TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
+    slope_ptr = slope_t->data_ptr<float>();
+  }
   TORCH_CHECK(dy.is_contiguous() && x.is_contiguous(), "dy/x must be contiguous");
   const int B = x.size(0), C = x.size(1);
   const long S = x.size(2);
@@ -212,22 +227,30 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
   TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32,
               "group_norm_act: dtype must be float32 or bfloat16");
   TORCH_CHECK(dy.scalar_type() == x.scalar_type(), "dy/x dtype mismatch");
+  auto slope_ws = torch::zeros({1}, fopt);
   launch_gn_bwd(dy.data_ptr(), x.data_ptr(), mean.data_ptr<float>(),
                 rstd.data_ptr<float>(), weight.data_ptr<float>(),
                 bias.data_ptr<float>(), row_ws.data_ptr<float>(),
-                chan_ws.data_ptr<float>(), dx.data_ptr(), rows, row_len, S, C,
-                (int)G, (int)act, (float)slope, bf16, stream());
+                chan_ws.data_ptr<float>(), slope_ws.data_ptr<float>(),
+                dx.data_ptr(), rows, row_len, S, C, (int)G, (int)act,
+                (float)slope, slope_ptr, bf16, stream());
   // chan_ws[:, 0] = dbias, chan_ws[:, 1] = dweight
   auto dbias = chan_ws.select(1, 0).contiguous();
   auto dweight = chan_ws.select(1, 1).contiguous();
-  return {dx, dweight, dbias};
+  return {dx, dweight, dbias, slope_ws};
 }
 
 // x (B, C, K, N); GN stats over full (K, N); returns pooled
 // {y (B,C,N), argmax (B,C,N) u8, mean, rstd}
 std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
     torch::Tensor x, int64_t G, torch::Tensor weight, torch::Tensor bias,
-    double eps, int64_t act, double slope) {
+    double eps, int64_t act, double slope,
+    c10::optional<torch::Tensor> slope_t) {
+  const float* slope_ptr = nullptr;
+  if (act == 2) {
+    TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
+    slope_ptr = slope_t->data_ptr<float>();
+  }
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4, "x must be contiguous (B,C,K,N)");
   check_f32(weight, "weight");
   check_f32(bias, "bias");
@@ -248,14 +271,19 @@ std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
                   ws.data_ptr<float>(), mean.data_ptr<float>(),
                   rstd.data_ptr<float>(), weight.data_ptr<float>(),
                   bias.data_ptr<float>(), rows, row_len, N, K, C, (int)G,
-                  (float)eps, (int)act, (float)slope, bf16, stream());
+                  (float)eps, (int)act, (float)slope, slope_ptr, bf16, stream());
   return {y, am, mean, rstd};
 }
 
 std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
     torch::Tensor dy, torch::Tensor x, torch::Tensor am, torch::Tensor mean,
     torch::Tensor rstd, int64_t G, torch::Tensor weight, torch::Tensor bias,
-    int64_t act, double slope) {
+    int64_t act, double slope, c10::optional<torch::Tensor> slope_t) {
+  const float* slope_ptr = nullptr;
+  if (act == 2) {
+    TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
+    slope_ptr = slope_t->data_ptr<float>();
+  }
   TORCH_CHECK(dy.is_contiguous() && x.is_contiguous(), "dy/x must be contiguous");
   const int B = x.size(0), C = x.size(1), K = x.size(2);
   const long N = x.size(3);
@@ -266,15 +294,17 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
   auto chan_ws = torch::zeros({C, 2}, fopt);
   auto dx = torch::empty_like(x);
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  auto slope_ws = torch::zeros({1}, fopt);
   launch_gnmp_bwd(dy.data_ptr(), x.data_ptr(), am.data_ptr<unsigned char>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
                   weight.data_ptr<float>(), bias.data_ptr<float>(),
                   row_ws.data_ptr<float>(), chan_ws.data_ptr<float>(),
-                  dx.data_ptr(), rows, row_len, N, K, C, (int)G, (int)act,
-                  (float)slope, bf16, stream());
+                  slope_ws.data_ptr<float>(), dx.data_ptr(), rows, row_len, N,
+                  K, C, (int)G, (int)act, (float)slope, slope_ptr, bf16,
+                  stream());
   auto dbias = chan_ws.select(1, 0).contiguous();
   auto dweight = chan_ws.select(1, 1).contiguous();
-  return {dx, dweight, dbias};
+  return {dx, dweight, dbias, slope_ws};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

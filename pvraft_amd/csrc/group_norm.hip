@@ -33,6 +33,11 @@ DEV_INLINE void st(T *p, float v) {
   *p = (T)v;
 }
 
+template <int ACT>
+DEV_INLINE float act_slope(float slope, const float *slope_ptr) {
+  return ACT == 2 ? *slope_ptr : slope;
+}
+
 DEV_INLINE float block_sum(float v) {
   v = wave_sum(v);
   __shared__ float sh[GN_THREADS / WAVE];
@@ -97,14 +102,15 @@ __global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
     const T *__restrict__ x, T *__restrict__ y, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
     const float *__restrict__ beta, long S, int C, int G, long total,
-    float slope) {
+    float slope, const float *__restrict__ slope_ptr) {
   const int Cg = C / G;
+  if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     const long c = (i / S) % C;
     const long row = i / (S * Cg);  // == b * G + g
     float v = (ld(x + i) - mean[row]) * rstd[row] * gamma[c] + beta[c];
-    if (ACT == 1) v = v > 0.f ? v : v * slope;
+    if (ACT >= 1) v = v > 0.f ? v : v * slope;
     st(y + i, v);
   }
 }
@@ -124,7 +130,8 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
     const float *__restrict__ gamma, const float *__restrict__ beta,
     float *__restrict__ row_ws,   // (rows, 2)
     float *__restrict__ chan_ws,  // (C, 2)
-    long S, int C, int G, float slope) {
+    float *__restrict__ slope_ws, // (1,) d slope accumulator (ACT == 2)
+    long S, int C, int G, float slope, const float *__restrict__ slope_ptr) {
   const int c = blockIdx.y;
   const int b = blockIdx.z;
   const int Cg = C / G;
@@ -132,16 +139,18 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
   const float m = mean[row];
   const float r = rstd[row];
   const float ga = gamma[c], be = beta[c];
+  if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const T *xb = x + ((long)b * C + c) * S;
   const T *dyb = dy + ((long)b * C + c) * S;
 
-  float sum_dx = 0.f, sum_dxx = 0.f, c_dg = 0.f, c_db = 0.f;
+  float sum_dx = 0.f, sum_dxx = 0.f, c_dg = 0.f, c_db = 0.f, d_sl = 0.f;
   for (long i = (long)blockIdx.x * GN_THREADS + threadIdx.x; i < S;
        i += (long)gridDim.x * GN_THREADS) {
     const float xhat = (ld(xb + i) - m) * r;
     float g = ld(dyb + i);
-    if (ACT == 1) {
+    if (ACT >= 1) {
       const float pre = xhat * ga + be;
+      if (ACT == 2 && pre <= 0.f) d_sl += g * pre;
       g = pre > 0.f ? g : g * slope;
     }
     c_db += g;
@@ -154,11 +163,13 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
   sum_dxx = block_sum(sum_dxx);
   c_db = block_sum(c_db);
   c_dg = block_sum(c_dg);
+  if (ACT == 2) d_sl = block_sum(d_sl);
   if (threadIdx.x == 0) {
     atomicAdd(&row_ws[row * 2 + 0], sum_dx);
     atomicAdd(&row_ws[row * 2 + 1], sum_dxx);
     atomicAdd(&chan_ws[c * 2 + 0], c_db);
     atomicAdd(&chan_ws[c * 2 + 1], c_dg);
+    if (ACT == 2) atomicAdd(slope_ws, d_sl);
   }
 }
 
@@ -168,9 +179,11 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
     const float *__restrict__ row_ws, T *__restrict__ dx, long S, int C,
-    int G, long row_len, long total, float slope) {
+    int G, long row_len, long total, float slope,
+    const float *__restrict__ slope_ptr) {
   const int Cg = C / G;
   const float inv_n = 1.0f / (float)row_len;
+  if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     const long c = (i / S) % C;
@@ -178,7 +191,7 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
     const float r = rstd[row];
     const float xhat = (ld(x + i) - mean[row]) * r;
     float g = ld(dy + i);
-    if (ACT == 1) {
+    if (ACT >= 1) {
       const float pre = xhat * gamma[c] + beta[c];
       g = pre > 0.f ? g : g * slope;
     }
@@ -207,8 +220,10 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_fwd_apply_kernel(
     const T *__restrict__ x, T *__restrict__ y, unsigned char *__restrict__ am,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta, long N,
-    int K, int C, int G, long total_out, float slope) {
+    int K, int C, int G, long total_out, float slope,
+    const float *__restrict__ slope_ptr) {
   const int Cg = C / G;
+  if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_out;
        i += (long)gridDim.x * blockDim.x) {
     const long n = i % N;
@@ -223,7 +238,7 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_fwd_apply_kernel(
     int bk = 0;
     for (int k = 0; k < K; ++k) {
       float v = (ld(base + (long)k * N) - m) * r * ga + be;
-      if (ACT == 1) v = v > 0.f ? v : v * slope;
+      if (ACT >= 1) v = v > 0.f ? v : v * slope;
       if (v > best) {
         best = v;
         bk = k;
@@ -243,7 +258,8 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_reduce_kernel(
     const unsigned char *__restrict__ am, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
     const float *__restrict__ beta, float *__restrict__ row_ws,
-    float *__restrict__ chan_ws, long N, int K, int C, int G, float slope) {
+    float *__restrict__ chan_ws, float *__restrict__ slope_ws, long N, int K,
+    int C, int G, float slope, const float *__restrict__ slope_ptr) {
   const int c = blockIdx.y;
   const int b = blockIdx.z;
   const int Cg = C / G;
@@ -251,17 +267,19 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_reduce_kernel(
   const float m = mean[row];
   const float r = rstd[row];
   const float ga = gamma[c], be = beta[c];
+  if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const T *xb = x + ((long)b * C + c) * K * N;
   const long pooled_base = ((long)b * C + c) * N;
 
-  float sum_dx = 0.f, sum_dxx = 0.f, c_dg = 0.f, c_db = 0.f;
+  float sum_dx = 0.f, sum_dxx = 0.f, c_dg = 0.f, c_db = 0.f, d_sl = 0.f;
   for (long n = (long)blockIdx.x * GN_THREADS + threadIdx.x; n < N;
        n += (long)gridDim.x * GN_THREADS) {
     const int k = am[pooled_base + n];
     const float xhat = (ld(xb + (long)k * N + n) - m) * r;
     float g = ld(dy + pooled_base + n);
-    if (ACT == 1) {
+    if (ACT >= 1) {
       const float pre = xhat * ga + be;
+      if (ACT == 2 && pre <= 0.f) d_sl += g * pre;
       g = pre > 0.f ? g : g * slope;
     }
     c_db += g;
@@ -274,15 +292,19 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_reduce_kernel(
   sum_dxx = block_sum(sum_dxx);
   c_db = block_sum(c_db);
   c_dg = block_sum(c_dg);
+  if (ACT == 2) d_sl = block_sum(d_sl);
   if (threadIdx.x == 0) {
     atomicAdd(&row_ws[row * 2 + 0], sum_dx);
     atomicAdd(&row_ws[row * 2 + 1], sum_dxx);
     atomicAdd(&chan_ws[c * 2 + 0], c_db);
     atomicAdd(&chan_ws[c * 2 + 1], c_dg);
+    if (ACT == 2) atomicAdd(slope_ws, d_sl);
   }
 }
 
-// backward pass 2: dense dx over the full (B, C, K, N) domain
+// backward pass 2: dense dx over the full (B, C, K, N) domain.
+// Grid (N chunks, C, B), one thread per n looping k: am/dy read once per
+// pooled position instead of once per (k, n), and no per-element divides.
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
@@ -290,31 +312,38 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
     const float *__restrict__ rstd, const float *__restrict__ gamma,
     const float *__restrict__ beta, const float *__restrict__ row_ws,
     T *__restrict__ dx, long N, int K, int C, int G, long row_len,
-    long total, float slope) {
+    float slope, const float *__restrict__ slope_ptr) {
+  const int c = blockIdx.y;
+  const int b = blockIdx.z;
   const int Cg = C / G;
+  const int row = b * G + c / Cg;
+  const float m = mean[row];
+  const float r = rstd[row];
+  const float ga = gamma[c], be = beta[c];
+  if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const float inv_n = 1.0f / (float)row_len;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
-    const long n = i % N;
-    const long k = i / N % K;
-    const long c = i / (N * K) % C;
-    const long b = i / (N * K * C);
-    const long row = b * G + c / Cg;
-    const float r = rstd[row];
-    const float xhat = (ld(x + i) - mean[row]) * r;
-    const long pooled = (b * C + c) * N + n;
-    float dxhat = 0.f;
-    if ((int)k == (int)am[pooled]) {
-      float g = ld(dy + pooled);
-      if (ACT == 1) {
-        const float pre = xhat * gamma[c] + beta[c];
-        g = pre > 0.f ? g : g * slope;
+  const float s1 = row_ws[row * 2 + 0];
+  const float s2 = row_ws[row * 2 + 1];
+  const long base = ((long)b * C + c) * K * N;
+  const long pooled_base = ((long)b * C + c) * N;
+  for (long n = (long)blockIdx.x * GN_THREADS + threadIdx.x; n < N;
+       n += (long)gridDim.x * GN_THREADS) {
+    const int ksel = am[pooled_base + n];
+    float g = ld(dy + pooled_base + n);
+    for (int k = 0; k < K; ++k) {
+      const long i = base + (long)k * N + n;
+      const float xhat = (ld(x + i) - m) * r;
+      float dxhat = 0.f;
+      if (k == ksel) {
+        float gs = g;
+        if (ACT >= 1) {
+          const float pre = xhat * ga + be;
+          gs = pre > 0.f ? gs : gs * slope;
+        }
+        dxhat = gs * ga;
       }
-      dxhat = g * gamma[c];
+      st(dx + i, (dxhat - (s1 + xhat * s2) * inv_n) * r);
     }
-    const float v =
-        (dxhat - (row_ws[row * 2 + 0] + xhat * row_ws[row * 2 + 1]) * inv_n) * r;
-    st(dx + i, v);
   }
 }
 
@@ -330,7 +359,7 @@ template <typename T>
 void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
                  const float *gamma, const float *beta, int rows, long row_len,
                  long S, int C, int G, float eps, int act, float slope,
-                 hipStream_t stream) {
+                 const float *slope_ptr, hipStream_t stream) {
   const long total = (long)rows * row_len;
   const int bpr = pick_blocks_per_row(row_len, rows);
   hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, dim3(rows * bpr),
@@ -338,14 +367,14 @@ void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
   hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
                      0, stream, ws, mean, rstd, row_len, rows, eps);
   const int apply_blocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
-  if (act == 1)
-    hipLaunchKernelGGL((gn_fwd_apply_kernel<T, 1>), dim3(apply_blocks),
-                       dim3(GN_THREADS), 0, stream, x, y, mean, rstd, gamma,
-                       beta, S, C, G, total, slope);
-  else
-    hipLaunchKernelGGL((gn_fwd_apply_kernel<T, 0>), dim3(apply_blocks),
-                       dim3(GN_THREADS), 0, stream, x, y, mean, rstd, gamma,
-                       beta, S, C, G, total, slope);
+#define GN_FWD_APPLY(A)                                                        \
+  hipLaunchKernelGGL((gn_fwd_apply_kernel<T, A>), dim3(apply_blocks),          \
+                     dim3(GN_THREADS), 0, stream, x, y, mean, rstd, gamma,     \
+                     beta, S, C, G, total, slope, slope_ptr)
+  if (act == 2) GN_FWD_APPLY(2);
+  else if (act == 1) GN_FWD_APPLY(1);
+  else GN_FWD_APPLY(0);
+#undef GN_FWD_APPLY
 }
 
 static int pick_chunks(long spatial, long bc) {
@@ -361,34 +390,35 @@ static int pick_chunks(long spatial, long bc) {
 template <typename T>
 void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
                  const float *gamma, const float *beta, float *row_ws,
-                 float *chan_ws, T *dx, int rows, long row_len, long S, int C,
-                 int G, int act, float slope, hipStream_t stream) {
+                 float *chan_ws, float *slope_ws, T *dx, int rows, long row_len,
+                 long S, int C, int G, int act, float slope,
+                 const float *slope_ptr, hipStream_t stream) {
   const long total = (long)rows * row_len;
   const int B = rows / G;
   const dim3 rgrid(pick_chunks(S, (long)B * C), C, B);
   const int apply_blocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
-  if (act == 1) {
-    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 1>), rgrid, dim3(GN_THREADS),
-                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws,
-                       chan_ws, S, C, G, slope);
-    hipLaunchKernelGGL((gn_bwd_apply_kernel<T, 1>), dim3(apply_blocks),
-                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
-                       beta, row_ws, dx, S, C, G, row_len, total, slope);
-  } else {
-    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, 0>), rgrid, dim3(GN_THREADS),
-                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws,
-                       chan_ws, S, C, G, slope);
-    hipLaunchKernelGGL((gn_bwd_apply_kernel<T, 0>), dim3(apply_blocks),
-                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,
-                       beta, row_ws, dx, S, C, G, row_len, total, slope);
-  }
+#define GN_BWD(A)                                                              \
+  do {                                                                         \
+    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, A>), rgrid, dim3(GN_THREADS),  \
+                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws,      \
+                       chan_ws, slope_ws, S, C, G, slope, slope_ptr);          \
+    hipLaunchKernelGGL((gn_bwd_apply_kernel<T, A>), dim3(apply_blocks),        \
+                       dim3(GN_THREADS), 0, stream, dy, x, mean, rstd, gamma,  \
+                       beta, row_ws, dx, S, C, G, row_len, total, slope,       \
+                       slope_ptr);                                             \
+  } while (0)
+  if (act == 2) GN_BWD(2);
+  else if (act == 1) GN_BWD(1);
+  else GN_BWD(0);
+#undef GN_BWD
 }
 
 template <typename T>
 void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
                    float *rstd, const float *gamma, const float *beta, int rows,
                    long row_len, long N, int K, int C, int G, float eps,
-                   int act, float slope, hipStream_t stream) {
+                   int act, float slope, const float *slope_ptr,
+                   hipStream_t stream) {
   const int B = rows / G;
   const long total_out = (long)B * C * N;
   const int bpr = pick_blocks_per_row(row_len, rows);
@@ -397,102 +427,106 @@ void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
   hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
                      0, stream, ws, mean, rstd, row_len, rows, eps);
   const int blocks = (int)min((total_out + GN_THREADS - 1) / GN_THREADS, (long)65535);
-  if (act == 1)
-    hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, 1>), dim3(blocks),
-                       dim3(GN_THREADS), 0, stream, x, y, am, mean, rstd,
-                       gamma, beta, N, K, C, G, total_out, slope);
-  else
-    hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, 0>), dim3(blocks),
-                       dim3(GN_THREADS), 0, stream, x, y, am, mean, rstd,
-                       gamma, beta, N, K, C, G, total_out, slope);
+#define GNMP_FWD(A)                                                            \
+  hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, A>), dim3(blocks),              \
+                     dim3(GN_THREADS), 0, stream, x, y, am, mean, rstd, gamma, \
+                     beta, N, K, C, G, total_out, slope, slope_ptr)
+  if (act == 2) GNMP_FWD(2);
+  else if (act == 1) GNMP_FWD(1);
+  else GNMP_FWD(0);
+#undef GNMP_FWD
 }
 
 template <typename T>
 void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                    const float *mean, const float *rstd, const float *gamma,
-                   const float *beta, float *row_ws, float *chan_ws, T *dx,
-                   int rows, long row_len, long N, int K, int C, int G,
-                   int act, float slope, hipStream_t stream) {
+                   const float *beta, float *row_ws, float *chan_ws,
+                   float *slope_ws, T *dx, int rows, long row_len, long N,
+                   int K, int C, int G, int act, float slope,
+                   const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
-  const long total = (long)rows * row_len;
   const dim3 rgrid(pick_chunks(N, (long)B * C), C, B);
-  const int ablocks = (int)min((total + GN_THREADS - 1) / GN_THREADS, (long)65535);
-  if (act == 1) {
-    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 1>), rgrid, dim3(GN_THREADS),
-                       0, stream, dy, x, am, mean, rstd, gamma, beta, row_ws,
-                       chan_ws, N, K, C, G, slope);
-    hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, 1>), dim3(ablocks),
-                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
-                       gamma, beta, row_ws, dx, N, K, C, G, row_len, total, slope);
-  } else {
-    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, 0>), rgrid, dim3(GN_THREADS),
-                       0, stream, dy, x, am, mean, rstd, gamma, beta, row_ws,
-                       chan_ws, N, K, C, G, slope);
-    hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, 0>), dim3(ablocks),
-                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,
-                       gamma, beta, row_ws, dx, N, K, C, G, row_len, total, slope);
-  }
+#define GNMP_BWD(A)                                                            \
+  do {                                                                         \
+    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, A>), rgrid,                  \
+                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,     \
+                       gamma, beta, row_ws, chan_ws, slope_ws, N, K, C, G,     \
+                       slope, slope_ptr);                                      \
+    hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, A>), rgrid,                   \
+                       dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,     \
+                       gamma, beta, row_ws, dx, N, K, C, G, row_len, slope,    \
+                       slope_ptr);                                             \
+  } while (0)
+  if (act == 2) GNMP_BWD(2);
+  else if (act == 1) GNMP_BWD(1);
+  else GNMP_BWD(0);
+#undef GNMP_BWD
 }
 
 // type-erased entry points (bindings.cpp is host-compiled, no HIP types)
 void launch_gnmp_fwd(const void *x, void *y, unsigned char *am, float *ws,
                      float *mean, float *rstd, const float *gamma,
                      const float *beta, int rows, long row_len, long N, int K,
-                     int C, int G, float eps, int act, float slope, bool bf16,
-                     hipStream_t stream) {
+                     int C, int G, float eps, int act, float slope,
+                     const float *slope_ptr, bool bf16, hipStream_t stream) {
   if (bf16)
     gnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)x,
                                   (__hip_bfloat16 *)y, am, ws, mean, rstd,
                                   gamma, beta, rows, row_len, N, K, C, G, eps,
-                                  act, slope, stream);
+                                  act, slope, slope_ptr, stream);
   else
     gnmp_fwd_impl<float>((const float *)x, (float *)y, am, ws, mean, rstd,
                          gamma, beta, rows, row_len, N, K, C, G, eps, act,
-                         slope, stream);
+                         slope, slope_ptr, stream);
 }
 
 void launch_gnmp_bwd(const void *dy, const void *x, const unsigned char *am,
                      const float *mean, const float *rstd, const float *gamma,
                      const float *beta, float *row_ws, float *chan_ws,
-                     void *dx, int rows, long row_len, long N, int K, int C,
-                     int G, int act, float slope, bool bf16,
-                     hipStream_t stream) {
+                     float *slope_ws, void *dx, int rows, long row_len, long N,
+                     int K, int C, int G, int act, float slope,
+                     const float *slope_ptr, bool bf16, hipStream_t stream) {
   if (bf16)
     gnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, am, mean, rstd,
-        gamma, beta, row_ws, chan_ws, (__hip_bfloat16 *)dx, rows, row_len, N,
-        K, C, G, act, slope, stream);
+        gamma, beta, row_ws, chan_ws, slope_ws, (__hip_bfloat16 *)dx, rows,
+        row_len, N, K, C, G, act, slope, slope_ptr, stream);
   else
     gnmp_bwd_impl<float>((const float *)dy, (const float *)x, am, mean, rstd,
-                         gamma, beta, row_ws, chan_ws, (float *)dx, rows,
-                         row_len, N, K, C, G, act, slope, stream);
+                         gamma, beta, row_ws, chan_ws, slope_ws, (float *)dx,
+                         rows, row_len, N, K, C, G, act, slope, slope_ptr,
+                         stream);
 }
 
 void launch_gn_fwd(const void *x, void *y, float *ws, float *mean, float *rstd,
                    const float *gamma, const float *beta, int rows,
                    long row_len, long S, int C, int G, float eps, int act,
-                   float slope, bool bf16, hipStream_t stream) {
+                   float slope, const float *slope_ptr, bool bf16,
+                   hipStream_t stream) {
   if (bf16)
     gn_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)x, (__hip_bfloat16 *)y,
                                 ws, mean, rstd, gamma, beta, rows, row_len, S,
-                                C, G, eps, act, slope, stream);
+                                C, G, eps, act, slope, slope_ptr, stream);
   else
     gn_fwd_impl<float>((const float *)x, (float *)y, ws, mean, rstd, gamma,
-                       beta, rows, row_len, S, C, G, eps, act, slope, stream);
+                       beta, rows, row_len, S, C, G, eps, act, slope,
+                       slope_ptr, stream);
 }
 
 void launch_gn_bwd(const void *dy, const void *x, const float *mean,
                    const float *rstd, const float *gamma, const float *beta,
-                   float *row_ws, float *chan_ws, void *dx, int rows,
-                   long row_len, long S, int C, int G, int act, float slope,
-                   bool bf16, hipStream_t stream) {
+                   float *row_ws, float *chan_ws, float *slope_ws, void *dx,
+                   int rows, long row_len, long S, int C, int G, int act,
+                   float slope, const float *slope_ptr, bool bf16,
+                   hipStream_t stream) {
   if (bf16)
     gn_bwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)dy,
                                 (const __hip_bfloat16 *)x, mean, rstd, gamma,
-                                beta, row_ws, chan_ws, (__hip_bfloat16 *)dx,
-                                rows, row_len, S, C, G, act, slope, stream);
+                                beta, row_ws, chan_ws, slope_ws,
+                                (__hip_bfloat16 *)dx, rows, row_len, S, C, G,
+                                act, slope, slope_ptr, stream);
   else
     gn_bwd_impl<float>((const float *)dy, (const float *)x, mean, rstd, gamma,
-                       beta, row_ws, chan_ws, (float *)dx, rows, row_len, S, C,
-                       G, act, slope, stream);
+                       beta, row_ws, chan_ws, slope_ws, (float *)dx, rows,
+                       row_len, S, C, G, act, slope, slope_ptr, stream);
 }

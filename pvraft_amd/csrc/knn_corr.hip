@@ -9,8 +9,10 @@
 // deterministic selection), and invalidates the winner with an unrolled
 // compare-select.  No LDS, no scratch, no (B,N,K) distance tensor.
 //
-// Output (B, 4, N, k): channel 0 = corr at the selected candidate,
-// channels 1..3 = candidate xyz - coords.  idx (B, N, k) is saved for
+// Output (B, 4, k, N): channel 0 = corr at the selected candidate,
+// channels 1..3 = candidate xyz - coords (k-major layout so the following
+// conv2d/GN/pool pipeline pools over dim 2, same as the SetConv stage and
+// reusable by the fused GN+act+maxpool kernel).  idx (B, N, k) is saved for
 // backward; selections within a row are unique so backward is a plain
 // scatter (no atomics), gradient flows to corr only (coords is detached by
 // the caller every GRU iteration, reference RAFTSceneFlow.py:41).
@@ -23,7 +25,7 @@ __global__ __launch_bounds__(256) void knn_corr_fwd_kernel(
     const float *__restrict__ corr,    // (B, N, K)
     const float *__restrict__ xyz,     // (B, N, K, 3)
     const float *__restrict__ coords,  // (B, N, 3)
-    float *__restrict__ out,           // (B, 4, N, k)
+    float *__restrict__ out,           // (B, 4, k, N)
     int *__restrict__ out_idx,         // (B, N, k)
     int B, int N, int K, int k) {
   const long p = (long)blockIdx.x * 4 + wave_id();
@@ -52,8 +54,8 @@ __global__ __launch_bounds__(256) void knn_corr_fwd_kernel(
     }
   }
 
-  float *dst = out + (((long)b * 4) * N + n) * k;
-  const long ch_stride = (long)N * k;
+  float *dst = out + ((long)b * 4) * k * N + n;
+  const long ch_stride = (long)k * N;
   int *idst = out_idx + p * k;
 
   for (int t = 0; t < k; ++t) {
@@ -72,10 +74,10 @@ __global__ __launch_bounds__(256) void knn_corr_fwd_kernel(
     // winner lane emits and invalidates its slot
     if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
       const int j = bidx;
-      dst[t] = cand_corr[j];
-      dst[t + ch_stride] = cand_xyz[j * 3 + 0] - cx;
-      dst[t + 2 * ch_stride] = cand_xyz[j * 3 + 1] - cy;
-      dst[t + 3 * ch_stride] = cand_xyz[j * 3 + 2] - cz;
+      dst[(long)t * N] = cand_corr[j];
+      dst[(long)t * N + ch_stride] = cand_xyz[j * 3 + 0] - cx;
+      dst[(long)t * N + 2 * ch_stride] = cand_xyz[j * 3 + 1] - cy;
+      dst[(long)t * N + 3 * ch_stride] = cand_xyz[j * 3 + 2] - cz;
       idst[t] = j;
       const int slot = bidx / WAVE;
 #pragma unroll
@@ -86,20 +88,20 @@ __global__ __launch_bounds__(256) void knn_corr_fwd_kernel(
 }
 
 __global__ void knn_corr_bwd_kernel(
-    const float *__restrict__ gout,  // (B, 4, N, k)
+    const float *__restrict__ gout,  // (B, 4, k, N)
     const int *__restrict__ idx,     // (B, N, k)
     float *__restrict__ gcorr,       // (B, N, K) pre-zeroed
     int B, int N, int K, int k) {
   const long gid = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long total = (long)B * N * k;
+  const long total = (long)B * k * N;
   if (gid >= total) return;
-  const int t = (int)(gid % k);
-  const long p = gid / k;  // b * N + n
-  const int b = (int)(p / N);
-  const int n = (int)(p % N);
+  const int n = (int)(gid % N);
+  const int t = (int)((gid / N) % k);
+  const int b = (int)(gid / ((long)N * k));
+  const long p = (long)b * N + n;
   const int j = idx[p * k + t];
   // channel 0 of gout feeds corr; rel-xyz channels carry no gradient
-  gcorr[p * K + j] = gout[(((long)b * 4) * N + n) * k + t];
+  gcorr[p * K + j] = gout[((long)b * 4 * k + t) * N + n];
 }
 
 void launch_knn_corr_fwd(const float *corr, const float *xyz,

@@ -83,10 +83,23 @@ class CorrBlock(nn.Module):
         feat = ops.voxel_corr(
             field.corr, field.xyz, coords, self.base_scale, self.num_levels, self.resolution
         )  # B, L*R^3, N
-        return self.out_conv(feat)
+        # out_conv = Sequential(conv, GN, PReLU, conv) by state-dict layout;
+        # executed with GN+PReLU fused (learnable slope = out_conv[2].weight)
+        conv1, gn, prelu, conv2 = self.out_conv[0], self.out_conv[1], self.out_conv[2], self.out_conv[3]
+        x = conv1(feat)
+        x = ops.group_norm_act(
+            x, gn.num_groups, gn.weight, gn.bias, gn.eps, act="prelu", slope_t=prelu.weight
+        )
+        return conv2(x)
 
     def _knn_feature(self, field: CorrField, coords: Tensor) -> Tensor:
-        feat = ops.knn_corr(field.corr, field.xyz, coords, self.knn)  # B, 4, N, k
-        feat = self.knn_conv(feat)
-        feat = feat.max(dim=3)[0]  # B, 64, N
+        feat = ops.knn_corr(field.corr, field.xyz, coords, self.knn)  # B, 4, k, N
+        # knn_conv = Sequential(conv, GN, PReLU) + max over the k axis;
+        # executed as conv -> fused GN+PReLU+maxpool (one kernel pipeline,
+        # the (B, 64, k, N) activation never materialises post-GN)
+        conv, gn, prelu = self.knn_conv[0], self.knn_conv[1], self.knn_conv[2]
+        feat = conv(feat)
+        feat = ops.group_norm_act_maxpool(
+            feat, gn.num_groups, gn.weight, gn.bias, gn.eps, act="prelu", slope_t=prelu.weight
+        )  # B, 64, N
         return self.knn_out(feat)

@@ -1,12 +1,18 @@
-"""1x1 convolutions as plain GEMMs.
+"""1x1 convolutions as plain GEMMs with split-K weight gradients.
 
 Every convolution in PV-RAFT has kernel size 1 (reference gconv.py:26-33,
 corr.py:15-29, update.py:11-29,60-66), i.e. it IS a GEMM over the flattened
 spatial dim.  MIOpen's conv path wraps it in NHWC transposes and igemm
-kernels (visible in rocprof as batched_transpose_* + igemm_wrw_*); routing
-through torch.matmul hits hipBLASLt directly and lets autograd produce
-plain GEMM backward.  Subclasses keep nn.Conv1d/Conv2d parameter layout so
-state dicts stay interchangeable with the reference.
+kernels; torch.matmul hits hipBLASLt directly.
+
+The weight gradient of these layers is a (Cout x Cin) output with a
+B*N ~ 16-65k reduction dim: hipBLASLt's heuristic picks a non-split-K
+kernel (a handful of workgroups on 256 CUs, measured ~210 us for a ~5 us
+problem), so _PwMatmul computes dW as a chunked batched GEMM (SPLITK
+partial products summed) instead of relying on matmul autograd.
+
+Subclasses keep nn.Conv1d/Conv2d parameter layout so state dicts stay
+interchangeable with the reference.
 """
 
 from __future__ import annotations
@@ -15,14 +21,54 @@ import torch
 import torch.nn as nn
 from torch import Tensor
 
+SPLITK = 16
+
+
+class _PwMatmul(torch.autograd.Function):
+    """y (B, Co, S) = w (Co, Ci) @ x (B, Ci, S) with split-K backward."""
+
+    @staticmethod
+    def forward(ctx, w: Tensor, x: Tensor) -> Tensor:
+        ctx.save_for_backward(w, x)
+        return torch.matmul(w, x)
+
+    @staticmethod
+    def backward(ctx, dy: Tensor):
+        w, x = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = torch.matmul(w.t(), dy) if ctx.needs_input_grad[1] else None
+        dw = None
+        if ctx.needs_input_grad[0]:
+            B, Ci, S = x.shape
+            Co = dy.shape[1]
+            if x.is_cuda and S % SPLITK == 0:
+                Sc = S // SPLITK
+                # (B*SPLITK, Co, Sc) @ (B*SPLITK, Sc, Ci) -> sum over batch
+                dy_c = dy.view(B, Co, SPLITK, Sc).permute(0, 2, 1, 3)
+                x_c = x.view(B, Ci, SPLITK, Sc).permute(0, 2, 3, 1)
+                dw = torch.matmul(dy_c, x_c).sum(dim=(0, 1)).to(w.dtype)
+            else:
+                dw = torch.einsum("bos,bis->oi", dy, x).to(w.dtype)
+        return dw, dx
+
+
+def pw_matmul(weight: Tensor, x: Tensor) -> Tensor:
+    w = weight
+    if x.is_cuda and torch.is_autocast_enabled():
+        dt = torch.get_autocast_dtype("cuda")
+        w = w.to(dt)
+        x = x.to(dt)
+    elif x.dtype != w.dtype:
+        x = x.to(w.dtype)
+    return _PwMatmul.apply(w, x.contiguous())
+
 
 class PwConv1d(nn.Conv1d):
     """nn.Conv1d(k=1) with a matmul forward."""
 
     def forward(self, x: Tensor) -> Tensor:
         # x (B, Cin, N) -> (B, Cout, N)
-        w = self.weight.squeeze(-1)  # (Cout, Cin)
-        y = torch.matmul(w, x)
+        y = pw_matmul(self.weight.squeeze(-1), x)
         if self.bias is not None:
             y = y + self.bias.view(1, -1, 1)
         return y
@@ -33,8 +79,7 @@ class PwConv2d(nn.Conv2d):
 
     def forward(self, x: Tensor) -> Tensor:
         B, C, H, W = x.shape
-        w = self.weight.view(self.out_channels, C)
-        y = torch.matmul(w, x.reshape(B, C, H * W))
+        y = pw_matmul(self.weight.view(self.out_channels, C), x.reshape(B, C, H * W))
         if self.bias is not None:
             y = y + self.bias.view(1, -1, 1)
         return y.view(B, self.out_channels, H, W)

@@ -124,25 +124,27 @@ class _KnnCorr(torch.autograd.Function):
 
 
 class _GroupNormAct(torch.autograd.Function):
-    """GroupNorm with optionally fused LeakyReLU (HIP, fp32/bf16 IO)."""
+    """GroupNorm with fused LeakyReLU / learnable-PReLU (HIP, fp32/bf16 IO)."""
 
     @staticmethod
-    def forward(ctx, x, num_groups, weight, bias, eps, act, slope):
+    def forward(ctx, x, num_groups, weight, bias, eps, act, slope, slope_t):
         w = weight.float().contiguous()
         b = bias.float().contiguous()
-        y, mean, rstd = _EXT.group_norm_act_fwd(x, num_groups, w, b, eps, act, slope)
-        ctx.save_for_backward(x, mean, rstd, w, b)
+        st_ = slope_t.float().reshape(1).contiguous() if slope_t is not None else None
+        y, mean, rstd = _EXT.group_norm_act_fwd(x, num_groups, w, b, eps, act, slope, st_)
+        ctx.save_for_backward(x, mean, rstd, w, b, st_)
         ctx.conf = (num_groups, act, slope, weight.dtype)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, mean, rstd, w, b = ctx.saved_tensors
+        x, mean, rstd, w, b, st_ = ctx.saved_tensors
         num_groups, act, slope, wdtype = ctx.conf
-        dx, dw, db = _EXT.group_norm_act_bwd(
-            dy.contiguous(), x, mean, rstd, num_groups, w, b, act, slope
+        dx, dw, db, dsl = _EXT.group_norm_act_bwd(
+            dy.contiguous(), x, mean, rstd, num_groups, w, b, act, slope, st_
         )
-        return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None
+        dslope = dsl.to(wdtype) if act == 2 else None
+        return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None, dslope
 
 
 def group_norm_act(
@@ -153,14 +155,17 @@ def group_norm_act(
     eps: float = 1e-5,
     act: str = "none",
     slope: float = 0.1,
+    slope_t: Optional[Tensor] = None,
 ) -> Tensor:
-    """GroupNorm over (B, C, *spatial) with optional fused LeakyReLU.
+    """GroupNorm over (B, C, *spatial) with fused activation.
 
+    act: "none", "lrelu" (constant slope) or "prelu" (learnable scalar
+    slope tensor ``slope_t``, gradient included).
     GPU: single HIP pipeline (multi-workgroup reduction; ATen's GroupNorm
     uses one workgroup per (batch, group) which starves MI355X's 256 CUs).
     CPU / reference mode: F.group_norm (+ activation).
     """
-    act_id = {"none": 0, "lrelu": 1}[act]
+    act_id = {"none": 0, "lrelu": 1, "prelu": 2}[act]
     if _use_hip(x):
         shape = x.shape
         y = _GroupNormAct.apply(
@@ -171,11 +176,14 @@ def group_norm_act(
             eps,
             act_id,
             slope,
+            slope_t,
         )
         return y.view(shape)
     y = torch.nn.functional.group_norm(x, num_groups, weight, bias, eps)
     if act_id == 1:
         y = torch.nn.functional.leaky_relu(y, slope)
+    elif act_id == 2:
+        y = torch.nn.functional.prelu(y, slope_t.to(y.dtype))
     return y
 
 
@@ -187,22 +195,24 @@ class _GroupNormActMaxpool(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x, num_groups, weight, bias, eps, act, slope):
+    def forward(ctx, x, num_groups, weight, bias, eps, act, slope, slope_t):
         w = weight.float().contiguous()
         b = bias.float().contiguous()
-        y, am, mean, rstd = _EXT.group_norm_act_maxpool_fwd(x, num_groups, w, b, eps, act, slope)
-        ctx.save_for_backward(x, am, mean, rstd, w, b)
+        st_ = slope_t.float().reshape(1).contiguous() if slope_t is not None else None
+        y, am, mean, rstd = _EXT.group_norm_act_maxpool_fwd(x, num_groups, w, b, eps, act, slope, st_)
+        ctx.save_for_backward(x, am, mean, rstd, w, b, st_)
         ctx.conf = (num_groups, act, slope, weight.dtype)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, am, mean, rstd, w, b = ctx.saved_tensors
+        x, am, mean, rstd, w, b, st_ = ctx.saved_tensors
         num_groups, act, slope, wdtype = ctx.conf
-        dx, dw, db = _EXT.group_norm_act_maxpool_bwd(
-            dy.contiguous(), x, am, mean, rstd, num_groups, w, b, act, slope
+        dx, dw, db, dsl = _EXT.group_norm_act_maxpool_bwd(
+            dy.contiguous(), x, am, mean, rstd, num_groups, w, b, act, slope, st_
         )
-        return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None
+        dslope = dsl.to(wdtype) if act == 2 else None
+        return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None, dslope
 
 
 def group_norm_act_maxpool(
@@ -213,16 +223,19 @@ def group_norm_act_maxpool(
     eps: float = 1e-5,
     act: str = "lrelu",
     slope: float = 0.1,
+    slope_t: Optional[Tensor] = None,
 ) -> Tensor:
     """(B, C, K, N) -> (B, C, N): GroupNorm -> activation -> max over K."""
-    act_id = {"none": 0, "lrelu": 1}[act]
+    act_id = {"none": 0, "lrelu": 1, "prelu": 2}[act]
     if _use_hip(x):
         return _GroupNormActMaxpool.apply(
-            x.contiguous(), num_groups, weight, bias, eps, act_id, slope
+            x.contiguous(), num_groups, weight, bias, eps, act_id, slope, slope_t
         )
     y = torch.nn.functional.group_norm(x, num_groups, weight, bias, eps)
     if act_id == 1:
         y = torch.nn.functional.leaky_relu(y, slope)
+    elif act_id == 2:
+        y = torch.nn.functional.prelu(y, slope_t.to(y.dtype))
     return y.max(dim=2)[0]
 
 
@@ -282,6 +295,9 @@ class _CorrTruncate(torch.autograd.Function):
                 fmap1, fmap2, xyz2, truncate_k, chunk=_CorrTruncate.CHUNK
             )
         ctx.save_for_backward(fmap1, fmap2, idx)
+        # backward GEMMs run bf16 when the surrounding step is bf16 autocast
+        # (standard amp gradient precision; fp32 master weights untouched)
+        ctx.bf16_bwd = fmap1.is_cuda and torch.is_autocast_enabled()
         return corr, idx, txyz
 
     @staticmethod
@@ -293,10 +309,13 @@ class _CorrTruncate(torch.autograd.Function):
         # top-K indices are unique within each row, so the sparse gradient
         # expands to a dense (B, N, M) buffer with a CONFLICT-FREE scatter
         # (no atomics), and both fmap gradients become plain rocBLAS GEMMs.
-        gfull = torch.zeros(B, N, M, dtype=g_corr.dtype, device=g_corr.device)
-        gfull.scatter_(2, idx, g_corr * scale)
-        g1 = torch.bmm(fmap2, gfull.transpose(1, 2))  # (B,C,M)@(B,M,N) -> (B,C,N)
-        g2 = torch.bmm(fmap1, gfull)  # (B,C,N)@(B,N,M) -> (B,C,M)
+        dt = torch.bfloat16 if ctx.bf16_bwd else g_corr.dtype
+        gfull = torch.zeros(B, N, M, dtype=dt, device=g_corr.device)
+        gfull.scatter_(2, idx, (g_corr * scale).to(dt))
+        f1 = fmap1.to(dt)
+        f2 = fmap2.to(dt)
+        g1 = torch.bmm(f2, gfull.transpose(1, 2).contiguous()).float()  # (B,C,N)
+        g2 = torch.bmm(f1, gfull).float()  # (B,C,M)
         return g1, g2, None, None
 
 

@@ -91,7 +91,8 @@ def corr_truncate(
     M = fmap2.shape[2]
     K = min(truncate_k, M)
     scale = 1.0 / math.sqrt(C)
-    f1t = fmap1.transpose(1, 2)  # B, N, C
+    # contiguous LHS: hipBLASLt's strided-A path measured ~10x slower here
+    f1t = fmap1.transpose(1, 2).contiguous()  # B, N, C
     vals, idxs = [], []
     for s in range(0, N, chunk):
         c = torch.bmm(f1t[:, s : s + chunk], fmap2) * scale  # B, n, M
@@ -160,9 +161,13 @@ def knn_corr(corr: Tensor, xyz: Tensor, coords: Tensor, k: int) -> Tensor:
     """k nearest of the K truncated candidates around each point.
 
     corr: (B, N, K), xyz: (B, N, K, 3), coords: (B, N, 3).
-    Returns (B, 4, N, k): channel 0 = gathered correlation, channels 1..3 =
-    candidate position relative to coords.  Selection by squared distance
-    (reference corr.py:78-81: topk(-dist)); indices are constants to
+    Returns (B, 4, k, N): channel 0 = gathered correlation, channels 1..3 =
+    candidate position relative to coords.  The reference lays this out
+    (B, 4, N, k) and pools over the last dim (corr.py:84-92); here the k
+    axis is dim 2 so the conv/GN/pool pipeline matches the SetConv stage
+    and can use the fused GN+act+maxpool kernel -- GN statistics and the
+    max are permutation-invariant, values identical.  Selection by squared
+    distance (corr.py:78-81: topk(-dist)); indices are constants to
     autograd, gradients flow into ``corr`` via the gather only (coords is
     detached by the caller each GRU iteration, RAFTSceneFlow.py:41).
     """
@@ -172,7 +177,7 @@ def knn_corr(corr: Tensor, xyz: Tensor, coords: Tensor, k: int) -> Tensor:
         d = xyz - coords.unsqueeze(2)
         dist = (d * d).sum(-1)  # B, N, K
         nbr = dist.topk(k, dim=2, largest=False).indices  # B, N, k
-    kc = corr.gather(2, nbr).unsqueeze(1)  # B, 1, N, k
+    kc = corr.gather(2, nbr).transpose(1, 2).unsqueeze(1)  # B, 1, k, N
     kx = xyz.gather(2, nbr.unsqueeze(-1).expand(B, N, k, 3))  # B, N, k, 3
-    rel = (kx - coords.unsqueeze(2)).permute(0, 3, 1, 2)  # B, 3, N, k
+    rel = (kx - coords.unsqueeze(2)).permute(0, 3, 2, 1)  # B, 3, k, N
     return torch.cat([kc, rel], dim=1)

@@ -82,10 +82,11 @@ def test_knn_corr_fwd_bwd(B, N, K, k):
 
     out = ops.knn_corr(corr, xyz, coords, k)
     out_ref = R.knn_corr(corr_ref, xyz, coords, k)
-    # selections may differ on exact distance ties; compare sorted per-point
+    assert out.shape == (B, 4, k, N)
+    # selections may differ on exact distance ties; compare sorted over k
     assert torch.allclose(
-        out.sort(dim=-1).values, out_ref.sort(dim=-1).values, atol=1e-4
-    ), (out.sort(dim=-1).values - out_ref.sort(dim=-1).values).abs().max()
+        out.sort(dim=2).values, out_ref.sort(dim=2).values, atol=1e-4
+    ), (out.sort(dim=2).values - out_ref.sort(dim=2).values).abs().max()
 
     # backward: channel-0 grads scatter into corr at the selected slots
     out[:, 0].sum().backward()

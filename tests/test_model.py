@@ -103,3 +103,37 @@ def test_iteration_zero_of_longer_run_matches_shorter_run():
         f4 = model(p, num_iters=4)
     assert torch.allclose(f2[0], f4[0], atol=1e-5)
     assert torch.allclose(f2[1], f4[1], atol=1e-5)
+
+
+def test_pw_conv_matches_nn_conv():
+    """PwConv1d/2d (matmul + split-K wgrad) vs stock nn.Conv forward/backward."""
+    from pvraft_amd.model.pointwise import PwConv1d, PwConv2d
+
+    torch.manual_seed(0)
+    conv = torch.nn.Conv1d(6, 10, 1)
+    pw = PwConv1d(6, 10, 1)
+    pw.load_state_dict(conv.state_dict())
+    x1 = torch.randn(2, 6, 32, requires_grad=True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    y1 = conv(x1)
+    y2 = pw(x2)
+    assert torch.allclose(y1, y2, atol=1e-6)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, atol=1e-6)
+    assert torch.allclose(conv.weight.grad.squeeze(-1), pw.weight.grad.squeeze(-1), atol=1e-5)
+    assert torch.allclose(conv.bias.grad, pw.bias.grad, atol=1e-5)
+
+    conv2 = torch.nn.Conv2d(5, 7, 1, bias=False)
+    pw2 = PwConv2d(5, 7, 1, bias=False)
+    pw2.load_state_dict(conv2.state_dict())
+    x = torch.randn(2, 5, 4, 16, requires_grad=True)
+    xb = x.detach().clone().requires_grad_(True)
+    ya = conv2(x)
+    yb = pw2(xb)
+    assert torch.allclose(ya, yb, atol=1e-6)
+    ya.sum().backward()
+    yb.sum().backward()
+    assert torch.allclose(x.grad, xb.grad, atol=1e-6)
+    assert torch.allclose(conv2.weight.grad, pw2.weight.grad, atol=1e-5)

@@ -121,14 +121,14 @@ def test_knn_corr_values():
     coords = torch.randn(B, N, 3)
     xyz = coords.unsqueeze(2) + torch.randn(B, N, K, 3)
     out = R.knn_corr(corr, xyz, coords, k)
-    assert out.shape == (B, 4, N, k)
+    assert out.shape == (B, 4, k, N)
     # check point 0: k nearest candidates by distance
     d = ((xyz[0, 0] - coords[0, 0]) ** 2).sum(-1)
     nbr = d.topk(k, largest=False).indices
-    assert torch.allclose(out[0, 0, 0].sort().values, corr[0, 0, nbr].sort().values)
+    assert torch.allclose(out[0, 0, :, 0].sort().values, corr[0, 0, nbr].sort().values)
     rel = xyz[0, 0, nbr] - coords[0, 0]
     assert torch.allclose(
-        out[0, 1:, 0].t().sort(0).values, rel.sort(0).values
+        out[0, 1:, :, 0].t().sort(0).values, rel.sort(0).values
     )
 
 
