@@ -239,7 +239,15 @@ class Trainer:
         run_loss, run_epe, seen = 0.0, 0.0, 0
         t0 = time.time()
         loader = CudaPrefetcher(self.train_loader, self.device)
-        for i, batch in enumerate(loader):
+        iterator = enumerate(loader)
+        if self.dist.is_main:
+            try:
+                from tqdm import tqdm
+
+                iterator = enumerate(tqdm(loader, desc=f"epoch {epoch}", leave=False))
+            except ImportError:
+                pass
+        for i, batch in iterator:
             loss, final_flow = self.train_step(batch)
             with torch.no_grad():
                 epe = compute_epe_train(final_flow.float(), batch)
