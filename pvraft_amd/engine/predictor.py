@@ -56,7 +56,6 @@ class Predictor:
             self._in1.copy_(xyz1)
             self._in2.copy_(xyz2)
             self._capture()
-            return self._out.clone()
         self._in1.copy_(xyz1, non_blocking=True)
         self._in2.copy_(xyz2, non_blocking=True)
         self._graph.replay()
