@@ -45,7 +45,7 @@ class Graph:
             flat = self.idx.reshape(B, N * k)
             order = flat.argsort(dim=1)
             targets = flat.gather(1, order)
-            bounds = torch.arange(N + 1, device=flat.device).expand(B, N + 1)
+            bounds = torch.arange(N + 1, device=flat.device).expand(B, N + 1).contiguous()
             offsets = torch.searchsorted(targets, bounds, side="left")
             self._csr = (order.to(torch.int32).contiguous(), offsets.to(torch.int32).contiguous())
         return self._csr
