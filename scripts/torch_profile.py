@@ -1,0 +1,43 @@
+#!/usr/bin/env python3
+"""Op-level attribution of the train step (torch.profiler, 2 steps)."""
+
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from pvraft_amd.data import synthetic_batch
+from pvraft_amd.model import PVRaft
+from pvraft_amd.utils import sequence_loss
+
+
+def main():
+    device = torch.device("cuda:0")
+    model = PVRaft(truncate_k=512).to(device)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    batch = synthetic_batch(2, 8192, device=device)
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            flows = model(batch["sequence"], num_iters=8)
+            loss = sequence_loss(flows, batch, gamma=0.8)
+        loss.backward()
+        opt.step()
+
+    for _ in range(3):
+        step()
+    torch.cuda.synchronize()
+    from torch.profiler import ProfilerActivity, profile
+
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
+        for _ in range(2):
+            step()
+        torch.cuda.synchronize()
+    print(prof.key_averages().table(sort_by="cuda_time_total", row_limit=32, max_name_column_width=55))
+
+
+if __name__ == "__main__":
+    main()
