@@ -31,6 +31,8 @@ void launch_gnmp_fwd(const void*, void*, unsigned char*, float*, float*,
                      float*, const float*, const float*, int, long, long, int,
                      int, int, float, int, float, const float*, bool,
                      hipStream_t);
+void launch_pw_wgrad(const void*, const void*, float*, int, int, int, long,
+                     hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
                      float*, float*, float*, void*, int, long, long, int, int,
@@ -307,7 +309,23 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
   return {dx, dweight, dbias, slope_ws};
 }
 
+// dy (B, Co, S) bf16, x (B, Ci, S) bf16 -> dW (Co, Ci) fp32 (split-K MFMA)
+torch::Tensor pw_wgrad(torch::Tensor dy, torch::Tensor x) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 3);
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 && x.scalar_type() == torch::kBFloat16,
+              "pw_wgrad is bf16-only");
+  TORCH_CHECK(dy.size(0) == x.size(0) && dy.size(2) == x.size(2), "shape mismatch");
+  const int B = dy.size(0), Co = dy.size(1), Ci = x.size(1);
+  const long S = dy.size(2);
+  auto dw = torch::zeros({Co, Ci}, dy.options().dtype(torch::kFloat32));
+  launch_pw_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), B, Co,
+                  Ci, S, stream());
+  return dw;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("pw_wgrad", &pw_wgrad);
   m.def("group_norm_act_fwd", &group_norm_act_fwd);
   m.def("group_norm_act_bwd", &group_norm_act_bwd);
   m.def("group_norm_act_maxpool_fwd", &group_norm_act_maxpool_fwd);

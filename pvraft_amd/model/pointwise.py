@@ -39,14 +39,13 @@ class _PwMatmul(torch.autograd.Function):
         dx = torch.matmul(w.t(), dy) if ctx.needs_input_grad[1] else None
         dw = None
         if ctx.needs_input_grad[0]:
-            B, Ci, S = x.shape
-            Co = dy.shape[1]
-            if x.is_cuda and S % SPLITK == 0:
-                Sc = S // SPLITK
-                # (B*SPLITK, Co, Sc) @ (B*SPLITK, Sc, Ci) -> sum over batch
-                dy_c = dy.view(B, Co, SPLITK, Sc).permute(0, 2, 1, 3)
-                x_c = x.view(B, Ci, SPLITK, Sc).permute(0, 2, 3, 1)
-                dw = torch.matmul(dy_c, x_c).sum(dim=(0, 1)).to(w.dtype)
+            if x.is_cuda and x.dtype == torch.bfloat16:
+                try:
+                    from pvraft_amd import _C
+
+                    dw = _C.pw_wgrad(dy, x).to(w.dtype)
+                except (ImportError, AttributeError):
+                    dw = torch.einsum("bos,bis->oi", dy, x).to(w.dtype)
             else:
                 dw = torch.einsum("bos,bis->oi", dy, x).to(w.dtype)
         return dw, dx

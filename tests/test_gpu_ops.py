@@ -216,3 +216,17 @@ def test_group_norm_act_maxpool_fwd_bwd(dtype):
     wtol = 2e-1 if dtype == torch.float32 else 1.5
     assert torch.allclose(w.grad, wr.grad, rtol=5e-2, atol=wtol), (w.grad - wr.grad).abs().max()
     assert torch.allclose(b.grad, br.grad, rtol=5e-2, atol=wtol), (b.grad - br.grad).abs().max()
+
+
+@pytest.mark.parametrize("B,Co,Ci,S", [(2, 64, 192, 16384), (2, 61, 128, 8192), (1, 128, 131, 524288 // 4), (3, 16, 9, 1000)])
+def test_pw_wgrad_mfma_matches_einsum(B, Co, Ci, S):
+    from pvraft_amd import _C
+
+    dy = torch.randn(B, Co, S, device=dev(), dtype=torch.bfloat16)
+    x = torch.randn(B, Ci, S, device=dev(), dtype=torch.bfloat16)
+    got = _C.pw_wgrad(dy, x)
+    want = torch.einsum("bos,bis->oi", dy.float(), x.float())
+    # bf16 inputs, fp32 accumulation both sides; atomic split-K ordering
+    err = (got - want).abs().max().item()
+    denom = want.abs().max().item()
+    assert err < 0.02 * max(denom, 1.0), (err, denom)
