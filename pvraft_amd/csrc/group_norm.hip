@@ -516,11 +516,14 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
 // --------------------------------------------------------------- launchers
 
 static int pick_chunks(long spatial, long bc) {
+  // fill the chip (~4096 blocks) but keep >= ~16 vector iterations per
+  // block: excess blocks multiply the per-block workspace atomics, which
+  // serialize on the tiny (rows,2)/(C,2) arrays
   long want = 4096 / (bc > 0 ? bc : 1);
-  long cap = (spatial + GN_THREADS - 1) / GN_THREADS;
-  long chunks = want < 1 ? 1 : want;
-  if (chunks > cap) chunks = cap;
-  return (int)(chunks < 1 ? 1 : chunks);
+  long cap = spatial / ((long)GN_THREADS * 16);
+  long chunks = want < cap ? want : cap;
+  if (chunks < 1) chunks = 1;
+  return (int)chunks;
 }
 
 template <typename T>

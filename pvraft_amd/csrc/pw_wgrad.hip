@@ -123,10 +123,15 @@ void launch_pw_wgrad(const void *dy, const void *x, float *dw, int B, int Co,
                      int Ci, long S, hipStream_t stream) {
   const int to = (Co + TILE - 1) / TILE;
   const int ti = (Ci + TILE - 1) / TILE;
-  // enough K-splits to fill 256 CUs; each chunk should cover >= ~1 KB rows
-  long want = 1024 / ((long)to * ti * B) + 1;
-  long cap = (S + 255) / 256;
-  int schunks = (int)(want < 1 ? 1 : (want > cap ? (cap < 1 ? 1 : cap) : want));
+  // K-splits: fill CUs but cap the per-output atomic depth (B * schunks
+  // blocks all atomicAdd the same dW tile -- deep splits serialize) and
+  // keep >= 16 K-blocks of work per chunk
+  long want = 512 / ((long)to * ti * B) + 1;
+  if (want * B > 32) want = 32 / (B > 0 ? B : 1) + 1;
+  long cap = S / (KB * 16);
+  long sc = want < cap ? want : cap;
+  if (sc < 1) sc = 1;
+  int schunks = (int)sc;
   dim3 grid(to, ti, B * schunks);
   hipLaunchKernelGGL(pw_wgrad_kernel, grid, dim3(WG_THREADS), 0, stream,
                      (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, dw,
