@@ -32,7 +32,7 @@ void launch_gnmp_fwd(const void*, void*, unsigned char*, float*, float*,
                      int, int, float, int, float, const float*, bool,
                      hipStream_t);
 void launch_pw_wgrad(const void*, const void*, float*, int, int, int, long,
-                     hipStream_t);
+                     int, hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
                      float*, float*, float*, void*, int, long, long, int, int,
@@ -310,7 +310,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
 }
 
 // dy (B, Co, S) bf16, x (B, Ci, S) bf16 -> dW (Co, Ci) fp32 (split-K MFMA)
-torch::Tensor pw_wgrad(torch::Tensor dy, torch::Tensor x) {
+torch::Tensor pw_wgrad(torch::Tensor dy, torch::Tensor x, int64_t schunks = 0) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 3);
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
   TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 && x.scalar_type() == torch::kBFloat16,
@@ -320,12 +320,12 @@ torch::Tensor pw_wgrad(torch::Tensor dy, torch::Tensor x) {
   const long S = dy.size(2);
   auto dw = torch::zeros({Co, Ci}, dy.options().dtype(torch::kFloat32));
   launch_pw_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), B, Co,
-                  Ci, S, stream());
+                  Ci, S, (int)schunks, stream());
   return dw;
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("pw_wgrad", &pw_wgrad);
+  m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0);
   m.def("group_norm_act_fwd", &group_norm_act_fwd);
   m.def("group_norm_act_bwd", &group_norm_act_bwd);
   m.def("group_norm_act_maxpool_fwd", &group_norm_act_maxpool_fwd);

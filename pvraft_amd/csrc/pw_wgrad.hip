@@ -120,17 +120,17 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
 }
 
 void launch_pw_wgrad(const void *dy, const void *x, float *dw, int B, int Co,
-                     int Ci, long S, hipStream_t stream) {
+                     int Ci, long S, int schunks_opt, hipStream_t stream) {
   const int to = (Co + TILE - 1) / TILE;
   const int ti = (Ci + TILE - 1) / TILE;
-  // K-splits: fill CUs but cap the per-output atomic depth (B * schunks
-  // blocks all atomicAdd the same dW tile -- deep splits serialize) and
-  // keep >= 16 K-blocks of work per chunk
-  long want = 512 / ((long)to * ti * B) + 1;
-  if (want * B > 32) want = 32 / (B > 0 ? B : 1) + 1;
-  long cap = S / (KB * 16);
-  long sc = want < cap ? want : cap;
-  if (sc < 1) sc = 1;
+  long sc = schunks_opt;
+  if (sc <= 0) {
+    // default: fill ~512 workgroups, but keep >= 4 K-blocks per chunk
+    sc = 512 / ((long)to * ti * B) + 1;
+    long cap = S / (KB * 4);
+    if (sc > cap) sc = cap;
+    if (sc < 1) sc = 1;
+  }
   int schunks = (int)sc;
   dim3 grid(to, ti, B * schunks);
   hipLaunchKernelGGL(pw_wgrad_kernel, grid, dim3(WG_THREADS), 0, stream,

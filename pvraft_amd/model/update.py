@@ -5,6 +5,14 @@ flow appended (update.py:15-20).  ConvGRU: pointwise gates over B x C x N
 (update.py:31-40).  FlowHead: Conv1d branch || SetConv branch on the context
 graph, concat -> Conv1d 128->64 -> ReLU -> Conv1d 64->3 (update.py:57-72).
 Attribute names match reference state dicts.
+
+Concat-free execution: every concatenation in this block feeds a 1x1 conv,
+and W @ cat(a, b) == W_a @ a + W_b @ b with W_a/W_b column slices of the
+weight -- so the big per-iteration cat tensors (hx, [r*h; x], [cor; flo],
+[out_set; out], each ~2-6 MB x 8 GRU iterations) are never materialised;
+the gate preactivations are sums of partial GEMMs on the existing tensors.
+Weight slicing is a view (no copy); values and state dicts are identical
+to the reference formulation up to fp summation order.
 """
 
 from __future__ import annotations
@@ -15,8 +23,27 @@ import torch.nn.functional as F
 from torch import Tensor
 
 from .graph import Graph
-from .pointwise import PwConv1d
+from .pointwise import PwConv1d, pw_matmul
 from .setconv import SetConv
+
+
+def _split_mm(weight: Tensor, bias, parts):
+    """sum_i W[:, lo_i:hi_i] @ parts[i] (+ bias): the concat-free 1x1 conv.
+
+    weight: (Co, Ci_total, 1) conv weight; parts: list of (B, Ci_i, N)
+    tensors whose channel sizes sum to Ci_total.
+    """
+    w = weight.squeeze(-1)
+    lo = 0
+    out = None
+    for p in parts:
+        hi = lo + p.shape[1]
+        term = pw_matmul(w[:, lo:hi].contiguous(), p)
+        out = term if out is None else out + term
+        lo = hi
+    if bias is not None:
+        out = out + bias.view(1, -1, 1)
+    return out
 
 
 class MotionEncoder(nn.Module):
@@ -26,13 +53,17 @@ class MotionEncoder(nn.Module):
         self.conv_flow = PwConv1d(3, 64, 1)
         self.conv = PwConv1d(64 + 64, 64 - 3, 1)
 
-    def forward(self, flow: Tensor, corr: Tensor) -> Tensor:
-        """flow (B, N, 3), corr (B, 64, N) -> (B, 64, N)."""
+    def forward(self, flow: Tensor, corr: Tensor):
+        """flow (B, N, 3), corr (B, 64, N) -> (motion61 (B, 61, N), flow_t (B, 3, N)).
+
+        The reference returns cat([motion61, flow_t]) (update.py:19-20); the
+        parts are kept separate here and consumed slice-wise downstream.
+        """
         flow_t = flow.transpose(1, 2).contiguous()
         cor = F.relu(self.conv_corr(corr))
         flo = F.relu(self.conv_flow(flow_t))
-        out = F.relu(self.conv(torch.cat([cor, flo], dim=1)))
-        return torch.cat([out, flow_t], dim=1)
+        out = F.relu(_split_mm(self.conv.weight, self.conv.bias, [cor, flo]))
+        return out, flow_t
 
 
 class ConvGRU(nn.Module):
@@ -43,10 +74,15 @@ class ConvGRU(nn.Module):
         self.convq = PwConv1d(input_dim + hidden_dim, hidden_dim, 1)
 
     def forward(self, h: Tensor, x: Tensor) -> Tensor:
-        hx = torch.cat([h, x], dim=1)
-        z = torch.sigmoid(self.convz(hx))
-        r = torch.sigmoid(self.convr(hx))
-        q = torch.tanh(self.convq(torch.cat([r * h, x], dim=1)))
+        """Reference formulation (update.py:31-40) for direct use."""
+        return self.forward_parts(h, [x])
+
+    def forward_parts(self, h: Tensor, x_parts) -> Tensor:
+        """Gates from the concat parts: conv(cat(h, *x_parts)) done as
+        summed partial GEMMs (weight column slices)."""
+        z = torch.sigmoid(_split_mm(self.convz.weight, self.convz.bias, [h] + list(x_parts)))
+        r = torch.sigmoid(_split_mm(self.convr.weight, self.convr.bias, [h] + list(x_parts)))
+        q = torch.tanh(_split_mm(self.convq.weight, self.convq.bias, [r * h] + list(x_parts)))
         return (1 - z) * h + z * q
 
 
@@ -77,7 +113,9 @@ class FlowHead(nn.Module):
     def forward(self, x: Tensor, graph: Graph) -> Tensor:
         out = self.conv1(x)
         out_set = self.setconv(x.transpose(1, 2).contiguous(), graph).transpose(1, 2).contiguous()
-        return self.out_conv(torch.cat([out_set, out], dim=1))
+        # out_conv(cat([out_set, out])) without the cat
+        mid = F.relu(_split_mm(self.out_conv[0].weight, self.out_conv[0].bias, [out_set, out]))
+        return self.out_conv[2](mid)
 
 
 class UpdateBlock(nn.Module):
@@ -88,7 +126,8 @@ class UpdateBlock(nn.Module):
         self.flow_head = FlowHead(input_dim=hidden_dim)
 
     def forward(self, net: Tensor, inp: Tensor, corr: Tensor, flow: Tensor, graph: Graph):
-        motion = self.motion_encoder(flow, corr)
-        net = self.gru(net, torch.cat([inp, motion], dim=1))
+        motion, flow_t = self.motion_encoder(flow, corr)
+        # gru input = cat(inp, motion, flow_t) (reference update.py:84), as parts
+        net = self.gru.forward_parts(net, [inp, motion, flow_t])
         delta_flow = self.flow_head(net, graph).transpose(1, 2).contiguous()
         return net, delta_flow
