@@ -43,9 +43,12 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
   const int tile_i = blockIdx.y * TILE;
   const int b = blockIdx.z / schunks;
   const int chunk = blockIdx.z % schunks;
-  const long per = (S + schunks - 1) / schunks;
+  // chunk bounds KB-aligned so in-range stage loads are whole 16B vectors
+  const long per = (((S + schunks - 1) / schunks + KB - 1) / KB) * KB;
   const long s_lo = chunk * per;
   const long s_hi = min(s_lo + per, S);
+  if (s_lo >= S) return;
+  const bool vec_ok = (S % 8 == 0);
 
   const int lane = lane_id();
   const int wid = wave_id();
@@ -70,21 +73,30 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
     {
       const int o = tile_o + ldr;
       const long s = s0 + ldc;
+      const bool full = vec_ok && (s + 8 <= s_hi);
       bf16x8 av = (bf16x8)(__bf16)0.0f;
       if (o < Co) {
         const __hip_bfloat16 *src = dyb + (long)o * S + s;
+        if (full) {
+          av = *(const bf16x8 *)src;  // one coalesced 16 B load
+        } else {
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          if (s + e < s_hi) ((__bf16 *)&av)[e] = *(const __bf16 *)(src + e);
+          for (int e = 0; e < 8; ++e)
+            if (s + e < s_hi) ((__bf16 *)&av)[e] = *(const __bf16 *)(src + e);
+        }
       }
       *(bf16x8 *)&sA[ldr][ldc] = av;
       const int i = tile_i + ldr;
       bf16x8 bv = (bf16x8)(__bf16)0.0f;
       if (i < Ci) {
         const __hip_bfloat16 *src = xb + (long)i * S + s;
+        if (full) {
+          bv = *(const bf16x8 *)src;
+        } else {
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          if (s + e < s_hi) ((__bf16 *)&bv)[e] = *(const __bf16 *)(src + e);
+          for (int e = 0; e < 8; ++e)
+            if (s + e < s_hi) ((__bf16 *)&bv)[e] = *(const __bf16 *)(src + e);
+        }
       }
       *(bf16x8 *)&sB[ldr][ldc] = bv;
     }
