@@ -3,8 +3,7 @@
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
 
-void launch_knn_graph(const float*, float*, int*, int*, int, int, int, int,
-                      hipStream_t);
+void launch_knn_graph(const float*, int*, int, int, int, hipStream_t);
 void launch_gather_edge_fwd(const void*, const int*, const float*, void*,
                             int, int, int, int, bool, hipStream_t);
 void launch_gather_edge_bwd(const void*, const int*, float*, int, int, int,
@@ -55,16 +54,9 @@ torch::Tensor knn_graph(torch::Tensor xyz, int64_t k) {
   TORCH_CHECK(xyz.dim() == 3 && xyz.size(2) == 3, "xyz must be (B,N,3)");
   const int B = xyz.size(0), N = xyz.size(1);
   TORCH_CHECK(k >= 1 && k <= 48 && k <= N, "knn_graph requires 1 <= k <= 48, k <= N");
-  // candidate-range splits: enough phase-1 blocks to fill 256 CUs
-  const long base_blocks = (long)((N + 127) / 128) * B;
-  int splits = 1;
-  while (splits < 16 && base_blocks * splits < 1024) splits *= 2;
   auto out = torch::empty({B, N, k}, xyz.options().dtype(torch::kInt32));
-  auto part_dist = torch::empty({B, N, splits, k}, xyz.options());
-  auto part_idx = torch::empty({B, N, splits, k}, xyz.options().dtype(torch::kInt32));
-  launch_knn_graph(xyz.data_ptr<float>(), part_dist.data_ptr<float>(),
-                   part_idx.data_ptr<int>(), out.data_ptr<int>(), B, N, (int)k,
-                   splits, stream());
+  launch_knn_graph(xyz.data_ptr<float>(), out.data_ptr<int>(), B, N, (int)k,
+                   stream());
   return out;
 }
 
