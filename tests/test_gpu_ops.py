@@ -230,3 +230,24 @@ def test_pw_wgrad_mfma_matches_einsum(B, Co, Ci, S):
     err = (got - want).abs().max().item()
     denom = want.abs().max().item()
     assert err < 0.02 * max(denom, 1.0), (err, denom)
+
+
+def test_knn_graph_degenerate_ties():
+    """All-identical points: every distance ties; the kernel must still emit
+    k valid indices per query (histogram refine + boundary overflow path)."""
+    xyz = torch.zeros(1, 300, 3, device=dev())
+    idx = ops.knn_graph(xyz, 16)
+    assert idx.shape == (1, 300, 16)
+    assert (idx >= 0).all() and (idx < 300).all()
+
+    # lattice: many exact ties at each shell
+    g = torch.stack(torch.meshgrid(
+        torch.arange(8.0), torch.arange(8.0), torch.arange(8.0), indexing="ij"
+    ), -1).reshape(1, -1, 3).to(dev())
+    idx = ops.knn_graph(g, 8)
+    want = R.knn_idx(g, 8)
+    B, N, k = idx.shape
+    sq = lambda i: (
+        (g.unsqueeze(2) - g.gather(1, i.reshape(B, -1, 1).expand(B, N * k, 3).long()).view(B, N, k, 3)) ** 2
+    ).sum(-1).sort(-1).values
+    assert torch.allclose(sq(idx), sq(want), atol=1e-4)
