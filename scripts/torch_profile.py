@@ -32,11 +32,17 @@ def main():
     torch.cuda.synchronize()
     from torch.profiler import ProfilerActivity, profile
 
-    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA], record_shapes=True) as prof:
         for _ in range(2):
             step()
         torch.cuda.synchronize()
-    print(prof.key_averages().table(sort_by="cuda_time_total", row_limit=32, max_name_column_width=55))
+    print(prof.key_averages().table(sort_by="cuda_time_total", row_limit=25, max_name_column_width=45))
+    print("=== by shape (copy-ish ops) ===")
+    ka = prof.key_averages(group_by_input_shape=True)
+    rows = [e for e in ka if e.key in ("aten::copy_", "aten::contiguous", "aten::clone", "aten::cat")]
+    rows.sort(key=lambda e: -e.self_device_time_total)
+    for e in rows[:25]:
+        print(f"{e.key:18s} {str(e.input_shapes)[:90]:90s} {e.count:>5} {e.self_device_time_total/1000.0:8.2f}ms")
 
 
 if __name__ == "__main__":
