@@ -32,7 +32,7 @@ def main():
     torch.cuda.synchronize()
     from torch.profiler import ProfilerActivity, profile
 
-    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA], record_shapes=True) as prof:
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA], record_shapes=True, with_stack=os.environ.get("PVRAFT_PROF_STACK", "0") == "1") as prof:
         for _ in range(2):
             step()
         torch.cuda.synchronize()
@@ -43,6 +43,15 @@ def main():
     rows.sort(key=lambda e: -e.self_device_time_total)
     for e in rows[:25]:
         print(f"{e.key:18s} {str(e.input_shapes)[:90]:90s} {e.count:>5} {e.self_device_time_total/1000.0:8.2f}ms")
+    if os.environ.get("PVRAFT_PROF_STACK", "0") == "1":
+        print("=== stacks for hot copies ===")
+        ka2 = prof.key_averages(group_by_stack_n=6)
+        rows2 = [e for e in ka2 if e.key == "aten::copy_"]
+        rows2.sort(key=lambda e: -e.self_device_time_total)
+        for e in rows2[:8]:
+            print(f"--- {e.count} calls, {e.self_device_time_total/1000.0:.2f}ms")
+            for line in (e.stack or [])[:6]:
+                print("   ", line)
 
 
 if __name__ == "__main__":
