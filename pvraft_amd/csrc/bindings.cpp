@@ -30,6 +30,8 @@ void launch_gnmp_fwd(const void*, void*, unsigned char*, float*, float*,
                      float*, const float*, const float*, int, long, long, int,
                      int, int, float, int, float, const float*, bool,
                      hipStream_t);
+void launch_transpose(const void*, void*, long, int, long, long, bool,
+                      hipStream_t);
 void launch_pw_wgrad(const void*, const void*, float*, int, int, int, long,
                      int, hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
@@ -316,8 +318,24 @@ torch::Tensor pw_wgrad(torch::Tensor dy, torch::Tensor x, int64_t schunks = 0) {
   return dw;
 }
 
+// (B, R, C) view (row-contiguous, arbitrary batch stride) -> (B, C, R)
+torch::Tensor batched_transpose(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 3, "x must be 3-D GPU");
+  TORCH_CHECK(x.stride(2) == 1 && x.stride(1) == x.size(2),
+              "rows must be contiguous");
+  const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32, "fp32/bf16 only");
+  const int B = x.size(0);
+  const long R = x.size(1), C = x.size(2);
+  auto out = torch::empty({B, C, R}, x.options());
+  launch_transpose(x.data_ptr(), out.data_ptr(), x.stride(0), B, R, C, bf16,
+                   stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0);
+  m.def("batched_transpose", &batched_transpose);
   m.def("group_norm_act_fwd", &group_norm_act_fwd);
   m.def("group_norm_act_bwd", &group_norm_act_bwd);
   m.def("group_norm_act_maxpool_fwd", &group_norm_act_maxpool_fwd);

@@ -55,17 +55,18 @@ __global__ void gather_edge_fwd_kernel(
 }
 
 // CSR backward: deterministic, atomic-free.  The caller transposes the
-// first C channels of gout to gT (B, N*K, C) (each edge's gradient is a
-// contiguous C-vector) and provides the inverse adjacency in CSR form:
-// order (B, N*K) = edge ids sorted by target node, offsets (B, N+1).
+// first C channels of gout to gT (B, K*N, C) (each edge's gradient is a
+// contiguous C-vector; edge id = j*N + n) and provides the inverse
+// adjacency in CSR form: order (B, K*N) = edge ids sorted by target node,
+// offsets (B, N+1).
 //   grad[b, m, c] = sum_{t in [off[m], off[m+1])} gT[b, order[t], c]
-//                 - sum_j gT[b, m*K + j, c]
+//                 - sum_j gT[b, j*N + m, c]
 // Thread = (b, m, c) with c fastest: both loops read contiguous C-vectors,
 // wave-coalesced; writes are unique.
 template <typename T>
 __global__ void gather_edge_bwd_csr_kernel(
-    const T *__restrict__ gT,          // (B, N*K, C)
-    const int *__restrict__ order,     // (B, N*K)
+    const T *__restrict__ gT,          // (B, K*N, C)
+    const int *__restrict__ order,     // (B, K*N)
     const int *__restrict__ offsets,   // (B, N+1)
     T *__restrict__ grad,              // (B, N, C)
     int B, int N, int K, int C) {
@@ -83,9 +84,8 @@ __global__ void gather_edge_bwd_csr_kernel(
   const int hi = offsets[(long)b * (N + 1) + m + 1];
   float acc = 0.f;
   for (int t = lo; t < hi; ++t) acc += ldf(gbase + (long)ord[t] * C + c);
-  const T *own = gbase + ((long)m * K) * C + c;
   float cen = 0.f;
-  for (int j = 0; j < K; ++j) cen += ldf(own + (long)j * C);
+  for (int j = 0; j < K; ++j) cen += ldf(gbase + ((long)j * N + m) * C + c);
   stf(grad + gid, acc - cen);
 }
 

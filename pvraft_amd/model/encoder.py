@@ -33,7 +33,9 @@ class PointEncoder(nn.Module):
         """
         if graph is None:
             graph = Graph.build(pc, self.num_neighbors)
+        from pvraft_amd import ops
+
         x = self.feat_conv1(pc, graph)
         x = self.feat_conv2(x, graph)
         x = self.feat_conv3(x, graph)
-        return x.transpose(1, 2).contiguous(), graph
+        return ops.transpose_last2(x), graph

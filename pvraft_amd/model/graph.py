@@ -42,7 +42,9 @@ class Graph:
             return None
         if self._csr is None:
             B, N, k = self.idx.shape
-            flat = self.idx.reshape(B, N * k)
+            # edge ids in (j, n) order: id = j*N + n, matching the
+            # (B, C, K*N) -> (B, K*N, C) transpose of the gradient tensor
+            flat = self.idx.permute(0, 2, 1).reshape(B, k * N)
             order = flat.argsort(dim=1)
             targets = flat.gather(1, order)
             bounds = torch.arange(N + 1, device=flat.device).expand(B, N + 1).contiguous()

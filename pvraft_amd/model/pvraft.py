@@ -75,11 +75,12 @@ class PVRaft(nn.Module):
 
         coords1, coords2 = xyz1, xyz1
         flow_predictions = []
+        wcache = self.update_block.make_wcache()
         for _ in range(num_iters):
             coords2 = coords2.detach()
             corr = self.corr_block(field, coords2)
             flow = coords2 - coords1
-            net, delta_flow = self.update_block(net, inp, corr, flow, graph_context)
+            net, delta_flow = self.update_block(net, inp, corr, flow, graph_context, wcache)
             coords2 = coords2 + delta_flow
             flow_predictions.append(coords2 - coords1)
         return flow_predictions
@@ -142,11 +143,12 @@ class PVRaftRefine(nn.Module):
             inp = torch.relu(inp)
 
             coords1, coords2 = xyz1, xyz1
+            wcache = self.update_block.make_wcache()
             for _ in range(num_iters):
                 coords2 = coords2.detach()
                 corr = self.corr_block(field, coords2)
                 flow = coords2 - coords1
-                net, delta_flow = self.update_block(net, inp, corr, flow, graph_context)
+                net, delta_flow = self.update_block(net, inp, corr, flow, graph_context, wcache)
                 coords2 = coords2 + delta_flow
         return self.refine_block(coords2 - coords1, graph1)
 

@@ -27,20 +27,31 @@ from .pointwise import PwConv1d, pw_matmul
 from .setconv import SetConv
 
 
-def _split_mm(weight: Tensor, bias, parts):
+def _slice_weight(weight: Tensor, sizes):
+    """Contiguous column slices of a (Co, Ci, 1) conv weight."""
+    w = weight.squeeze(-1)
+    out, lo = [], 0
+    for s_ in sizes:
+        out.append(w[:, lo : lo + s_].contiguous())
+        lo += s_
+    return out
+
+
+def _split_mm(weight: Tensor, bias, parts, wparts=None):
     """sum_i W[:, lo_i:hi_i] @ parts[i] (+ bias): the concat-free 1x1 conv.
 
     weight: (Co, Ci_total, 1) conv weight; parts: list of (B, Ci_i, N)
-    tensors whose channel sizes sum to Ci_total.
+    tensors whose channel sizes sum to Ci_total.  ``wparts`` supplies
+    pre-sliced contiguous weights (cached once per forward -- the GRU loop
+    reuses the same slices for all iterations, which also lets autocast's
+    weight-cast cache hit).
     """
-    w = weight.squeeze(-1)
-    lo = 0
+    if wparts is None:
+        wparts = _slice_weight(weight, [p.shape[1] for p in parts])
     out = None
-    for p in parts:
-        hi = lo + p.shape[1]
-        term = pw_matmul(w[:, lo:hi].contiguous(), p)
+    for w_i, p in zip(wparts, parts):
+        term = pw_matmul(w_i, p)
         out = term if out is None else out + term
-        lo = hi
     if bias is not None:
         out = out + bias.view(1, -1, 1)
     return out
@@ -53,16 +64,19 @@ class MotionEncoder(nn.Module):
         self.conv_flow = PwConv1d(3, 64, 1)
         self.conv = PwConv1d(64 + 64, 64 - 3, 1)
 
-    def forward(self, flow: Tensor, corr: Tensor):
+    def forward(self, flow: Tensor, corr: Tensor, wcache=None):
         """flow (B, N, 3), corr (B, 64, N) -> (motion61 (B, 61, N), flow_t (B, 3, N)).
 
         The reference returns cat([motion61, flow_t]) (update.py:19-20); the
         parts are kept separate here and consumed slice-wise downstream.
         """
-        flow_t = flow.transpose(1, 2).contiguous()
+        from pvraft_amd import ops
+
+        flow_t = ops.transpose_last2(flow)
         cor = F.relu(self.conv_corr(corr))
         flo = F.relu(self.conv_flow(flow_t))
-        out = F.relu(_split_mm(self.conv.weight, self.conv.bias, [cor, flo]))
+        wp = wcache.get("motion") if wcache else None
+        out = F.relu(_split_mm(self.conv.weight, self.conv.bias, [cor, flo], wp))
         return out, flow_t
 
 
@@ -77,12 +91,15 @@ class ConvGRU(nn.Module):
         """Reference formulation (update.py:31-40) for direct use."""
         return self.forward_parts(h, [x])
 
-    def forward_parts(self, h: Tensor, x_parts) -> Tensor:
+    def forward_parts(self, h: Tensor, x_parts, wcache=None) -> Tensor:
         """Gates from the concat parts: conv(cat(h, *x_parts)) done as
         summed partial GEMMs (weight column slices)."""
-        z = torch.sigmoid(_split_mm(self.convz.weight, self.convz.bias, [h] + list(x_parts)))
-        r = torch.sigmoid(_split_mm(self.convr.weight, self.convr.bias, [h] + list(x_parts)))
-        q = torch.tanh(_split_mm(self.convq.weight, self.convq.bias, [r * h] + list(x_parts)))
+        wz = wcache.get("z") if wcache else None
+        wr = wcache.get("r") if wcache else None
+        wq = wcache.get("q") if wcache else None
+        z = torch.sigmoid(_split_mm(self.convz.weight, self.convz.bias, [h] + list(x_parts), wz))
+        r = torch.sigmoid(_split_mm(self.convr.weight, self.convr.bias, [h] + list(x_parts), wr))
+        q = torch.tanh(_split_mm(self.convq.weight, self.convq.bias, [r * h] + list(x_parts), wq))
         return (1 - z) * h + z * q
 
 
@@ -110,11 +127,14 @@ class FlowHead(nn.Module):
             PwConv1d(64, 3, 1),
         )
 
-    def forward(self, x: Tensor, graph: Graph) -> Tensor:
+    def forward(self, x: Tensor, graph: Graph, wcache=None) -> Tensor:
+        from pvraft_amd import ops
+
         out = self.conv1(x)
-        out_set = self.setconv(x.transpose(1, 2).contiguous(), graph).transpose(1, 2).contiguous()
+        out_set = ops.transpose_last2(self.setconv(ops.transpose_last2(x), graph))
         # out_conv(cat([out_set, out])) without the cat
-        mid = F.relu(_split_mm(self.out_conv[0].weight, self.out_conv[0].bias, [out_set, out]))
+        wp = wcache.get("flowhead") if wcache else None
+        mid = F.relu(_split_mm(self.out_conv[0].weight, self.out_conv[0].bias, [out_set, out], wp))
         return self.out_conv[2](mid)
 
 
@@ -125,9 +145,24 @@ class UpdateBlock(nn.Module):
         self.gru = ConvGRU(input_dim=input_dim, hidden_dim=hidden_dim)
         self.flow_head = FlowHead(input_dim=hidden_dim)
 
-    def forward(self, net: Tensor, inp: Tensor, corr: Tensor, flow: Tensor, graph: Graph):
-        motion, flow_t = self.motion_encoder(flow, corr)
+    def make_wcache(self):
+        """Slice the concat-consuming conv weights once per forward (the GRU
+        loop reuses them across all iterations)."""
+        gate_sizes = [64, 64, 61, 3]  # [h, inp, motion61, flow3]
+        return {
+            "z": _slice_weight(self.gru.convz.weight, gate_sizes),
+            "r": _slice_weight(self.gru.convr.weight, gate_sizes),
+            "q": _slice_weight(self.gru.convq.weight, gate_sizes),
+            "motion": _slice_weight(self.motion_encoder.conv.weight, [64, 64]),
+            "flowhead": _slice_weight(self.flow_head.out_conv[0].weight, [64, 64]),
+        }
+
+    def forward(self, net: Tensor, inp: Tensor, corr: Tensor, flow: Tensor, graph: Graph,
+                wcache=None):
+        from pvraft_amd import ops
+
+        motion, flow_t = self.motion_encoder(flow, corr, wcache)
         # gru input = cat(inp, motion, flow_t) (reference update.py:84), as parts
-        net = self.gru.forward_parts(net, [inp, motion, flow_t])
-        delta_flow = self.flow_head(net, graph).transpose(1, 2).contiguous()
+        net = self.gru.forward_parts(net, [inp, motion, flow_t], wcache)
+        delta_flow = ops.transpose_last2(self.flow_head(net, graph, wcache))
         return net, delta_flow

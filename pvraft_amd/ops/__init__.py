@@ -65,6 +65,33 @@ def hip_available() -> bool:
 # ---------------------------------------------------------------------------
 
 
+class _BatchedTranspose(torch.autograd.Function):
+    """LDS-tiled (B, R, C) -> (B, C, R) transpose (ATen's strided copy
+    path measured ~145 GB/s on this pattern)."""
+
+    @staticmethod
+    def forward(ctx, x: Tensor) -> Tensor:
+        return _EXT.batched_transpose(x)
+
+    @staticmethod
+    def backward(ctx, dy: Tensor):
+        return _EXT.batched_transpose(dy.contiguous())
+
+
+def transpose_last2(x: Tensor) -> Tensor:
+    """(B, R, C) -> (B, C, R), contiguous result."""
+    if (
+        x.is_cuda
+        and x.dtype in (torch.float32, torch.bfloat16)
+        and x.stride(2) == 1
+        and x.stride(1) == x.shape[2]
+        and os.environ.get("PVRAFT_REF_OPS", "0") != "1"
+        and _load_ext() is not None
+    ):
+        return _BatchedTranspose.apply(x)
+    return x.transpose(1, 2).contiguous()
+
+
 class _GatherEdgeConcat(torch.autograd.Function):
     """out (B,C+3,K,N) = concat(feat[nbr]-feat[center], xyz[nbr]-xyz[center]).
 
@@ -86,7 +113,9 @@ class _GatherEdgeConcat(torch.autograd.Function):
         grad_out = grad_out.contiguous()
         if order is not None:
             B, _, K, N = grad_out.shape
-            gT = grad_out[:, :C].permute(0, 3, 2, 1).reshape(B, N * K, C).contiguous()
+            # edge id = j*N + n: a plain last-two-dims transpose of the
+            # (B, C, K*N) channel-slice view (LDS-tiled kernel)
+            gT = _EXT.batched_transpose(grad_out[:, :C].reshape(B, C, K * N))
             g = _EXT.gather_edge_bwd_csr(gT, order, offsets, K)
         else:
             g = _EXT.gather_edge_concat_bwd(grad_out, idx, C)

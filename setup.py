@@ -21,6 +21,7 @@ SOURCES = [
     "pvraft_amd/csrc/knn_corr.hip",
     "pvraft_amd/csrc/group_norm.hip",
     "pvraft_amd/csrc/pw_wgrad.hip",
+    "pvraft_amd/csrc/transpose.hip",
 ]
 
 setup(
