@@ -149,13 +149,18 @@ class UpdateBlock(nn.Module):
         """Slice the concat-consuming conv weights once per forward (the GRU
         loop reuses them across all iterations)."""
         gate_sizes = [64, 64, 61, 3]  # [h, inp, motion61, flow3]
-        return {
+        cache = {
             "z": _slice_weight(self.gru.convz.weight, gate_sizes),
             "r": _slice_weight(self.gru.convr.weight, gate_sizes),
             "q": _slice_weight(self.gru.convq.weight, gate_sizes),
             "motion": _slice_weight(self.motion_encoder.conv.weight, [64, 64]),
             "flowhead": _slice_weight(self.flow_head.out_conv[0].weight, [64, 64]),
         }
+        # pre-cast once under autocast so per-iteration GEMMs skip the cast
+        if torch.is_autocast_enabled():
+            dt = torch.get_autocast_dtype("cuda")
+            cache = {k: [w.to(dt) for w in v] for k, v in cache.items()}
+        return cache
 
     def forward(self, net: Tensor, inp: Tensor, corr: Tensor, flow: Tensor, graph: Graph,
                 wcache=None):

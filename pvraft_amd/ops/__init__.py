@@ -80,6 +80,9 @@ class _BatchedTranspose(torch.autograd.Function):
 
 def transpose_last2(x: Tensor) -> Tensor:
     """(B, R, C) -> (B, C, R), contiguous result."""
+    # transpose-view of a contiguous tensor: the result IS the underlying
+    if x.stride(1) == 1 and x.stride(2) == x.shape[1]:
+        return x.transpose(1, 2)
     if (
         x.is_cuda
         and x.dtype in (torch.float32, torch.bfloat16)
@@ -287,7 +290,21 @@ def gather_edge_concat(feats: Tensor, idx: Tensor, xyz: Tensor, csr=None) -> Ten
     ``csr`` = (order, offsets) from Graph.csr() selects the deterministic
     atomic-free backward.
     """
-    feats = feats.contiguous()
+    if not feats.is_contiguous():
+        if (
+            feats.is_cuda
+            and feats.dim() == 3
+            and feats.stride(1) == 1
+            and feats.stride(2) == feats.shape[1]
+            and feats.dtype in (torch.float32, torch.bfloat16)
+            and os.environ.get("PVRAFT_REF_OPS", "0") != "1"
+            and _load_ext() is not None
+        ):
+            # (B, N, C) transpose-view of contiguous (B, C, N): transpose
+            # through the LDS-tiled kernel instead of ATen's strided copy
+            feats = _BatchedTranspose.apply(feats.transpose(1, 2))
+        else:
+            feats = feats.contiguous()
     if _use_hip(feats):
         if feats.dtype not in (torch.float32, torch.bfloat16):
             feats = feats.float()
