@@ -69,7 +69,9 @@ class PwConv1d(nn.Conv1d):
         # x (B, Cin, N) -> (B, Cout, N)
         y = pw_matmul(self.weight.squeeze(-1), x)
         if self.bias is not None:
-            y = y + self.bias.view(1, -1, 1)
+            # keep the GEMM dtype (an fp32 bias would promote the whole
+            # downstream chain out of bf16 under autocast)
+            y = y + self.bias.view(1, -1, 1).to(y.dtype)
         return y
 
 
@@ -80,5 +82,5 @@ class PwConv2d(nn.Conv2d):
         B, C, H, W = x.shape
         y = pw_matmul(self.weight.view(self.out_channels, C), x.reshape(B, C, H * W))
         if self.bias is not None:
-            y = y + self.bias.view(1, -1, 1)
+            y = y + self.bias.view(1, -1, 1).to(y.dtype)
         return y.view(B, self.out_channels, H, W)

@@ -53,7 +53,7 @@ def _split_mm(weight: Tensor, bias, parts, wparts=None):
         term = pw_matmul(w_i, p)
         out = term if out is None else out + term
     if bias is not None:
-        out = out + bias.view(1, -1, 1)
+        out = out + bias.view(1, -1, 1).to(out.dtype)
     return out
 
 
