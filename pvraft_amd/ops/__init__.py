@@ -360,7 +360,11 @@ class _CorrTruncate(torch.autograd.Function):
         gfull.scatter_(2, idx, (g_corr * scale).to(dt))
         f1 = fmap1.to(dt)
         f2 = fmap2.to(dt)
-        g1 = torch.bmm(f2, gfull.transpose(1, 2).contiguous()).float()  # (B,C,N)
+        if gfull.is_cuda and _load_ext() is not None:
+            gfull_t = _EXT.batched_transpose(gfull)
+        else:
+            gfull_t = gfull.transpose(1, 2).contiguous()
+        g1 = torch.bmm(f2, gfull_t).float()  # (B,C,N)
         g2 = torch.bmm(f1, gfull).float()  # (B,C,M)
         return g1, g2, None, None
 

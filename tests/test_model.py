@@ -137,3 +137,25 @@ def test_pw_conv_matches_nn_conv():
     yb.sum().backward()
     assert torch.allclose(x.grad, xb.grad, atol=1e-6)
     assert torch.allclose(conv2.weight.grad, pw2.weight.grad, atol=1e-5)
+
+
+def test_concat_free_gru_matches_cat_formulation():
+    """forward_parts (summed weight-slice GEMMs) == conv on torch.cat."""
+    from pvraft_amd.model.update import ConvGRU, _split_mm
+
+    torch.manual_seed(1)
+    gru = ConvGRU(input_dim=128, hidden_dim=64)
+    h = torch.randn(2, 64, 40)
+    inp = torch.randn(2, 64, 40)
+    motion = torch.randn(2, 61, 40)
+    flow_t = torch.randn(2, 3, 40)
+
+    out_parts = gru.forward_parts(h, [inp, motion, flow_t])
+    out_cat = gru(h, torch.cat([inp, motion, flow_t], dim=1))
+    assert torch.allclose(out_parts, out_cat, atol=1e-5)
+
+    # _split_mm == conv1d on the concatenation
+    conv = torch.nn.Conv1d(128, 61, 1)
+    got = _split_mm(conv.weight, conv.bias, [inp, motion, flow_t])
+    want = conv(torch.cat([inp, motion, flow_t], dim=1))
+    assert torch.allclose(got, want, atol=1e-5)
