@@ -30,6 +30,12 @@ void launch_gnmp_fwd(const void*, void*, unsigned char*, float*, float*,
                      float*, const float*, const float*, int, long, long, int,
                      int, int, float, int, float, const float*, bool,
                      hipStream_t);
+void launch_pv_corr_fused_fwd(const float*, const float*, const float*,
+                              float*, float*, int*, int, int, int, int, int,
+                              float, hipStream_t);
+void launch_pv_corr_fused_bwd(const float*, const float*, const float*,
+                              const float*, const int*, float*, int, int, int,
+                              int, int, float, hipStream_t);
 void launch_transpose(const void*, void*, long, int, long, long, bool,
                       hipStream_t);
 void launch_pw_wgrad(const void*, const void*, float*, int, int, int, long,
@@ -333,7 +339,45 @@ torch::Tensor batched_transpose(torch::Tensor x) {
   return out;
 }
 
+std::vector<torch::Tensor> pv_corr_fused_fwd(torch::Tensor corr,
+                                             torch::Tensor xyz,
+                                             torch::Tensor coords,
+                                             double base_scale,
+                                             int64_t num_levels, int64_t k) {
+  check_f32(corr, "corr");
+  check_f32(xyz, "xyz");
+  check_f32(coords, "coords");
+  const int B = corr.size(0), N = corr.size(1), K = corr.size(2);
+  TORCH_CHECK(K <= 512 && num_levels >= 1 && num_levels <= 4 && k >= 1 && k <= 64 && k <= K);
+  auto vox = torch::empty({B, num_levels * 27, N}, corr.options());
+  auto knn = torch::empty({B, 4, k, N}, corr.options());
+  auto idx = torch::empty({B, N, k}, corr.options().dtype(torch::kInt32));
+  launch_pv_corr_fused_fwd(corr.data_ptr<float>(), xyz.data_ptr<float>(),
+                           coords.data_ptr<float>(), vox.data_ptr<float>(),
+                           knn.data_ptr<float>(), idx.data_ptr<int>(), B, N, K,
+                           (int)num_levels, (int)k, (float)base_scale, stream());
+  return {vox, knn, idx};
+}
+
+torch::Tensor pv_corr_fused_bwd(torch::Tensor g_vox, torch::Tensor g_knn,
+                                torch::Tensor xyz, torch::Tensor coords,
+                                torch::Tensor knn_idx, int64_t num_levels,
+                                int64_t k, double base_scale) {
+  check_f32(g_vox, "g_vox");
+  check_f32(g_knn, "g_knn");
+  const int B = xyz.size(0), N = xyz.size(1), K = xyz.size(2);
+  auto gcorr = torch::empty({B, N, K}, g_vox.options());
+  launch_pv_corr_fused_bwd(g_vox.data_ptr<float>(), g_knn.data_ptr<float>(),
+                           xyz.data_ptr<float>(), coords.data_ptr<float>(),
+                           knn_idx.data_ptr<int>(), gcorr.data_ptr<float>(), B,
+                           N, K, (int)num_levels, (int)k, (float)base_scale,
+                           stream());
+  return gcorr;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("pv_corr_fused_fwd", &pv_corr_fused_fwd);
+  m.def("pv_corr_fused_bwd", &pv_corr_fused_bwd);
   m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0);
   m.def("batched_transpose", &batched_transpose);
   m.def("group_norm_act_fwd", &group_norm_act_fwd);

@@ -76,13 +76,24 @@ class CorrBlock(nn.Module):
         return CorrField(corr=corr, xyz=txyz)
 
     def forward(self, field: CorrField, coords: Tensor) -> Tensor:
-        """Query the field at coords (B, N, 3) -> (B, 64, N)."""
-        return self._voxel_feature(field, coords) + self._knn_feature(field, coords)
+        """Query the field at coords (B, N, 3) -> (B, 64, N).
+
+        One fused kernel pass produces both raw lookups (voxel pyramid +
+        kNN candidates); the two small conv pipelines then run on them.
+        """
+        vox, knn_raw = ops.pv_corr_lookup(
+            field.corr, field.xyz, coords, self.base_scale, self.num_levels,
+            self.knn, self.resolution,
+        )
+        return self._voxel_conv(vox) + self._knn_conv(knn_raw)
 
     def _voxel_feature(self, field: CorrField, coords: Tensor) -> Tensor:
         feat = ops.voxel_corr(
             field.corr, field.xyz, coords, self.base_scale, self.num_levels, self.resolution
         )  # B, L*R^3, N
+        return self._voxel_conv(feat)
+
+    def _voxel_conv(self, feat: Tensor) -> Tensor:
         # out_conv = Sequential(conv, GN, PReLU, conv) by state-dict layout;
         # executed with GN+PReLU fused (learnable slope = out_conv[2].weight)
         conv1, gn, prelu, conv2 = self.out_conv[0], self.out_conv[1], self.out_conv[2], self.out_conv[3]
@@ -94,6 +105,9 @@ class CorrBlock(nn.Module):
 
     def _knn_feature(self, field: CorrField, coords: Tensor) -> Tensor:
         feat = ops.knn_corr(field.corr, field.xyz, coords, self.knn)  # B, 4, k, N
+        return self._knn_conv(feat)
+
+    def _knn_conv(self, feat: Tensor) -> Tensor:
         # knn_conv = Sequential(conv, GN, PReLU) + max over the k axis;
         # executed as conv -> fused GN+PReLU+maxpool (one kernel pipeline,
         # the (B, 64, k, N) activation never materialises post-GN)

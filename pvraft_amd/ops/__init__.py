@@ -403,6 +403,52 @@ def voxel_corr(
     return reference.voxel_corr(corr, xyz, coords, base_scale, num_levels, resolution)
 
 
+class _PVCorrFused(torch.autograd.Function):
+    """Fused per-iteration lookup: voxel pyramid + kNN branch in one kernel
+    pass over the truncated correlation field (see csrc/pv_corr_fused.hip)."""
+
+    @staticmethod
+    def forward(ctx, corr, xyz, coords, base_scale, num_levels, k):
+        vox, knn, idx = _EXT.pv_corr_fused_fwd(corr, xyz, coords, base_scale, num_levels, k)
+        ctx.save_for_backward(xyz, coords, idx)
+        ctx.conf = (base_scale, num_levels, k)
+        return vox, knn
+
+    @staticmethod
+    def backward(ctx, g_vox, g_knn):
+        xyz, coords, idx = ctx.saved_tensors
+        base_scale, num_levels, k = ctx.conf
+        g = _EXT.pv_corr_fused_bwd(
+            g_vox.contiguous(), g_knn.contiguous(), xyz, coords, idx, num_levels, k, base_scale
+        )
+        return g, None, None, None, None, None
+
+
+def pv_corr_lookup(
+    corr: Tensor,
+    xyz: Tensor,
+    coords: Tensor,
+    base_scale: float,
+    num_levels: int,
+    k: int,
+    resolution: int = 3,
+):
+    """(voxel (B, L*27, N), knn (B, 4, k, N)) in one fused pass on GPU."""
+    if _use_hip(corr) and resolution == 3:
+        return _PVCorrFused.apply(
+            corr.contiguous().float(),
+            xyz.contiguous().float(),
+            coords.contiguous().float(),
+            float(base_scale),
+            int(num_levels),
+            int(k),
+        )
+    return (
+        reference.voxel_corr(corr, xyz, coords, base_scale, num_levels, resolution),
+        reference.knn_corr(corr, xyz, coords, k),
+    )
+
+
 def knn_corr(corr: Tensor, xyz: Tensor, coords: Tensor, k: int) -> Tensor:
     """(B,N,K),(B,N,K,3),(B,N,3) -> (B,4,N,k)."""
     if _use_hip(corr):
