@@ -46,20 +46,19 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
   const float *cand_xyz = xyz + p * K * 3;
   const float *cand_corr = corr + p * K;
 
-  // candidates in registers: rel-offsets, corr value, squared distance
-  float rx[MAXC], ry[MAXC], rz[MAXC], cv[MAXC], d[MAXC];
+  // distances for the kNN phase stay in registers; candidate xyz/corr are
+  // re-read per phase (L1-resident -- caching them in VGPRs cost ~40
+  // registers and measured SLOWER via occupancy)
+  float d[MAXC];
 #pragma unroll
   for (int t = 0; t < MAXC; ++t) {
     const int j = lane + t * WAVE;
     if (j < K) {
-      rx[t] = cand_xyz[j * 3 + 0] - cx;
-      ry[t] = cand_xyz[j * 3 + 1] - cy;
-      rz[t] = cand_xyz[j * 3 + 2] - cz;
-      cv[t] = cand_corr[j];
-      d[t] = rx[t] * rx[t] + ry[t] * ry[t] + rz[t] * rz[t];
+      const float dx = cand_xyz[j * 3 + 0] - cx;
+      const float dy = cand_xyz[j * 3 + 1] - cy;
+      const float dz = cand_xyz[j * 3 + 2] - cz;
+      d[t] = dx * dx + dy * dy + dz * dz;
     } else {
-      rx[t] = ry[t] = rz[t] = 0.f;
-      cv[t] = 0.f;
       d[t] = INFINITY;
     }
   }
@@ -73,18 +72,16 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
       s[q] = 0.f;
       c[q] = 0.f;
     }
-#pragma unroll
-    for (int t = 0; t < MAXC; ++t) {
-      const int j = lane + t * WAVE;
-      if (j < K) {
-        const float dx = rintf(rx[t] * inv_r);
-        const float dy = rintf(ry[t] * inv_r);
-        const float dz = rintf(rz[t] * inv_r);
+    for (int j = lane; j < K; j += WAVE) {
+      {
+        const float dx = rintf((cand_xyz[j * 3 + 0] - cx) * inv_r);
+        const float dy = rintf((cand_xyz[j * 3 + 1] - cy) * inv_r);
+        const float dz = rintf((cand_xyz[j * 3 + 2] - cz) * inv_r);
         const bool valid = fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
                            fabsf(dz) <= (R / 2);
         const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
                          ((int)dz + R / 2);
-        const float v = cv[t];
+        const float v = cand_corr[j];
 #pragma unroll
         for (int q = 0; q < CELLS; ++q) {
           const bool hit = valid && (cell == q);
@@ -128,10 +125,11 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     wave_argmin(best, bidx);
     if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
       const int s_ = bidx / WAVE;
-      dst[(long)t * N] = cv[s_];
-      dst[(long)t * N + ch_stride] = rx[s_];
-      dst[(long)t * N + 2 * ch_stride] = ry[s_];
-      dst[(long)t * N + 3 * ch_stride] = rz[s_];
+      const int j = bidx;
+      dst[(long)t * N] = cand_corr[j];
+      dst[(long)t * N + ch_stride] = cand_xyz[j * 3 + 0] - cx;
+      dst[(long)t * N + 2 * ch_stride] = cand_xyz[j * 3 + 1] - cy;
+      dst[(long)t * N + 3 * ch_stride] = cand_xyz[j * 3 + 2] - cz;
       idst[t] = bidx;
 #pragma unroll
       for (int ss = 0; ss < MAXC; ++ss)
@@ -172,19 +170,6 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
   const float cz = coords[p * 3 + 2];
   const float *cand_xyz = xyz + p * K * 3;
 
-  float rx[MAXC], ry[MAXC], rz[MAXC];
-#pragma unroll
-  for (int t = 0; t < MAXC; ++t) {
-    const int j = lane + t * WAVE;
-    if (j < K) {
-      rx[t] = cand_xyz[j * 3 + 0] - cx;
-      ry[t] = cand_xyz[j * 3 + 1] - cy;
-      rz[t] = cand_xyz[j * 3 + 2] - cz;
-    } else {
-      rx[t] = ry[t] = rz[t] = 1e30f;
-    }
-  }
-
   // per-level counts + this lane's cell's (count, g_vox)
   float cnt_mine[MAXL], g_mine[MAXL];
 #pragma unroll
@@ -197,20 +182,16 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
     float c[CELLS];
 #pragma unroll
     for (int q = 0; q < CELLS; ++q) c[q] = 0.f;
+    for (int j = lane; j < K; j += WAVE) {
+      const float dx = rintf((cand_xyz[j * 3 + 0] - cx) * inv_r);
+      const float dy = rintf((cand_xyz[j * 3 + 1] - cy) * inv_r);
+      const float dz = rintf((cand_xyz[j * 3 + 2] - cz) * inv_r);
+      const bool valid = fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
+                         fabsf(dz) <= (R / 2);
+      const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
+                       ((int)dz + R / 2);
 #pragma unroll
-    for (int t = 0; t < MAXC; ++t) {
-      const int j = lane + t * WAVE;
-      if (j < K) {
-        const float dx = rintf(rx[t] * inv_r);
-        const float dy = rintf(ry[t] * inv_r);
-        const float dz = rintf(rz[t] * inv_r);
-        const bool valid = fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
-                           fabsf(dz) <= (R / 2);
-        const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
-                         ((int)dz + R / 2);
-#pragma unroll
-        for (int q = 0; q < CELLS; ++q) c[q] += (valid && cell == q) ? 1.f : 0.f;
-      }
+      for (int q = 0; q < CELLS; ++q) c[q] += (valid && cell == q) ? 1.f : 0.f;
     }
 #pragma unroll
     for (int q = 0; q < CELLS; ++q) c[q] = wave_sum(c[q]);
@@ -228,18 +209,18 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
       }
   }
 
-#pragma unroll
-  for (int t = 0; t < MAXC; ++t) {
-    const int j = lane + t * WAVE;
-    if (j >= K) continue;
+  for (int j = lane; j < K; j += WAVE) {
+    const float ox = cand_xyz[j * 3 + 0] - cx;
+    const float oy = cand_xyz[j * 3 + 1] - cy;
+    const float oz = cand_xyz[j * 3 + 2] - cz;
     float g = 0.f;
 #pragma unroll
     for (int l = 0; l < MAXL; ++l) {
       if (l < L) {
         const float inv_r = 1.0f / (base_scale * (float)(1 << l));
-        const float dx = rintf(rx[t] * inv_r);
-        const float dy = rintf(ry[t] * inv_r);
-        const float dz = rintf(rz[t] * inv_r);
+        const float dx = rintf(ox * inv_r);
+        const float dy = rintf(oy * inv_r);
+        const float dz = rintf(oz * inv_r);
         const bool valid = fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
                            fabsf(dz) <= (R / 2);
         const int cell = valid ? ((int)dx + R / 2) * (R * R) +
