@@ -31,3 +31,26 @@ print("equal:", torch.allclose(a, b), torch.allclose(a, c), torch.allclose(a, d,
 wt = w.t()
 clones("matmul(w.t().unsqueeze(0), y3d)", lambda: torch.matmul(wt.unsqueeze(0), a))
 clones("bmm(w.t().expand, y3d)", lambda: torch.bmm(wt.unsqueeze(0).expand(2, -1, -1), a))
+
+print("--- context probes ---")
+class F1(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, w_, x_):
+        return torch.matmul(w_, x_)
+    @staticmethod
+    def backward(ctx, g):
+        return None, None
+
+clones("inside Function", lambda: F1.apply(w, x))
+def under_ac():
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        return torch.matmul(w, x)
+clones("under autocast", under_ac)
+def both():
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        return F1.apply(w, x)
+clones("Function under autocast", both)
+xg = x.clone().requires_grad_(True)
+clones("matmul requires_grad", lambda: torch.matmul(w, xg))
+wg = w.clone().requires_grad_(True)
+clones("matmul both grad", lambda: torch.matmul(wg, xg))
