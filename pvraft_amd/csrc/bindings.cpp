@@ -38,8 +38,8 @@ void launch_pv_corr_fused_bwd(const float*, const float*, const float*,
                               int, int, float, hipStream_t);
 void launch_transpose(const void*, void*, long, int, long, long, bool,
                       hipStream_t);
-void launch_pw_wgrad(const void*, const void*, float*, int, int, int, long,
-                     int, hipStream_t);
+void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
+                     long, int, hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
                      float*, float*, float*, void*, int, long, long, int, int,
@@ -222,24 +222,26 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
   const int rows = B * G;
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto row_ws = torch::zeros({rows, 2}, fopt);
-  auto chan_ws = torch::zeros({C, 2}, fopt);
+  // one zeroed workspace: [rows*2 row sums | C*2 channel sums | 1 d slope]
+  auto ws = torch::zeros({rows * 2 + C * 2 + 1}, fopt);
+  float* row_ws = ws.data_ptr<float>();
+  float* chan_ws = row_ws + rows * 2;
+  float* slope_ws = chan_ws + C * 2;
   auto dx = torch::empty_like(x);
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32,
               "group_norm_act: dtype must be float32 or bfloat16");
   TORCH_CHECK(dy.scalar_type() == x.scalar_type(), "dy/x dtype mismatch");
-  auto slope_ws = torch::zeros({1}, fopt);
   launch_gn_bwd(dy.data_ptr(), x.data_ptr(), mean.data_ptr<float>(),
                 rstd.data_ptr<float>(), weight.data_ptr<float>(),
-                bias.data_ptr<float>(), row_ws.data_ptr<float>(),
-                chan_ws.data_ptr<float>(), slope_ws.data_ptr<float>(),
+                bias.data_ptr<float>(), row_ws, chan_ws, slope_ws,
                 dx.data_ptr(), rows, row_len, S, C, (int)G, (int)act,
                 (float)slope, slope_ptr, bf16, stream());
-  // chan_ws[:, 0] = dbias, chan_ws[:, 1] = dweight
-  auto dbias = chan_ws.select(1, 0).contiguous();
-  auto dweight = chan_ws.select(1, 1).contiguous();
-  return {dx, dweight, dbias, slope_ws};
+  auto cw = ws.narrow(0, rows * 2, C * 2).view({C, 2});
+  auto dbias = cw.select(1, 0).contiguous();
+  auto dweight = cw.select(1, 1).contiguous();
+  auto dslope = ws.narrow(0, rows * 2 + C * 2, 1);
+  return {dx, dweight, dbias, dslope};
 }
 
 // x (B, C, K, N); GN stats over full (K, N); returns pooled
@@ -292,25 +294,27 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto row_ws = torch::zeros({rows, 2}, fopt);
-  auto chan_ws = torch::zeros({C, 2}, fopt);
+  auto ws = torch::zeros({rows * 2 + C * 2 + 1}, fopt);
+  float* row_ws = ws.data_ptr<float>();
+  float* chan_ws = row_ws + rows * 2;
+  float* slope_ws = chan_ws + C * 2;
   auto dx = torch::empty_like(x);
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
-  auto slope_ws = torch::zeros({1}, fopt);
   launch_gnmp_bwd(dy.data_ptr(), x.data_ptr(), am.data_ptr<unsigned char>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                  weight.data_ptr<float>(), bias.data_ptr<float>(),
-                  row_ws.data_ptr<float>(), chan_ws.data_ptr<float>(),
-                  slope_ws.data_ptr<float>(), dx.data_ptr(), rows, row_len, N,
-                  K, C, (int)G, (int)act, (float)slope, slope_ptr, bf16,
-                  stream());
-  auto dbias = chan_ws.select(1, 0).contiguous();
-  auto dweight = chan_ws.select(1, 1).contiguous();
-  return {dx, dweight, dbias, slope_ws};
+                  weight.data_ptr<float>(), bias.data_ptr<float>(), row_ws,
+                  chan_ws, slope_ws, dx.data_ptr(), rows, row_len, N, K, C,
+                  (int)G, (int)act, (float)slope, slope_ptr, bf16, stream());
+  auto cw = ws.narrow(0, rows * 2, C * 2).view({C, 2});
+  auto dbias = cw.select(1, 0).contiguous();
+  auto dweight = cw.select(1, 1).contiguous();
+  auto dslope = ws.narrow(0, rows * 2 + C * 2, 1);
+  return {dx, dweight, dbias, dslope};
 }
 
 // dy (B, Co, S) bf16, x (B, Ci, S) bf16 -> dW (Co, Ci) fp32 (split-K MFMA)
-torch::Tensor pw_wgrad(torch::Tensor dy, torch::Tensor x, int64_t schunks = 0) {
+std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
+                                    int64_t schunks = 0, bool with_bias = false) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 3);
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
   TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 && x.scalar_type() == torch::kBFloat16,
@@ -319,9 +323,17 @@ torch::Tensor pw_wgrad(torch::Tensor dy, torch::Tensor x, int64_t schunks = 0) {
   const int B = dy.size(0), Co = dy.size(1), Ci = x.size(1);
   const long S = dy.size(2);
   auto dw = torch::zeros({Co, Ci}, dy.options().dtype(torch::kFloat32));
-  launch_pw_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), B, Co,
-                  Ci, S, (int)schunks, stream());
-  return dw;
+  torch::Tensor dbias;
+  float* dbias_ptr = nullptr;
+  if (with_bias) {
+    dbias = torch::zeros({Co}, dy.options().dtype(torch::kFloat32));
+    dbias_ptr = dbias.data_ptr<float>();
+  } else {
+    dbias = torch::Tensor();
+  }
+  launch_pw_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), dbias_ptr,
+                  B, Co, Ci, S, (int)schunks, stream());
+  return {dw, dbias};
 }
 
 // (B, R, C) view (row-contiguous, arbitrary batch stride) -> (B, C, R)
@@ -378,7 +390,7 @@ torch::Tensor pv_corr_fused_bwd(torch::Tensor g_vox, torch::Tensor g_knn,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pv_corr_fused_fwd", &pv_corr_fused_fwd);
   m.def("pv_corr_fused_bwd", &pv_corr_fused_bwd);
-  m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0);
+  m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0, pybind11::arg("with_bias") = false);
   m.def("batched_transpose", &batched_transpose);
   m.def("group_norm_act_fwd", &group_norm_act_fwd);
   m.def("group_norm_act_bwd", &group_norm_act_bwd);

@@ -46,24 +46,9 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
   const float *cand_xyz = xyz + p * K * 3;
   const float *cand_corr = corr + p * K;
 
-  // distances for the kNN phase stay in registers; candidate xyz/corr are
-  // re-read per phase (L1-resident -- caching them in VGPRs cost ~40
-  // registers and measured SLOWER via occupancy)
-  float d[MAXC];
-#pragma unroll
-  for (int t = 0; t < MAXC; ++t) {
-    const int j = lane + t * WAVE;
-    if (j < K) {
-      const float dx = cand_xyz[j * 3 + 0] - cx;
-      const float dy = cand_xyz[j * 3 + 1] - cy;
-      const float dz = cand_xyz[j * 3 + 2] - cz;
-      d[t] = dx * dx + dy * dy + dz * dz;
-    } else {
-      d[t] = INFINITY;
-    }
-  }
-
-  // ---- voxel pyramid
+  // ---- voxel pyramid (candidates re-read per phase: L1-resident;
+  // caching them in VGPRs or keeping the kNN distances live across this
+  // loop measured SLOWER via occupancy)
   for (int l = 0; l < L; ++l) {
     const float inv_r = 1.0f / (base_scale * (float)(1 << l));
     float s[CELLS], c[CELLS];
@@ -107,7 +92,21 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     }
   }
 
-  // ---- kNN branch (argmin rounds on the register distances)
+  // ---- kNN branch (argmin rounds on register distances, computed here so
+  // they are not live during the voxel phase)
+  float d[MAXC];
+#pragma unroll
+  for (int t = 0; t < MAXC; ++t) {
+    const int j = lane + t * WAVE;
+    if (j < K) {
+      const float dx = cand_xyz[j * 3 + 0] - cx;
+      const float dy = cand_xyz[j * 3 + 1] - cy;
+      const float dz = cand_xyz[j * 3 + 2] - cz;
+      d[t] = dx * dx + dy * dy + dz * dz;
+    } else {
+      d[t] = INFINITY;
+    }
+  }
   float *dst = knn + ((long)b * 4) * k * N + n;
   const long ch_stride = (long)k * N;
   int *idst = knn_idx + p * k;

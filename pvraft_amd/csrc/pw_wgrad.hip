@@ -35,6 +35,7 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
     const __hip_bfloat16 *__restrict__ dy,  // (B, Co, S)
     const __hip_bfloat16 *__restrict__ x,   // (B, Ci, S)
     float *__restrict__ dw,                 // (Co, Ci) pre-zeroed fp32
+    float *__restrict__ dbias,              // (Co) pre-zeroed fp32 or null
     int B, int Co, int Ci, long S, int schunks) {
   __shared__ __hip_bfloat16 sA[TILE][KB + LDS_PAD];
   __shared__ __hip_bfloat16 sB[TILE][KB + LDS_PAD];
@@ -68,6 +69,7 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
   // loads: 2048 elems per tile, 256 threads -> 8 contiguous bf16 per thread
   const int ldr = threadIdx.x / 4;        // row 0..63
   const int ldc = (threadIdx.x % 4) * 8;  // col 0,8,16,24
+  float bias_part = 0.f;                  // this thread's dy row partial
   for (long s0 = s_lo; s0 < s_hi; s0 += KB) {
     __syncthreads();
     {
@@ -86,6 +88,10 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
         }
       }
       *(bf16x8 *)&sA[ldr][ldc] = av;
+      if (dbias != nullptr && tile_i == 0) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) bias_part += (float)((__bf16 *)&av)[e];
+      }
       const int i = tile_i + ldr;
       bf16x8 bv = (bf16x8)(__bf16)0.0f;
       if (i < Ci) {
@@ -116,6 +122,18 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
     }
   }
 
+  // bias gradient: reduce the per-thread dy-row partials (4 threads per
+  // row share ldr) and accumulate -- replaces a separate slow ATen reduce
+  if (dbias != nullptr && tile_i == 0) {
+    __shared__ float s_bias[TILE];
+    if (threadIdx.x < TILE) s_bias[threadIdx.x] = 0.f;
+    __syncthreads();
+    atomicAdd(&s_bias[ldr], bias_part);
+    __syncthreads();
+    if (threadIdx.x < TILE && tile_o + threadIdx.x < Co)
+      atomicAdd(&dbias[tile_o + threadIdx.x], s_bias[threadIdx.x]);
+  }
+
   // C/D map: col = lane&15, row = (lane>>4)*4 + reg
   const int crow = (lane >> 4) * 4;
   const int ccol = lane & 15;
@@ -131,8 +149,9 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
       }
 }
 
-void launch_pw_wgrad(const void *dy, const void *x, float *dw, int B, int Co,
-                     int Ci, long S, int schunks_opt, hipStream_t stream) {
+void launch_pw_wgrad(const void *dy, const void *x, float *dw, float *dbias,
+                     int B, int Co, int Ci, long S, int schunks_opt,
+                     hipStream_t stream) {
   const int to = (Co + TILE - 1) / TILE;
   const int ti = (Ci + TILE - 1) / TILE;
   long sc = schunks_opt;
@@ -147,5 +166,5 @@ void launch_pw_wgrad(const void *dy, const void *x, float *dw, int B, int Co,
   dim3 grid(to, ti, B * schunks);
   hipLaunchKernelGGL(pw_wgrad_kernel, grid, dim3(WG_THREADS), 0, stream,
                      (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, dw,
-                     B, Co, Ci, S, schunks);
+                     dbias, B, Co, Ci, S, schunks);
 }

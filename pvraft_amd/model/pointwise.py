@@ -34,9 +34,13 @@ class _PwMatmul(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, w: Tensor, x: Tensor) -> Tensor:
+    def forward(ctx, w: Tensor, x: Tensor, bias) -> Tensor:
         ctx.save_for_backward(w, x)
-        return torch.bmm(w.unsqueeze(0).expand(x.shape[0], -1, -1), x)
+        ctx.has_bias = bias is not None
+        y = torch.bmm(w.unsqueeze(0).expand(x.shape[0], -1, -1), x)
+        if bias is not None:
+            y = y + bias.view(1, -1, 1)
+        return y
 
     @staticmethod
     def backward(ctx, dy: Tensor):
@@ -48,41 +52,49 @@ class _PwMatmul(torch.autograd.Function):
             else None
         )
         dw = None
+        dbias = None
+        need_bias = ctx.has_bias and ctx.needs_input_grad[2]
         if ctx.needs_input_grad[0]:
             if x.is_cuda and x.dtype == torch.bfloat16:
                 try:
                     from pvraft_amd import _C
 
-                    dw = _C.pw_wgrad(dy, x).to(w.dtype)
+                    dw, db = _C.pw_wgrad(dy, x, 0, need_bias)
+                    dw = dw.to(w.dtype)
+                    if need_bias:
+                        dbias = db.to(dy.dtype)  # grad dtype must match input
+                        need_bias = False
                 except (ImportError, AttributeError):
                     dw = torch.einsum("bos,bis->oi", dy, x).to(w.dtype)
             else:
                 dw = torch.einsum("bos,bis->oi", dy, x).to(w.dtype)
-        return dw, dx
+        if need_bias:
+            dbias = dy.sum(dim=(0, 2))
+        return dw, dx, dbias
 
 
-def pw_matmul(weight: Tensor, x: Tensor) -> Tensor:
+def pw_matmul(weight: Tensor, x: Tensor, bias: Tensor = None) -> Tensor:
     w = weight
     if x.is_cuda and torch.is_autocast_enabled():
         dt = torch.get_autocast_dtype("cuda")
         w = w.to(dt)
         x = x.to(dt)
+        if bias is not None:
+            bias = bias.to(dt)
     elif x.dtype != w.dtype:
         x = x.to(w.dtype)
-    return _PwMatmul.apply(w, x.contiguous())
+    if bias is not None and bias.dtype != x.dtype:
+        bias = bias.to(x.dtype)
+    return _PwMatmul.apply(w, x.contiguous(), bias)
 
 
 class PwConv1d(nn.Conv1d):
     """nn.Conv1d(k=1) with a matmul forward."""
 
     def forward(self, x: Tensor) -> Tensor:
-        # x (B, Cin, N) -> (B, Cout, N)
-        y = pw_matmul(self.weight.squeeze(-1), x)
-        if self.bias is not None:
-            # keep the GEMM dtype (an fp32 bias would promote the whole
-            # downstream chain out of bf16 under autocast)
-            y = y + self.bias.view(1, -1, 1).to(y.dtype)
-        return y
+        # x (B, Cin, N) -> (B, Cout, N); bias fused into the GEMM Function
+        # (its gradient comes from the wgrad kernel, not a slow ATen reduce)
+        return pw_matmul(self.weight.squeeze(-1), x, self.bias)
 
 
 class PwConv2d(nn.Conv2d):
@@ -90,7 +102,5 @@ class PwConv2d(nn.Conv2d):
 
     def forward(self, x: Tensor) -> Tensor:
         B, C, H, W = x.shape
-        y = pw_matmul(self.weight.view(self.out_channels, C), x.reshape(B, C, H * W))
-        if self.bias is not None:
-            y = y + self.bias.view(1, -1, 1).to(y.dtype)
+        y = pw_matmul(self.weight.view(self.out_channels, C), x.reshape(B, C, H * W), self.bias)
         return y.view(B, self.out_channels, H, W)

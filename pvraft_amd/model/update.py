@@ -49,11 +49,10 @@ def _split_mm(weight: Tensor, bias, parts, wparts=None):
     if wparts is None:
         wparts = _slice_weight(weight, [p.shape[1] for p in parts])
     out = None
-    for w_i, p in zip(wparts, parts):
-        term = pw_matmul(w_i, p)
+    for i, (w_i, p) in enumerate(zip(wparts, parts)):
+        # bias rides the first partial GEMM (fused fwd add + kernel dbias)
+        term = pw_matmul(w_i, p, bias if i == 0 else None)
         out = term if out is None else out + term
-    if bias is not None:
-        out = out + bias.view(1, -1, 1).to(out.dtype)
     return out
 
 

@@ -224,12 +224,15 @@ def test_pw_wgrad_mfma_matches_einsum(B, Co, Ci, S):
 
     dy = torch.randn(B, Co, S, device=dev(), dtype=torch.bfloat16)
     x = torch.randn(B, Ci, S, device=dev(), dtype=torch.bfloat16)
-    got = _C.pw_wgrad(dy, x)
+    got, dbias = _C.pw_wgrad(dy, x, 0, True)
     want = torch.einsum("bos,bis->oi", dy.float(), x.float())
     # bf16 inputs, fp32 accumulation both sides; atomic split-K ordering
     err = (got - want).abs().max().item()
     denom = want.abs().max().item()
     assert err < 0.02 * max(denom, 1.0), (err, denom)
+    want_b = dy.float().sum(dim=(0, 2))
+    errb = (dbias - want_b).abs().max().item()
+    assert errb < 0.02 * max(want_b.abs().max().item(), 1.0), errb
 
 
 def test_knn_graph_degenerate_ties():
