@@ -79,12 +79,18 @@ def main():
     model.train()
     use_graph = args.graph and cuda
     if use_graph:
-        from pvraft_amd.engine.graphed import build_graphed_step
+        try:
+            from pvraft_amd.engine.graphed import build_graphed_step
 
-        reducer.hooks_enabled = False
-        graphed = build_graphed_step(
-            model, batch, num_iters=args.iters, gamma=0.8, reducer=reducer, amp=amp
-        )
+            reducer.hooks_enabled = False
+            graphed = build_graphed_step(
+                model, batch, num_iters=args.iters, gamma=0.8, reducer=reducer, amp=amp
+            )
+        except Exception as e:  # pragma: no cover - capture-env specific
+            print(f"[bench] hipGraph capture failed ({e!r}); falling back to eager", file=sys.stderr)
+            use_graph = False
+            reducer.hooks_enabled = True
+    if use_graph:
 
         def step():
             loss = graphed.replay()

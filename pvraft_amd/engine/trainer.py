@@ -201,15 +201,22 @@ class Trainer:
             key: [t.clone() for t in batch.data[key]] for key in batch.data
         }
         self.reducer.hooks_enabled = False
-        self._graph_step = build_graphed_step(
-            self.model,
-            self._static_batch,
-            num_iters=self.args.iters,
-            gamma=getattr(self.args, "gamma", 0.8),
-            reducer=self.reducer,
-            amp=self.amp,
-            loss_fn=None if self.loss_is_sequence else (lambda flows, b: self._loss(flows, b)),
-        )
+        try:
+            self._graph_step = build_graphed_step(
+                self.model,
+                self._static_batch,
+                num_iters=self.args.iters,
+                gamma=getattr(self.args, "gamma", 0.8),
+                reducer=self.reducer,
+                amp=self.amp,
+                loss_fn=None if self.loss_is_sequence else (lambda flows, b: self._loss(flows, b)),
+            )
+        except Exception as e:  # pragma: no cover - capture-env specific
+            self.log.warning(f"hipGraph capture failed ({e!r}); falling back to eager steps")
+            self.reducer.hooks_enabled = True
+            self.args.hipgraph = False
+            self._graph_step = None
+            return False
         self._graph_shape = shape
         self.log.info("train step captured into a hipGraph")
         return True
