@@ -76,11 +76,15 @@ class PVRaft(nn.Module):
         coords1, coords2 = xyz1, xyz1
         flow_predictions = []
         wcache = self.update_block.make_wcache()
+        # context-feature gate contribution is iteration-constant: hoist it
+        inp_pre = self.update_block.gru.precompute_inp(inp, wcache["gru"])
         for _ in range(num_iters):
             coords2 = coords2.detach()
             corr = self.corr_block(field, coords2)
             flow = coords2 - coords1
-            net, delta_flow = self.update_block(net, inp, corr, flow, graph_context, wcache)
+            net, delta_flow = self.update_block(
+                net, inp, corr, flow, graph_context, wcache, inp_pre=inp_pre
+            )
             coords2 = coords2 + delta_flow
             flow_predictions.append(coords2 - coords1)
         return flow_predictions
@@ -144,11 +148,14 @@ class PVRaftRefine(nn.Module):
 
             coords1, coords2 = xyz1, xyz1
             wcache = self.update_block.make_wcache()
+            inp_pre = self.update_block.gru.precompute_inp(inp, wcache["gru"])
             for _ in range(num_iters):
                 coords2 = coords2.detach()
                 corr = self.corr_block(field, coords2)
                 flow = coords2 - coords1
-                net, delta_flow = self.update_block(net, inp, corr, flow, graph_context, wcache)
+                net, delta_flow = self.update_block(
+                    net, inp, corr, flow, graph_context, wcache, inp_pre=inp_pre
+                )
                 coords2 = coords2 + delta_flow
         return self.refine_block(coords2 - coords1, graph1)
 

@@ -90,15 +90,49 @@ class ConvGRU(nn.Module):
         """Reference formulation (update.py:31-40) for direct use."""
         return self.forward_parts(h, [x])
 
-    def forward_parts(self, h: Tensor, x_parts, wcache=None) -> Tensor:
-        """Gates from the concat parts: conv(cat(h, *x_parts)) done as
-        summed partial GEMMs (weight column slices)."""
-        wz = wcache.get("z") if wcache else None
-        wr = wcache.get("r") if wcache else None
-        wq = wcache.get("q") if wcache else None
-        z = torch.sigmoid(_split_mm(self.convz.weight, self.convz.bias, [h] + list(x_parts), wz))
-        r = torch.sigmoid(_split_mm(self.convr.weight, self.convr.bias, [h] + list(x_parts), wr))
-        q = torch.tanh(_split_mm(self.convq.weight, self.convq.bias, [r * h] + list(x_parts), wq))
+    def gate_weights(self, part_sizes):
+        """Cross-gate fused weight slices.
+
+        All three gates consume the same non-hidden inputs, so their weight
+        column slices stack row-wise into ONE GEMM per input part
+        (z/r/q = rows [0:64/64:128/128:192]); only the hidden operand
+        differs (z,r read h; q reads r*h), so the h columns fuse z+r only.
+        """
+        hd = self.convz.out_channels
+        wz = self.convz.weight.squeeze(-1)
+        wr = self.convr.weight.squeeze(-1)
+        wq = self.convq.weight.squeeze(-1)
+        out = {"zr_h": torch.cat([wz[:, :hd], wr[:, :hd]], dim=0).contiguous(),
+               "q_h": wq[:, :hd].contiguous(),
+               "b_zr": torch.cat([self.convz.bias, self.convr.bias], dim=0),
+               "b_q": self.convq.bias,
+               "parts": []}
+        lo = hd
+        for sz in part_sizes:
+            out["parts"].append(
+                torch.cat([wz[:, lo : lo + sz], wr[:, lo : lo + sz], wq[:, lo : lo + sz]], dim=0).contiguous()
+            )
+            lo += sz
+        return out
+
+    def precompute_inp(self, inp: Tensor, gw) -> Tensor:
+        """Contribution of the iteration-constant context features to all
+        three gate preactivations -- hoisted out of the GRU loop."""
+        return pw_matmul(gw["parts"][0], inp)
+
+    def forward_parts(self, h: Tensor, x_parts, wcache=None, gw=None, pre=None) -> Tensor:
+        """Gates from the concat parts (cross-gate fused GEMMs)."""
+        if gw is None:
+            gw = self.gate_weights([p.shape[1] for p in x_parts])
+        hd = self.convz.out_channels
+        # shared-input contribution for all gates: (B, 3*hd, N)
+        m = pre if pre is not None else pw_matmul(gw["parts"][0], x_parts[0])
+        for w_i, p in zip(gw["parts"][1:], x_parts[1:]):
+            m = m + pw_matmul(w_i, p)
+        zr = torch.sigmoid(pw_matmul(gw["zr_h"], h, gw["b_zr"]) + m[:, : 2 * hd])
+        z = zr[:, :hd]
+        r = zr[:, hd:]
+        q = torch.tanh(pw_matmul(gw["q_h"], r * h, gw["b_q"]) + m[:, 2 * hd :])
         return (1 - z) * h + z * q
 
 
@@ -145,28 +179,37 @@ class UpdateBlock(nn.Module):
         self.flow_head = FlowHead(input_dim=hidden_dim)
 
     def make_wcache(self):
-        """Slice the concat-consuming conv weights once per forward (the GRU
-        loop reuses them across all iterations)."""
-        gate_sizes = [64, 64, 61, 3]  # [h, inp, motion61, flow3]
+        """Prepare the concat-free weights once per forward (the GRU loop
+        reuses them across all iterations): cross-gate fused GRU weights
+        plus the motion/flow-head column slices."""
         cache = {
-            "z": _slice_weight(self.gru.convz.weight, gate_sizes),
-            "r": _slice_weight(self.gru.convr.weight, gate_sizes),
-            "q": _slice_weight(self.gru.convq.weight, gate_sizes),
+            "gru": self.gru.gate_weights([64, 61, 3]),  # [inp, motion61, flow3]
             "motion": _slice_weight(self.motion_encoder.conv.weight, [64, 64]),
             "flowhead": _slice_weight(self.flow_head.out_conv[0].weight, [64, 64]),
         }
         # pre-cast once under autocast so per-iteration GEMMs skip the cast
         if torch.is_autocast_enabled():
             dt = torch.get_autocast_dtype("cuda")
-            cache = {k: [w.to(dt) for w in v] for k, v in cache.items()}
+
+            def cast(v):
+                if isinstance(v, torch.Tensor):
+                    return v.to(dt)
+                if isinstance(v, list):
+                    return [cast(x) for x in v]
+                if isinstance(v, dict):
+                    return {k: cast(x) for k, x in v.items()}
+                return v
+
+            cache = cast(cache)
         return cache
 
     def forward(self, net: Tensor, inp: Tensor, corr: Tensor, flow: Tensor, graph: Graph,
-                wcache=None):
+                wcache=None, inp_pre=None):
         from pvraft_amd import ops
 
         motion, flow_t = self.motion_encoder(flow, corr, wcache)
         # gru input = cat(inp, motion, flow_t) (reference update.py:84), as parts
-        net = self.gru.forward_parts(net, [inp, motion, flow_t], wcache)
+        gw = wcache.get("gru") if wcache else None
+        net = self.gru.forward_parts(net, [inp, motion, flow_t], wcache, gw=gw, pre=inp_pre)
         delta_flow = ops.transpose_last2(self.flow_head(net, graph, wcache))
         return net, delta_flow
