@@ -36,6 +36,8 @@ void launch_pv_corr_fused_fwd(const float*, const float*, const float*,
 void launch_pv_corr_fused_bwd(const float*, const float*, const float*,
                               const float*, const int*, float*, int, int, int,
                               int, int, float, hipStream_t);
+void launch_topk_rows(const float*, float*, int*, long, int, int,
+                      hipStream_t);
 void launch_transpose(const void*, void*, long, int, long, long, bool,
                       hipStream_t);
 void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
@@ -387,7 +389,23 @@ torch::Tensor pv_corr_fused_bwd(torch::Tensor g_vox, torch::Tensor g_knn,
   return gcorr;
 }
 
+// vals (R, M) fp32 -> top-K largest per row: values (R, K), idx (R, K)
+std::vector<torch::Tensor> topk_rows(torch::Tensor vals, int64_t K) {
+  check_f32(vals, "vals");
+  TORCH_CHECK(vals.dim() == 2, "vals must be (R, M)");
+  const long R = vals.size(0);
+  const int M = vals.size(1);
+  TORCH_CHECK(M <= 8192, "topk_rows supports M <= 8192");
+  TORCH_CHECK(K >= 1 && K <= M, "need 1 <= K <= M");
+  auto out_v = torch::empty({R, K}, vals.options());
+  auto out_i = torch::empty({R, K}, vals.options().dtype(torch::kInt32));
+  launch_topk_rows(vals.data_ptr<float>(), out_v.data_ptr<float>(),
+                   out_i.data_ptr<int>(), R, M, (int)K, stream());
+  return {out_v, out_i};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("topk_rows", &topk_rows);
   m.def("pv_corr_fused_fwd", &pv_corr_fused_fwd);
   m.def("pv_corr_fused_bwd", &pv_corr_fused_bwd);
   m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0, pybind11::arg("with_bias") = false);

@@ -1,0 +1,166 @@
+// K3: row-wise top-K (largest) selection for the correlation truncation
+// (reference model/corr.py:37: torch.topk(corr, k=512, dim=2)).
+//
+// Same histogram-select idea as the kNN graph kernel, keyed on the
+// order-preserving unsigned map of floats, MSB-first 8-bit rounds:
+// correlation values cluster within one exponent, so up to three refine
+// rounds (exponent byte, then two mantissa bytes) narrow the threshold
+// until the boundary bin fits the wave-select buffer.  One workgroup per
+// row; the row is staged once into LDS (M <= 8192 -> 32 KB) and every
+// pass reads it from there.
+//
+// Output is the top-K SET with matching indices (unsorted: every consumer
+// -- the voxel/kNN lookups -- is order-invariant; the reference's
+// sorted=True is an implementation detail of torch.topk).
+#include <hip/hip_runtime.h>
+#include "common.h"
+
+#define TK_THREADS 256
+#define TK_MAXM 8192  // LDS row cache (32 KB)
+#define TK_CAP 128    // boundary-buffer capacity
+
+// order-preserving map: a < b (as floats, incl. negatives) <=> key(a) < key(b)
+DEV_INLINE unsigned fkey(float v) {
+  unsigned b = __float_as_uint(v);
+  return (b & 0x80000000u) ? ~b : (b | 0x80000000u);
+}
+
+__global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
+    const float *__restrict__ vals,  // (R, M)
+    float *__restrict__ out_v,       // (R, K)
+    int *__restrict__ out_i,         // (R, K)
+    int M, int K) {
+  __shared__ float s_row[TK_MAXM];
+  __shared__ unsigned s_hist[256];
+  __shared__ float s_bv[TK_CAP];
+  __shared__ int s_bi[TK_CAP];
+  __shared__ unsigned s_acc, s_bcnt, s_state[3];
+  // s_state: [0] = threshold prefix (bytes chosen so far, left-aligned)
+  //          [1] = number of refined bytes (1..3)
+  //          [2] = remaining slots to take at the exact threshold byte
+
+  const int row = blockIdx.x;
+  const float *src = vals + (long)row * M;
+  for (int i = threadIdx.x; i < M; i += TK_THREADS) s_row[i] = src[i];
+  if (threadIdx.x == 0) {
+    s_acc = 0;
+    s_bcnt = 0;
+  }
+
+  // MSB-first refine rounds: keep candidates whose key's examined prefix
+  // equals the running threshold prefix; histogram the next byte.
+  unsigned prefix = 0;  // broadcast via s_state after each round
+  int need = K;
+  int nbytes = 0;
+  for (int round = 0; round < 3; ++round) {
+    __syncthreads();
+    for (int i = threadIdx.x; i < 256; i += TK_THREADS) s_hist[i] = 0;
+    __syncthreads();
+    const int shift = 24 - 8 * round;
+    const unsigned mask = round == 0 ? 0u : (0xFFFFFFFFu << (shift + 8));
+    for (int i = threadIdx.x; i < M; i += TK_THREADS) {
+      const unsigned key = fkey(s_row[i]);
+      if ((key & mask) == prefix)
+        atomicAdd(&s_hist[(key >> shift) & 0xFFu], 1u);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      // walk bins from the TOP (largest values first)
+      unsigned cum = 0;
+      int T = 0;
+      for (int bin = 255; bin >= 0; --bin) {
+        const unsigned nxt = cum + s_hist[bin];
+        if (nxt >= (unsigned)need) {
+          T = bin;
+          break;
+        }
+        cum = nxt;
+      }
+      s_state[0] = prefix | ((unsigned)T << shift);
+      s_state[1] = (unsigned)(round + 1);
+      s_state[2] = (unsigned)(need - (int)cum);
+      // stop refining once the threshold bin fits the boundary buffer
+      if (s_hist[T] <= TK_CAP - 8 || round == 2) s_state[1] |= 0x100u;
+    }
+    __syncthreads();
+    prefix = s_state[0];
+    need = (int)s_state[2];
+    nbytes = (int)(s_state[1] & 0xFFu);
+    if (s_state[1] & 0x100u) break;
+  }
+
+  // collect: accepted = key prefix strictly greater than the threshold
+  // prefix at the examined depth; boundary = exactly equal.
+  const int shift = 32 - 8 * nbytes;
+  __syncthreads();
+  for (int i = threadIdx.x; i < M; i += TK_THREADS) {
+    const float v = s_row[i];
+    const unsigned kp = shift == 32 ? 0u : (fkey(v) >> shift);
+    const unsigned tp = shift == 32 ? 0u : (prefix >> shift);
+    if (kp > tp) {
+      const unsigned slot = atomicAdd(&s_acc, 1u);
+      out_v[(long)row * K + slot] = v;
+      out_i[(long)row * K + slot] = i;
+    } else if (kp == tp) {
+      const unsigned p = atomicAdd(&s_bcnt, 1u);
+      if (p < TK_CAP) {
+        s_bv[p] = v;
+        s_bi[p] = i;
+      }
+    }
+  }
+  __syncthreads();
+
+  // wave 0 selects the remaining `need` LARGEST from the boundary buffer
+  if (wave_id() == 0) {
+    const int lane = lane_id();
+    const int L = (int)min(s_bcnt, (unsigned)TK_CAP);
+    int take = need;
+    if (take > L) take = L;
+    const int base = (int)s_acc;
+    float dv[TK_CAP / WAVE];
+    int iv[TK_CAP / WAVE];
+#pragma unroll
+    for (int s_ = 0; s_ < TK_CAP / WAVE; ++s_) {
+      const int p = lane + s_ * WAVE;
+      dv[s_] = p < L ? -s_bv[p] : INFINITY;  // argmin on negated = argmax
+      iv[s_] = p < L ? s_bi[p] : 0x7fffffff;
+    }
+    int written = 0;
+    for (int r = 0; r < take; ++r) {
+      float best = INFINITY;
+      int bslot = 0;
+#pragma unroll
+      for (int s_ = 0; s_ < TK_CAP / WAVE; ++s_)
+        if (dv[s_] < best) {
+          best = dv[s_];
+          bslot = s_;
+        }
+      int bidx = lane + bslot * WAVE;
+      if (best == INFINITY) bidx = 0x7fffffff;
+      float bv = best;
+      wave_argmin(bv, bidx);
+      if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
+        const int s_ = bidx / WAVE;
+        out_v[(long)row * K + base + r] = -dv[s_];
+        out_i[(long)row * K + base + r] = iv[s_];
+#pragma unroll
+        for (int ss = 0; ss < TK_CAP / WAVE; ++ss)
+          if (ss == s_) dv[ss] = INFINITY;
+      }
+      if (bidx != 0x7fffffff) ++written;
+    }
+    // degenerate ties beyond CAP: pad with the first boundary entry
+    if (lane == 0)
+      for (int r = base + written; r < K; ++r) {
+        out_v[(long)row * K + r] = L > 0 ? s_bv[0] : -INFINITY;
+        out_i[(long)row * K + r] = L > 0 ? s_bi[0] : 0;
+      }
+  }
+}
+
+void launch_topk_rows(const float *vals, float *out_v, int *out_i, long R,
+                      int M, int K, hipStream_t stream) {
+  hipLaunchKernelGGL(topk_rows_kernel, dim3((unsigned)R), dim3(TK_THREADS), 0,
+                     stream, vals, out_v, out_i, M, K);
+}

@@ -337,13 +337,41 @@ class _CorrTruncate(torch.autograd.Function):
         # quality and value precision of the correlation field follow the
         # reference, corr.py:95-99); autocast would otherwise hijack the bmm
         with torch.no_grad(), torch.autocast("cuda", enabled=False):
-            corr, idx, txyz = reference.corr_truncate(
-                fmap1, fmap2, xyz2, truncate_k, chunk=_CorrTruncate.CHUNK
-            )
+            M = fmap2.shape[2]
+            if fmap1.is_cuda and M <= 8192 and _load_ext() is not None:
+                corr, idx, txyz = _CorrTruncate._gpu_forward(fmap1, fmap2, xyz2, truncate_k)
+            else:
+                corr, idx, txyz = reference.corr_truncate(
+                    fmap1, fmap2, xyz2, truncate_k, chunk=_CorrTruncate.CHUNK
+                )
         ctx.save_for_backward(fmap1, fmap2, idx)
         # backward GEMMs run bf16 when the surrounding step is bf16 autocast
         # (standard amp gradient precision; fp32 master weights untouched)
         ctx.bf16_bwd = fmap1.is_cuda and torch.is_autocast_enabled()
+        return corr, idx, txyz
+
+    @staticmethod
+    def _gpu_forward(fmap1: Tensor, fmap2: Tensor, xyz2: Tensor, truncate_k: int):
+        """Chunked rocBLAS GEMM + histogram-select top-K kernel (the
+        torch.topk path sorts and was the next profiler hotspot; the top-K
+        SET is order-free for every consumer)."""
+        B, C, N = fmap1.shape
+        M = fmap2.shape[2]
+        K = min(truncate_k, M)
+        scale = 1.0 / math.sqrt(C)
+        f1t = fmap1.transpose(1, 2).contiguous()
+        chunk = _CorrTruncate.CHUNK
+        vals, idxs = [], []
+        for s in range(0, N, chunk):
+            c = torch.bmm(f1t[:, s : s + chunk], fmap2)
+            c = c * scale
+            n_rows = c.shape[1]
+            v, i = _EXT.topk_rows(c.reshape(B * n_rows, M), K)
+            vals.append(v.view(B, n_rows, K))
+            idxs.append(i.view(B, n_rows, K).long())
+        corr = torch.cat(vals, dim=1)
+        idx = torch.cat(idxs, dim=1)
+        txyz = xyz2.gather(1, idx.reshape(B, N * K).unsqueeze(-1).expand(B, N * K, 3)).view(B, N, K, 3)
         return corr, idx, txyz
 
     @staticmethod

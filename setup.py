@@ -23,6 +23,7 @@ SOURCES = [
     "pvraft_amd/csrc/pw_wgrad.hip",
     "pvraft_amd/csrc/transpose.hip",
     "pvraft_amd/csrc/pv_corr_fused.hip",
+    "pvraft_amd/csrc/topk_rows.hip",
 ]
 
 setup(

@@ -281,3 +281,27 @@ def test_pv_corr_fused_matches_separate(B, N, K, L, k):
     knn[:, 0].sum().backward()
     knn_ref[:, 0].sum().backward()
     assert torch.allclose(corr.grad.sum(), corr2.grad.sum())
+
+
+@pytest.mark.parametrize("R,M,K", [(64, 8192, 512), (33, 1000, 100), (16, 512, 512), (8, 777, 32)])
+def test_topk_rows_matches_torch(R, M, K):
+    from pvraft_amd import _C
+
+    vals = torch.randn(R, M, device=dev())
+    got_v, got_i = _C.topk_rows(vals, K)
+    want_v, want_i = vals.topk(K, dim=1)
+    # set comparison (kernel output unsorted; ties may swap at the boundary)
+    assert torch.allclose(got_v.sort(dim=1).values, want_v.sort(dim=1).values, atol=1e-5), (
+        (got_v.sort(dim=1).values - want_v.sort(dim=1).values).abs().max()
+    )
+    # values must match the gathered indices exactly
+    assert torch.equal(got_v, vals.gather(1, got_i.long()))
+
+
+def test_topk_rows_degenerate_ties():
+    from pvraft_amd import _C
+
+    vals = torch.zeros(4, 2048, device=dev())
+    got_v, got_i = _C.topk_rows(vals, 256)
+    assert (got_v == 0).all()
+    assert (got_i >= 0).all() and (got_i < 2048).all()
