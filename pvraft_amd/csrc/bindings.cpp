@@ -38,6 +38,15 @@ void launch_pv_corr_fused_bwd(const float*, const float*, const float*,
                               int, int, float, hipStream_t);
 void launch_topk_rows(const float*, float*, int*, long, int, int,
                       hipStream_t);
+void launch_gru_zr_fwd(const void*, const void*, void*, void*, void*, long,
+                       long, bool, hipStream_t);
+void launch_gru_zr_bwd(const void*, const void*, const void*, const void*,
+                       const void*, void*, void*, long, long, bool,
+                       hipStream_t);
+void launch_gru_q_fwd(const void*, const void*, const void*, void*, void*,
+                      long, long, bool, hipStream_t);
+void launch_gru_q_bwd(const void*, const void*, const void*, const void*,
+                      void*, void*, void*, long, long, bool, hipStream_t);
 void launch_transpose(const void*, void*, long, int, long, long, bool,
                       hipStream_t);
 void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
@@ -404,8 +413,78 @@ std::vector<torch::Tensor> topk_rows(torch::Tensor vals, int64_t K) {
   return {out_v, out_i};
 }
 
+namespace {
+
+void check_gru(const torch::Tensor& t, const torch::Tensor& like,
+               const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous(), name,
+              " must be a contiguous GPU tensor");
+  TORCH_CHECK(t.scalar_type() == like.scalar_type(), name,
+              " dtype mismatch in fused GRU gates");
+}
+
+}  // namespace
+
+// pre (B,2H,N) [z|r preactivations], h (B,H,N) -> z, r, rh (each (B,H,N))
+std::vector<torch::Tensor> gru_zr_fwd(torch::Tensor pre, torch::Tensor h) {
+  TORCH_CHECK(pre.is_cuda() && pre.is_contiguous() && h.is_contiguous());
+  TORCH_CHECK(pre.dim() == 3 && h.dim() == 3 && pre.size(0) == h.size(0) &&
+              pre.size(1) == 2 * h.size(1) && pre.size(2) == h.size(2));
+  const bool bf16 = pre.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || pre.scalar_type() == torch::kFloat32);
+  check_gru(h, pre, "h");
+  auto z = torch::empty_like(h), r = torch::empty_like(h),
+       rh = torch::empty_like(h);
+  launch_gru_zr_fwd(pre.data_ptr(), h.data_ptr(), z.data_ptr(), r.data_ptr(),
+                    rh.data_ptr(), h.size(0), (long)h.size(1) * h.size(2),
+                    bf16, stream());
+  return {z, r, rh};
+}
+
+std::vector<torch::Tensor> gru_zr_bwd(torch::Tensor dz, torch::Tensor drh,
+                                      torch::Tensor z, torch::Tensor r,
+                                      torch::Tensor h) {
+  const bool bf16 = z.scalar_type() == torch::kBFloat16;
+  for (auto* t : {&dz, &drh, &r, &h}) check_gru(*t, z, "gru_zr_bwd input");
+  auto dpre = torch::empty({h.size(0), 2 * h.size(1), h.size(2)}, h.options());
+  auto dh = torch::empty_like(h);
+  launch_gru_zr_bwd(dz.data_ptr(), drh.data_ptr(), z.data_ptr(), r.data_ptr(),
+                    h.data_ptr(), dpre.data_ptr(), dh.data_ptr(), h.size(0),
+                    (long)h.size(1) * h.size(2), bf16, stream());
+  return {dpre, dh};
+}
+
+// pre_q, z, h (B,H,N) -> q, hnew
+std::vector<torch::Tensor> gru_q_fwd(torch::Tensor pre, torch::Tensor z,
+                                     torch::Tensor h) {
+  const bool bf16 = pre.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || pre.scalar_type() == torch::kFloat32);
+  for (auto* t : {&pre, &z, &h}) check_gru(*t, pre, "gru_q_fwd input");
+  auto q = torch::empty_like(h), hnew = torch::empty_like(h);
+  launch_gru_q_fwd(pre.data_ptr(), z.data_ptr(), h.data_ptr(), q.data_ptr(),
+                   hnew.data_ptr(), h.size(0), (long)h.size(1) * h.size(2),
+                   bf16, stream());
+  return {q, hnew};
+}
+
+std::vector<torch::Tensor> gru_q_bwd(torch::Tensor dhnew, torch::Tensor q,
+                                     torch::Tensor z, torch::Tensor h) {
+  const bool bf16 = q.scalar_type() == torch::kBFloat16;
+  for (auto* t : {&dhnew, &z, &h}) check_gru(*t, q, "gru_q_bwd input");
+  auto dpre = torch::empty_like(h), dz = torch::empty_like(h),
+       dh = torch::empty_like(h);
+  launch_gru_q_bwd(dhnew.data_ptr(), q.data_ptr(), z.data_ptr(), h.data_ptr(),
+                   dpre.data_ptr(), dz.data_ptr(), dh.data_ptr(), h.size(0),
+                   (long)h.size(1) * h.size(2), bf16, stream());
+  return {dpre, dz, dh};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_rows", &topk_rows);
+  m.def("gru_zr_fwd", &gru_zr_fwd);
+  m.def("gru_zr_bwd", &gru_zr_bwd);
+  m.def("gru_q_fwd", &gru_q_fwd);
+  m.def("gru_q_bwd", &gru_q_bwd);
   m.def("pv_corr_fused_fwd", &pv_corr_fused_fwd);
   m.def("pv_corr_fused_bwd", &pv_corr_fused_bwd);
   m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0, pybind11::arg("with_bias") = false);

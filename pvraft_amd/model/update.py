@@ -121,7 +121,11 @@ class ConvGRU(nn.Module):
         return pw_matmul(gw["parts"][0], inp)
 
     def forward_parts(self, h: Tensor, x_parts, wcache=None, gw=None, pre=None) -> Tensor:
-        """Gates from the concat parts (cross-gate fused GEMMs)."""
+        """Gates from the concat parts (cross-gate fused GEMMs); the gate
+        elementwise math (sigmoid/tanh/r*h/blend + backward) runs as two
+        fused HIP kernels per direction (ops.gru_zr / ops.gru_q)."""
+        from pvraft_amd import ops
+
         if gw is None:
             gw = self.gate_weights([p.shape[1] for p in x_parts])
         hd = self.convz.out_channels
@@ -129,11 +133,8 @@ class ConvGRU(nn.Module):
         m = pre if pre is not None else pw_matmul(gw["parts"][0], x_parts[0])
         for w_i, p in zip(gw["parts"][1:], x_parts[1:]):
             m = m + pw_matmul(w_i, p)
-        zr = torch.sigmoid(pw_matmul(gw["zr_h"], h, gw["b_zr"]) + m[:, : 2 * hd])
-        z = zr[:, :hd]
-        r = zr[:, hd:]
-        q = torch.tanh(pw_matmul(gw["q_h"], r * h, gw["b_q"]) + m[:, 2 * hd :])
-        return (1 - z) * h + z * q
+        z, rh = ops.gru_zr(pw_matmul(gw["zr_h"], h, gw["b_zr"]) + m[:, : 2 * hd], h)
+        return ops.gru_q(pw_matmul(gw["q_h"], rh, gw["b_q"]) + m[:, 2 * hd :], z, h)
 
 
 class ConvRNN(nn.Module):

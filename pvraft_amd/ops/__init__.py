@@ -487,3 +487,60 @@ def knn_corr(corr: Tensor, xyz: Tensor, coords: Tensor, k: int) -> Tensor:
             int(k),
         )
     return reference.knn_corr(corr, xyz, coords, k)
+
+
+class _GruZR(torch.autograd.Function):
+    """Fused GRU z/r gates (csrc/gru_gates.hip): one kernel for the
+    sigmoid over the stacked z|r preactivations plus r*h (reference
+    model/update.py:34-37), one kernel for the full backward."""
+
+    @staticmethod
+    def forward(ctx, pre: Tensor, h: Tensor):
+        z, r, rh = _EXT.gru_zr_fwd(pre, h)
+        ctx.save_for_backward(z, r, h)
+        return z, rh
+
+    @staticmethod
+    def backward(ctx, dz: Tensor, drh: Tensor):
+        z, r, h = ctx.saved_tensors
+        if dz is None:
+            dz = torch.zeros_like(z)
+        if drh is None:
+            drh = torch.zeros_like(z)
+        dpre, dh = _EXT.gru_zr_bwd(dz.contiguous(), drh.contiguous(), z, r, h)
+        return dpre, dh
+
+
+class _GruQ(torch.autograd.Function):
+    """Fused GRU candidate gate + blend: h' = (1-z) h + z tanh(pre_q)
+    (reference model/update.py:38-40)."""
+
+    @staticmethod
+    def forward(ctx, pre: Tensor, z: Tensor, h: Tensor) -> Tensor:
+        q, hnew = _EXT.gru_q_fwd(pre, z, h)
+        ctx.save_for_backward(q, z, h)
+        return hnew
+
+    @staticmethod
+    def backward(ctx, dhnew: Tensor):
+        q, z, h = ctx.saved_tensors
+        dpre, dz, dh = _EXT.gru_q_bwd(dhnew.contiguous(), q, z, h)
+        return dpre, dz, dh
+
+
+def gru_zr(pre_zr: Tensor, h: Tensor):
+    """pre_zr (B,2H,N) stacked z|r preactivations, h (B,H,N) -> (z, r*h)."""
+    if _use_hip(pre_zr):
+        return _GruZR.apply(pre_zr.contiguous(), h.to(pre_zr.dtype).contiguous())
+    hd = h.shape[1]
+    zr = torch.sigmoid(pre_zr)
+    return zr[:, :hd], zr[:, hd:] * h
+
+
+def gru_q(pre_q: Tensor, z: Tensor, h: Tensor) -> Tensor:
+    """h' = (1-z)*h + z*tanh(pre_q), all (B,H,N)."""
+    if _use_hip(pre_q):
+        return _GruQ.apply(
+            pre_q.contiguous(), z.contiguous(), h.to(pre_q.dtype).contiguous()
+        )
+    return (1 - z) * h + z * torch.tanh(pre_q)

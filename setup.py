@@ -24,6 +24,7 @@ SOURCES = [
     "pvraft_amd/csrc/transpose.hip",
     "pvraft_amd/csrc/pv_corr_fused.hip",
     "pvraft_amd/csrc/topk_rows.hip",
+    "pvraft_amd/csrc/gru_gates.hip",
 ]
 
 setup(
