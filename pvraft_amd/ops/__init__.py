@@ -530,7 +530,7 @@ class _GruQ(torch.autograd.Function):
 
 def gru_zr(pre_zr: Tensor, h: Tensor):
     """pre_zr (B,2H,N) stacked z|r preactivations, h (B,H,N) -> (z, r*h)."""
-    if _use_hip(pre_zr):
+    if _use_hip(pre_zr) and os.environ.get("PVRAFT_NO_GRU_FUSION", "0") != "1":
         return _GruZR.apply(pre_zr.contiguous(), h.to(pre_zr.dtype).contiguous())
     hd = h.shape[1]
     zr = torch.sigmoid(pre_zr)
@@ -539,7 +539,7 @@ def gru_zr(pre_zr: Tensor, h: Tensor):
 
 def gru_q(pre_q: Tensor, z: Tensor, h: Tensor) -> Tensor:
     """h' = (1-z)*h + z*tanh(pre_q), all (B,H,N)."""
-    if _use_hip(pre_q):
+    if _use_hip(pre_q) and os.environ.get("PVRAFT_NO_GRU_FUSION", "0") != "1":
         return _GruQ.apply(
             pre_q.contiguous(), z.contiguous(), h.to(pre_q.dtype).contiguous()
         )

@@ -162,8 +162,10 @@ def test_predictor_graphed_matches_eager_eval():
 
 
 def test_trainer_end_to_end_on_gpu(tmp_path):
-    """One tiny epoch of the real Trainer on the HIP path (graph capture,
-    prefetcher, checkpointing)."""
+    """Two tiny epochs of the real Trainer on the HIP path (graph capture,
+    prefetcher, checkpointing) with val interleaved -- the epoch-2 loop
+    replays the captured graph after an eager eval pass, exercising the
+    graph/eager/prefetcher transition."""
     import argparse
 
     from pvraft_amd.engine import Trainer
@@ -172,15 +174,16 @@ def test_trainer_end_to_end_on_gpu(tmp_path):
     args = argparse.Namespace(
         root=str(tmp_path), exp_path="gpu_exp", dataset="SYNTH", max_points=512,
         corr_levels=3, base_scales=0.25, truncate_k=64, iters=2, gamma=0.8,
-        batch_size=2, gpus="", num_epochs=1, weights=None, checkpoint_interval=5,
-        refine=False, num_workers=0, amp=True, synth_len=6, hipgraph=True,
+        batch_size=2, gpus="", num_epochs=2, weights=None, checkpoint_interval=5,
+        refine=False, num_workers=2, amp=True, synth_len=6, hipgraph=True,
     )
     old = trainer_mod.VAL_ITERS
     trainer_mod.VAL_ITERS = 2
     try:
         t = Trainer(args)
-        t.training(1)
-        results = t.val_test(1, mode="val")
+        for epoch in (1, 2):
+            t.training(epoch)
+            results = t.val_test(epoch, mode="val")
     finally:
         trainer_mod.VAL_ITERS = old
     assert results["epe"] >= 0
