@@ -66,6 +66,17 @@ void check_f32(const torch::Tensor& t, const char* name) {
 
 hipStream_t stream() { return at::hip::getCurrentHIPStream().stream(); }
 
+// zero-filled workspace via hipMemsetAsync (a memset node instead of an
+// ATen fill kernel -- these workspaces are allocated on every call and the
+// fill launches were ~2 ms/step in aggregate)
+torch::Tensor zeros_fast(at::IntArrayRef sizes,
+                         const torch::TensorOptions& opt) {
+  auto t = torch::empty(sizes, opt);
+  if (t.numel() > 0)
+    hipMemsetAsync(t.data_ptr(), 0, t.numel() * t.element_size(), stream());
+  return t;
+}
+
 }  // namespace
 
 torch::Tensor knn_graph(torch::Tensor xyz, int64_t k) {
@@ -171,7 +182,7 @@ std::vector<torch::Tensor> knn_corr_fwd(torch::Tensor corr, torch::Tensor xyz,
 torch::Tensor knn_corr_bwd(torch::Tensor gout, torch::Tensor idx, int64_t K) {
   check_f32(gout, "gout");
   const int B = gout.size(0), k = gout.size(2), N = gout.size(3);
-  auto gcorr = torch::zeros({B, N, K}, gout.options());
+  auto gcorr = zeros_fast({B, N, K}, gout.options());
   launch_knn_corr_bwd(gout.data_ptr<float>(), idx.data_ptr<int>(),
                       gcorr.data_ptr<float>(), B, N, (int)K, k, stream());
   return gcorr;
@@ -200,7 +211,7 @@ std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
   const int rows = B * G;
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto ws = torch::zeros({rows, 2}, fopt);
+  auto ws = zeros_fast({rows, 2}, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty_like(x);
@@ -234,7 +245,7 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
   // one zeroed workspace: [rows*2 row sums | C*2 channel sums | 1 d slope]
-  auto ws = torch::zeros({rows * 2 + C * 2 + 1}, fopt);
+  auto ws = zeros_fast({rows * 2 + C * 2 + 1}, fopt);
   float* row_ws = ws.data_ptr<float>();
   float* chan_ws = row_ws + rows * 2;
   float* slope_ws = chan_ws + C * 2;
@@ -275,7 +286,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto ws = torch::zeros({rows, 2}, fopt);
+  auto ws = zeros_fast({rows, 2}, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty({B, C, N}, x.options());
@@ -305,7 +316,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto ws = torch::zeros({rows * 2 + C * 2 + 1}, fopt);
+  auto ws = zeros_fast({rows * 2 + C * 2 + 1}, fopt);
   float* row_ws = ws.data_ptr<float>();
   float* chan_ws = row_ws + rows * 2;
   float* slope_ws = chan_ws + C * 2;
@@ -333,14 +344,15 @@ std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
   TORCH_CHECK(dy.size(0) == x.size(0) && dy.size(2) == x.size(2), "shape mismatch");
   const int B = dy.size(0), Co = dy.size(1), Ci = x.size(1);
   const long S = dy.size(2);
-  auto dw = torch::zeros({Co, Ci}, dy.options().dtype(torch::kFloat32));
+  // dw and dbias share one zeroed allocation (both are atomic-accumulated)
+  auto flat = zeros_fast({(long)Co * Ci + (with_bias ? Co : 0)},
+                         dy.options().dtype(torch::kFloat32));
+  auto dw = flat.narrow(0, 0, (long)Co * Ci).view({Co, Ci});
   torch::Tensor dbias;
   float* dbias_ptr = nullptr;
   if (with_bias) {
-    dbias = torch::zeros({Co}, dy.options().dtype(torch::kFloat32));
+    dbias = flat.narrow(0, (long)Co * Ci, Co);
     dbias_ptr = dbias.data_ptr<float>();
-  } else {
-    dbias = torch::Tensor();
   }
   launch_pw_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), dbias_ptr,
                   B, Co, Ci, S, (int)schunks, stream());

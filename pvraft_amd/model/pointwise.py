@@ -23,6 +23,30 @@ from torch import Tensor
 
 SPLITK = 16
 
+# Per-forward cache of autocast weight/bias casts.  PVRaft/PVRaftRefine
+# clear it at forward entry, so within one step the 8-iteration GRU loop
+# casts each weight once instead of once per pw_matmul call (~300 extra
+# bf16 copy kernels per step measured).  Values hold the source tensor so
+# ids cannot be reused while a cache entry lives.  Inside hipGraph capture
+# the clear happens at the top of the captured fn, so the cast kernels are
+# recorded and re-read the fp32 weights on every replay.
+_STEP_CACHE: dict = {}
+
+
+def clear_step_cache() -> None:
+    _STEP_CACHE.clear()
+
+
+def _cast_cached(t: Tensor, dt) -> Tensor:
+    if t.dtype == dt:
+        return t
+    key = (id(t), dt)
+    hit = _STEP_CACHE.get(key)
+    if hit is None:
+        hit = (t, t.to(dt))
+        _STEP_CACHE[key] = hit
+    return hit[1]
+
 
 class _PwMatmul(torch.autograd.Function):
     """y (B, Co, S) = w (Co, Ci) @ x (B, Ci, S) with split-K backward.
@@ -77,10 +101,10 @@ def pw_matmul(weight: Tensor, x: Tensor, bias: Tensor = None) -> Tensor:
     w = weight
     if x.is_cuda and torch.is_autocast_enabled():
         dt = torch.get_autocast_dtype("cuda")
-        w = w.to(dt)
+        w = _cast_cached(w, dt)
         x = x.to(dt)
         if bias is not None:
-            bias = bias.to(dt)
+            bias = _cast_cached(bias, dt)
     elif x.dtype != w.dtype:
         x = x.to(w.dtype)
     if bias is not None and bias.dtype != x.dtype:

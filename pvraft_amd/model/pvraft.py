@@ -60,6 +60,9 @@ class PVRaft(nn.Module):
         )
 
     def forward(self, p, num_iters: int = 12) -> List[Tensor]:
+        from .pointwise import clear_step_cache
+
+        clear_step_cache()  # fresh autocast weight casts each forward
         xyz1, xyz2 = p
         graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
         fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
@@ -134,6 +137,9 @@ class PVRaftRefine(nn.Module):
                 pmt.requires_grad_(False)
 
     def forward(self, p, num_iters: int = 32) -> Tensor:
+        from .pointwise import clear_step_cache
+
+        clear_step_cache()
         with torch.no_grad():
             xyz1, xyz2 = p
             graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
