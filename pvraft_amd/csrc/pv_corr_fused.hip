@@ -1,30 +1,40 @@
 // K4+K5 fused: the per-GRU-iteration point-voxel correlation lookup
 // (reference model/corr.py:44-93) as ONE kernel per direction.  Both the
 // voxel pyramid and the kNN branch read the same (corr, xyz, coords)
-// candidate field; fusing them loads each point's K candidates into
-// registers once per iteration instead of twice (+ once per pyramid level).
+// candidate field; fusing them keeps the field's HBM traffic to one pass
+// per phase and the intermediate (B,N,K) distance / (B,N,27) scatter
+// tensors never exist.
 //
-// Geometry: one wave per point (block = 256 = 4 points), lanes own
-// ceil(K/64) <= 8 candidates in statically-indexed VGPRs.
+// Geometry: one wave per point (block = 256 = 4 points).
+//
+// Voxel binning uses per-wave LDS histograms (ds_add_f32): a first cut
+// kept 27 (sum, count) accumulators per LANE in statically-indexed VGPRs
+// and reduced them with 54 wave butterflies per level -- ~2300 VALU ops
+// per wave per level; the LDS-atomic histogram replaces that with 2
+// LDS atomics per candidate (conflicts serialize inside the LDS pipe,
+// bounded by the cell occupancy) and no cross-lane reduction at all,
+// since the histogram is already wave-wide.
 //
 // forward outputs:
-//   voxel (B, L*27, N): per-level 3^3 mean of corr over quantised offsets
-//     (27 statically-indexed accumulators per lane, wave butterfly reduce);
+//   voxel (B, L*27, N): per-level 3^3 mean of corr over quantised offsets;
 //   knn (B, 4, k, N): [corr; rel-xyz] of the k nearest candidates
 //     (wave argmin rounds, ties to the smallest index);
 //   knn_idx (B, N, k) for backward.
 // backward (d corr only; quantisation/selection indices are constants to
 // autograd, reference corr.py:52-62 and coords detached per iteration):
 //   d corr[j] = sum_l [valid_l(j)] g_vox[l*27+cell_l(j)] / cnt_l(cell)
-//             + [j selected at slot t] g_knn[0, t].
+//             + [j selected at slot t] g_knn[0, t]
+// with the kNN term resolved by an LDS scatter over the candidate row
+// (selected slots are unique within a row) instead of a k-way scan.
 #include <hip/hip_runtime.h>
 #include "common.h"
 
 #define R 3
 #define CELLS 27
 #define MAXL 4
-#define MAXC 8   // K <= 512
-#define MAXKN 64 // selected neighbours <= 64
+#define MAXC 8    // K <= 512 candidate slots per lane
+#define MAXKN 64  // selected neighbours <= 64
+#define MAXK 512  // candidate field width (backward LDS scatter row)
 
 __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     const float *__restrict__ corr,    // (B, N, K)
@@ -34,11 +44,15 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     float *__restrict__ knn,           // (B, 4, k, N)
     int *__restrict__ knn_idx,         // (B, N, k)
     int B, int N, int K, int L, int k, float base_scale) {
+  __shared__ float s_sum[4][CELLS];
+  __shared__ float s_cnt[4][CELLS];
+
   const long p = (long)blockIdx.x * 4 + wave_id();
   if (p >= (long)B * N) return;
   const int b = (int)(p / N);
   const int n = (int)(p % N);
   const int lane = lane_id();
+  const int w = wave_id();
 
   const float cx = coords[p * 3 + 0];
   const float cy = coords[p * 3 + 1];
@@ -46,50 +60,31 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
   const float *cand_xyz = xyz + p * K * 3;
   const float *cand_corr = corr + p * K;
 
-  // ---- voxel pyramid (candidates re-read per phase: L1-resident;
-  // caching them in VGPRs or keeping the kNN distances live across this
-  // loop measured SLOWER via occupancy)
+  // ---- voxel pyramid: per-wave LDS histogram per level
   for (int l = 0; l < L; ++l) {
     const float inv_r = 1.0f / (base_scale * (float)(1 << l));
-    float s[CELLS], c[CELLS];
-#pragma unroll
-    for (int q = 0; q < CELLS; ++q) {
-      s[q] = 0.f;
-      c[q] = 0.f;
+    if (lane < CELLS) {
+      s_sum[w][lane] = 0.f;
+      s_cnt[w][lane] = 0.f;
     }
+    __threadfence_block();
     for (int j = lane; j < K; j += WAVE) {
-      {
-        const float dx = rintf((cand_xyz[j * 3 + 0] - cx) * inv_r);
-        const float dy = rintf((cand_xyz[j * 3 + 1] - cy) * inv_r);
-        const float dz = rintf((cand_xyz[j * 3 + 2] - cz) * inv_r);
-        const bool valid = fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
-                           fabsf(dz) <= (R / 2);
+      const float dx = rintf((cand_xyz[j * 3 + 0] - cx) * inv_r);
+      const float dy = rintf((cand_xyz[j * 3 + 1] - cy) * inv_r);
+      const float dz = rintf((cand_xyz[j * 3 + 2] - cz) * inv_r);
+      if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) && fabsf(dz) <= (R / 2)) {
         const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
                          ((int)dz + R / 2);
-        const float v = cand_corr[j];
-#pragma unroll
-        for (int q = 0; q < CELLS; ++q) {
-          const bool hit = valid && (cell == q);
-          s[q] += hit ? v : 0.f;
-          c[q] += hit ? 1.f : 0.f;
-        }
+        atomicAdd(&s_sum[w][cell], cand_corr[j]);
+        atomicAdd(&s_cnt[w][cell], 1.f);
       }
     }
-#pragma unroll
-    for (int q = 0; q < CELLS; ++q) {
-      s[q] = wave_sum(s[q]);
-      c[q] = wave_sum(c[q]);
-    }
-    if (lane < CELLS) {
-      float sv = 0.f, cc = 0.f;
-#pragma unroll
-      for (int q = 0; q < CELLS; ++q)
-        if (lane == q) {
-          sv = s[q];
-          cc = c[q];
-        }
-      vox[((long)b * L * CELLS + l * CELLS + lane) * N + n] = sv / fmaxf(cc, 1.f);
-    }
+    __threadfence_block();
+    if (lane < CELLS)
+      vox[((long)b * L * CELLS + l * CELLS + lane) * N + n] =
+          s_sum[w][lane] / fmaxf(s_cnt[w][lane], 1.f);
+    // next level's zeroing is ordered behind these reads by instruction
+    // order within the wave (LDS ops issue in order from one wave)
   }
 
   // ---- kNN branch (argmin rounds on register distances, computed here so
@@ -145,74 +140,57 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
     const int *__restrict__ knn_idx,   // (B, N, k)
     float *__restrict__ gcorr,         // (B, N, K)
     int B, int N, int K, int L, int k, float base_scale) {
-  __shared__ int s_sel[4][MAXKN];
-  __shared__ float s_g0[4][MAXKN];
+  __shared__ float s_cnt[4][MAXL][CELLS];
+  __shared__ float s_gv[4][MAXL][CELLS];
+  __shared__ float s_kg[4][MAXK];  // per-candidate kNN grad (scattered)
 
   const long p = (long)blockIdx.x * 4 + wave_id();
-  const bool active = p < (long)B * N;
-  const int b = active ? (int)(p / N) : 0;
-  const int n = active ? (int)(p % N) : 0;
+  if (p >= (long)B * N) return;
+  const int b = (int)(p / N);
+  const int n = (int)(p % N);
   const int lane = lane_id();
   const int w = wave_id();
-
-  // knn selections + channel-0 grads into LDS (per wave); every thread
-  // reaches the barrier (inactive tail waves included)
-  if (active && lane < k) {
-    s_sel[w][lane] = knn_idx[p * k + lane];
-    s_g0[w][lane] = g_knn[(((long)b * 4) * k + lane) * N + n];
-  }
-  __syncthreads();
-  if (!active) return;
 
   const float cx = coords[p * 3 + 0];
   const float cy = coords[p * 3 + 1];
   const float cz = coords[p * 3 + 2];
   const float *cand_xyz = xyz + p * K * 3;
 
-  // per-level counts + this lane's cell's (count, g_vox)
-  float cnt_mine[MAXL], g_mine[MAXL];
-#pragma unroll
-  for (int l = 0; l < MAXL; ++l) {
-    cnt_mine[l] = 0.f;
-    g_mine[l] = 0.f;
+  // scatter the kNN channel-0 grads onto their candidate slots (unique)
+  for (int j = lane; j < K; j += WAVE) s_kg[w][j] = 0.f;
+  __threadfence_block();
+  if (lane < k) {
+    const int sel = knn_idx[p * k + lane];
+    const float g0 = g_knn[(((long)b * 4) * k + lane) * N + n];
+    s_kg[w][sel] = g0;  // slots unique within the row: plain store
   }
+
+  // per-level cell counts (LDS histogram) + g_vox row into LDS
   for (int l = 0; l < L; ++l) {
     const float inv_r = 1.0f / (base_scale * (float)(1 << l));
-    float c[CELLS];
-#pragma unroll
-    for (int q = 0; q < CELLS; ++q) c[q] = 0.f;
+    if (lane < CELLS) {
+      s_cnt[w][l][lane] = 0.f;
+      s_gv[w][l][lane] =
+          g_vox[((long)b * L * CELLS + l * CELLS + lane) * N + n];
+    }
+    __threadfence_block();
     for (int j = lane; j < K; j += WAVE) {
       const float dx = rintf((cand_xyz[j * 3 + 0] - cx) * inv_r);
       const float dy = rintf((cand_xyz[j * 3 + 1] - cy) * inv_r);
       const float dz = rintf((cand_xyz[j * 3 + 2] - cz) * inv_r);
-      const bool valid = fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
-                         fabsf(dz) <= (R / 2);
-      const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
-                       ((int)dz + R / 2);
-#pragma unroll
-      for (int q = 0; q < CELLS; ++q) c[q] += (valid && cell == q) ? 1.f : 0.f;
+      if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) && fabsf(dz) <= (R / 2))
+        atomicAdd(&s_cnt[w][l][(int)dx * (R * R) + (int)dy * R + (int)dz +
+                               (R / 2) * (R * R + R + 1)],
+                  1.f);
     }
-#pragma unroll
-    for (int q = 0; q < CELLS; ++q) c[q] = wave_sum(c[q]);
-    float cc = 0.f;
-#pragma unroll
-    for (int q = 0; q < CELLS; ++q)
-      if (lane == q) cc = c[q];
-#pragma unroll
-    for (int ll = 0; ll < MAXL; ++ll)
-      if (ll == l) {
-        cnt_mine[ll] = cc;
-        g_mine[ll] = (lane < CELLS)
-                         ? g_vox[((long)b * L * CELLS + l * CELLS + lane) * N + n]
-                         : 0.f;
-      }
   }
+  __threadfence_block();
 
   for (int j = lane; j < K; j += WAVE) {
     const float ox = cand_xyz[j * 3 + 0] - cx;
     const float oy = cand_xyz[j * 3 + 1] - cy;
     const float oz = cand_xyz[j * 3 + 2] - cz;
-    float g = 0.f;
+    float g = s_kg[w][j];
 #pragma unroll
     for (int l = 0; l < MAXL; ++l) {
       if (l < L) {
@@ -220,18 +198,14 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
         const float dx = rintf(ox * inv_r);
         const float dy = rintf(oy * inv_r);
         const float dz = rintf(oz * inv_r);
-        const bool valid = fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
-                           fabsf(dz) <= (R / 2);
-        const int cell = valid ? ((int)dx + R / 2) * (R * R) +
-                                     ((int)dy + R / 2) * R + ((int)dz + R / 2)
-                               : 0;
-        const float gq = __shfl(g_mine[l], cell, WAVE);
-        const float cq = __shfl(cnt_mine[l], cell, WAVE);
-        g += valid ? gq / fmaxf(cq, 1.f) : 0.f;
+        if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
+            fabsf(dz) <= (R / 2)) {
+          const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
+                           ((int)dz + R / 2);
+          g += s_gv[w][l][cell] / fmaxf(s_cnt[w][l][cell], 1.f);
+        }
       }
     }
-    // knn contribution: selected slots are unique within the row
-    for (int t2 = 0; t2 < k; ++t2) g += (s_sel[w][t2] == j) ? s_g0[w][t2] : 0.f;
     gcorr[p * K + j] = g;
   }
 }
