@@ -66,11 +66,27 @@ void check_f32(const torch::Tensor& t, const char* name) {
 
 hipStream_t stream() { return at::hip::getCurrentHIPStream().stream(); }
 
-// zero-filled workspace via hipMemsetAsync (a memset node instead of an
-// ATen fill kernel -- these workspaces are allocated on every call and the
-// fill launches were ~2 ms/step in aggregate)
+// zero-filled workspace via hipMemsetAsync (cheaper than an ATen fill
+// kernel; these workspaces are allocated on every call and the fill
+// launches were ~2 ms/step in aggregate).
+//
+// NEVER memset while the stream is capturing into a hipGraph: on ROCm 7.2
+// a memset enqueued during capture is mis-ordered at replay against the
+// kernels consuming the workspace -- replays then intermittently read
+// partially-zeroed GroupNorm/wgrad workspaces and produce garbage
+// gradients (reproduced 6/6 at tiny config, OK 6/6 with the memset
+// disabled; see profiles/README.md).  Inside capture the plain fill
+// kernel is captured instead, which replays correctly.
 torch::Tensor zeros_fast(at::IntArrayRef sizes,
                          const torch::TensorOptions& opt) {
+  static const bool disabled = [] {
+    const char* e = getenv("PVRAFT_NO_MEMSET");
+    return e && e[0] == '1';
+  }();
+  hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+  hipStreamIsCapturing(stream(), &cap);
+  if (disabled || cap != hipStreamCaptureStatusNone)
+    return torch::zeros(sizes, opt);
   auto t = torch::empty(sizes, opt);
   if (t.numel() > 0)
     hipMemsetAsync(t.data_ptr(), 0, t.numel() * t.element_size(), stream());

@@ -40,6 +40,10 @@ def clear_step_cache() -> None:
 def _cast_cached(t: Tensor, dt) -> Tensor:
     if t.dtype == dt:
         return t
+    import os
+
+    if os.environ.get("PVRAFT_NO_CAST_CACHE", "0") == "1":
+        return t.to(dt)
     key = (id(t), dt)
     hit = _STEP_CACHE.get(key)
     if hit is None:
