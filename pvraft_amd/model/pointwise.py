@@ -59,23 +59,33 @@ class _PwMatmul(torch.autograd.Function):
     decomposition clones activation-sized tensors whenever gradients are
     required (measured ~14 ms/step of hidden aten::copy_); bmm is a native
     batched op with no fold path.
+
+    The autocast weight cast happens INSIDE the Function (``compute_dt``):
+    the Function's inputs stay the fp32 leaves, so the fp32 weight/bias
+    grads from the wgrad kernel are returned directly -- no dw->bf16 copy
+    + ToCopyBackward cast-back pair per weight per iteration (~250 copy
+    kernels/step measured when the cast was an autograd-visible op).
     """
 
     @staticmethod
-    def forward(ctx, w: Tensor, x: Tensor, bias) -> Tensor:
-        ctx.save_for_backward(w, x)
+    def forward(ctx, w: Tensor, x: Tensor, bias, compute_dt) -> Tensor:
+        wc = w if compute_dt is None else _cast_cached(w, compute_dt)
+        bc = bias if (bias is None or compute_dt is None) else _cast_cached(bias, compute_dt)
+        ctx.save_for_backward(wc, x)
         ctx.has_bias = bias is not None
-        y = torch.bmm(w.unsqueeze(0).expand(x.shape[0], -1, -1), x)
-        if bias is not None:
-            y = y + bias.view(1, -1, 1)
+        ctx.grad_dtypes = (w.dtype, bias.dtype if bias is not None else None)
+        y = torch.bmm(wc.unsqueeze(0).expand(x.shape[0], -1, -1), x)
+        if bc is not None:
+            y = y + bc.view(1, -1, 1)
         return y
 
     @staticmethod
     def backward(ctx, dy: Tensor):
-        w, x = ctx.saved_tensors
+        wc, x = ctx.saved_tensors
+        w_dtype, b_dtype = ctx.grad_dtypes
         dy = dy.contiguous()
         dx = (
-            torch.bmm(w.t().unsqueeze(0).expand(dy.shape[0], -1, -1), dy)
+            torch.bmm(wc.t().unsqueeze(0).expand(dy.shape[0], -1, -1), dy)
             if ctx.needs_input_grad[1]
             else None
         )
@@ -88,32 +98,35 @@ class _PwMatmul(torch.autograd.Function):
                     from pvraft_amd import _C
 
                     dw, db = _C.pw_wgrad(dy, x, 0, need_bias)
-                    dw = dw.to(w.dtype)
                     if need_bias:
-                        dbias = db.to(dy.dtype)  # grad dtype must match input
+                        dbias = db.to(b_dtype) if b_dtype != db.dtype else db
                         need_bias = False
                 except (ImportError, AttributeError):
-                    dw = torch.einsum("bos,bis->oi", dy, x).to(w.dtype)
+                    dw = torch.einsum("bos,bis->oi", dy, x)
             else:
-                dw = torch.einsum("bos,bis->oi", dy, x).to(w.dtype)
+                dw = torch.einsum("bos,bis->oi", dy, x)
+            if dw.dtype != w_dtype:
+                dw = dw.to(w_dtype)
         if need_bias:
             dbias = dy.sum(dim=(0, 2))
-        return dw, dx, dbias
+            if dbias.dtype != b_dtype:
+                dbias = dbias.to(b_dtype)
+        return dw, dx, dbias, None
 
 
 def pw_matmul(weight: Tensor, x: Tensor, bias: Tensor = None) -> Tensor:
     w = weight
+    dt = None
     if x.is_cuda and torch.is_autocast_enabled():
         dt = torch.get_autocast_dtype("cuda")
-        w = _cast_cached(w, dt)
         x = x.to(dt)
-        if bias is not None:
-            bias = _cast_cached(bias, dt)
+        if w.dtype == dt:
+            dt = None  # already compute-dtype (e.g. pre-cast slice)
     elif x.dtype != w.dtype:
         x = x.to(w.dtype)
-    if bias is not None and bias.dtype != x.dtype:
+    if bias is not None and dt is None and bias.dtype != x.dtype:
         bias = bias.to(x.dtype)
-    return _PwMatmul.apply(w, x.contiguous(), bias)
+    return _PwMatmul.apply(w, x.contiguous(), bias, dt)
 
 
 class PwConv1d(nn.Conv1d):

@@ -183,26 +183,15 @@ class UpdateBlock(nn.Module):
         """Prepare the concat-free weights once per forward (the GRU loop
         reuses them across all iterations): cross-gate fused GRU weights
         plus the motion/flow-head column slices."""
-        cache = {
+        # slices stay in the weights' dtype: pw_matmul's per-forward cast
+        # cache dedupes the autocast cast across the GRU iterations, and
+        # keeping the Function inputs fp32 keeps the weight grads fp32
+        # end-to-end (no bf16 round-trip per iteration)
+        return {
             "gru": self.gru.gate_weights([64, 61, 3]),  # [inp, motion61, flow3]
             "motion": _slice_weight(self.motion_encoder.conv.weight, [64, 64]),
             "flowhead": _slice_weight(self.flow_head.out_conv[0].weight, [64, 64]),
         }
-        # pre-cast once under autocast so per-iteration GEMMs skip the cast
-        if torch.is_autocast_enabled():
-            dt = torch.get_autocast_dtype("cuda")
-
-            def cast(v):
-                if isinstance(v, torch.Tensor):
-                    return v.to(dt)
-                if isinstance(v, list):
-                    return [cast(x) for x in v]
-                if isinstance(v, dict):
-                    return {k: cast(x) for k, x in v.items()}
-                return v
-
-            cache = cast(cache)
-        return cache
 
     def forward(self, net: Tensor, inp: Tensor, corr: Tensor, flow: Tensor, graph: Graph,
                 wcache=None, inp_pre=None):
