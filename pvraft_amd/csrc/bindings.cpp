@@ -93,6 +93,18 @@ torch::Tensor zeros_fast(at::IntArrayRef sizes,
   return t;
 }
 
+// persistent zero workspace for the GN forward reduction: the finalize
+// kernel re-zeroes it after consuming, so one buffer per size serves every
+// call and every hipGraph replay (stream-ordered) with no per-call
+// allocation or fill.  Held forever (a few KB total).
+torch::Tensor& gn_fwd_ws(int rows, const torch::TensorOptions& opt) {
+  static std::unordered_map<int, torch::Tensor> cache;
+  auto it = cache.find(rows);
+  if (it == cache.end())
+    it = cache.emplace(rows, torch::zeros({rows, 2}, opt)).first;
+  return it->second;
+}
+
 }  // namespace
 
 torch::Tensor knn_graph(torch::Tensor xyz, int64_t k) {
@@ -227,7 +239,7 @@ std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
   const int rows = B * G;
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto ws = zeros_fast({rows, 2}, fopt);
+  auto& ws = gn_fwd_ws(rows, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty_like(x);
@@ -302,7 +314,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto ws = zeros_fast({rows, 2}, fopt);
+  auto& ws = gn_fwd_ws(rows, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty({B, C, N}, x.options());

@@ -110,8 +110,11 @@ __global__ __launch_bounds__(GN_THREADS) void gn_fwd_reduce_kernel(
   }
 }
 
-// pass 2: rows threads -> mean/rstd
-__global__ void gn_fwd_finalize_kernel(const float *__restrict__ ws,
+// pass 2: rows threads -> mean/rstd.  The workspace is self-cleaning:
+// finalize is its only consumer and writes it back to zero, so the SAME
+// persistent buffer serves every call (and every hipGraph replay) with no
+// per-call allocation or fill kernel -- calls are ordered on the stream.
+__global__ void gn_fwd_finalize_kernel(float *__restrict__ ws,
                                        float *__restrict__ mean,
                                        float *__restrict__ rstd, long row_len,
                                        int rows, float eps) {
@@ -119,6 +122,8 @@ __global__ void gn_fwd_finalize_kernel(const float *__restrict__ ws,
   if (row >= rows) return;
   const float m = ws[row * 2 + 0] / (float)row_len;
   const float var = ws[row * 2 + 1] / (float)row_len - m * m;
+  ws[row * 2 + 0] = 0.f;
+  ws[row * 2 + 1] = 0.f;
   mean[row] = m;
   rstd[row] = rsqrtf(fmaxf(var, 0.f) + eps);
 }
