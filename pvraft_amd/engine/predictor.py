@@ -27,13 +27,15 @@ class Predictor:
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self._in1 = torch.zeros(batch, points, 3, device=self.device)
         self._in2 = torch.zeros(batch, points, 3, device=self.device)
-        self._out: Optional[torch.Tensor] = None
+        self._flows = None
+        self.last_flows = None
 
-    def _forward(self) -> torch.Tensor:
+    def _forward(self):
         with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.amp):
             flows = self.model([self._in1, self._in2], num_iters=self.iters)
-        out = flows[-1] if isinstance(flows, (list, tuple)) else flows
-        return out.float()
+        if not isinstance(flows, (list, tuple)):
+            flows = [flows]
+        return [f.float() for f in flows]
 
     def _capture(self) -> None:
         s = torch.cuda.Stream()
@@ -45,13 +47,17 @@ class Predictor:
         torch.cuda.synchronize()
         self._graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self._graph):
-            self._out = self._forward()
+            self._flows = self._forward()
 
     @torch.no_grad()
     def __call__(self, xyz1: torch.Tensor, xyz2: torch.Tensor) -> torch.Tensor:
+        """Returns the FINAL flow (B, N, 3); ``last_flows`` holds the whole
+        per-iteration list (static graph outputs -- consume before the next
+        call) for sequence-loss evaluation."""
         if not self.use_graph:
             self._in1, self._in2 = xyz1.to(self.device), xyz2.to(self.device)
-            return self._forward()
+            self.last_flows = self._forward()
+            return self.last_flows[-1]
         if self._graph is None:
             self._in1.copy_(xyz1)
             self._in2.copy_(xyz2)
@@ -59,4 +65,5 @@ class Predictor:
         self._in1.copy_(xyz1, non_blocking=True)
         self._in2.copy_(xyz2, non_blocking=True)
         self._graph.replay()
-        return self._out.clone()
+        self.last_flows = self._flows
+        return self._flows[-1].clone()

@@ -281,6 +281,19 @@ class Trainer:
     def _eval_iters(self) -> int:
         return VAL_ITERS
 
+    def _val_predictor(self, points: int):
+        """Graph-captured eval forward (shared by val and test: both run
+        bs=1 x max_points); weights are re-read at every replay, so the
+        same graph serves all epochs."""
+        pred = getattr(self, "_val_pred", None)
+        if pred is None or pred._in1.shape[1] != points or pred.iters != self._eval_iters():
+            from .predictor import Predictor
+
+            pred = Predictor(self.model, points=points, batch=1,
+                             iters=self._eval_iters(), amp=self.amp)
+            self._val_pred = pred
+        return pred
+
     @torch.no_grad()
     def val_test(self, epoch: Optional[int] = None, mode: str = "val"):
         if mode == "test" and epoch is None:
@@ -291,13 +304,20 @@ class Trainer:
             if os.path.isfile(best):
                 load_checkpoint(best, self.model, strict=True)
         self.model.eval()
+        use_graph = self._graph_enabled()
         loader = self.val_loader if mode == "val" else self.test_loader
         sums = torch.zeros(6, dtype=torch.float64, device=self.device)  # loss,epe,s,r,out,count
         for batch in loader:
             batch = batch.to(self.device, non_blocking=True)
-            with torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.amp):
-                est_flow = self.model(batch["sequence"], num_iters=self._eval_iters())
-            final = self._final_flow(est_flow).float()
+            pc1, pc2 = batch["sequence"]
+            if use_graph and pc1.shape[0] == 1 and pc1.shape[1] == pc2.shape[1]:
+                pred = self._val_predictor(pc1.shape[1])
+                final = pred(pc1, pc2)
+                est_flow = pred.last_flows if self.loss_is_sequence else pred.last_flows[-1]
+            else:
+                with torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.amp):
+                    est_flow = self.model(batch["sequence"], num_iters=self._eval_iters())
+                final = self._final_flow(est_flow).float()
             loss = self._loss(est_flow, batch)
             epe3d, accs, accr, outl = compute_epe(final, batch)
             sums += torch.tensor(
