@@ -38,6 +38,12 @@ void launch_pv_corr_fused_bwd(const float*, const float*, const float*,
                               int, int, float, hipStream_t);
 void launch_topk_rows(const float*, float*, int*, long, int, int,
                       hipStream_t);
+void launch_seq_loss_fwd(const float* const*, const float*, const float*,
+                         float*, float*, float*, long, int, int, float,
+                         hipStream_t);
+void launch_seq_loss_bwd(const float* const*, float* const*, const float*,
+                         const float*, const float*, const float*, long, int,
+                         int, float, hipStream_t);
 void launch_gru_zr_fwd(const void*, const void*, void*, void*, void*, long,
                        long, bool, hipStream_t);
 void launch_gru_zr_bwd(const void*, const void*, const void*, const void*,
@@ -98,10 +104,12 @@ torch::Tensor zeros_fast(at::IntArrayRef sizes,
 // call and every hipGraph replay (stream-ordered) with no per-call
 // allocation or fill.  Held forever (a few KB total).
 torch::Tensor& gn_fwd_ws(int rows, const torch::TensorOptions& opt) {
-  static std::unordered_map<int, torch::Tensor> cache;
-  auto it = cache.find(rows);
-  if (it == cache.end())
-    it = cache.emplace(rows, torch::zeros({rows, 2}, opt)).first;
+  // leaked on purpose: a static map of CUDA tensors must not run its
+  // destructor during process teardown (races CUDA context destruction)
+  static auto* cache = new std::unordered_map<int, torch::Tensor>();
+  auto it = cache->find(rows);
+  if (it == cache->end())
+    it = cache->emplace(rows, torch::zeros({rows, 2}, opt)).first;
   return it->second;
 }
 
@@ -519,8 +527,64 @@ std::vector<torch::Tensor> gru_q_bwd(torch::Tensor dhnew, torch::Tensor q,
   return {dpre, dz, dh};
 }
 
+// fused gamma-weighted sequence loss over T flows -> {loss, denom}
+std::vector<torch::Tensor> seq_loss_fwd(std::vector<torch::Tensor> flows,
+                                        torch::Tensor gt, torch::Tensor mask,
+                                        double gamma) {
+  const int T = (int)flows.size();
+  TORCH_CHECK(T >= 1 && T <= 32, "seq_loss supports 1..32 flows");
+  check_f32(gt, "gt");
+  check_f32(mask, "mask");
+  const long BN = gt.numel() / 3;
+  TORCH_CHECK(mask.numel() == BN, "mask/gt shape mismatch");
+  const float* ptrs[32];
+  for (int t = 0; t < T; ++t) {
+    check_f32(flows[t], "flow");
+    TORCH_CHECK(flows[t].numel() == BN * 3, "flow shape mismatch");
+    ptrs[t] = flows[t].data_ptr<float>();
+  }
+  auto fopt = gt.options();
+  // persistent self-cleaning workspace (finalize re-zeroes); leaked on
+  // purpose so its destructor never races CUDA teardown at exit
+  static torch::Tensor* ws_p = new torch::Tensor();
+  torch::Tensor& ws = *ws_p;
+  if (!ws.defined()) ws = torch::zeros({33}, fopt);
+  auto loss = torch::empty({}, fopt);
+  auto denom = torch::empty({}, fopt);
+  launch_seq_loss_fwd(ptrs, gt.data_ptr<float>(), mask.data_ptr<float>(),
+                      ws.data_ptr<float>(), loss.data_ptr<float>(),
+                      denom.data_ptr<float>(), BN, T, 1, (float)gamma,
+                      stream());
+  return {loss, denom};
+}
+
+std::vector<torch::Tensor> seq_loss_bwd(std::vector<torch::Tensor> flows,
+                                        torch::Tensor gt, torch::Tensor mask,
+                                        torch::Tensor dloss,
+                                        torch::Tensor denom, double gamma) {
+  const int T = (int)flows.size();
+  TORCH_CHECK(T >= 1 && T <= 32);
+  const long BN = gt.numel() / 3;
+  const float* ptrs[32];
+  float* gptrs[32];
+  std::vector<torch::Tensor> grads;
+  grads.reserve(T);
+  for (int t = 0; t < T; ++t) {
+    ptrs[t] = flows[t].data_ptr<float>();
+    grads.push_back(torch::empty_like(flows[t]));
+    gptrs[t] = grads[t].data_ptr<float>();
+  }
+  launch_seq_loss_bwd(ptrs, gptrs, gt.data_ptr<float>(),
+                      mask.data_ptr<float>(), dloss.data_ptr<float>(),
+                      denom.data_ptr<float>(), BN, T, 1, (float)gamma,
+                      stream());
+  return grads;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_rows", &topk_rows);
+  m.def("seq_loss_fwd", &seq_loss_fwd);
+  m.def("seq_loss_bwd", &seq_loss_bwd);
   m.def("gru_zr_fwd", &gru_zr_fwd);
   m.def("gru_zr_bwd", &gru_zr_bwd);
   m.def("gru_q_fwd", &gru_q_fwd);
