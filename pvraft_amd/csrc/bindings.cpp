@@ -57,6 +57,8 @@ void launch_transpose(const void*, void*, long, int, long, long, bool,
                       hipStream_t);
 void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
                      long, int, hipStream_t);
+void launch_gn_bwd_extract(float*, float*, float*, float*, int, int,
+                           hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
                      float*, float*, float*, void*, int, long, long, int, int,
@@ -103,13 +105,13 @@ torch::Tensor zeros_fast(at::IntArrayRef sizes,
 // kernel re-zeroes it after consuming, so one buffer per size serves every
 // call and every hipGraph replay (stream-ordered) with no per-call
 // allocation or fill.  Held forever (a few KB total).
-torch::Tensor& gn_fwd_ws(int rows, const torch::TensorOptions& opt) {
+torch::Tensor& persistent_ws(long len, const torch::TensorOptions& opt) {
   // leaked on purpose: a static map of CUDA tensors must not run its
   // destructor during process teardown (races CUDA context destruction)
-  static auto* cache = new std::unordered_map<int, torch::Tensor>();
-  auto it = cache->find(rows);
+  static auto* cache = new std::unordered_map<long, torch::Tensor>();
+  auto it = cache->find(len);
   if (it == cache->end())
-    it = cache->emplace(rows, torch::zeros({rows, 2}, opt)).first;
+    it = cache->emplace(len, torch::zeros({len}, opt)).first;
   return it->second;
 }
 
@@ -247,7 +249,7 @@ std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
   const int rows = B * G;
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto& ws = gn_fwd_ws(rows, fopt);
+  auto& ws = persistent_ws((long)rows * 2, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty_like(x);
@@ -281,7 +283,7 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
   // one zeroed workspace: [rows*2 row sums | C*2 channel sums | 1 d slope]
-  auto ws = zeros_fast({rows * 2 + C * 2 + 1}, fopt);
+  auto& ws = persistent_ws((long)rows * 2 + C * 2 + 1, fopt);
   float* row_ws = ws.data_ptr<float>();
   float* chan_ws = row_ws + rows * 2;
   float* slope_ws = chan_ws + C * 2;
@@ -295,10 +297,12 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
                 bias.data_ptr<float>(), row_ws, chan_ws, slope_ws,
                 dx.data_ptr(), rows, row_len, S, C, (int)G, (int)act,
                 (float)slope, slope_ptr, bf16, stream());
-  auto cw = ws.narrow(0, rows * 2, C * 2).view({C, 2});
-  auto dbias = cw.select(1, 0).contiguous();
-  auto dweight = cw.select(1, 1).contiguous();
-  auto dslope = ws.narrow(0, rows * 2 + C * 2, 1);
+  auto dweight = torch::empty({C}, fopt);
+  auto dbias = torch::empty({C}, fopt);
+  auto dslope = torch::empty({1}, fopt);
+  launch_gn_bwd_extract(ws.data_ptr<float>(), dweight.data_ptr<float>(),
+                        dbias.data_ptr<float>(), dslope.data_ptr<float>(),
+                        rows, C, stream());
   return {dx, dweight, dbias, dslope};
 }
 
@@ -322,7 +326,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto& ws = gn_fwd_ws(rows, fopt);
+  auto& ws = persistent_ws((long)rows * 2, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty({B, C, N}, x.options());
@@ -352,7 +356,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto ws = zeros_fast({rows * 2 + C * 2 + 1}, fopt);
+  auto& ws = persistent_ws((long)rows * 2 + C * 2 + 1, fopt);
   float* row_ws = ws.data_ptr<float>();
   float* chan_ws = row_ws + rows * 2;
   float* slope_ws = chan_ws + C * 2;
@@ -363,10 +367,12 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
                   weight.data_ptr<float>(), bias.data_ptr<float>(), row_ws,
                   chan_ws, slope_ws, dx.data_ptr(), rows, row_len, N, K, C,
                   (int)G, (int)act, (float)slope, slope_ptr, bf16, stream());
-  auto cw = ws.narrow(0, rows * 2, C * 2).view({C, 2});
-  auto dbias = cw.select(1, 0).contiguous();
-  auto dweight = cw.select(1, 1).contiguous();
-  auto dslope = ws.narrow(0, rows * 2 + C * 2, 1);
+  auto dweight = torch::empty({C}, fopt);
+  auto dbias = torch::empty({C}, fopt);
+  auto dslope = torch::empty({1}, fopt);
+  launch_gn_bwd_extract(ws.data_ptr<float>(), dweight.data_ptr<float>(),
+                        dbias.data_ptr<float>(), dslope.data_ptr<float>(),
+                        rows, C, stream());
   return {dx, dweight, dbias, dslope};
 }
 

@@ -692,3 +692,36 @@ void launch_gnmp_bwd(const void *dy, const void *x, const unsigned char *am,
                          rows, row_len, N, K, C, G, act, slope, slope_ptr,
                          stream);
 }
+
+// extract dbias/dweight/dslope from the backward workspace and re-zero the
+// WHOLE workspace, so a persistent buffer serves every call (the fill +
+// two strided-copy + clone launches per backward become this one kernel)
+__global__ void gn_bwd_extract_kernel(float *__restrict__ ws,
+                                      float *__restrict__ dweight,
+                                      float *__restrict__ dbias,
+                                      float *__restrict__ dslope, int rows,
+                                      int C) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < C) {
+    dbias[i] = ws[rows * 2 + 2 * i];
+    dweight[i] = ws[rows * 2 + 2 * i + 1];
+    ws[rows * 2 + 2 * i] = 0.f;
+    ws[rows * 2 + 2 * i + 1] = 0.f;
+  }
+  if (i < rows) {
+    ws[2 * i] = 0.f;
+    ws[2 * i + 1] = 0.f;
+  }
+  if (i == 0) {
+    dslope[0] = ws[rows * 2 + C * 2];
+    ws[rows * 2 + C * 2] = 0.f;
+  }
+}
+
+void launch_gn_bwd_extract(float *ws, float *dweight, float *dbias,
+                           float *dslope, int rows, int C,
+                           hipStream_t stream) {
+  const int n = rows > C ? rows : C;
+  hipLaunchKernelGGL(gn_bwd_extract_kernel, dim3((n + 255) / 256), dim3(256),
+                     0, stream, ws, dweight, dbias, dslope, rows, C);
+}
