@@ -55,9 +55,8 @@ void launch_gru_q_bwd(const void*, const void*, const void*, const void*,
                       void*, void*, void*, long, long, bool, hipStream_t);
 void launch_transpose(const void*, void*, long, int, long, long, bool,
                       hipStream_t);
-void launch_pw_wgrad(const void*, const void*, float*, float*, float*,
-                     float*, int, int, int, long, int, hipStream_t);
-int pw_wgrad_schunks(int, int, int, long, int);
+void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
+                     long, int, hipStream_t);
 void launch_gn_bwd_extract(float*, float*, float*, float*, int, int,
                            hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
@@ -398,25 +397,21 @@ std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
   TORCH_CHECK(dy.size(0) == x.size(0) && dy.size(2) == x.size(2), "shape mismatch");
   const int B = dy.size(0), Co = dy.size(1), Ci = x.size(1);
   const long S = dy.size(2);
-  auto fopt = dy.options().dtype(torch::kFloat32);
-  // two-stage: per-block partial tiles land in a persistent scratch that
-  // every call fully overwrites (never zeroed, no atomics), then a thin
-  // reduce writes dw/dbias into plain torch::empty outputs
-  const int sc = pw_wgrad_schunks(B, Co, Ci, S, (int)schunks);
-  const long to64 = ((Co + 63) / 64) * 64, ti64 = ((Ci + 63) / 64) * 64;
-  const long z = (long)B * sc;
-  auto& scratch = scratch_ws(z * to64 * ti64 + z * to64, fopt);
-  float* part = scratch.data_ptr<float>();
-  float* bias_part = part + z * to64 * ti64;
-  auto dw = torch::empty({Co, Ci}, fopt);
+  // dw and dbias share one zeroed allocation (both are atomic-accumulated;
+  // a two-stage partial-store + reduce variant measured 2x SLOWER -- the
+  // partial-buffer write+read traffic dwarfs the L2 atomics on this tiny
+  // output, scripts/kernel_bench.py)
+  auto flat = zeros_fast({(long)Co * Ci + (with_bias ? Co : 0)},
+                         dy.options().dtype(torch::kFloat32));
+  auto dw = flat.narrow(0, 0, (long)Co * Ci).view({Co, Ci});
   torch::Tensor dbias;
   float* dbias_ptr = nullptr;
   if (with_bias) {
-    dbias = torch::empty({Co}, fopt);
+    dbias = flat.narrow(0, (long)Co * Ci, Co);
     dbias_ptr = dbias.data_ptr<float>();
   }
   launch_pw_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), dbias_ptr,
-                  part, bias_part, B, Co, Ci, S, sc, stream());
+                  B, Co, Ci, S, (int)schunks, stream());
   return {dw, dbias};
 }
 

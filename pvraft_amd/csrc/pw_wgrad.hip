@@ -7,16 +7,9 @@
 // of workgroups on 256 CUs, measured ~210 us for ~5 us of work), and the
 // batched-matmul reformulation forces permute copies (~25 ms/step of
 // aten::copy_).  This kernel reads both operands IN PLACE (row-major over
-// s, k-contiguous) and splits the reduction across B x SCHUNKS workgroups.
-//
-// TWO-STAGE accumulation: stage 1 writes each block's fp32 partial tile to
-// a persistent scratch with PLAIN stores (the padded (B*sc, to*64, ti*64)
-// region is fully overwritten every call, so it is never zeroed and never
-// atomically contended -- the earlier one-stage version did ~4096 global
-// atomicAdds per block onto the tiny dW and needed a zero-fill launch per
-// call, together ~40% of the kernel's time at the per-iteration shapes);
-// stage 2 is a thin reduction producing dW (and the fused bias grad) with
-// one thread per output element.
+// s, k-contiguous) and splits the reduction across B x SCHUNKS workgroups,
+// accumulating fp32 partial tiles into dW with one atomicAdd per output
+// element per block.
 //
 // Geometry: 256 threads = 4 waves per block; 64x64 output tile, each wave
 // one 32x32 quadrant = 2x2 v_mfma_f32_16x16x32_bf16 fragments (16 fp32
@@ -38,12 +31,12 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define KB 32        // K-block per MFMA
 #define LDS_PAD 8    // bf16 elements of row padding (16 B)
 
-__global__ __launch_bounds__(WG_THREADS) void pw_wgrad_partial_kernel(
+__global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
     const __hip_bfloat16 *__restrict__ dy,  // (B, Co, S)
     const __hip_bfloat16 *__restrict__ x,   // (B, Ci, S)
-    float *__restrict__ part,               // (B*sc, to*64, ti*64)
-    float *__restrict__ bias_part_out,      // (B*sc, to*64) or null
-    int B, int Co, int Ci, long S, int schunks, int to64, int ti64) {
+    float *__restrict__ dw,                 // (Co, Ci) pre-zeroed fp32
+    float *__restrict__ dbias,              // (Co) pre-zeroed fp32 or null
+    int B, int Co, int Ci, long S, int schunks) {
   __shared__ __hip_bfloat16 sA[TILE][KB + LDS_PAD];
   __shared__ __hip_bfloat16 sB[TILE][KB + LDS_PAD];
 
@@ -55,8 +48,7 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_partial_kernel(
   const long per = (((S + schunks - 1) / schunks + KB - 1) / KB) * KB;
   const long s_lo = chunk * per;
   const long s_hi = min(s_lo + per, S);
-  // NOTE: blocks with an empty range still fall through and store their
-  // (zero) partial tile -- the scratch is never pre-zeroed.
+  if (s_lo >= S) return;
   const bool vec_ok = (S % 8 == 0);
 
   const int lane = lane_id();
@@ -96,7 +88,7 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_partial_kernel(
         }
       }
       *(bf16x8 *)&sA[ldr][ldc] = av;
-      if (bias_part_out != nullptr && tile_i == 0) {
+      if (dbias != nullptr && tile_i == 0) {
 #pragma unroll
         for (int e = 0; e < 8; ++e) bias_part += (float)((__bf16 *)&av)[e];
       }
@@ -130,22 +122,19 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_partial_kernel(
     }
   }
 
-  // bias-grad partials: reduce this block's per-thread dy-row sums into
-  // one row vector and store it (only the tile_i == 0 column of blocks)
-  if (bias_part_out != nullptr && tile_i == 0) {
+  // bias gradient: reduce the per-thread dy-row partials (4 threads per
+  // row share ldr) and accumulate -- replaces a separate slow ATen reduce
+  if (dbias != nullptr && tile_i == 0) {
     __shared__ float s_bias[TILE];
     if (threadIdx.x < TILE) s_bias[threadIdx.x] = 0.f;
     __syncthreads();
-    atomicAdd(&s_bias[ldr], bias_part);  // LDS, 4 lanes per row
+    atomicAdd(&s_bias[ldr], bias_part);
     __syncthreads();
-    if (threadIdx.x < TILE)
-      bias_part_out[(long)blockIdx.z * to64 + tile_o + threadIdx.x] =
-          s_bias[threadIdx.x];
+    if (threadIdx.x < TILE && tile_o + threadIdx.x < Co)
+      atomicAdd(&dbias[tile_o + threadIdx.x], s_bias[threadIdx.x]);
   }
 
-  // C/D map: col = lane&15, row = (lane>>4)*4 + reg; plain stores into the
-  // padded partial tile (no guards: o_pad < to*64, i_pad < ti*64 always)
-  float *pt = part + (long)blockIdx.z * to64 * ti64;
+  // C/D map: col = lane&15, row = (lane>>4)*4 + reg
   const int crow = (lane >> 4) * 4;
   const int ccol = lane & 15;
 #pragma unroll
@@ -156,63 +145,26 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_partial_kernel(
       for (int e = 0; e < 4; ++e) {
         const int o = tile_o + wo + a * 16 + crow + e;
         const int i = tile_i + wi + bb * 16 + ccol;
-        pt[(long)o * ti64 + i] = acc[a][bb][e];
+        if (o < Co && i < Ci) atomicAdd(&dw[(long)o * Ci + i], acc[a][bb][e]);
       }
 }
 
-// stage 2: dw[o,i] = sum_z part[z,o,i]; threads with i==0 also reduce the
-// bias partials.  Z = B*schunks <= a few hundred; reads are i-coalesced.
-__global__ __launch_bounds__(WG_THREADS) void pw_wgrad_reduce_kernel(
-    const float *__restrict__ part, const float *__restrict__ bias_part,
-    float *__restrict__ dw, float *__restrict__ dbias, int Co, int Ci, int Z,
-    int to64, int ti64) {
-  const long t = (long)blockIdx.x * WG_THREADS + threadIdx.x;
-  const long total = (long)Co * Ci;
-  if (t < total) {
-    const int o = (int)(t / Ci);
-    const int i = (int)(t - (long)o * Ci);
-    float s = 0.f;
-    const long stride = (long)to64 * ti64;
-    const float *p = part + (long)o * ti64 + i;
-    for (int z = 0; z < Z; ++z) s += p[(long)z * stride];
-    dw[t] = s;
-  }
-  if (dbias != nullptr && t < Co) {
-    float s = 0.f;
-    const float *p = bias_part + t;
-    for (int z = 0; z < Z; ++z) s += p[(long)z * to64];
-    dbias[t] = s;
-  }
-}
-
 void launch_pw_wgrad(const void *dy, const void *x, float *dw, float *dbias,
-                     float *part, float *bias_part, int B, int Co, int Ci,
-                     long S, int schunks, hipStream_t stream) {
+                     int B, int Co, int Ci, long S, int schunks_opt,
+                     hipStream_t stream) {
   const int to = (Co + TILE - 1) / TILE;
   const int ti = (Ci + TILE - 1) / TILE;
-  const int to64 = to * TILE, ti64 = ti * TILE;
+  long sc = schunks_opt;
+  if (sc <= 0) {
+    // default: fill ~512 workgroups, but keep >= 4 K-blocks per chunk
+    sc = 512 / ((long)to * ti * B) + 1;
+    long cap = S / (KB * 4);
+    if (sc > cap) sc = cap;
+    if (sc < 1) sc = 1;
+  }
+  int schunks = (int)sc;
   dim3 grid(to, ti, B * schunks);
-  hipLaunchKernelGGL(pw_wgrad_partial_kernel, grid, dim3(WG_THREADS), 0,
-                     stream, (const __hip_bfloat16 *)dy,
-                     (const __hip_bfloat16 *)x, part,
-                     dbias != nullptr ? bias_part : nullptr, B, Co, Ci, S,
-                     schunks, to64, ti64);
-  const long total = (long)Co * Ci;
-  hipLaunchKernelGGL(pw_wgrad_reduce_kernel,
-                     dim3((unsigned)((total + WG_THREADS - 1) / WG_THREADS)),
-                     dim3(WG_THREADS), 0, stream, part, bias_part, dw, dbias,
-                     Co, Ci, B * schunks, to64, ti64);
-}
-
-int pw_wgrad_schunks(int B, int Co, int Ci, long S, int schunks_opt) {
-  if (schunks_opt > 0) return schunks_opt;
-  const int to = (Co + TILE - 1) / TILE;
-  const int ti = (Ci + TILE - 1) / TILE;
-  // atomic-free split: fill the chip, bounded by one KB-block per chunk
-  long sc = 768 / ((long)to * ti * B) + 1;
-  long cap = (S + KB - 1) / KB;
-  if (sc > cap) sc = cap;
-  if (sc > 256) sc = 256;
-  if (sc < 1) sc = 1;
-  return (int)sc;
+  hipLaunchKernelGGL(pw_wgrad_kernel, grid, dim3(WG_THREADS), 0, stream,
+                     (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, dw,
+                     dbias, B, Co, Ci, S, schunks);
 }
