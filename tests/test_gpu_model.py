@@ -191,3 +191,30 @@ def test_trainer_end_to_end_on_gpu(tmp_path):
 
     assert os.path.exists(os.path.join(str(tmp_path), "experiments", "gpu_exp", "checkpoints", "last_checkpoint.params"))
     assert t._graph_step is not None  # the step really ran through the graph
+
+
+def test_val_predictor_matches_eager_val(tmp_path):
+    """Trainer.val_test served by the graphed Predictor must reproduce the
+    eager eval metrics (same weights, same loader)."""
+    import argparse
+
+    from pvraft_amd.engine import Trainer
+    import pvraft_amd.engine.trainer as trainer_mod
+
+    args = argparse.Namespace(
+        root=str(tmp_path), exp_path="vp", dataset="SYNTH", max_points=512,
+        corr_levels=3, base_scales=0.25, truncate_k=64, iters=2, gamma=0.8,
+        batch_size=2, gpus="", num_epochs=1, weights=None, checkpoint_interval=5,
+        refine=False, num_workers=0, amp=False, synth_len=6, hipgraph=True,
+    )
+    old = trainer_mod.VAL_ITERS
+    trainer_mod.VAL_ITERS = 4
+    try:
+        t = Trainer(args)
+        graphed = t.val_test(None, mode="val")
+        t.args.hipgraph = False  # forces the eager branch in val_test
+        eager = t.val_test(None, mode="val")
+    finally:
+        trainer_mod.VAL_ITERS = old
+    for key in ("loss", "epe", "outlier"):
+        assert abs(graphed[key] - eager[key]) < 2e-3, (key, graphed[key], eager[key])
