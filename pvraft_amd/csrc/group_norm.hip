@@ -520,11 +520,20 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
 
 // --------------------------------------------------------------- launchers
 
-static int pick_chunks(long spatial, long bc) {
+static int pick_chunks(long spatial, long bc, int cg = 0) {
   // fill the chip (~4096 blocks) but keep >= ~16 vector iterations per
   // block: excess blocks multiply the per-block workspace atomics, which
-  // serialize on the tiny (rows,2)/(C,2) arrays
+  // serialize on the tiny (rows,2)/(C,2) arrays.  When the channels-per-
+  // group count is known, additionally bound the ATOMICS PER ROW ADDRESS
+  // (chunks * cg blocks hit each (row,2) slot): the kNN-branch shape
+  // (C=64, G=8, S=262144) ran 32 chunks x 8 channels = 512 serialized
+  // adds per address and measured 85 us for ~8 us of reads.
   long want = 4096 / (bc > 0 ? bc : 1);
+  if (cg > 0) {
+    long row_cap = 64 / cg;
+    if (row_cap < 4) row_cap = 4;
+    if (want > row_cap) want = row_cap;
+  }
   long cap = spatial / ((long)GN_THREADS * 16);
   long chunks = want < cap ? want : cap;
   if (chunks < 1) chunks = 1;
@@ -537,8 +546,10 @@ void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
                  long S, int C, int G, float eps, int act, float slope,
                  const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
+  // reduce grid bounds atomics per row; apply has no atomics -> fill freely
+  const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
   const dim3 grid(pick_chunks(S, (long)B * C), C, B);
-  hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, grid, dim3(GN_THREADS), 0,
+  hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, rgrid, dim3(GN_THREADS), 0,
                      stream, x, ws, S, C, G);
   hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
                      0, stream, ws, mean, rstd, row_len, rows, eps);
@@ -559,10 +570,11 @@ void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
                  long S, int C, int G, int act, float slope,
                  const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
+  const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
   const dim3 grid(pick_chunks(S, (long)B * C), C, B);
 #define GN_BWD(A)                                                              \
   do {                                                                         \
-    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, A>), grid, dim3(GN_THREADS),   \
+    hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, A>), rgrid, dim3(GN_THREADS),  \
                        0, stream, dy, x, mean, rstd, gamma, beta, row_ws,      \
                        chan_ws, slope_ws, S, C, G, slope, slope_ptr);          \
     hipLaunchKernelGGL((gn_bwd_apply_kernel<T, A>), grid, dim3(GN_THREADS),    \
@@ -583,7 +595,7 @@ void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
                    hipStream_t stream) {
   const int B = rows / G;
   const long S = (long)K * N;
-  const dim3 rgrid(pick_chunks(S, (long)B * C), C, B);
+  const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
   hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, rgrid, dim3(GN_THREADS), 0,
                      stream, x, ws, S, C, G);
   hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
@@ -607,10 +619,11 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                    int K, int C, int G, int act, float slope,
                    const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
+  const dim3 rgrid(pick_chunks(N, (long)B * C, C / G), C, B);
   const dim3 grid(pick_chunks(N, (long)B * C), C, B);
 #define GNMP_BWD(A)                                                            \
   do {                                                                         \
-    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, A>), grid,                   \
+    hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, A>), rgrid,                  \
                        dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,     \
                        gamma, beta, row_ws, chan_ws, slope_ws, N, K, C, G,     \
                        slope, slope_ptr);                                      \
