@@ -60,6 +60,20 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
   const float *cand_xyz = xyz + p * K * 3;
   const float *cand_corr = corr + p * K;
 
+  // stage the wave's whole candidate field in LDS ONCE (xyz stored
+  // relative to the query point): the voxel levels, the distance pass and
+  // both emit paths re-read it, and the HBM re-reads (~6 x 8 KB per
+  // point) dominated the kernel (~100 us of a 187 us call)
+  __shared__ float s_cxyz[4][MAXK][3];
+  __shared__ float s_ccor[4][MAXK];
+  for (int j = lane; j < K; j += WAVE) {
+    s_ccor[w][j] = cand_corr[j];
+    s_cxyz[w][j][0] = cand_xyz[j * 3 + 0] - cx;
+    s_cxyz[w][j][1] = cand_xyz[j * 3 + 1] - cy;
+    s_cxyz[w][j][2] = cand_xyz[j * 3 + 2] - cz;
+  }
+  __threadfence_block();
+
   // ---- voxel pyramid: per-wave LDS histogram per level
   for (int l = 0; l < L; ++l) {
     const float inv_r = 1.0f / (base_scale * (float)(1 << l));
@@ -69,13 +83,13 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     }
     __threadfence_block();
     for (int j = lane; j < K; j += WAVE) {
-      const float dx = rintf((cand_xyz[j * 3 + 0] - cx) * inv_r);
-      const float dy = rintf((cand_xyz[j * 3 + 1] - cy) * inv_r);
-      const float dz = rintf((cand_xyz[j * 3 + 2] - cz) * inv_r);
+      const float dx = rintf(s_cxyz[w][j][0] * inv_r);
+      const float dy = rintf(s_cxyz[w][j][1] * inv_r);
+      const float dz = rintf(s_cxyz[w][j][2] * inv_r);
       if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) && fabsf(dz) <= (R / 2)) {
         const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
                          ((int)dz + R / 2);
-        atomicAdd(&s_sum[w][cell], cand_corr[j]);
+        atomicAdd(&s_sum[w][cell], s_ccor[w][j]);
         atomicAdd(&s_cnt[w][cell], 1.f);
       }
     }
@@ -107,9 +121,9 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
   for (int t = 0; t < MAXC; ++t) {
     const int j = lane + t * WAVE;
     if (j < K) {
-      const float dx = cand_xyz[j * 3 + 0] - cx;
-      const float dy = cand_xyz[j * 3 + 1] - cy;
-      const float dz = cand_xyz[j * 3 + 2] - cz;
+      const float dx = s_cxyz[w][j][0];
+      const float dy = s_cxyz[w][j][1];
+      const float dz = s_cxyz[w][j][2];
       d[t] = dx * dx + dy * dy + dz * dz;
     } else {
       d[t] = INFINITY;
@@ -174,10 +188,10 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
       const unsigned kp = __float_as_uint(d[t]) >> shc;
       if (kp < tp) {
         const int slot = (int)atomicAdd(&s_st[w][2], 1u);
-        dst[(long)slot * N] = cand_corr[j];
-        dst[(long)slot * N + ch_stride] = cand_xyz[j * 3 + 0] - cx;
-        dst[(long)slot * N + 2 * ch_stride] = cand_xyz[j * 3 + 1] - cy;
-        dst[(long)slot * N + 3 * ch_stride] = cand_xyz[j * 3 + 2] - cz;
+        dst[(long)slot * N] = s_ccor[w][j];
+        dst[(long)slot * N + ch_stride] = s_cxyz[w][j][0];
+        dst[(long)slot * N + 2 * ch_stride] = s_cxyz[w][j][1];
+        dst[(long)slot * N + 3 * ch_stride] = s_cxyz[w][j][2];
         idst[slot] = j;
       } else if (kp == tp) {
         const unsigned bp = atomicAdd(&s_st[w][3], 1u);
@@ -201,10 +215,10 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     int j = bj;
     wave_argmin(v, j);
     if (j != 0x7fffffff && j == bj) {
-      dst[(long)(base + r) * N] = cand_corr[j];
-      dst[(long)(base + r) * N + ch_stride] = cand_xyz[j * 3 + 0] - cx;
-      dst[(long)(base + r) * N + 2 * ch_stride] = cand_xyz[j * 3 + 1] - cy;
-      dst[(long)(base + r) * N + 3 * ch_stride] = cand_xyz[j * 3 + 2] - cz;
+      dst[(long)(base + r) * N] = s_ccor[w][j];
+      dst[(long)(base + r) * N + ch_stride] = s_cxyz[w][j][0];
+      dst[(long)(base + r) * N + 2 * ch_stride] = s_cxyz[w][j][1];
+      dst[(long)(base + r) * N + 3 * ch_stride] = s_cxyz[w][j][2];
       idst[base + r] = j;
       bv = INFINITY;
       bj = 0x7fffffff;
@@ -236,6 +250,15 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
   const float cz = coords[p * 3 + 2];
   const float *cand_xyz = xyz + p * K * 3;
 
+  // stage relative candidate xyz in LDS (read by the count histograms and
+  // again by the gradient pass)
+  __shared__ float s_cxyz[4][MAXK][3];
+  for (int j = lane; j < K; j += WAVE) {
+    s_cxyz[w][j][0] = cand_xyz[j * 3 + 0] - cx;
+    s_cxyz[w][j][1] = cand_xyz[j * 3 + 1] - cy;
+    s_cxyz[w][j][2] = cand_xyz[j * 3 + 2] - cz;
+  }
+
   // scatter the kNN channel-0 grads onto their candidate slots (unique)
   for (int j = lane; j < K; j += WAVE) s_kg[w][j] = 0.f;
   __threadfence_block();
@@ -255,9 +278,9 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
     }
     __threadfence_block();
     for (int j = lane; j < K; j += WAVE) {
-      const float dx = rintf((cand_xyz[j * 3 + 0] - cx) * inv_r);
-      const float dy = rintf((cand_xyz[j * 3 + 1] - cy) * inv_r);
-      const float dz = rintf((cand_xyz[j * 3 + 2] - cz) * inv_r);
+      const float dx = rintf(s_cxyz[w][j][0] * inv_r);
+      const float dy = rintf(s_cxyz[w][j][1] * inv_r);
+      const float dz = rintf(s_cxyz[w][j][2] * inv_r);
       if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) && fabsf(dz) <= (R / 2))
         atomicAdd(&s_cnt[w][l][(int)dx * (R * R) + (int)dy * R + (int)dz +
                                (R / 2) * (R * R + R + 1)],
@@ -267,9 +290,9 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
   __threadfence_block();
 
   for (int j = lane; j < K; j += WAVE) {
-    const float ox = cand_xyz[j * 3 + 0] - cx;
-    const float oy = cand_xyz[j * 3 + 1] - cy;
-    const float oz = cand_xyz[j * 3 + 2] - cz;
+    const float ox = s_cxyz[w][j][0];
+    const float oy = s_cxyz[w][j][1];
+    const float oz = s_cxyz[w][j][2];
     float g = s_kg[w][j];
 #pragma unroll
     for (int l = 0; l < MAXL; ++l) {
