@@ -115,17 +115,6 @@ torch::Tensor& persistent_ws(long len, const torch::TensorOptions& opt) {
   return it->second;
 }
 
-// persistent DIRTY scratch (contents undefined between calls; every user
-// must fully overwrite its region before reading) -- separate from
-// persistent_ws, whose buffers must stay zero between uses.
-torch::Tensor& scratch_ws(long len, const torch::TensorOptions& opt) {
-  static auto* cache = new std::unordered_map<long, torch::Tensor>();
-  auto it = cache->find(len);
-  if (it == cache->end())
-    it = cache->emplace(len, torch::empty({len}, opt)).first;
-  return it->second;
-}
-
 }  // namespace
 
 torch::Tensor knn_graph(torch::Tensor xyz, int64_t k) {
