@@ -447,9 +447,14 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
     const float *__restrict__ rstd, const float *__restrict__ gamma,
     const float *__restrict__ beta, const float *__restrict__ row_ws,
     T *__restrict__ dx, long N, int K, int C, int G, long row_len,
-    float slope, const float *__restrict__ slope_ptr) {
+    int ksplit, float slope, const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
+  // blockIdx.x jointly spans spatial chunks and a strided K split (the
+  // K-loop per thread made the kernel latency-bound at ~128 blocks)
+  const int kc = blockIdx.x % ksplit;
+  const int nchunk = blockIdx.x / ksplit;
+  const int nchunks = gridDim.x / ksplit;
   const int c = blockIdx.y;
   const int b = blockIdx.z;
   const int Cg = C / G;
@@ -465,8 +470,8 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
   const long pooled = ((long)b * C + c) * N;
   if (N % W == 0) {
     const long NV = N / W;
-    for (long u = (long)blockIdx.x * GN_THREADS + threadIdx.x; u < NV;
-         u += (long)gridDim.x * GN_THREADS) {
+    for (long u = (long)nchunk * GN_THREADS + threadIdx.x; u < NV;
+         u += (long)nchunks * GN_THREADS) {
       unsigned char ks[W];
       float g[W];
 #pragma unroll
@@ -474,7 +479,7 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
         ks[e] = am[pooled + u * W + e];
         g[e] = ld(dy + pooled + u * W + e);
       }
-      for (int k = 0; k < K; ++k) {
+      for (int k = kc; k < K; k += ksplit) {
         const long i = base + (long)k * N + u * W;
         const V xv = *(const V *)(x + i);
         V ov;
@@ -496,11 +501,11 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
       }
     }
   } else {
-    for (long n = (long)blockIdx.x * GN_THREADS + threadIdx.x; n < N;
-         n += (long)gridDim.x * GN_THREADS) {
+    for (long n = (long)nchunk * GN_THREADS + threadIdx.x; n < N;
+         n += (long)nchunks * GN_THREADS) {
       const int ksel = am[pooled + n];
       const float g0 = ld(dy + pooled + n);
-      for (int k = 0; k < K; ++k) {
+      for (int k = kc; k < K; k += ksplit) {
         const long i = base + (long)k * N + n;
         const float xhat = (ld(x + i) - m) * r;
         float dxhat = 0.f;
@@ -620,7 +625,11 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                    const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
   const dim3 rgrid(pick_chunks(N, (long)B * C, C / G), C, B);
-  const dim3 grid(pick_chunks(N, (long)B * C), C, B);
+  const int nchunks = pick_chunks(N, (long)B * C);
+  int ksplit = (int)(1024 / ((long)nchunks * B * C));
+  if (ksplit > K) ksplit = K;
+  if (ksplit < 1) ksplit = 1;
+  const dim3 grid(nchunks * ksplit, C, B);
 #define GNMP_BWD(A)                                                            \
   do {                                                                         \
     hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, A>), rgrid,                  \
@@ -629,8 +638,8 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                        slope, slope_ptr);                                      \
     hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, A>), grid,                    \
                        dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,     \
-                       gamma, beta, row_ws, dx, N, K, C, G, row_len, slope,    \
-                       slope_ptr);                                             \
+                       gamma, beta, row_ws, dx, N, K, C, G, row_len, ksplit,   \
+                       slope, slope_ptr);                                      \
   } while (0)
   if (act == 2) GNMP_BWD(2);
   else if (act == 1) GNMP_BWD(1);
