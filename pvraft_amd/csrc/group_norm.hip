@@ -605,7 +605,14 @@ void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
                      stream, x, ws, S, C, G);
   hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
                      0, stream, ws, mean, rstd, row_len, rows, eps);
-  const dim3 pgrid(pick_chunks(N, (long)B * C), C, B);
+  // the apply loops all K per thread (argmax), so a thread's work is K x
+  // its vector count: allow chunking down to ~1 vector per thread instead
+  // of pick_chunks' 16-iteration floor (C*B is small -> it underfilled)
+  long pv = N / VecT<T>::W;
+  long pch = 4096 / ((long)B * C);
+  if (pch > pv / GN_THREADS) pch = pv / GN_THREADS;
+  if (pch < 1) pch = 1;
+  const dim3 pgrid((unsigned)pch, C, B);
 #define GNMP_FWD(A)                                                            \
   hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, A>), pgrid, dim3(GN_THREADS),   \
                      0, stream, x, y, am, mean, rstd, gamma, beta, N, K, C, G, \
