@@ -31,7 +31,10 @@ __global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
     int *__restrict__ out_i,         // (R, K)
     int M, int K) {
   __shared__ float s_row[TK_MAXM];
-  __shared__ unsigned s_hist[256];
+  // per-wave sub-histograms: correlation values cluster into a handful of
+  // exponent bins, so a single shared histogram serializes the LDS
+  // atomics ~wave-wide; each wave bins privately and the counts merge once
+  __shared__ unsigned s_hist[TK_THREADS / WAVE][256];
   __shared__ float s_bv[TK_CAP];
   __shared__ int s_bi[TK_CAP];
   __shared__ unsigned s_acc, s_bcnt, s_state[3];
@@ -52,16 +55,26 @@ __global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
   unsigned prefix = 0;  // broadcast via s_state after each round
   int need = K;
   int nbytes = 0;
+  const int wv = wave_id();
   for (int round = 0; round < 3; ++round) {
     __syncthreads();
-    for (int i = threadIdx.x; i < 256; i += TK_THREADS) s_hist[i] = 0;
+    for (int i = threadIdx.x; i < 256 * (TK_THREADS / WAVE); i += TK_THREADS)
+      ((unsigned *)s_hist)[i] = 0;
     __syncthreads();
     const int shift = 24 - 8 * round;
     const unsigned mask = round == 0 ? 0u : (0xFFFFFFFFu << (shift + 8));
     for (int i = threadIdx.x; i < M; i += TK_THREADS) {
       const unsigned key = fkey(s_row[i]);
       if ((key & mask) == prefix)
-        atomicAdd(&s_hist[(key >> shift) & 0xFFu], 1u);
+        atomicAdd(&s_hist[wv][(key >> shift) & 0xFFu], 1u);
+    }
+    __syncthreads();
+    // merge the per-wave counts into s_hist[0]
+    for (int i = threadIdx.x; i < 256; i += TK_THREADS) {
+      unsigned t = s_hist[0][i];
+#pragma unroll
+      for (int ww = 1; ww < TK_THREADS / WAVE; ++ww) t += s_hist[ww][i];
+      s_hist[0][i] = t;
     }
     __syncthreads();
     if (threadIdx.x == 0) {
@@ -69,7 +82,7 @@ __global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
       unsigned cum = 0;
       int T = 0;
       for (int bin = 255; bin >= 0; --bin) {
-        const unsigned nxt = cum + s_hist[bin];
+        const unsigned nxt = cum + s_hist[0][bin];
         if (nxt >= (unsigned)need) {
           T = bin;
           break;
@@ -80,7 +93,7 @@ __global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
       s_state[1] = (unsigned)(round + 1);
       s_state[2] = (unsigned)(need - (int)cum);
       // stop refining once the threshold bin fits the boundary buffer
-      if (s_hist[T] <= TK_CAP - 8 || round == 2) s_state[1] |= 0x100u;
+      if (s_hist[0][T] <= TK_CAP - 8 || round == 2) s_state[1] |= 0x100u;
     }
     __syncthreads();
     prefix = s_state[0];
