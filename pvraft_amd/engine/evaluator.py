@@ -4,11 +4,10 @@ Builds the FT3D test split or KITTI, runs the (optionally refined) model
 with 32 GRU iterations (test.py:120 hard-codes 32 regardless of --iters)
 and reports mean EPE3D / Acc3DS / Acc3DR / Outlier.
 
-On GPU the forward is served by the hipGraph Predictor (fixed shapes).
-All metrics are computed on the final flow exactly as the reference; the
-informational "loss" column is then the final-flow masked L1 rather than
-the gamma-weighted sequence loss (the intermediate flows stay on-graph).
-Disable with --no_hipgraph for the eager path.
+On GPU the forward is served by the hipGraph Predictor (fixed shapes); it
+exposes every per-iteration flow, so the logged loss is the same
+gamma-weighted sequence loss the reference's test.py computes
+(test.py:122).  Disable with --no_hipgraph for the eager path.
 """
 
 from __future__ import annotations
@@ -81,7 +80,13 @@ def evaluate(args):
         batch = batch.to(device, non_blocking=True)
         if predictor is not None and batch["sequence"][0].shape[1] == args.max_points:
             final = predictor(batch["sequence"][0], batch["sequence"][1])
-            loss = compute_loss(final, batch)
+            flows = predictor.last_flows
+            # reference test.py:122 logs the gamma-weighted sequence loss
+            # for the stage-1 model (single refined flow -> plain loss)
+            if len(flows) > 1:
+                loss = sequence_loss(flows, batch, gamma=args.gamma if hasattr(args, "gamma") else 0.8)
+            else:
+                loss = compute_loss(final, batch)
         else:
             est_flow = model(batch["sequence"], num_iters=TEST_ITERS)
             if isinstance(est_flow, (list, tuple)):
