@@ -159,3 +159,21 @@ def test_concat_free_gru_matches_cat_formulation():
     got = _split_mm(conv.weight, conv.bias, [inp, motion, flow_t])
     want = conv(torch.cat([inp, motion, flow_t], dim=1))
     assert torch.allclose(got, want, atol=1e-5)
+
+
+def test_predictor_cpu_fallback_exposes_flows():
+    """Predictor without a GPU serves the eager path and exposes the full
+    per-iteration flow list for sequence-loss evaluation."""
+    import torch
+
+    from pvraft_amd.engine import Predictor
+    from pvraft_amd.model import PVRaft
+
+    model = PVRaft(truncate_k=16)
+    pred = Predictor(model, points=64, batch=1, iters=3, use_graph=False)
+    xyz1 = torch.randn(1, 64, 3)
+    xyz2 = xyz1 + 0.01 * torch.randn(1, 64, 3)
+    final = pred(xyz1, xyz2)
+    assert final.shape == (1, 64, 3)
+    assert len(pred.last_flows) == 3
+    assert torch.equal(pred.last_flows[-1], final)
