@@ -23,7 +23,13 @@ class Predictor:
         self.iters = iters
         self.amp = amp
         self.device = next(model.parameters()).device
-        self.use_graph = use_graph and self.device.type == "cuda"
+        # KNOWN ISSUE (under investigation, profiles/README.md): graph
+        # capture+replay of the inference forward intermittently raises an
+        # illegal memory access for batch >= 5 (eager is clean at every
+        # batch size, all per-op GPU tests pass, and batch <= 4 graphs are
+        # exercised throughout training/eval).  Serve large batches eagerly
+        # until root-caused.
+        self.use_graph = use_graph and self.device.type == "cuda" and batch <= 4
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self._in1 = torch.zeros(batch, points, 3, device=self.device)
         self._in2 = torch.zeros(batch, points, 3, device=self.device)
