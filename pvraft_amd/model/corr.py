@@ -38,7 +38,8 @@ from .pointwise import PwConv1d, PwConv2d
 
 @dataclass
 class CorrField:
-    corr: Tensor  # (B, N, K) truncated correlation values (top-K, sorted)
+    corr: Tensor  # (B, N, K) truncated correlation values (top-K SET,
+    #               unordered: every consumer is order-invariant)
     xyz: Tensor   # (B, N, K, 3) positions of the selected candidates in pc2
 
 
