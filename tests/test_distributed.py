@@ -104,3 +104,52 @@ def _full_model_step(info):
 
 def test_distributed_train_step_keeps_ranks_in_sync():
     spawn(_full_model_step, 29603)
+
+
+def _multibucket_reduce(info):
+    """Force several buckets (tiny cap) and mix hook-driven + reduce_all
+    paths; also leave one submodule unused so finalize must flush its
+    bucket late."""
+    torch.manual_seed(7)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(64, 64), torch.nn.Linear(64, 64), torch.nn.Linear(64, 8)
+    )
+    broadcast_module(model)
+    reducer = GradReducer(model, bucket_cap_mb=0.01)  # ~10 KB -> multiple buckets
+    assert len(reducer.buckets) >= 3
+
+    # hook path
+    reducer.zero_grad()
+    x = torch.randn(4, 64) + info.rank
+    model(x).sum().backward()
+    reducer.finalize()
+    g1 = torch.cat([p.grad.flatten() for p in model.parameters()])
+    gs = [torch.empty_like(g1) for _ in range(WORLD)]
+    dist.all_gather(gs, g1)
+    assert torch.allclose(gs[0], gs[1], atol=1e-6)
+
+    # graph-style path: hooks off, grads accumulate, reduce_all afterwards
+    reducer.hooks_enabled = False
+    reducer.zero_grad()
+    model(x).sum().backward()
+    reducer.reduce_all()
+    g2 = torch.cat([p.grad.flatten() for p in model.parameters()])
+    gs2 = [torch.empty_like(g2) for _ in range(WORLD)]
+    dist.all_gather(gs2, g2)
+    assert torch.allclose(gs2[0], gs2[1], atol=1e-6)
+    assert torch.allclose(g1, g2, atol=1e-6)  # both paths: mean over ranks
+
+    # partial-use model: last layer never runs -> its bucket flushes in finalize
+    reducer.hooks_enabled = True
+    reducer.zero_grad()
+    h = model[0](x)
+    h.sum().backward()
+    reducer.finalize()  # must not deadlock; unused grads stay zero, agree across ranks
+    g3 = torch.cat([p.grad.flatten() for p in model.parameters()])
+    gs3 = [torch.empty_like(g3) for _ in range(WORLD)]
+    dist.all_gather(gs3, g3)
+    assert torch.allclose(gs3[0], gs3[1], atol=1e-6)
+
+
+def test_gradreducer_multibucket_and_partial_use():
+    spawn(_multibucket_reduce, 29603)
