@@ -152,4 +152,49 @@ def _multibucket_reduce(info):
 
 
 def test_gradreducer_multibucket_and_partial_use():
-    spawn(_multibucket_reduce, 29603)
+    spawn(_multibucket_reduce, 29605)
+
+
+def _full_trainer_epoch(info):
+    """The whole Trainer under a 2-rank group: DistributedSampler sharding,
+    per-rank loaders, val metric all-reduce, rank-0-only checkpointing."""
+    import argparse
+    import tempfile
+
+    import pvraft_amd.engine.trainer as trainer_mod
+    from pvraft_amd.engine import Trainer
+
+    tmp = tempfile.mkdtemp() if info.rank == 0 else None
+    obj = [tmp]
+    dist.broadcast_object_list(obj, src=0)
+    tmp = obj[0]
+
+    args = argparse.Namespace(
+        root=tmp, exp_path="dist_exp", dataset="SYNTH", max_points=48,
+        corr_levels=3, base_scales=0.25, truncate_k=16, iters=2, gamma=0.8,
+        batch_size=2, gpus="", num_epochs=1, weights=None, checkpoint_interval=5,
+        refine=False, num_workers=0, amp=False, synth_len=4, hipgraph=False,
+    )
+    old = trainer_mod.VAL_ITERS
+    trainer_mod.VAL_ITERS = 2
+    try:
+        t = Trainer(args)
+        assert t.dist.world_size == WORLD
+        t.training(1)
+        results = t.val_test(1, mode="val")
+    finally:
+        trainer_mod.VAL_ITERS = old
+    assert results["epe"] >= 0
+    # params in sync after the epoch
+    flat = torch.cat([p.detach().flatten() for p in t.model.parameters()])
+    flats = [torch.empty_like(flat) for _ in range(WORLD)]
+    dist.all_gather(flats, flat)
+    assert torch.allclose(flats[0], flats[1], atol=1e-7)
+    # rank 0 wrote checkpoints; rank 1 did not race it
+    ckpt = os.path.join(tmp, "experiments", "dist_exp", "checkpoints", "last_checkpoint.params")
+    if info.rank == 0:
+        assert os.path.exists(ckpt)
+
+
+def test_distributed_trainer_epoch():
+    spawn(_full_trainer_epoch, 29607)
