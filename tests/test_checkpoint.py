@@ -3,6 +3,7 @@
 import argparse
 import os
 
+import pytest
 import torch
 
 from pvraft_amd.model import PVRaft
@@ -69,3 +70,40 @@ def test_train_state_roundtrip(tmp_path):
     s2 = opt2.state_dict()["state"]
     for k in s1:
         assert torch.equal(s1[k]["exp_avg"], s2[k]["exp_avg"])
+
+
+@pytest.mark.parametrize("refine", [False, True])
+def test_state_dict_parity_with_reference_model(refine):
+    """Key-for-key, shape-for-shape state-dict parity with the ACTUAL
+    reference model classes (checkpoints interchangeable both ways).
+    Runs only where the reference repo is mounted (read-only import)."""
+    import argparse
+    import importlib
+    import sys
+
+    if not os.path.isdir("/root/reference/model"):
+        pytest.skip("reference repo not available")
+    sys.path.insert(0, "/root/reference")
+    try:
+        try:
+            mod = importlib.import_module(
+                "model.RAFTSceneFlowRefine" if refine else "model.RAFTSceneFlow"
+            )
+        except ImportError as e:  # missing reference-era deps (torch_scatter)
+            pytest.skip(f"reference model unimportable: {e}")
+        cls = getattr(mod, "RSF_refine" if refine else "RSF")
+        args = argparse.Namespace(corr_levels=3, base_scales=0.25, truncate_k=64)
+        ref_sd = {k: tuple(v.shape) for k, v in cls(args).state_dict().items()}
+    finally:
+        sys.path.remove("/root/reference")
+
+    from pvraft_amd.model import PVRaft, PVRaftRefine
+
+    ours = (PVRaftRefine if refine else PVRaft)(
+        corr_levels=3, base_scales=0.25, truncate_k=64
+    )
+    our_sd = {k: tuple(v.shape) for k, v in ours.state_dict().items()}
+    assert set(ref_sd) == set(our_sd), (
+        sorted(set(ref_sd) - set(our_sd)), sorted(set(our_sd) - set(ref_sd)))
+    mismatch = {k: (ref_sd[k], our_sd[k]) for k in ref_sd if ref_sd[k] != our_sd[k]}
+    assert not mismatch, mismatch
