@@ -6,16 +6,18 @@ the HPLFlowNet KITTI_mapping.txt (142 usable scenes), removes ground points
 (y < -1.4 in both clouds) and points at z >= 35 m, and uses flow =
 pc2 - pc1 with an all-ones mask.
 
-The mapping file is third-party data shipped with HPLFlowNet, not with this
-framework: pass its path via ``mapping_file`` or place KITTI_mapping.txt in
-the dataset root.  Without it all scenes are used (a warning is emitted,
-metrics then cover 200 instead of 142 scenes).
+The mapping file (third-party data originating from HPLFlowNet's KITTI
+raw-drive index, same data the reference ships as
+datasets/KITTI_mapping.txt) is bundled at ``pvraft_amd/data/KITTI_mapping.txt``
+and used by default, so KITTI metrics always cover the 142-scene protocol.
+A mapping file placed in the dataset root or passed via ``mapping_file``
+overrides the bundled copy; ``mapping_file=""`` explicitly opts out and
+evaluates all 200 scenes (non-comparable numbers -- opt-in only).
 """
 
 from __future__ import annotations
 
 import os
-import warnings
 
 import numpy as np
 
@@ -24,6 +26,7 @@ from .base import SceneFlowDataset
 KITTI_SCENES = 200
 GROUND_Y = -1.4
 MAX_DEPTH = 35.0
+BUNDLED_MAPPING = os.path.join(os.path.dirname(__file__), "KITTI_mapping.txt")
 
 
 class Kitti(SceneFlowDataset):
@@ -50,12 +53,8 @@ class Kitti(SceneFlowDataset):
 
         if mapping_file is None:
             cand = os.path.join(root, "KITTI_mapping.txt")
-            mapping_file = cand if os.path.exists(cand) else None
-        if mapping_file is None:
-            warnings.warn(
-                "KITTI_mapping.txt not found: using all scenes. For parity with "
-                "HPLFlowNet evaluation, provide the mapping file (142 scenes)."
-            )
+            mapping_file = cand if os.path.exists(cand) else BUNDLED_MAPPING
+        if mapping_file == "":  # explicit opt-out: all 200 scenes
             return useful_paths
 
         with open(mapping_file) as fd:

@@ -89,7 +89,7 @@ def test_kitti_filters(tmp_path):
         pc2[5:8, 2] = 40.0
         np.save(os.path.join(d, "pc1.npy"), pc1)
         np.save(os.path.join(d, "pc2.npy"), pc2)
-    ds = Kitti(str(tmp_path), nb_points=16, strict_sizes=False)
+    ds = Kitti(str(tmp_path), nb_points=16, strict_sizes=False, mapping_file="")
     assert len(ds) == 5
     seq, gt = ds.load_sequence(0)
     assert seq[0].shape[0] <= 64 - 5 - 3
@@ -97,6 +97,29 @@ def test_kitti_filters(tmp_path):
     # no point is ground (y < -1.4) in both clouds
     assert not np.logical_and(seq[0][:, 1] < -1.4, seq[1][:, 1] < -1.4).any()
     assert gt[0].shape == (seq[0].shape[0], 1)
+
+
+def test_kitti_bundled_mapping_default(tmp_path):
+    """The HPLFlowNet 142-scene protocol is the default (reference
+    datasets/KITTI_mapping.txt consumed at kitti_hplflownet.py:44-50)."""
+    from pvraft_amd.data.kitti import BUNDLED_MAPPING
+
+    with open(BUNDLED_MAPPING) as fd:
+        lines = [line.strip() for line in fd.readlines()]
+    assert len(lines) == 200
+    assert sum(1 for line in lines if line) == 142
+
+    rng = np.random.default_rng(2)
+    for i in range(200):
+        d = os.path.join(str(tmp_path), f"{i:06d}")
+        os.makedirs(d)
+        pc = rng.normal(size=(32, 3)).astype(np.float32)
+        np.save(os.path.join(d, "pc1.npy"), pc)
+        np.save(os.path.join(d, "pc2.npy"), pc + 0.01)
+    ds = Kitti(str(tmp_path), nb_points=16)  # no mapping arg: bundled default
+    assert len(ds) == 142
+    kept = {int(os.path.split(p)[-1]) for p in ds.paths}
+    assert kept == {i for i, line in enumerate(lines) if line}
 
 
 def test_subsample_skips_small_samples():
