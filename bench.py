@@ -44,6 +44,8 @@ def parse_args():
     p.add_argument("--no-graph", dest="graph", action="store_false",
                    help="disable hipGraph capture of the train step")
     p.add_argument("--device", type=str, default=None, help="force device (cpu for plumbing tests)")
+    p.add_argument("--master_port", type=int, default=0,
+                   help="rendezvous port for the self-launch (default: free port)")
     return p.parse_args()
 
 
@@ -52,10 +54,13 @@ def main():
 
     # self-launch one rank per GPU if asked for N>1 outside torchrun
     if args.gpus > 1 and "RANK" not in os.environ:
+        from pvraft_amd.cli import free_port
+
+        port = args.master_port or free_port()
         cmd = [
             sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
             f"--nproc-per-node={args.gpus}", "--master-addr=127.0.0.1",
-            "--master-port=29533", os.path.abspath(__file__),
+            f"--master-port={port}", os.path.abspath(__file__),
         ] + sys.argv[1:]
         os.execvpe(cmd[0], cmd, dict(os.environ))
 

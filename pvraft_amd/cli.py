@@ -13,7 +13,19 @@ from __future__ import annotations
 
 import argparse
 import os
+import socket
 import sys
+
+
+def free_port() -> int:
+    """Pick a currently-free TCP port on 127.0.0.1 for the rendezvous.
+
+    Hardcoded ports (29531/29533 in round 1) collide when two multi-GPU
+    jobs share a host; binding port 0 lets the kernel choose.
+    """
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
 def add_common_args(parser: argparse.ArgumentParser, training: bool) -> None:
@@ -36,6 +48,12 @@ def add_common_args(parser: argparse.ArgumentParser, training: bool) -> None:
     parser.add_argument("--amp", help="bf16 autocast compute", action="store_true")
     parser.add_argument("--no_hipgraph", dest="hipgraph", help="disable hipGraph train-step capture", action="store_false")
     parser.add_argument("--synth_len", help="synthetic dataset length", default=256, type=int)
+    parser.add_argument(
+        "--master_port",
+        help="rendezvous port for multi-GPU launch (default: pick a free port)",
+        default=0,
+        type=int,
+    )
     if training:
         parser.add_argument("--gamma", help="exponential weights", default=0.8, type=float)
         parser.add_argument("--batch_size", help="global mini-batch size", default=1, type=int)
@@ -56,6 +74,7 @@ def maybe_relaunch_distributed(args, script: str) -> bool:
             os.environ.setdefault("HIP_VISIBLE_DEVICES", gpus[0])
         return False
     os.environ["HIP_VISIBLE_DEVICES"] = ",".join(gpus)
+    port = int(getattr(args, "master_port", 0)) or free_port()
     cmd = [
         sys.executable,
         "-m",
@@ -63,7 +82,7 @@ def maybe_relaunch_distributed(args, script: str) -> bool:
         "--nnodes=1",
         f"--nproc-per-node={len(gpus)}",
         "--master-addr=127.0.0.1",
-        "--master-port=29531",
+        f"--master-port={port}",
         script,
     ] + [a for a in sys.argv[1:]]
     os.execvpe(cmd[0], cmd, os.environ)

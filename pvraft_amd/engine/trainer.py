@@ -55,6 +55,28 @@ from pvraft_amd.utils import (
 VAL_ITERS = 32  # reference engine.py:198 / test.py:120
 
 
+class EvalShardSampler(torch.utils.data.Sampler):
+    """Pad-free eval sharding: rank r sees indices r, r+W, r+2W, ...
+
+    DistributedSampler(drop_last=False) pads non-divisible datasets by
+    duplicating samples, and the duplicates enter the all-reduced metric
+    sums -- multi-GPU val metrics then deviate from the reference's
+    every-sample-once bs=1 protocol (engine.py:176).  Here shards are
+    uneven (sizes differ by at most 1) and the count-weighted all-reduce
+    in val_test makes the distributed mean bit-equal to the single-rank
+    protocol.
+    """
+
+    def __init__(self, dataset, rank: int, world_size: int):
+        self.indices = list(range(rank, len(dataset), world_size))
+
+    def __iter__(self):
+        return iter(self.indices)
+
+    def __len__(self):
+        return len(self.indices)
+
+
 class Trainer:
     loss_is_sequence = True
 
@@ -124,11 +146,7 @@ class Trainer:
         self.test_loader = self._eval_loader(self.test_dataset, workers)
 
     def _eval_loader(self, dataset, workers):
-        sampler = (
-            dist_data.DistributedSampler(dataset, shuffle=False, drop_last=False)
-            if self.dist.distributed
-            else None
-        )
+        sampler = EvalShardSampler(dataset, self.dist.rank, self.dist.world_size) if self.dist.distributed else None
         return DataLoader(
             dataset,
             batch_size=1,

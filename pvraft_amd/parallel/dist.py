@@ -58,6 +58,10 @@ def init_distributed(backend: Optional[str] = None, timeout_s: int = 600) -> Dis
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29531")
+    # host driver on this pool only supports dmabuf IPC; legacy-mode IPC
+    # makes RCCL cross-process tensor sharing fail with
+    # hipIpcGetMemHandle: invalid argument
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s))

@@ -198,3 +198,77 @@ def _full_trainer_epoch(info):
 
 def test_distributed_trainer_epoch():
     spawn(_full_trainer_epoch, 29607)
+
+
+def test_eval_shard_sampler_no_duplication():
+    from pvraft_amd.engine.trainer import EvalShardSampler
+
+    for n in (1, 5, 7, 142):
+        for world in (1, 2, 3, 8):
+            shards = [list(EvalShardSampler(range(n), r, world)) for r in range(world)]
+            flat = sorted(i for s in shards for i in s)
+            assert flat == list(range(n))  # every sample exactly once
+            sizes = [len(s) for s in shards]
+            assert max(sizes) - min(sizes) <= 1
+
+
+def _padfree_val_metrics(info):
+    """Distributed val metrics on a non-divisible dataset must match the
+    single-process every-sample-once protocol (no duplicated samples in
+    the all-reduced sums)."""
+    import argparse
+    import tempfile
+
+    import pvraft_amd.engine.trainer as trainer_mod
+    from pvraft_amd.data import Batch
+    from pvraft_amd.engine import Trainer
+    from pvraft_amd.utils import compute_epe
+
+    tmp = tempfile.mkdtemp() if info.rank == 0 else None
+    obj = [tmp]
+    dist.broadcast_object_list(obj, src=0)
+    tmp = obj[0]
+
+    args = argparse.Namespace(
+        root=tmp, exp_path="padfree_exp", dataset="SYNTH", max_points=48,
+        corr_levels=3, base_scales=0.25, truncate_k=16, iters=2, gamma=0.8,
+        batch_size=2, gpus="", num_epochs=1, weights=None, checkpoint_interval=5,
+        refine=False, num_workers=0, amp=False, synth_len=24, hipgraph=False,
+    )
+    old = trainer_mod.VAL_ITERS
+    trainer_mod.VAL_ITERS = 2
+    try:
+        t = Trainer(args)
+        n_val = len(t.val_dataset)
+        assert n_val % WORLD != 0  # the padding bug only shows on non-divisible sets
+        results = t.val_test(None, mode="val")
+
+        if info.rank == 0:
+            # single-process protocol: every sample once, bs=1
+            t.model.eval()
+            with torch.no_grad():
+                vals = []
+                for i in range(n_val):
+                    batch = Batch([t.val_dataset[i]])
+                    est = t.model(batch["sequence"], num_iters=2)
+                    loss = t._loss(est, batch)
+                    epe3d, accs, accr, outl = compute_epe(est[-1].float(), batch)
+                    vals.append([loss.item(), epe3d, accs, accr, outl])
+            import numpy as np
+
+            ref = np.mean(vals, axis=0)
+            got = [results["loss"], results["epe"], results["acc3d_strict"],
+                   results["acc3d_relax"], results["outlier"]]
+            ref = [ref[0], ref[1], ref[2], ref[3], ref[4]]
+            # each __getitem__ re-permutes the points (random subsample),
+            # so two reads differ by reduction-order noise (~1e-7); a
+            # duplicated sample in a 3-sample mean would shift the result
+            # by O(10%).  1e-4 separates the two regimes cleanly.
+            for g, r in zip(got, ref):
+                assert abs(g - r) < 1e-4, (got, ref)
+    finally:
+        trainer_mod.VAL_ITERS = old
+
+
+def test_distributed_val_padfree_metrics():
+    spawn(_padfree_val_metrics, 29609)
