@@ -44,7 +44,12 @@ def _cast_cached(t: Tensor, dt) -> Tensor:
 
     if os.environ.get("PVRAFT_NO_CAST_CACHE", "0") == "1":
         return t.to(dt)
-    key = (id(t), dt)
+    # key includes the tensor version so an optimizer step (in-place
+    # update bumps _version) invalidates the entry even when the cache is
+    # not cleared between steps (direct sub-block calls outside
+    # PVRaft.forward); inside hipGraph capture _version is stable, so the
+    # recorded cast kernels re-read the fp32 weights on every replay.
+    key = (id(t), t._version, dt)
     hit = _STEP_CACHE.get(key)
     if hit is None:
         hit = (t, t.to(dt))

@@ -177,3 +177,19 @@ def test_predictor_cpu_fallback_exposes_flows():
     assert final.shape == (1, 64, 3)
     assert len(pred.last_flows) == 3
     assert torch.equal(pred.last_flows[-1], final)
+
+
+def test_cast_cache_invalidated_by_inplace_update():
+    """_STEP_CACHE must not serve stale bf16 copies after an
+    optimizer step mutates the fp32 weight in place (version-keyed)."""
+    from pvraft_amd.model.pointwise import _cast_cached, clear_step_cache
+
+    clear_step_cache()
+    w = torch.randn(4, 4)
+    c1 = _cast_cached(w, torch.bfloat16)
+    assert torch.equal(c1, w.to(torch.bfloat16))
+    w.add_(1.0)  # in-place update bumps w._version
+    c2 = _cast_cached(w, torch.bfloat16)
+    assert torch.equal(c2, w.to(torch.bfloat16))
+    assert not torch.equal(c1, c2)
+    clear_step_cache()
