@@ -59,6 +59,15 @@ void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
                      long, int, hipStream_t);
 void launch_gn_bwd_extract(float*, float*, float*, float*, int, int,
                            hipStream_t);
+void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
+                      const float*, const float*, void*, unsigned char*, int,
+                      long, int, int, int, float, int, float, const float*,
+                      bool, hipStream_t);
+void launch_egnmp_bwd(const void*, const void*, const int*,
+                      const unsigned char*, const int*, const int*,
+                      const float*, const float*, const float*, const float*,
+                      float*, float*, float*, void*, int, long, int, int, int,
+                      int, float, const float*, bool, hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
                      float*, float*, float*, void*, int, long, long, int, int,
@@ -376,6 +385,87 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
   return {dx, dweight, dbias, dslope};
 }
 
+// SetConv stage 1 on the linearly-restructured operands: wg (B, N, M)
+// point-major = (fc1 W @ [feats; xyz]) per point; pooled output y and u8
+// argmax are point-major too.  GN stats over the full (M/G, K, N) edge
+// extent, identical to group_norm_act_maxpool.
+std::vector<torch::Tensor> edge_gnmp_fwd(torch::Tensor wg, torch::Tensor idx,
+                                         int64_t G, torch::Tensor weight,
+                                         torch::Tensor bias, double eps,
+                                         int64_t act, double slope,
+                                         c10::optional<torch::Tensor> slope_t) {
+  const float* slope_ptr = nullptr;
+  if (act == 2) {
+    TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
+    slope_ptr = slope_t->data_ptr<float>();
+  }
+  TORCH_CHECK(wg.is_cuda() && wg.is_contiguous() && wg.dim() == 3, "wg must be contiguous (B,N,M)");
+  TORCH_CHECK(idx.scalar_type() == torch::kInt32 && idx.is_contiguous(), "idx must be contiguous int32");
+  check_f32(weight, "weight");
+  check_f32(bias, "bias");
+  const int B = wg.size(0), M = wg.size(2), K = idx.size(2);
+  const long N = wg.size(1);
+  TORCH_CHECK(idx.size(0) == B && idx.size(1) == N, "idx/wg shape mismatch");
+  TORCH_CHECK(M % G == 0 && G <= 8 && M <= 256, "edge_gnmp: need M % G == 0, G <= 8, M <= 256");
+  TORCH_CHECK(K <= 255, "edge_gnmp: K must fit u8 argmax");
+  const bool bf16 = wg.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || wg.scalar_type() == torch::kFloat32, "fp32/bf16 only");
+  const int rows = B * (int)G;
+  auto fopt = wg.options().dtype(torch::kFloat32);
+  auto& ws = persistent_ws((long)rows * 2, fopt);
+  auto mean = torch::empty({rows}, fopt);
+  auto rstd = torch::empty({rows}, fopt);
+  auto y = torch::empty_like(wg);
+  auto am = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
+  launch_egnmp_fwd(wg.data_ptr(), idx.data_ptr<int>(), ws.data_ptr<float>(),
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                   weight.data_ptr<float>(), bias.data_ptr<float>(),
+                   y.data_ptr(), am.data_ptr<unsigned char>(), B, N, K, M,
+                   (int)G, (float)eps, (int)act, (float)slope, slope_ptr,
+                   bf16, stream());
+  return {y, am, mean, rstd};
+}
+
+std::vector<torch::Tensor> edge_gnmp_bwd(
+    torch::Tensor dy, torch::Tensor wg, torch::Tensor idx, torch::Tensor am,
+    torch::Tensor order, torch::Tensor offsets, torch::Tensor mean,
+    torch::Tensor rstd, int64_t G, torch::Tensor weight, torch::Tensor bias,
+    int64_t act, double slope, c10::optional<torch::Tensor> slope_t) {
+  const float* slope_ptr = nullptr;
+  if (act == 2) {
+    TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
+    slope_ptr = slope_t->data_ptr<float>();
+  }
+  TORCH_CHECK(dy.is_contiguous() && wg.is_contiguous(), "dy/wg must be contiguous");
+  TORCH_CHECK(dy.scalar_type() == wg.scalar_type(), "dy/wg dtype mismatch");
+  TORCH_CHECK(order.scalar_type() == torch::kInt32 && order.is_contiguous());
+  TORCH_CHECK(offsets.scalar_type() == torch::kInt32 && offsets.is_contiguous());
+  const int B = wg.size(0), M = wg.size(2), K = idx.size(2);
+  const long N = wg.size(1);
+  const bool bf16 = wg.scalar_type() == torch::kBFloat16;
+  const int rows = B * (int)G;
+  auto fopt = wg.options().dtype(torch::kFloat32);
+  auto& ws = persistent_ws((long)rows * 2 + M * 2 + 1, fopt);
+  float* row_ws = ws.data_ptr<float>();
+  float* chan_ws = row_ws + rows * 2;
+  float* slope_ws = chan_ws + M * 2;
+  auto dwg = torch::empty_like(wg);
+  launch_egnmp_bwd(dy.data_ptr(), wg.data_ptr(), idx.data_ptr<int>(),
+                   am.data_ptr<unsigned char>(), order.data_ptr<int>(),
+                   offsets.data_ptr<int>(), mean.data_ptr<float>(),
+                   rstd.data_ptr<float>(), weight.data_ptr<float>(),
+                   bias.data_ptr<float>(), row_ws, chan_ws, slope_ws,
+                   dwg.data_ptr(), B, N, K, M, (int)G, (int)act, (float)slope,
+                   slope_ptr, bf16, stream());
+  auto dweight = torch::empty({M}, fopt);
+  auto dbias = torch::empty({M}, fopt);
+  auto dslope = torch::empty({1}, fopt);
+  launch_gn_bwd_extract(ws.data_ptr<float>(), dweight.data_ptr<float>(),
+                        dbias.data_ptr<float>(), dslope.data_ptr<float>(),
+                        rows, M, stream());
+  return {dwg, dweight, dbias, dslope};
+}
+
 // dy (B, Co, S) bf16, x (B, Ci, S) bf16 -> dW (Co, Ci) fp32 (split-K MFMA)
 std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
                                     int64_t schunks = 0, bool with_bias = false) {
@@ -606,6 +696,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_act_bwd", &group_norm_act_bwd);
   m.def("group_norm_act_maxpool_fwd", &group_norm_act_maxpool_fwd);
   m.def("group_norm_act_maxpool_bwd", &group_norm_act_maxpool_bwd);
+  m.def("edge_gnmp_fwd", &edge_gnmp_fwd);
+  m.def("edge_gnmp_bwd", &edge_gnmp_bwd);
   m.def("knn_graph", &knn_graph, "fused kNN graph (CDNA4)");
   m.def("gather_edge_concat_fwd", &gather_edge_concat_fwd);
   m.def("gather_edge_concat_bwd", &gather_edge_concat_bwd);

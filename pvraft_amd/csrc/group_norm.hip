@@ -747,6 +747,15 @@ __global__ void gn_bwd_extract_kernel(float *__restrict__ ws,
   }
 }
 
+// exported for the edge_gnmp (gathered SetConv stage 1) kernels, which
+// share the same workspace protocol
+void launch_gn_finalize(float *ws, float *mean, float *rstd, long row_len,
+                        int rows, float eps, hipStream_t stream) {
+  hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256),
+                     dim3(256), 0, stream, ws, mean, rstd, row_len, rows,
+                     eps);
+}
+
 void launch_gn_bwd_extract(float *ws, float *dweight, float *dbias,
                            float *dslope, int rows, int C,
                            hipStream_t stream) {

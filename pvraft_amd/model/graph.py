@@ -23,6 +23,14 @@ class Graph:
     idx: Tensor  # (B, N, k) int64 neighbour indices (self included)
     xyz: Tensor  # (B, N, 3) the cloud the graph was built on
     _csr: Optional[Tuple[Tensor, Tensor]] = field(default=None, repr=False)
+    _idx32: Optional[Tensor] = field(default=None, repr=False)
+
+    @property
+    def idx32(self) -> Tensor:
+        """int32 copy of idx, cached (shared by every SetConv on the graph)."""
+        if self._idx32 is None:
+            self._idx32 = self.idx.to(torch.int32).contiguous()
+        return self._idx32
 
     @property
     def k(self) -> int:

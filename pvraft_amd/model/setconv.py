@@ -14,6 +14,8 @@ gconv.py:64-66.
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 from torch import Tensor
@@ -39,16 +41,46 @@ class SetConv(nn.Module):
         """feats (B, N, C) -> (B, N, out_ch).
 
         Reference order gconv.py:71-83: conv -> GN -> lrelu (-> max over
-        neighbours after stage 1).  LeakyReLU is fused into each GroupNorm;
-        stage 1 additionally fuses the neighbour max-pool, so the
-        (B, C, K, N) activation never materialises post-conv.
+        neighbours after stage 1).  LeakyReLU is fused into each GroupNorm.
+
+        GPU path (stage 1): fc1 is linear in the edge vector and the edge
+        vector is a difference of per-point [feats; xyz] vectors, so
+            fc1(edge(n, j)) = Wg[idx[n, j]] - Wg[n],  Wg = fc1_W @ [feats; xyz]
+        -- the GEMM runs over N points (not K*N edges, 32x less) and the
+        gather-diff + GN + lrelu + max-pool fuse into the edge_gnmp kernels
+        over the (B, N, mid) Wg tensor; the reference's (B, C+3, K, N) and
+        (B, mid, K, N) tensors (gconv.py:64-68) never exist.  CPU /
+        reference mode keeps the explicit composition.
         """
-        x = ops.gather_edge_concat(feats, graph.idx, graph.xyz, csr=graph.csr())  # B, C+3, K, N
-        x = self.fc1(x)
-        x = ops.group_norm_act_maxpool(
-            x, self.gn1.num_groups, self.gn1.weight, self.gn1.bias,
-            self.gn1.eps, act="lrelu", slope=0.1,
-        )  # B, mid, N
+        if (
+            feats.is_cuda
+            and ops.hip_available()
+            and os.environ.get("PVRAFT_REF_OPS", "0") != "1"
+        ):
+            f_cn = feats.transpose(1, 2)  # usually a free view of (B, C, N)
+            dt = (
+                torch.get_autocast_dtype("cuda")
+                if feats.is_cuda and torch.is_autocast_enabled()
+                else feats.dtype
+            )
+            g = torch.cat([f_cn.to(dt), graph.xyz.transpose(1, 2).to(dt)], dim=1)
+            from .pointwise import pw_matmul
+
+            wg = pw_matmul(self.fc1.weight.view(self.fc1.out_channels, -1), g)
+            wg_t = ops.transpose_last2(wg)  # (B, N, mid)
+            y_t = ops.edge_gnmp(
+                wg_t, graph.idx32, graph.csr(), self.gn1.num_groups,
+                self.gn1.weight, self.gn1.bias, self.gn1.eps,
+                act="lrelu", slope=0.1,
+            )
+            x = ops.transpose_last2(y_t)  # (B, mid, N)
+        else:
+            x = ops.gather_edge_concat(feats, graph.idx, graph.xyz, csr=graph.csr())
+            x = self.fc1(x)
+            x = ops.group_norm_act_maxpool(
+                x, self.gn1.num_groups, self.gn1.weight, self.gn1.bias,
+                self.gn1.eps, act="lrelu", slope=0.1,
+            )  # B, mid, N
         x = self.gn2(self.fc2(x))
         x = self.gn3(self.fc3(x))
         return x.transpose(1, 2)

@@ -271,6 +271,61 @@ def group_norm_act_maxpool(
     return y.max(dim=2)[0]
 
 
+class _EdgeGNMP(torch.autograd.Function):
+    """SetConv stage 1 on linearly-restructured operands (see
+    csrc/edge_gnmp.hip): input wg (B, N, M) point-major = fc1-W @ [feats;
+    xyz] per point; output = pooled max_j act(GN(wg[nbr] - wg[center]))
+    (B, N, M).  The (B, M, K, N) edge tensors of the reference formulation
+    (gconv.py:64-75) never exist, forward or backward; the backward is the
+    deterministic CSR walk."""
+
+    @staticmethod
+    def forward(ctx, wg_t, idx, order, offsets, num_groups, weight, bias, eps, act, slope, slope_t):
+        w = weight.float().contiguous()
+        b = bias.float().contiguous()
+        st_ = slope_t.float().reshape(1).contiguous() if slope_t is not None else None
+        y, am, mean, rstd = _EXT.edge_gnmp_fwd(wg_t, idx, num_groups, w, b, eps, act, slope, st_)
+        ctx.save_for_backward(wg_t, idx, am, order, offsets, mean, rstd, w, b, st_)
+        ctx.conf = (num_groups, act, slope, weight.dtype)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        wg_t, idx, am, order, offsets, mean, rstd, w, b, st_ = ctx.saved_tensors
+        num_groups, act, slope, wdtype = ctx.conf
+        dwg, dw, db, dsl = _EXT.edge_gnmp_bwd(
+            dy.contiguous(), wg_t, idx, am, order, offsets, mean, rstd,
+            num_groups, w, b, act, slope, st_,
+        )
+        dslope = dsl.to(wdtype) if act == 2 else None
+        return (dwg, None, None, None, None, dw.to(wdtype), db.to(wdtype),
+                None, None, None, dslope)
+
+
+def edge_gnmp(
+    wg_t: Tensor,
+    idx: Tensor,
+    csr,
+    num_groups: int,
+    weight: Tensor,
+    bias: Tensor,
+    eps: float = 1e-5,
+    act: str = "lrelu",
+    slope: float = 0.1,
+    slope_t: Optional[Tensor] = None,
+) -> Tensor:
+    """(B, N, M) wg + (B, N, K) neighbours -> (B, N, M) pooled SetConv
+    stage 1 (gather-diff -> GN -> act -> max over K), GPU only."""
+    if not _use_hip(wg_t):
+        raise RuntimeError("edge_gnmp is a GPU-only fused op")
+    act_id = {"none": 0, "lrelu": 1, "prelu": 2}[act]
+    order, offsets = csr
+    return _EdgeGNMP.apply(
+        wg_t.contiguous(), idx, order, offsets, num_groups, weight, bias,
+        eps, act_id, slope, slope_t,
+    )
+
+
 # ---------------------------------------------------------------------------
 # public functional API (model code calls these)
 # ---------------------------------------------------------------------------
