@@ -60,14 +60,15 @@ void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
 void launch_gn_bwd_extract(float*, float*, float*, float*, int, int,
                            hipStream_t);
 void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
-                      const float*, const float*, void*, unsigned char*, int,
-                      long, int, int, int, float, int, float, const float*,
-                      bool, hipStream_t);
+                      float*, const float*, const float*, void*,
+                      unsigned char*, int, long, int, int, int, float, int,
+                      float, const float*, bool, int, hipStream_t);
 void launch_egnmp_bwd(const void*, const void*, const int*,
                       const unsigned char*, const int*, const int*,
                       const float*, const float*, const float*, const float*,
-                      float*, float*, float*, void*, int, long, int, int, int,
-                      int, float, const float*, bool, hipStream_t);
+                      float*, float*, void*, int, long, int, int, int, int,
+                      float, const float*, bool, int, hipStream_t);
+int egnmp_reduce_chunks(long, int, int);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
                      float*, float*, float*, void*, int, long, long, int, int,
@@ -406,23 +407,27 @@ std::vector<torch::Tensor> edge_gnmp_fwd(torch::Tensor wg, torch::Tensor idx,
   const int B = wg.size(0), M = wg.size(2), K = idx.size(2);
   const long N = wg.size(1);
   TORCH_CHECK(idx.size(0) == B && idx.size(1) == N, "idx/wg shape mismatch");
-  TORCH_CHECK(M % G == 0 && G <= 8 && M <= 256, "edge_gnmp: need M % G == 0, G <= 8, M <= 256");
+  TORCH_CHECK(M % G == 0 && G <= 8 && M <= 256 && M % 4 == 0,
+              "edge_gnmp: need M % G == 0, M % 4 == 0, G <= 8, M <= 256");
   TORCH_CHECK(K <= 255, "edge_gnmp: K must fit u8 argmax");
   const bool bf16 = wg.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || wg.scalar_type() == torch::kFloat32, "fp32/bf16 only");
   const int rows = B * (int)G;
   auto fopt = wg.options().dtype(torch::kFloat32);
   auto& ws = persistent_ws((long)rows * 2, fopt);
+  const int rchunks = egnmp_reduce_chunks(N, M, B);
+  auto scratch = torch::empty({(long)rows * 2, (long)rchunks * B}, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty_like(wg);
   auto am = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
-  launch_egnmp_fwd(wg.data_ptr(), idx.data_ptr<int>(), ws.data_ptr<float>(),
+  launch_egnmp_fwd(wg.data_ptr(), idx.data_ptr<int>(),
+                   scratch.data_ptr<float>(), ws.data_ptr<float>(),
                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
                    weight.data_ptr<float>(), bias.data_ptr<float>(),
                    y.data_ptr(), am.data_ptr<unsigned char>(), B, N, K, M,
                    (int)G, (float)eps, (int)act, (float)slope, slope_ptr,
-                   bf16, stream());
+                   bf16, rchunks, stream());
   return {y, am, mean, rstd};
 }
 
@@ -445,18 +450,19 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
   const bool bf16 = wg.scalar_type() == torch::kBFloat16;
   const int rows = B * (int)G;
   auto fopt = wg.options().dtype(torch::kFloat32);
-  auto& ws = persistent_ws((long)rows * 2 + M * 2 + 1, fopt);
-  float* row_ws = ws.data_ptr<float>();
-  float* chan_ws = row_ws + rows * 2;
-  float* slope_ws = chan_ws + M * 2;
+  const long ws_len = (long)rows * 2 + M * 2 + 1;
+  auto& ws = persistent_ws(ws_len, fopt);
+  const int rchunks = egnmp_reduce_chunks(N, M, B);
+  auto scratch = torch::empty({ws_len, (long)rchunks * B}, fopt);
   auto dwg = torch::empty_like(wg);
   launch_egnmp_bwd(dy.data_ptr(), wg.data_ptr(), idx.data_ptr<int>(),
                    am.data_ptr<unsigned char>(), order.data_ptr<int>(),
                    offsets.data_ptr<int>(), mean.data_ptr<float>(),
                    rstd.data_ptr<float>(), weight.data_ptr<float>(),
-                   bias.data_ptr<float>(), row_ws, chan_ws, slope_ws,
-                   dwg.data_ptr(), B, N, K, M, (int)G, (int)act, (float)slope,
-                   slope_ptr, bf16, stream());
+                   bias.data_ptr<float>(), scratch.data_ptr<float>(),
+                   ws.data_ptr<float>(), dwg.data_ptr(), B, N, K, M, (int)G,
+                   (int)act, (float)slope, slope_ptr, bf16, rchunks,
+                   stream());
   auto dweight = torch::empty({M}, fopt);
   auto dbias = torch::empty({M}, fopt);
   auto dslope = torch::empty({1}, fopt);

@@ -22,25 +22,26 @@
 // backward.
 //
 // Layout: Wg is stored point-major, wg[b, n, m] ("WgT"), so a neighbour
-// gather reads one contiguous M-vector and threads of the same point cover
-// adjacent channels (coalesced).  Outputs y/argmax are point-major too;
-// the caller transposes the 3 MB pooled result with the LDS-tiled
-// transpose kernel (trivial next to the 100 MB it replaces).
+// gather reads a contiguous M-vector; every thread owns a 4-channel quad
+// (8 B bf16 / 16 B fp32 vector loads) and threads of one point cover
+// adjacent quads (coalesced).  Outputs y/argmax are point-major too; the
+// caller transposes the 3 MB pooled result with the LDS-tiled transpose
+// kernel (trivial next to the 100 MB it replaces).
 //
-// Backward is DETERMINISTIC (atomic-free on the data path): the gradient
-// w.r.t. WgT at point p is
+// Reductions are DETERMINISTIC: each block writes its partial sums to a
+// per-block scratch slot (no global atomics) and a wave-per-output sum
+// kernel folds the partials into the workspace the shared finalize /
+// extract kernels consume.
+//
+// Backward data path is deterministic too: the gradient w.r.t. WgT at
+// point p is
 //   dWg[m, p] = sum_{edges e=(j,n): idx[n,j]=p} dx1[m, j, n]   (incoming)
 //             - sum_j dx1[m, j, p]                             (centre)
-// where dx1 is the standard GroupNorm+act+maxpool backward element,
-// recomputed on the fly from WgT / argmax / saved stats (exact same
-// formulas as gnmp_bwd_* in group_norm.hip).  The incoming sum walks the
-// same inverse-adjacency CSR (order/offsets) the round-1 CSR backward
-// used; every (p, m) output is written exactly once.
-//
-// GN statistics and their workspaces follow group_norm.hip exactly
-// (multi-block partial sums -> 2 atomics/block/row into a persistent
-// self-cleaning fp32 workspace -> shared finalize kernel), so numerics
-// match the unfused gnmp path bit-for-bit in fp32.
+// where dx1 is the standard GroupNorm+act+maxpool backward element
+// (exact same formulas as gnmp_bwd_* in group_norm.hip), recomputed on
+// the fly from WgT / argmax / saved stats.  The incoming sum walks the
+// same inverse-adjacency CSR (order/offsets, edge id = j*N + n) the
+// round-1 CSR backward used; every (p, m) output is written exactly once.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include "common.h"
@@ -61,53 +62,98 @@ DEV_INLINE void stg(T *p, float v) {
   *p = (T)v;
 }
 
+// 4-channel quad vector (8 B for bf16, 16 B for fp32)
+template <typename T>
+struct alignas(4 * sizeof(T)) Quad {
+  T v[4];
+};
+
+// e = j*N + n  ->  (j, n) without a hardware integer division
+DEV_INLINE void split_edge(int e, int N, float invN, int &j, int &n) {
+  j = (int)((float)e * invN);
+  n = e - j * N;
+  while (n < 0) {
+    --j;
+    n += N;
+  }
+  while (n >= N) {
+    ++j;
+    n -= N;
+  }
+}
+
 // ---------------------------------------------------------------- forward
 
-// pass 1: partial sum/sumsq per (b, group) over all (c, j, n) gather-diff
-// elements.  Thread (p_l, c) covers channel c of one point per iteration;
-// LDS bins bound the global atomics to 2 per (block, group).
+// pass 1: per-block partial sum/sumsq per group over all (c, j, n)
+// gather-diff elements -> scratch[(out)*(gridX*B) + gx*B + b], out in
+// [0, G*2).  Deterministic (no atomics past the LDS bins).
 template <typename T>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
     const T *__restrict__ wg,      // (B, N, M)
     const int *__restrict__ idx,   // (B, N, K)
-    float *__restrict__ ws,        // (B*G, 2) zeroed
+    float *__restrict__ scratch,   // (B*G*2, gridX*B)
     long N, int K, int M, int G) {
   const int b = blockIdx.z;
-  const int ppb = EG_THREADS / M;          // points per block iteration
-  const int p_l = (int)threadIdx.x / M;    // local point slot
-  const int c = (int)threadIdx.x % M;
+  const int B = gridDim.z;
+  const int tpc = M / 4;                 // threads per point (quad each)
+  const int ppb = EG_THREADS / tpc;      // points per block iteration
+  const int p_l = (int)threadIdx.x / tpc;
+  const int c4 = (int)threadIdx.x % tpc;
   const bool active = p_l < ppb;
   const int Cg = M / G;
-  const int g = c / Cg;
+  const int n_out = B * G * 2;
 
-  __shared__ float bins[8 * 2];  // G <= 8
-  if (threadIdx.x < (unsigned)(G * 2)) bins[threadIdx.x] = 0.f;
+  extern __shared__ float bins[];  // (B*G*2); only this block's b is used
+  for (unsigned i = threadIdx.x; i < (unsigned)n_out; i += EG_THREADS)
+    bins[i] = 0.f;
   __syncthreads();
 
   const T *wgb = wg + (long)b * N * M;
   const int *idxb = idx + (long)b * N * K;
-  float s = 0.f, ss = 0.f;
+  float s[4] = {0.f, 0.f, 0.f, 0.f}, ss[4] = {0.f, 0.f, 0.f, 0.f};
   if (active) {
     for (long n = (long)blockIdx.x * ppb + p_l; n < N;
          n += (long)gridDim.x * ppb) {
-      const float center = ldg(wgb + n * M + c);
+      const Quad<T> cq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
       const int *row = idxb + n * K;
       for (int j = 0; j < K; ++j) {
-        const float v = ldg(wgb + (long)row[j] * M + c) - center;
-        s += v;
-        ss += v * v;
+        const Quad<T> nq = *(const Quad<T> *)(wgb + (long)row[j] * M + c4 * 4);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const float v = (float)nq.v[e] - (float)cq.v[e];
+          s[e] += v;
+          ss[e] += v * v;
+        }
       }
     }
-    atomicAdd(&bins[g * 2 + 0], s);
-    atomicAdd(&bins[g * 2 + 1], ss);
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const int g = (c4 * 4 + e) / Cg;
+      atomicAdd(&bins[(b * G + g) * 2 + 0], s[e]);
+      atomicAdd(&bins[(b * G + g) * 2 + 1], ss[e]);
+    }
   }
   __syncthreads();
-  if (threadIdx.x < (unsigned)(G * 2))
-    atomicAdd(&ws[(long)(b * G) * 2 + threadIdx.x],
-              bins[threadIdx.x]);
+  const long col = (long)blockIdx.x * B + b;
+  const long stride = (long)gridDim.x * B;
+  for (unsigned i = threadIdx.x; i < (unsigned)n_out; i += EG_THREADS)
+    scratch[i * stride + col] = bins[i];
 }
 
-// pass 2 (finalize) is shared with group_norm.hip: launch_gn_finalize.
+// fold per-block partials into ws: one wave per output slot.  Slots that
+// belong to another batch's rows were written as zero, so the sum is
+// correct and deterministic (fixed grid -> fixed summation order).
+__global__ void egnmp_sum_partials_kernel(const float *__restrict__ scratch,
+                                          float *__restrict__ ws, long cols,
+                                          int n_out) {
+  const int out = blockIdx.x * (blockDim.x / WAVE) + wave_id();
+  if (out >= n_out) return;
+  const float *src = scratch + (long)out * cols;
+  float acc = 0.f;
+  for (long i = lane_id(); i < cols; i += WAVE) acc += src[i];
+  acc = wave_sum(acc);
+  if (lane_id() == 0) ws[out] = acc;
+}
 
 // pass 3: normalized+act gather-diff values, max over j -> pooled yT
 // (B, N, M) + u8 argmax
@@ -116,110 +162,154 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_apply_kernel(
     const T *__restrict__ wg, const int *__restrict__ idx,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
-    T *__restrict__ y,            // (B, N, M)
+    T *__restrict__ y,               // (B, N, M)
     unsigned char *__restrict__ am,  // (B, N, M)
     long N, int K, int M, int G, float slope,
     const float *__restrict__ slope_ptr) {
   const int b = blockIdx.z;
-  const int ppb = EG_THREADS / M;
-  const int p_l = (int)threadIdx.x / M;
-  const int c = (int)threadIdx.x % M;
+  const int tpc = M / 4;
+  const int ppb = EG_THREADS / tpc;
+  const int p_l = (int)threadIdx.x / tpc;
+  const int c4 = (int)threadIdx.x % tpc;
   if (p_l >= ppb) return;
   const int Cg = M / G;
-  const int row = b * G + c / Cg;
-  const float m = mean[row];
-  const float r = rstd[row];
-  const float ga = gamma[c], be = beta[c];
+  float m[4], r[4], ga[4], be[4];
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const int c = c4 * 4 + e;
+    const int row = b * G + c / Cg;
+    m[e] = mean[row];
+    r[e] = rstd[row];
+    ga[e] = gamma[c];
+    be[e] = beta[c];
+  }
   if (ACT == 2) slope = *slope_ptr;
 
   const T *wgb = wg + (long)b * N * M;
   const int *idxb = idx + (long)b * N * K;
   for (long n = (long)blockIdx.x * ppb + p_l; n < N;
        n += (long)gridDim.x * ppb) {
-    const float center = ldg(wgb + n * M + c);
+    const Quad<T> cq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
     const int *irow = idxb + n * K;
-    float best = -INFINITY;
-    int bk = 0;
+    float best[4];
+    int bk[4];
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      best[e] = -INFINITY;
+      bk[e] = 0;
+    }
     for (int j = 0; j < K; ++j) {
-      float v = (ldg(wgb + (long)irow[j] * M + c) - center - m) * r * ga + be;
-      if (ACT >= 1) v = v > 0.f ? v : v * slope;
-      if (v > best) {
-        best = v;
-        bk = j;
+      const Quad<T> nq = *(const Quad<T> *)(wgb + (long)irow[j] * M + c4 * 4);
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        float v = ((float)nq.v[e] - (float)cq.v[e] - m[e]) * r[e] * ga[e] + be[e];
+        if (ACT >= 1) v = v > 0.f ? v : v * slope;
+        if (v > best[e]) {
+          best[e] = v;
+          bk[e] = j;
+        }
       }
     }
-    stg(y + ((long)b * N + n) * M + c, best);
-    am[((long)b * N + n) * M + c] = (unsigned char)bk;
+    Quad<T> oq;
+    uchar4 aq;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) oq.v[e] = (T)best[e];
+    aq.x = (unsigned char)bk[0];
+    aq.y = (unsigned char)bk[1];
+    aq.z = (unsigned char)bk[2];
+    aq.w = (unsigned char)bk[3];
+    *(Quad<T> *)(y + ((long)b * N + n) * M + c4 * 4) = oq;
+    *(uchar4 *)(am + ((long)b * N + n) * M + c4 * 4) = aq;
   }
 }
 
 // ---------------------------------------------------------------- backward
 
-// pass 1: row sums {sum dxhat, sum dxhat*xhat} and channel sums
-// {sum dy_act, sum dy_act*xhat} over the POOLED domain (only the argmax
-// element of each (n, c) carries dy).  LDS bins bound global atomics.
+// pass 1: row sums {sum dxhat, sum dxhat*xhat}, channel sums
+// {sum dy_act, sum dy_act*xhat} and d slope over the POOLED domain (only
+// the argmax element of each (n, c) carries dy).  Partials land in
+// scratch exactly like the forward reduce; out layout matches the ws
+// layout [rows*2 | M*2 | 1] so one fold kernel serves both.
 template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_reduce_kernel(
-    const T *__restrict__ dy,     // (B, N, M) pooled grad (point-major)
+    const T *__restrict__ dy,  // (B, N, M) pooled grad (point-major)
     const T *__restrict__ wg, const int *__restrict__ idx,
     const unsigned char *__restrict__ am, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
-    const float *__restrict__ beta, float *__restrict__ row_ws,
-    float *__restrict__ chan_ws, float *__restrict__ slope_ws, long N, int K,
-    int M, int G, float slope, const float *__restrict__ slope_ptr) {
+    const float *__restrict__ beta, float *__restrict__ scratch, long N,
+    int K, int M, int G, float slope, const float *__restrict__ slope_ptr) {
   const int b = blockIdx.z;
-  const int ppb = EG_THREADS / M;
-  const int p_l = (int)threadIdx.x / M;
-  const int c = (int)threadIdx.x % M;
+  const int B = gridDim.z;
+  const int tpc = M / 4;
+  const int ppb = EG_THREADS / tpc;
+  const int p_l = (int)threadIdx.x / tpc;
+  const int c4 = (int)threadIdx.x % tpc;
   const bool active = p_l < ppb;
   const int Cg = M / G;
-  const int g = c / Cg;
-  const float m = mean[b * G + g];
-  const float r = rstd[b * G + g];
-  const float ga = gamma[c], be = beta[c];
   if (ACT == 2) slope = *slope_ptr;
 
-  extern __shared__ float sbins[];  // [G*2 rows | M*2 chans | 1 slope]
-  for (unsigned i = threadIdx.x; i < (unsigned)(G * 2 + M * 2 + 1);
-       i += EG_THREADS)
+  extern __shared__ float sbins[];  // [G*2*B rows | M*2 chans | 1 slope]
+  const int n_out = G * 2 * B + M * 2 + 1;
+  for (unsigned i = threadIdx.x; i < (unsigned)n_out; i += EG_THREADS)
     sbins[i] = 0.f;
   __syncthreads();
 
   const T *wgb = wg + (long)b * N * M;
   const int *idxb = idx + (long)b * N * K;
-  float sum_dx = 0.f, sum_dxx = 0.f, c_dg = 0.f, c_db = 0.f, d_sl = 0.f;
+  float sum_dx[4] = {}, sum_dxx[4] = {}, c_dg[4] = {}, c_db[4] = {};
+  float d_sl = 0.f;
+  float m[4], r[4], ga[4], be[4];
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const int c = c4 * 4 + e;
+    m[e] = mean[b * G + c / Cg];
+    r[e] = rstd[b * G + c / Cg];
+    ga[e] = gamma[c];
+    be[e] = beta[c];
+  }
   if (active) {
     for (long n = (long)blockIdx.x * ppb + p_l; n < N;
          n += (long)gridDim.x * ppb) {
-      const long pi = ((long)b * N + n) * M + c;
-      const int k = am[pi];
-      const int nb = idxb[n * K + k];
-      const float v = ldg(wgb + (long)nb * M + c) - ldg(wgb + n * M + c);
-      const float xhat = (v - m) * r;
-      float gv = ldg(dy + pi);
-      if (ACT >= 1) {
-        const float pre = xhat * ga + be;
-        if (ACT == 2 && pre <= 0.f) d_sl += gv * pre;
-        gv = pre > 0.f ? gv : gv * slope;
+      const long pi = ((long)b * N + n) * M + c4 * 4;
+      const uchar4 aq = *(const uchar4 *)(am + pi);
+      const Quad<T> gq = *(const Quad<T> *)(dy + pi);
+      const Quad<T> cq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
+      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const int nb = idxb[n * K + ks[e]];
+        const float v =
+            ldg(wgb + (long)nb * M + c4 * 4 + e) - (float)cq.v[e];
+        const float xhat = (v - m[e]) * r[e];
+        float gv = (float)gq.v[e];
+        if (ACT >= 1) {
+          const float pre = xhat * ga[e] + be[e];
+          if (ACT == 2 && pre <= 0.f) d_sl += gv * pre;
+          gv = pre > 0.f ? gv : gv * slope;
+        }
+        c_db[e] += gv;
+        c_dg[e] += gv * xhat;
+        const float dxhat = gv * ga[e];
+        sum_dx[e] += dxhat;
+        sum_dxx[e] += dxhat * xhat;
       }
-      c_db += gv;
-      c_dg += gv * xhat;
-      const float dxhat = gv * ga;
-      sum_dx += dxhat;
-      sum_dxx += dxhat * xhat;
     }
-    atomicAdd(&sbins[g * 2 + 0], sum_dx);
-    atomicAdd(&sbins[g * 2 + 1], sum_dxx);
-    atomicAdd(&sbins[G * 2 + c * 2 + 0], c_db);
-    atomicAdd(&sbins[G * 2 + c * 2 + 1], c_dg);
-    if (ACT == 2) atomicAdd(&sbins[G * 2 + M * 2], d_sl);
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const int c = c4 * 4 + e;
+      const int g = c / Cg;
+      atomicAdd(&sbins[(b * G + g) * 2 + 0], sum_dx[e]);
+      atomicAdd(&sbins[(b * G + g) * 2 + 1], sum_dxx[e]);
+      atomicAdd(&sbins[G * 2 * B + c * 2 + 0], c_db[e]);
+      atomicAdd(&sbins[G * 2 * B + c * 2 + 1], c_dg[e]);
+    }
+    if (ACT == 2) atomicAdd(&sbins[G * 2 * B + M * 2], d_sl);
   }
   __syncthreads();
-  for (unsigned i = threadIdx.x; i < (unsigned)(G * 2); i += EG_THREADS)
-    atomicAdd(&row_ws[(long)(b * G) * 2 + i], sbins[i]);
-  for (unsigned i = threadIdx.x; i < (unsigned)(M * 2); i += EG_THREADS)
-    atomicAdd(&chan_ws[i], sbins[G * 2 + i]);
-  if (ACT == 2 && threadIdx.x == 0) atomicAdd(slope_ws, sbins[G * 2 + M * 2]);
+  const long col = (long)blockIdx.x * B + b;
+  const long stride = (long)gridDim.x * B;
+  for (unsigned i = threadIdx.x; i < (unsigned)n_out; i += EG_THREADS)
+    scratch[i * stride + col] = sbins[i];
 }
 
 // pass 2: dWgT (B, N, M), deterministic.  dx1 elements are recomputed on
@@ -237,19 +327,27 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     int M, int G, long row_len, float slope,
     const float *__restrict__ slope_ptr) {
   const int b = blockIdx.z;
-  const int ppb = EG_THREADS / M;
-  const int p_l = (int)threadIdx.x / M;
-  const int c = (int)threadIdx.x % M;
+  const int tpc = M / 4;
+  const int ppb = EG_THREADS / tpc;
+  const int p_l = (int)threadIdx.x / tpc;
+  const int c4 = (int)threadIdx.x % tpc;
   if (p_l >= ppb) return;
   const int Cg = M / G;
-  const int row = b * G + c / Cg;
-  const float m = mean[row];
-  const float r = rstd[row];
-  const float ga = gamma[c], be = beta[c];
-  if (ACT == 2) slope = *slope_ptr;
   const float inv_n = 1.0f / (float)row_len;
-  const float s1 = row_ws[row * 2 + 0];
-  const float s2 = row_ws[row * 2 + 1];
+  const float invN = 1.0f / (float)N;
+  float m[4], r[4], ga[4], be[4], s1[4], s2[4];
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const int c = c4 * 4 + e;
+    const int row = b * G + c / Cg;
+    m[e] = mean[row];
+    r[e] = rstd[row];
+    ga[e] = gamma[c];
+    be[e] = beta[c];
+    s1[e] = row_ws[row * 2 + 0];
+    s2[e] = row_ws[row * 2 + 1];
+  }
+  if (ACT == 2) slope = *slope_ptr;
 
   const T *wgb = wg + (long)b * N * M;
   const T *dyb = dy + (long)b * N * M;
@@ -260,58 +358,70 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
 
   for (long p = (long)blockIdx.x * ppb + p_l; p < N;
        p += (long)gridDim.x * ppb) {
-    const float wg_p = ldg(wgb + p * M + c);
-    // centre term: sum_j dx1[c, j, p]
-    float acc = 0.f;
+    const Quad<T> pq = *(const Quad<T> *)(wgb + p * M + c4 * 4);
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    // centre term: -sum_j dx1[c, j, p]
     {
-      const int ksel = amb[p * M + c];
-      float g0 = ldg(dyb + p * M + c);
+      const uchar4 aq = *(const uchar4 *)(amb + p * M + c4 * 4);
+      const Quad<T> gq = *(const Quad<T> *)(dyb + p * M + c4 * 4);
+      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
       const int *irow = idxb + p * K;
       for (int j = 0; j < K; ++j) {
-        const float v = ldg(wgb + (long)irow[j] * M + c) - wg_p;
-        const float xhat = (v - m) * r;
-        float dxhat = 0.f;
-        if (j == ksel) {
-          float gs = g0;
-          if (ACT >= 1) {
-            const float pre = xhat * ga + be;
-            gs = pre > 0.f ? gs : gs * slope;
+        const Quad<T> nq =
+            *(const Quad<T> *)(wgb + (long)irow[j] * M + c4 * 4);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const float v = (float)nq.v[e] - (float)pq.v[e];
+          const float xhat = (v - m[e]) * r[e];
+          float dxhat = 0.f;
+          if (j == ks[e]) {
+            float gs = (float)gq.v[e];
+            if (ACT >= 1) {
+              const float pre = xhat * ga[e] + be[e];
+              gs = pre > 0.f ? gs : gs * slope;
+            }
+            dxhat = gs * ga[e];
           }
-          dxhat = gs * ga;
+          acc[e] -= (dxhat - (s1[e] + xhat * s2[e]) * inv_n) * r[e];
         }
-        acc -= (dxhat - (s1 + xhat * s2) * inv_n) * r;
       }
     }
     // incoming term: edges whose neighbour is p
     const int lo = offb[p], hi = offb[p + 1];
     for (int t = lo; t < hi; ++t) {
-      const int e = ordb[t];
-      const int j = e / (int)N;
-      const long n = e % (int)N;
-      const float v = wg_p - ldg(wgb + n * M + c);
-      const float xhat = (v - m) * r;
-      float dxhat = 0.f;
-      if (j == (int)amb[n * M + c]) {
-        float gs = ldg(dyb + n * M + c);
-        if (ACT >= 1) {
-          const float pre = xhat * ga + be;
-          gs = pre > 0.f ? gs : gs * slope;
+      int j, n;
+      split_edge(ordb[t], (int)N, invN, j, n);
+      const Quad<T> nq = *(const Quad<T> *)(wgb + (long)n * M + c4 * 4);
+      const uchar4 aq = *(const uchar4 *)(amb + (long)n * M + c4 * 4);
+      const Quad<T> gq = *(const Quad<T> *)(dyb + (long)n * M + c4 * 4);
+      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const float v = (float)pq.v[e] - (float)nq.v[e];
+        const float xhat = (v - m[e]) * r[e];
+        float dxhat = 0.f;
+        if (j == ks[e]) {
+          float gs = (float)gq.v[e];
+          if (ACT >= 1) {
+            const float pre = xhat * ga[e] + be[e];
+            gs = pre > 0.f ? gs : gs * slope;
+          }
+          dxhat = gs * ga[e];
         }
-        dxhat = gs * ga;
+        acc[e] += (dxhat - (s1[e] + xhat * s2[e]) * inv_n) * r[e];
       }
-      acc += (dxhat - (s1 + xhat * s2) * inv_n) * r;
     }
-    stg(dwg + ((long)b * N + p) * M + c, acc);
+    Quad<T> oq;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) oq.v[e] = (T)acc[e];
+    *(Quad<T> *)(dwg + ((long)b * N + p) * M + c4 * 4) = oq;
   }
 }
 
 // --------------------------------------------------------------- launchers
 
-static int eg_chunks(long N, int ppb, int B, long cap_atomics) {
-  // enough blocks to fill the chip, but bounded so the per-block workspace
-  // atomics stay cheap (cap_atomics blocks hit each row address)
-  long want = 2048 / (B > 0 ? B : 1);
-  if (want > cap_atomics) want = cap_atomics;
+static int eg_chunks(long N, int ppb, int B, int cap) {
+  long want = cap / (B > 0 ? B : 1);
   long blocks = (N + ppb - 1) / ppb;
   if (want > blocks) want = blocks;
   if (want < 1) want = 1;
@@ -319,18 +429,26 @@ static int eg_chunks(long N, int ppb, int B, long cap_atomics) {
 }
 
 template <typename T>
-void egnmp_fwd_impl(const T *wg, const int *idx, float *ws, float *mean,
-                    float *rstd, const float *gamma, const float *beta,
-                    T *y, unsigned char *am, int B, long N, int K, int M,
-                    int G, float eps, int act, float slope,
-                    const float *slope_ptr, hipStream_t stream) {
-  const int ppb = EG_THREADS / M;
-  const dim3 rgrid(eg_chunks(N, ppb, B, 64), 1, B);
-  hipLaunchKernelGGL(egnmp_fwd_reduce_kernel<T>, rgrid, dim3(EG_THREADS), 0,
-                     stream, wg, idx, ws, N, K, M, G);
+void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
+                    float *mean, float *rstd, const float *gamma,
+                    const float *beta, T *y, unsigned char *am, int B, long N,
+                    int K, int M, int G, float eps, int act, float slope,
+                    const float *slope_ptr, int rchunks, hipStream_t stream) {
+  const int tpc = M / 4;
+  const int ppb = EG_THREADS / tpc;
+  const dim3 rgrid(rchunks, 1, B);
+  const int n_out_f = B * G * 2;
+  hipLaunchKernelGGL(egnmp_fwd_reduce_kernel<T>, rgrid, dim3(EG_THREADS),
+                     (size_t)n_out_f * sizeof(float), stream, wg, idx,
+                     scratch, N, K, M, G);
+  const int waves_per_block = EG_THREADS / WAVE;
+  hipLaunchKernelGGL(egnmp_sum_partials_kernel,
+                     dim3((n_out_f + waves_per_block - 1) / waves_per_block),
+                     dim3(EG_THREADS), 0, stream, scratch, ws,
+                     (long)rchunks * B, n_out_f);
   launch_gn_finalize(ws, mean, rstd, (long)(M / G) * K * N, B * G, eps,
                      stream);
-  const dim3 agrid(eg_chunks(N, ppb, B, 1 << 20), 1, B);
+  const dim3 agrid(eg_chunks(N, ppb, B, 2048), 1, B);
 #define EG_FWD(A)                                                             \
   hipLaunchKernelGGL((egnmp_fwd_apply_kernel<T, A>), agrid, dim3(EG_THREADS), \
                      0, stream, wg, idx, mean, rstd, gamma, beta, y, am, N,   \
@@ -345,25 +463,32 @@ template <typename T>
 void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
                     const unsigned char *am, const int *order,
                     const int *offsets, const float *mean, const float *rstd,
-                    const float *gamma, const float *beta, float *row_ws,
-                    float *chan_ws, float *slope_ws, T *dwg, int B, long N,
-                    int K, int M, int G, int act, float slope,
-                    const float *slope_ptr, hipStream_t stream) {
-  const int ppb = EG_THREADS / M;
-  const dim3 rgrid(eg_chunks(N, ppb, B, 64), 1, B);
-  const dim3 agrid(eg_chunks(N, ppb, B, 1 << 20), 1, B);
-  const size_t shmem = (size_t)(G * 2 + M * 2 + 1) * sizeof(float);
+                    const float *gamma, const float *beta, float *scratch,
+                    float *ws, T *dwg, int B, long N, int K, int M, int G,
+                    int act, float slope, const float *slope_ptr, int rchunks,
+                    hipStream_t stream) {
+  const int tpc = M / 4;
+  const int ppb = EG_THREADS / tpc;
+  const dim3 rgrid(rchunks, 1, B);
+  const dim3 agrid(eg_chunks(N, ppb, B, 2048), 1, B);
+  const int n_out = B * G * 2 + M * 2 + 1;
+  const size_t shmem = (size_t)n_out * sizeof(float);
   const long row_len = (long)(M / G) * K * N;
+  const int waves_per_block = EG_THREADS / WAVE;
 #define EG_BWD(A)                                                              \
   do {                                                                         \
     hipLaunchKernelGGL((egnmp_bwd_reduce_kernel<T, A>), rgrid,                 \
                        dim3(EG_THREADS), shmem, stream, dy, wg, idx, am, mean, \
-                       rstd, gamma, beta, row_ws, chan_ws, slope_ws, N, K, M,  \
-                       G, slope, slope_ptr);                                   \
+                       rstd, gamma, beta, scratch, N, K, M, G, slope,          \
+                       slope_ptr);                                             \
+    hipLaunchKernelGGL(egnmp_sum_partials_kernel,                              \
+                       dim3((n_out + waves_per_block - 1) / waves_per_block),  \
+                       dim3(EG_THREADS), 0, stream, scratch, ws,               \
+                       (long)rchunks * B, n_out);                              \
     hipLaunchKernelGGL((egnmp_bwd_apply_kernel<T, A>), agrid,                  \
                        dim3(EG_THREADS), 0, stream, dy, wg, idx, am, order,    \
-                       offsets, mean, rstd, gamma, beta, row_ws, dwg, N, K, M, \
-                       G, row_len, slope, slope_ptr);                          \
+                       offsets, mean, rstd, gamma, beta, ws, dwg, N, K, M, G,  \
+                       row_len, slope, slope_ptr);                             \
   } while (0)
   if (act == 2) EG_BWD(2);
   else if (act == 1) EG_BWD(1);
@@ -371,37 +496,48 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
 #undef EG_BWD
 }
 
-void launch_egnmp_fwd(const void *wg, const int *idx, float *ws, float *mean,
-                      float *rstd, const float *gamma, const float *beta,
-                      void *y, unsigned char *am, int B, long N, int K, int M,
-                      int G, float eps, int act, float slope,
-                      const float *slope_ptr, bool bf16, hipStream_t stream) {
+int egnmp_reduce_chunks(long N, int M, int B) {
+  const int ppb = EG_THREADS / (M / 4);
+  long want = 1024 / (B > 0 ? B : 1);
+  long blocks = (N + ppb - 1) / ppb;
+  if (want > blocks) want = blocks;
+  if (want < 1) want = 1;
+  return (int)want;
+}
+
+void launch_egnmp_fwd(const void *wg, const int *idx, float *scratch,
+                      float *ws, float *mean, float *rstd, const float *gamma,
+                      const float *beta, void *y, unsigned char *am, int B,
+                      long N, int K, int M, int G, float eps, int act,
+                      float slope, const float *slope_ptr, bool bf16,
+                      int rchunks, hipStream_t stream) {
   if (bf16)
-    egnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)wg, idx, ws, mean,
-                                   rstd, gamma, beta, (__hip_bfloat16 *)y, am,
-                                   B, N, K, M, G, eps, act, slope, slope_ptr,
+    egnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)wg, idx, scratch,
+                                   ws, mean, rstd, gamma, beta,
+                                   (__hip_bfloat16 *)y, am, B, N, K, M, G,
+                                   eps, act, slope, slope_ptr, rchunks,
                                    stream);
   else
-    egnmp_fwd_impl<float>((const float *)wg, idx, ws, mean, rstd, gamma, beta,
-                          (float *)y, am, B, N, K, M, G, eps, act, slope,
-                          slope_ptr, stream);
+    egnmp_fwd_impl<float>((const float *)wg, idx, scratch, ws, mean, rstd,
+                          gamma, beta, (float *)y, am, B, N, K, M, G, eps,
+                          act, slope, slope_ptr, rchunks, stream);
 }
 
 void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
                       const unsigned char *am, const int *order,
                       const int *offsets, const float *mean, const float *rstd,
-                      const float *gamma, const float *beta, float *row_ws,
-                      float *chan_ws, float *slope_ws, void *dwg, int B,
-                      long N, int K, int M, int G, int act, float slope,
-                      const float *slope_ptr, bool bf16, hipStream_t stream) {
+                      const float *gamma, const float *beta, float *scratch,
+                      float *ws, void *dwg, int B, long N, int K, int M, int G,
+                      int act, float slope, const float *slope_ptr, bool bf16,
+                      int rchunks, hipStream_t stream) {
   if (bf16)
     egnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)wg, idx, am, order,
-        offsets, mean, rstd, gamma, beta, row_ws, chan_ws, slope_ws,
-        (__hip_bfloat16 *)dwg, B, N, K, M, G, act, slope, slope_ptr, stream);
+        offsets, mean, rstd, gamma, beta, scratch, ws, (__hip_bfloat16 *)dwg,
+        B, N, K, M, G, act, slope, slope_ptr, rchunks, stream);
   else
     egnmp_bwd_impl<float>((const float *)dy, (const float *)wg, idx, am, order,
-                          offsets, mean, rstd, gamma, beta, row_ws, chan_ws,
-                          slope_ws, (float *)dwg, B, N, K, M, G, act, slope,
-                          slope_ptr, stream);
+                          offsets, mean, rstd, gamma, beta, scratch, ws,
+                          (float *)dwg, B, N, K, M, G, act, slope, slope_ptr,
+                          rchunks, stream);
 }
