@@ -19,6 +19,17 @@
 // Fragment maps (gfx950, K=32): A and B operands are k-contiguous 8-vectors
 // at row (lane & 15), k-offset (lane >> 4) * 8; C/D: col = lane & 15,
 // row = (lane >> 4) * 4 + reg.
+//
+// Two entry points share the tile body:
+//  * pw_wgrad_kernel        -- one (dy, x, dw) problem per launch;
+//  * pw_wgrad_batched_kernel-- MANY problems in ONE launch.  The training
+//    step runs ~180 such weight gradients (the 8-iteration GRU loop's
+//    conv stacks); launched one-by-one they are launch/atomic-floor-bound
+//    (~20 us each for ~1 us of math).  The batched kernel takes a device
+//    array of job descriptors plus a block->(job, tile, chunk) map and
+//    accumulates DIRECTLY into each weight's fp32 grad buffer, so the
+//    per-call zero-fill + autograd accumulate-add pairs disappear with
+//    the launches.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include "common.h"
@@ -31,19 +42,14 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define KB 32        // K-block per MFMA
 #define LDS_PAD 8    // bf16 elements of row padding (16 B)
 
-__global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
-    const __hip_bfloat16 *__restrict__ dy,  // (B, Co, S)
-    const __hip_bfloat16 *__restrict__ x,   // (B, Ci, S)
-    float *__restrict__ dw,                 // (Co, Ci) pre-zeroed fp32
-    float *__restrict__ dbias,              // (Co) pre-zeroed fp32 or null
-    int B, int Co, int Ci, long S, int schunks) {
-  __shared__ __hip_bfloat16 sA[TILE][KB + LDS_PAD];
-  __shared__ __hip_bfloat16 sB[TILE][KB + LDS_PAD];
+#include "pw_wgrad_job.h"
 
-  const int tile_o = blockIdx.x * TILE;
-  const int tile_i = blockIdx.y * TILE;
-  const int b = blockIdx.z / schunks;
-  const int chunk = blockIdx.z % schunks;
+// One 64x64 output tile x one (b, chunk) slice of the reduction.
+__device__ __forceinline__ void pw_wgrad_tile(
+    const __hip_bfloat16 *__restrict__ dy, const __hip_bfloat16 *__restrict__ x,
+    float *__restrict__ dw, float *__restrict__ dbias, int Co, int Ci, long S,
+    int schunks, int tile_o, int tile_i, int b, int chunk,
+    __hip_bfloat16 (*sA)[KB + LDS_PAD], __hip_bfloat16 (*sB)[KB + LDS_PAD]) {
   // chunk bounds KB-aligned so in-range stage loads are whole 16B vectors
   const long per = (((S + schunks - 1) / schunks + KB - 1) / KB) * KB;
   const long s_lo = chunk * per;
@@ -149,6 +155,33 @@ __global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
       }
 }
 
+__global__ __launch_bounds__(WG_THREADS) void pw_wgrad_kernel(
+    const __hip_bfloat16 *__restrict__ dy, const __hip_bfloat16 *__restrict__ x,
+    float *__restrict__ dw, float *__restrict__ dbias, int B, int Co, int Ci,
+    long S, int schunks) {
+  __shared__ __hip_bfloat16 sA[TILE][KB + LDS_PAD];
+  __shared__ __hip_bfloat16 sB[TILE][KB + LDS_PAD];
+  const int b = blockIdx.z / schunks;
+  const int chunk = blockIdx.z % schunks;
+  pw_wgrad_tile(dy, x, dw, dbias, Co, Ci, S, schunks, blockIdx.x * TILE,
+                blockIdx.y * TILE, b, chunk, sA, sB);
+}
+
+// map[blockIdx.x] = job(12b) | bz(12b) | to(4b) | ti(4b)
+__global__ __launch_bounds__(WG_THREADS) void pw_wgrad_batched_kernel(
+    const PwWgradJob *__restrict__ jobs, const unsigned int *__restrict__ map) {
+  __shared__ __hip_bfloat16 sA[TILE][KB + LDS_PAD];
+  __shared__ __hip_bfloat16 sB[TILE][KB + LDS_PAD];
+  const unsigned int m = map[blockIdx.x];
+  const PwWgradJob j = jobs[m >> 20];
+  const int bz = (m >> 8) & 0xFFF;
+  const int to = (m >> 4) & 0xF;
+  const int ti = m & 0xF;
+  pw_wgrad_tile((const __hip_bfloat16 *)j.dy, (const __hip_bfloat16 *)j.x,
+                j.dw, j.dbias, j.Co, j.Ci, j.S, j.schunks, to * TILE,
+                ti * TILE, bz / j.schunks, bz % j.schunks, sA, sB);
+}
+
 void launch_pw_wgrad(const void *dy, const void *x, float *dw, float *dbias,
                      int B, int Co, int Ci, long S, int schunks_opt,
                      hipStream_t stream) {
@@ -167,4 +200,11 @@ void launch_pw_wgrad(const void *dy, const void *x, float *dw, float *dbias,
   hipLaunchKernelGGL(pw_wgrad_kernel, grid, dim3(WG_THREADS), 0, stream,
                      (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, dw,
                      dbias, B, Co, Ci, S, schunks);
+}
+
+void launch_pw_wgrad_batched(const void *jobs_dev, const unsigned int *map_dev,
+                             int n_blocks, hipStream_t stream) {
+  hipLaunchKernelGGL(pw_wgrad_batched_kernel, dim3(n_blocks),
+                     dim3(WG_THREADS), 0, stream,
+                     (const PwWgradJob *)jobs_dev, map_dev);
 }

@@ -68,8 +68,11 @@ def build_graphed_step(model, batch, num_iters: int, gamma: float,
     from pvraft_amd.utils import sequence_loss
 
     def fn():
+        from pvraft_amd.model import pointwise
+
         for b in reducer.buckets:
             b.flat.zero_()
+        pointwise.wgrad_defer_begin()
         with torch.autocast("cuda", dtype=torch.bfloat16, enabled=amp):
             flows = model(batch["sequence"], num_iters=num_iters)
             if loss_fn is not None:
@@ -77,6 +80,10 @@ def build_graphed_step(model, batch, num_iters: int, gamma: float,
             else:
                 loss = sequence_loss(flows, batch, gamma=gamma)
         loss.backward()
+        # deferred wgrad batch + stacked-weight scatter, recorded in the
+        # graph (at replay the captured kernels rerun; the Python queue
+        # stays empty)
+        pointwise.wgrad_flush()
         final = flows[-1] if isinstance(flows, (list, tuple)) else flows
         return loss, final
 

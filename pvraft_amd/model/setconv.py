@@ -66,7 +66,8 @@ class SetConv(nn.Module):
             g = torch.cat([f_cn.to(dt), graph.xyz.transpose(1, 2).to(dt)], dim=1)
             from .pointwise import pw_matmul
 
-            wg = pw_matmul(self.fc1.weight.view(self.fc1.out_channels, -1), g)
+            wg = pw_matmul(self.fc1.weight.view(self.fc1.out_channels, -1), g,
+                           targets=(self.fc1.weight, None))
             wg_t = ops.transpose_last2(wg)  # (B, N, mid)
             y_t = ops.edge_gnmp(
                 wg_t, graph.idx32, graph.csr(), self.gn1.num_groups,

@@ -28,11 +28,27 @@ from .setconv import SetConv
 
 
 def _slice_weight(weight: Tensor, sizes):
-    """Contiguous column slices of a (Co, Ci, 1) conv weight."""
+    """Contiguous column slices of a (Co, Ci, 1) conv weight, each paired
+    with a deferred-wgrad target: a zeroed fp32 buffer whose flush-time
+    post-callback adds it into the right column block of the parameter's
+    .grad (see pointwise.defer_buffer).  Returns [(w_slice, target|None)]."""
+    from . import pointwise
+
     w = weight.squeeze(-1)
+    defer = pointwise.wgrad_defer_active()
     out, lo = [], 0
     for s_ in sizes:
-        out.append(w[:, lo : lo + s_].contiguous())
+        sl = w[:, lo : lo + s_].contiguous()
+        tgt = None
+        if defer:
+            lo_c, sz = lo, s_
+            tgt = pointwise.defer_buffer(
+                sl.shape, weight.device,
+                lambda buf, lo_c=lo_c, sz=sz: pointwise._grad_buffer(weight)
+                .view(weight.shape[0], -1)[:, lo_c : lo_c + sz]
+                .add_(buf),
+            )
+        out.append((sl, tgt))
         lo += s_
     return out
 
@@ -49,9 +65,10 @@ def _split_mm(weight: Tensor, bias, parts, wparts=None):
     if wparts is None:
         wparts = _slice_weight(weight, [p.shape[1] for p in parts])
     out = None
-    for i, (w_i, p) in enumerate(zip(wparts, parts)):
+    for i, ((w_i, tgt), p) in enumerate(zip(wparts, parts)):
         # bias rides the first partial GEMM (fused fwd add + kernel dbias)
-        term = pw_matmul(w_i, p, bias if i == 0 else None)
+        b = bias if i == 0 else None
+        term = pw_matmul(w_i, p, b, targets=(tgt, bias) if tgt is not None else None)
         out = term if out is None else out + term
     return out
 
@@ -97,7 +114,13 @@ class ConvGRU(nn.Module):
         column slices stack row-wise into ONE GEMM per input part
         (z/r/q = rows [0:64/64:128/128:192]); only the hidden operand
         differs (z,r read h; q reads r*h), so the h columns fuse z+r only.
+
+        When deferred wgrad is active every stacked tensor is paired with a
+        fp32 buffer ("t_*") whose flush-time callback scatters it back into
+        convz/convr/convq's .grad column blocks.
         """
+        from . import pointwise
+
         hd = self.convz.out_channels
         wz = self.convz.weight.squeeze(-1)
         wr = self.convr.weight.squeeze(-1)
@@ -106,19 +129,49 @@ class ConvGRU(nn.Module):
                "q_h": wq[:, :hd].contiguous(),
                "b_zr": torch.cat([self.convz.bias, self.convr.bias], dim=0),
                "b_q": self.convq.bias,
-               "parts": []}
+               "parts": [],
+               "t_zr_h": None, "t_q_h": None, "t_b_zr": None, "t_parts": []}
         lo = hd
         for sz in part_sizes:
             out["parts"].append(
                 torch.cat([wz[:, lo : lo + sz], wr[:, lo : lo + sz], wq[:, lo : lo + sz]], dim=0).contiguous()
             )
             lo += sz
+        if pointwise.wgrad_defer_active():
+            dev = wz.device
+            gbuf = pointwise._grad_buffer
+            zw, rw, qw = self.convz.weight, self.convr.weight, self.convq.weight
+
+            def scatter_h(buf):
+                gbuf(zw).view(hd, -1)[:, :hd].add_(buf[:hd])
+                gbuf(rw).view(hd, -1)[:, :hd].add_(buf[hd:])
+
+            out["t_zr_h"] = pointwise.defer_buffer((2 * hd, hd), dev, scatter_h)
+            out["t_q_h"] = pointwise.defer_buffer(
+                (hd, hd), dev,
+                lambda buf: gbuf(qw).view(hd, -1)[:, :hd].add_(buf))
+            zb, rb = self.convz.bias, self.convr.bias
+            out["t_b_zr"] = pointwise.defer_buffer(
+                (2 * hd,), dev,
+                lambda buf: (gbuf(zb).add_(buf[:hd]), gbuf(rb).add_(buf[hd:])))
+            lo = hd
+            for sz in part_sizes:
+                def scatter_part(buf, lo_c=lo, s=sz):
+                    gbuf(zw).view(hd, -1)[:, lo_c:lo_c + s].add_(buf[:hd])
+                    gbuf(rw).view(hd, -1)[:, lo_c:lo_c + s].add_(buf[hd:2 * hd])
+                    gbuf(qw).view(hd, -1)[:, lo_c:lo_c + s].add_(buf[2 * hd:])
+                out["t_parts"].append(
+                    pointwise.defer_buffer((3 * hd, sz), dev, scatter_part))
+                lo += sz
+        else:
+            out["t_parts"] = [None] * len(part_sizes)
         return out
 
     def precompute_inp(self, inp: Tensor, gw) -> Tensor:
         """Contribution of the iteration-constant context features to all
         three gate preactivations -- hoisted out of the GRU loop."""
-        return pw_matmul(gw["parts"][0], inp)
+        t = gw["t_parts"][0]
+        return pw_matmul(gw["parts"][0], inp, targets=(t, None) if t is not None else None)
 
     def forward_parts(self, h: Tensor, x_parts, wcache=None, gw=None, pre=None) -> Tensor:
         """Gates from the concat parts (cross-gate fused GEMMs); the gate
@@ -129,12 +182,21 @@ class ConvGRU(nn.Module):
         if gw is None:
             gw = self.gate_weights([p.shape[1] for p in x_parts])
         hd = self.convz.out_channels
+
+        def tgt(t):
+            return (t, None) if t is not None else None
+
         # shared-input contribution for all gates: (B, 3*hd, N)
-        m = pre if pre is not None else pw_matmul(gw["parts"][0], x_parts[0])
-        for w_i, p in zip(gw["parts"][1:], x_parts[1:]):
-            m = m + pw_matmul(w_i, p)
-        z, rh = ops.gru_zr(pw_matmul(gw["zr_h"], h, gw["b_zr"]) + m[:, : 2 * hd], h)
-        return ops.gru_q(pw_matmul(gw["q_h"], rh, gw["b_q"]) + m[:, 2 * hd :], z, h)
+        m = pre if pre is not None else pw_matmul(
+            gw["parts"][0], x_parts[0], targets=tgt(gw["t_parts"][0]))
+        for w_i, t, p in zip(gw["parts"][1:], gw["t_parts"][1:], x_parts[1:]):
+            m = m + pw_matmul(w_i, p, targets=tgt(t))
+        zr_t = (gw["t_zr_h"], gw["t_b_zr"]) if gw["t_zr_h"] is not None else None
+        q_t = (gw["t_q_h"], gw["b_q"]) if gw["t_q_h"] is not None else None
+        z, rh = ops.gru_zr(
+            pw_matmul(gw["zr_h"], h, gw["b_zr"], targets=zr_t) + m[:, : 2 * hd], h)
+        return ops.gru_q(
+            pw_matmul(gw["q_h"], rh, gw["b_q"], targets=q_t) + m[:, 2 * hd :], z, h)
 
 
 class ConvRNN(nn.Module):

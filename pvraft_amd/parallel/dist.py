@@ -182,9 +182,18 @@ class GradReducer:
         if b.pending == 0 and self.enabled:
             b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM, group=self.pg, async_op=True)
 
+    def _flush_deferred(self) -> None:
+        # deferred wgrad jobs must land in the flat buffers before any
+        # remaining bucket reduction (buckets containing deferred params
+        # never fire their hooks early, so ordering is safe)
+        from pvraft_amd.model import pointwise
+
+        pointwise.wgrad_flush()
+
     def reduce_all(self) -> None:
         """Eager mean-all-reduce of every bucket (hipGraph mode: the
         backward ran inside a captured graph, hooks were disabled)."""
+        self._flush_deferred()
         if not self.enabled:
             return
         world = dist.get_world_size()
@@ -197,6 +206,7 @@ class GradReducer:
             b.flat /= world
 
     def finalize(self) -> None:
+        self._flush_deferred()
         world = dist.get_world_size() if self.enabled else 1
         for b in self.buckets:
             if b.work is None and self.enabled:
@@ -212,3 +222,9 @@ class GradReducer:
         for b in self.buckets:
             b.flat.zero_()
         self._reset_pending()
+        if self.buckets and self.buckets[0].flat.is_cuda:
+            # arm the deferred batched-wgrad path for this step (the flat
+            # views are the accumulation targets; flush in finalize)
+            from pvraft_amd.model import pointwise
+
+            pointwise.wgrad_defer_begin()
