@@ -577,10 +577,29 @@ std::vector<torch::Tensor> pw_wgrad_batched(std::vector<torch::Tensor> dys,
   memcpy(host.data_ptr(), jobs.data(), jbytes);
   memcpy((char*)host.data_ptr() + jbytes, map.data(), mbytes);
   auto dev = torch::empty({jbytes + mbytes}, dys[0].options().dtype(torch::kUInt8));
-  dev.copy_(host, /*non_blocking=*/true);
+  hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+  hipStreamIsCapturing(stream(), &cap);
+  if (cap != hipStreamCaptureStatusNone) {
+    // Descriptors are CONSTANT for a captured graph (pool addresses are
+    // stable across replays), so write them OUTSIDE the capture: a
+    // synchronous side-stream copy executes now, records nothing, and the
+    // captured kernel below only reads the buffer.  (A captured H2D copy
+    // via Tensor::copy_ trips hipErrorCapturedEvent on this stack.)
+    hipStream_t side;
+    hipStreamCreateWithFlags(&side, hipStreamNonBlocking);
+    hipMemcpyAsync(dev.data_ptr(), host.data_ptr(), jbytes + mbytes,
+                   hipMemcpyHostToDevice, side);
+    hipStreamSynchronize(side);
+    hipStreamDestroy(side);
+  } else {
+    hipMemcpyAsync(dev.data_ptr(), host.data_ptr(), jbytes + mbytes,
+                   hipMemcpyHostToDevice, stream());
+  }
   launch_pw_wgrad_batched(dev.data_ptr(),
                           (const unsigned int*)((char*)dev.data_ptr() + jbytes),
                           (int)map.size(), stream());
+  // caller keeps both alive: host until the async copy drains (eager) /
+  // dev for the lifetime of the captured graph
   return {host, dev};
 }
 

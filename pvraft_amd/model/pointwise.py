@@ -60,7 +60,8 @@ class _DeferState:
     on = False
     jobs: list = []   # (dy, x, dw_flat_fp32, dbias_fp32_or_None)
     posts: list = []  # callbacks run after the batched kernel
-    keep: list = []   # capture keep-alives (descriptor host/dev buffers)
+    keep_capture: list = []  # descriptor buffers owned by live hipGraphs
+    keep_eager: list = []    # recent eager descriptors (pinned-copy drain)
 
 
 _DEFER = _DeferState()
@@ -76,9 +77,13 @@ def wgrad_defer_begin() -> None:
     flat views, zeroed at step start)."""
     _DEFER.on = True
     # drop leftovers from an aborted step (stale tensors must not flush
-    # into this step's gradients)
+    # into this step's gradients); old eager descriptors (their async H2D
+    # copies drained many steps ago) can go -- captured graphs' descriptor
+    # buffers live in keep_capture and are never trimmed
     _DEFER.jobs.clear()
     _DEFER.posts.clear()
+    if len(_DEFER.keep_eager) > 64:
+        del _DEFER.keep_eager[:32]
 
 
 def wgrad_defer_end() -> None:
@@ -103,7 +108,9 @@ def wgrad_flush() -> None:
             [j[3] if j[3] is not None else empty for j in jobs],
         )
         if torch.cuda.is_current_stream_capturing():
-            _DEFER.keep.append(keep)  # replayed H2D copy re-reads these
+            _DEFER.keep_capture.append(keep)  # graph reads these at replay
+        else:
+            _DEFER.keep_eager.append(keep)  # until the async copy drains
     for fn in posts:
         fn()
 
