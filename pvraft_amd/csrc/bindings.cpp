@@ -512,6 +512,7 @@ std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
 #include "pw_wgrad_job.h"
 void launch_pw_wgrad_batched(const void*, const unsigned int*, int,
                              hipStream_t);
+void launch_desc_fill(void*, const void*, long, hipStream_t);
 
 std::vector<torch::Tensor> pw_wgrad_batched(std::vector<torch::Tensor> dys,
                                             std::vector<torch::Tensor> xs,
@@ -572,35 +573,21 @@ std::vector<torch::Tensor> pw_wgrad_batched(std::vector<torch::Tensor> dys,
   }
   const long jbytes = (long)n * sizeof(PwWgradJob);
   const long mbytes = (long)map.size() * sizeof(unsigned int);
-  auto host = torch::empty({jbytes + mbytes},
-                           torch::TensorOptions().dtype(torch::kUInt8).pinned_memory(true));
-  memcpy(host.data_ptr(), jobs.data(), jbytes);
-  memcpy((char*)host.data_ptr() + jbytes, map.data(), mbytes);
+  std::vector<unsigned char> blob(jbytes + mbytes);
+  memcpy(blob.data(), jobs.data(), jbytes);
+  memcpy(blob.data() + jbytes, map.data(), mbytes);
   auto dev = torch::empty({jbytes + mbytes}, dys[0].options().dtype(torch::kUInt8));
-  hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
-  hipStreamIsCapturing(stream(), &cap);
-  if (cap != hipStreamCaptureStatusNone) {
-    // Descriptors are CONSTANT for a captured graph (pool addresses are
-    // stable across replays), so write them OUTSIDE the capture: a
-    // synchronous side-stream copy executes now, records nothing, and the
-    // captured kernel below only reads the buffer.  (A captured H2D copy
-    // via Tensor::copy_ trips hipErrorCapturedEvent on this stack.)
-    hipStream_t side;
-    hipStreamCreateWithFlags(&side, hipStreamNonBlocking);
-    hipMemcpyAsync(dev.data_ptr(), host.data_ptr(), jbytes + mbytes,
-                   hipMemcpyHostToDevice, side);
-    hipStreamSynchronize(side);
-    hipStreamDestroy(side);
-  } else {
-    hipMemcpyAsync(dev.data_ptr(), host.data_ptr(), jbytes + mbytes,
-                   hipMemcpyHostToDevice, stream());
-  }
+  // descriptor bytes travel as kernel ARGUMENTS (desc_fill_kernel): plain
+  // launches that hipGraph capture records like any other kernel -- no
+  // H2D memcpy / events / pinned memory, which all invalidate capture on
+  // this stack.  Outside capture the same path is a handful of ~2 us
+  // launches.
+  launch_desc_fill(dev.data_ptr(), blob.data(), jbytes + mbytes, stream());
   launch_pw_wgrad_batched(dev.data_ptr(),
                           (const unsigned int*)((char*)dev.data_ptr() + jbytes),
                           (int)map.size(), stream());
-  // caller keeps both alive: host until the async copy drains (eager) /
-  // dev for the lifetime of the captured graph
-  return {host, dev};
+  // caller keeps dev alive for the lifetime of any capturing graph
+  return {dev};
 }
 
 // (B, R, C) view (row-contiguous, arbitrary batch stride) -> (B, C, R)

@@ -208,3 +208,30 @@ void launch_pw_wgrad_batched(const void *jobs_dev, const unsigned int *map_dev,
                      dim3(WG_THREADS), 0, stream,
                      (const PwWgradJob *)jobs_dev, map_dev);
 }
+
+// Capture-safe descriptor upload: the bytes ride in the kernel ARGUMENT
+// (by value, < 4 KB kernarg limit), so filling the device buffer is a
+// plain kernel launch that hipGraph capture records like any other --
+// no H2D memcpy, no events, no pinned memory.  Each replay rewrites the
+// same constants (descriptors are fixed for a captured graph).
+struct DescChunk {
+  unsigned char data[3584];
+};
+
+__global__ void desc_fill_kernel(unsigned char *__restrict__ dst, DescChunk c,
+                                 int n) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dst[i] = c.data[i];
+}
+
+void launch_desc_fill(void *dst, const void *src, long bytes,
+                      hipStream_t stream) {
+  const int CH = (int)sizeof(DescChunk);
+  for (long off = 0; off < bytes; off += CH) {
+    DescChunk c;
+    const int n = (int)((bytes - off) < CH ? (bytes - off) : CH);
+    __builtin_memcpy(c.data, (const char *)src + off, n);
+    hipLaunchKernelGGL(desc_fill_kernel, dim3((n + 255) / 256), dim3(256), 0,
+                       stream, (unsigned char *)dst + off, c, n);
+  }
+}
