@@ -549,12 +549,13 @@ std::vector<torch::Tensor> pw_wgrad_batched(std::vector<torch::Tensor> dys,
     base_blocks += (long)((Co + TILE - 1) / TILE) * ((Ci + TILE - 1) / TILE) *
                    dy.size(0);
   }
-  // split the reduction so the single launch still fills the chip
-  long want = 3072 / (base_blocks > 0 ? base_blocks : 1);
+  // split each job's reduction so every block gets ~96 K-loop iterations
+  // (~3072 reduction elements): a single global split starved the few
+  // huge-S jobs (S ~ 262k, the kNN-branch convs), whose straggler blocks
+  // then dominated the whole launch
+  (void)base_blocks;
   for (int j = 0; j < n; ++j) {
-    long cap = jobs[j].S / (32 * 4);
-    long sc = want;
-    if (sc > cap) sc = cap;
+    long sc = (jobs[j].S + 3071) / 3072;
     if (sc < 1) sc = 1;
     if ((long)jobs[j].B * sc > 4095) sc = 4095 / jobs[j].B;
     jobs[j].schunks = (int)sc;

@@ -8,7 +8,10 @@ comparison.
 """
 
 import copy
+import os
 import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
