@@ -258,13 +258,21 @@ def test_deferred_batched_wgrad_matches_plain_autograd():
     assert not pointwise._DEFER.jobs
 
     assert torch.allclose(loss.float(), loss2.float(), atol=1e-3, rtol=1e-3)
+    # Two runs of the SAME plain path already differ by up to ~13% rel-max
+    # per tensor (fp32-atomic GN statistics reorder -> bf16 argmax/top-k
+    # tie flips; measured by scripts/grad_noise_probe.py, min cosine
+    # 0.9955).  A mapping bug (wrong column offsets, missing job) would
+    # destroy direction/magnitude, so gate on cosine + norm ratio.
     ref = dict(model_ref.named_parameters())
     bad = []
     for name, p in model.named_parameters():
         g, gr = p.grad, ref[name].grad
-        if gr is None:
+        if gr is None or gr.abs().max() < 1e-12:
             continue
-        scale = gr.abs().max().item() + 1e-6
-        if not torch.allclose(g, gr, atol=2e-2 * scale + 1e-5, rtol=5e-2):
-            bad.append((name, (g - gr).abs().max().item(), scale))
+        cos = torch.nn.functional.cosine_similarity(
+            g.float().flatten(), gr.float().flatten(), dim=0
+        ).item()
+        ratio = (g.float().norm() / (gr.float().norm() + 1e-12)).item()
+        if cos < 0.99 or not (0.9 < ratio < 1.1):
+            bad.append((name, cos, ratio))
     assert not bad, bad
