@@ -2,57 +2,65 @@
 // model/flot/graph.py:53-60, which materialises the full B x N x N distance
 // matrix and argsorts it).
 //
-// Histogram-select design (the insert-heap variants measured ~2.3 ms for
-// B=2, N=8192: the first ~2k candidates of every query force a heap insert
-// in SOME lane, serialising whole waves).  Distances are non-negative
-// floats, so their bit patterns order monotonically:
+// Sample-threshold design (round 2; the round-1 histogram-select measured
+// 454 us at B=2, N=8192 -- one LDS atomic per (candidate, query) with
+// same-bin conflicts made it atomic-bound ~10x over its compute floor):
 //
-//   pass 1: histogram each query's N squared distances into 256 bins by
-//           float-exponent byte (bits >> 23); prefix-scan to find the
-//           threshold bin T where the k-th smallest falls;
-//   pass 1.5 (rare): if bin T is overfull (> CAP), refine it by the next 8
-//           mantissa bits;
-//   pass 2: candidates strictly below the threshold are accepted directly
-//           (slot via LDS counter); candidates AT the threshold go to a
-//           small boundary buffer; a per-query wave argmin selects the
-//           remaining slots from it.
+//   pass A: every query computes distances to a 1-in-ST sample of the
+//           cloud (ST = N/1024, so <= 1024 samples); a wave extracts the
+//           exact k'-th smallest sample distance (k' = ceil(2k/ST)) by
+//           argmin rounds -> threshold tau.  E[|{d <= tau}|] = 2k over the
+//           full cloud, so the collect buffer stays tiny.
+//   pass B: one full sweep; candidates with d <= tau go to a per-query
+//           (CAP=256) buffer -- ~2k LDS-atomic slot grabs per query
+//           instead of N.  Everything else is rejected with ZERO atomics.
+//   select: a wave extracts the exact k smallest from the buffer.
 //
-// No per-candidate branching storms: every pass is straight-line math +
-// one LDS atomic.  Candidate tiles (TILE_PTS x 3 fp32) are staged in LDS
-// and shared by QB=8 queries per workgroup; grid (ceil(N/QB), B) fills the
-// chip.  Neighbour order within a query is arbitrary (downstream max-pools
-// are order-invariant; reference relies on the set only).
+//   Underflow (buffer < k: ~1e-5 per query for k=32 by the Poisson tail)
+//   widens tau 4x and re-sweeps the active queries; after 3 widenings the
+//   remaining slots pad with the query itself (degenerate tie clouds --
+//   same semantics as round 1; downstream consumers are set/order
+//   invariant and edge features of self-pairs are zero).
+//
+//   For small clouds (ST == 1) the sample IS the cloud, tau is the exact
+//   min(2k, N)-th distance, and no underflow is possible.
+//
+// Candidate tiles (TILE_PTS x 3 fp32) are staged in LDS and shared by
+// QB=8 queries per workgroup; grid (ceil(N/QB), B) fills the chip.
 #include <hip/hip_runtime.h>
 #include "common.h"
 
 #define KNN_THREADS 256
 #define TILE_PTS 2048
-#define QB 8          // queries per workgroup
-#define CAP 256       // boundary-buffer capacity per query
-#define KNN_MAXK 48   // model uses 32 (reference extractor.py:10)
-
-DEV_INLINE unsigned dist_bits(float d) {
-  return __float_as_uint(d);  // d >= 0 -> monotonic
-}
+#define QB 8            // queries per workgroup
+#define CAP 256         // collect-buffer capacity per query
+#define SAMP 1024       // max sample size per query
+#define SPAD 4          // LDS row padding (bank spread across qi)
+#define KNN_MAXK 48     // model uses 32 (reference extractor.py:10)
 
 __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
     const float *__restrict__ xyz,  // (B, N, 3)
     int *__restrict__ out_idx,      // (B, N, k)
     int N, int k) {
   __shared__ float s_tile[TILE_PTS * 3];
-  __shared__ unsigned s_hist[QB][256];
+  __shared__ float s_samp[QB][SAMP + SPAD];
   __shared__ float s_bd[QB][CAP];
   __shared__ int s_bi[QB][CAP];
-  __shared__ unsigned s_acc[QB];   // accepted-slot counters
-  __shared__ unsigned s_bcnt[QB];  // boundary counters
-  __shared__ unsigned s_thr[QB];   // threshold bin T
-  __shared__ unsigned s_sub[QB];   // refined sub-bin (0xffffffff = no refine)
-  __shared__ int s_need[QB];       // slots to fill from the boundary buffer
+  __shared__ unsigned s_cnt[QB];  // collect counters
+  __shared__ float s_tau[QB];
+  __shared__ unsigned s_active;   // queries still needing a sweep
   __shared__ float s_q[QB][3];
 
   const int b = blockIdx.y;
   const int q0 = blockIdx.x * QB;
   const float *cloud = xyz + (long)b * N * 3;
+
+  const int st = N > SAMP ? (N + SAMP - 1) / SAMP : 1;
+  const int ns = (N + st - 1) / st;
+  int kp = 2 * k;  // k' such that E[collected] ~ 2k
+  if (st > 1) kp = (2 * k + st - 1) / st;
+  if (kp > ns) kp = ns;
+  if (kp < 1) kp = 1;
 
   if (threadIdx.x < QB) {
     const int q = q0 + threadIdx.x;
@@ -60,64 +68,66 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
     s_q[threadIdx.x][0] = cloud[qq * 3 + 0];
     s_q[threadIdx.x][1] = cloud[qq * 3 + 1];
     s_q[threadIdx.x][2] = cloud[qq * 3 + 2];
-    s_acc[threadIdx.x] = 0;
-    s_bcnt[threadIdx.x] = 0;
-    s_sub[threadIdx.x] = 0xffffffffu;
+    s_cnt[threadIdx.x] = 0;
   }
-  for (int i = threadIdx.x; i < QB * 256; i += KNN_THREADS)
-    ((unsigned *)s_hist)[i] = 0;
+  if (threadIdx.x == 0) s_active = (1u << QB) - 1;
   __syncthreads();
 
-  // ---- pass 1: exponent-byte histogram
-  for (int t0 = 0; t0 < N; t0 += TILE_PTS) {
-    const int tn = min(TILE_PTS, N - t0);
-    __syncthreads();
-    for (int i = threadIdx.x; i < tn * 3; i += KNN_THREADS)
-      s_tile[i] = cloud[(long)t0 * 3 + i];
-    __syncthreads();
-    for (int c = threadIdx.x; c < tn; c += KNN_THREADS) {
-      const float cx = s_tile[c * 3 + 0];
-      const float cy = s_tile[c * 3 + 1];
-      const float cz = s_tile[c * 3 + 2];
+  // ---- pass A: sample distances (cloud is L2-resident; strided gather)
+  for (int si = threadIdx.x; si < ns; si += KNN_THREADS) {
+    const long c = (long)si * st;
+    float cx = 0.f, cy = 0.f, cz = 0.f;
+    const bool ok = c < N;
+    if (ok) {
+      cx = cloud[c * 3 + 0];
+      cy = cloud[c * 3 + 1];
+      cz = cloud[c * 3 + 2];
+    }
 #pragma unroll
-      for (int qi = 0; qi < QB; ++qi) {
-        const float dx = cx - s_q[qi][0];
-        const float dy = cy - s_q[qi][1];
-        const float dz = cz - s_q[qi][2];
-        const float d = dx * dx + dy * dy + dz * dz;
-        atomicAdd(&s_hist[qi][dist_bits(d) >> 23], 1u);
-      }
+    for (int qi = 0; qi < QB; ++qi) {
+      const float dx = cx - s_q[qi][0];
+      const float dy = cy - s_q[qi][1];
+      const float dz = cz - s_q[qi][2];
+      s_samp[qi][si] = ok ? dx * dx + dy * dy + dz * dz : INFINITY;
     }
   }
   __syncthreads();
 
-  // ---- threshold scan (thread qi walks its query's 256 bins)
-  if (threadIdx.x < QB) {
-    const int qi = threadIdx.x;
-    unsigned cum = 0, T = 255;
-    for (int bin = 0; bin < 256; ++bin) {
-      const unsigned nxt = cum + s_hist[qi][bin];
-      if (nxt >= (unsigned)k) {
-        T = bin;
-        break;
+  // ---- exact k'-th smallest of each query's sample: wave argmin rounds
+  {
+    const int lane = lane_id();
+    for (int qi = wave_id(); qi < QB; qi += KNN_THREADS / WAVE) {
+      float dv[SAMP / WAVE];
+#pragma unroll
+      for (int s = 0; s < SAMP / WAVE; ++s) {
+        const int p = lane + s * WAVE;
+        dv[s] = p < ns ? s_samp[qi][p] : INFINITY;
       }
-      cum = nxt;
+      float tau = INFINITY;
+      for (int r = 0; r < kp; ++r) {
+        float best = INFINITY;
+        int bslot = 0;
+#pragma unroll
+        for (int s = 0; s < SAMP / WAVE; ++s)
+          if (dv[s] < best) {
+            best = dv[s];
+            bslot = s;
+          }
+        int bidx = lane + bslot * WAVE;
+        if (best == INFINITY) bidx = 0x7fffffff;
+        float bv = best;
+        wave_argmin(bv, bidx);
+        tau = bv;
+        if (bidx != 0x7fffffff && (bidx % WAVE) == lane) dv[bidx / WAVE] = INFINITY;
+      }
+      if (lane == 0) s_tau[qi] = tau;
     }
-    s_thr[qi] = T;
-    s_acc[qi] = 0;
-    s_need[qi] = k - (int)cum;  // slots to take from bin T
-    // overfull threshold bin -> refine by the next 8 bits below the exponent
-    if (s_hist[qi][T] > CAP - 8) s_sub[qi] = 0;  // mark: refine needed
   }
   __syncthreads();
 
-  bool any_refine = false;
-  for (int qi = 0; qi < QB; ++qi) any_refine |= (s_sub[qi] == 0u);
-  if (any_refine) {
-    // reuse the histograms for the sub-bins of each query's threshold bin
-    for (int i = threadIdx.x; i < QB * 256; i += KNN_THREADS)
-      ((unsigned *)s_hist)[i] = (((unsigned *)s_hist)[i] & 0u);
-    __syncthreads();
+  // ---- pass B: collect d <= tau (re-sweep with widened tau on underflow)
+  for (int iter = 0; iter < 3 && s_active != 0; ++iter) {
+    const unsigned active = s_active;
     for (int t0 = 0; t0 < N; t0 += TILE_PTS) {
       const int tn = min(TILE_PTS, N - t0);
       __syncthreads();
@@ -130,93 +140,44 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
         const float cz = s_tile[c * 3 + 2];
 #pragma unroll
         for (int qi = 0; qi < QB; ++qi) {
-          if (s_sub[qi] != 0u && s_sub[qi] != 0xfffffffeu) continue;
+          if (!(active >> qi & 1u)) continue;
           const float dx = cx - s_q[qi][0];
           const float dy = cy - s_q[qi][1];
           const float dz = cz - s_q[qi][2];
           const float d = dx * dx + dy * dy + dz * dz;
-          const unsigned bits = dist_bits(d);
-          if ((bits >> 23) == s_thr[qi])
-            atomicAdd(&s_hist[qi][(bits >> 15) & 0xff], 1u);
-        }
-      }
-    }
-    __syncthreads();
-    if (threadIdx.x < QB && s_sub[threadIdx.x] == 0u) {
-      const int qi = threadIdx.x;
-      unsigned cum = 0, T2 = 255;
-      const unsigned need = (unsigned)s_need[qi];
-      for (int bin = 0; bin < 256; ++bin) {
-        const unsigned nxt = cum + s_hist[qi][bin];
-        if (nxt >= need) {
-          T2 = bin;
-          break;
-        }
-        cum = nxt;
-      }
-      s_sub[qi] = T2;
-      s_need[qi] = (int)(need - cum);
-    }
-    __syncthreads();
-  }
-
-  // ---- pass 2: collect
-  for (int t0 = 0; t0 < N; t0 += TILE_PTS) {
-    const int tn = min(TILE_PTS, N - t0);
-    __syncthreads();
-    for (int i = threadIdx.x; i < tn * 3; i += KNN_THREADS)
-      s_tile[i] = cloud[(long)t0 * 3 + i];
-    __syncthreads();
-    for (int c = threadIdx.x; c < tn; c += KNN_THREADS) {
-      const float cx = s_tile[c * 3 + 0];
-      const float cy = s_tile[c * 3 + 1];
-      const float cz = s_tile[c * 3 + 2];
-#pragma unroll
-      for (int qi = 0; qi < QB; ++qi) {
-        const int q = q0 + qi;
-        if (q >= N) continue;
-        const float dx = cx - s_q[qi][0];
-        const float dy = cy - s_q[qi][1];
-        const float dz = cz - s_q[qi][2];
-        const float d = dx * dx + dy * dy + dz * dz;
-        const unsigned bits = dist_bits(d);
-        const unsigned bin = bits >> 23;
-        const unsigned T = s_thr[qi];
-        bool accept, boundary;
-        if (s_sub[qi] == 0xffffffffu) {
-          accept = bin < T;
-          boundary = bin == T;
-        } else {  // refined: threshold cut inside bin T at sub-bin T2
-          const unsigned sub = (bits >> 15) & 0xff;
-          accept = bin < T || (bin == T && sub < s_sub[qi]);
-          boundary = bin == T && sub == s_sub[qi];
-        }
-        if (accept) {
-          const unsigned slot = atomicAdd(&s_acc[qi], 1u);
-          out_idx[((long)b * N + q) * k + slot] = t0 + c;
-        } else if (boundary) {
-          const unsigned p = atomicAdd(&s_bcnt[qi], 1u);
-          if (p < CAP) {
-            s_bd[qi][p] = d;
-            s_bi[qi][p] = t0 + c;
+          if (d <= s_tau[qi]) {
+            const unsigned p = atomicAdd(&s_cnt[qi], 1u);
+            if (p < CAP) {
+              s_bd[qi][p] = d;
+              s_bi[qi][p] = t0 + c;
+            }
           }
         }
       }
     }
+    __syncthreads();
+    if (threadIdx.x < QB) {
+      const int qi = threadIdx.x;
+      if (active >> qi & 1u) {
+        if (s_cnt[qi] >= (unsigned)k || q0 + qi >= N) {
+          atomicAnd(&s_active, ~(1u << qi));
+        } else {
+          // underflow: widen (4x distance ~ 2x radius) and re-sweep
+          s_tau[qi] = s_tau[qi] * 4.f + 1e-30f;
+          s_cnt[qi] = 0;
+        }
+      }
+    }
+    __syncthreads();
   }
-  __syncthreads();
 
-  // ---- final: wave argmin rounds over each query's boundary buffer
-  // waves take queries round-robin (4 waves, 8 queries)
+  // ---- final: exact k smallest from each query's buffer
   const int lane = lane_id();
   for (int qi = wave_id(); qi < QB; qi += KNN_THREADS / WAVE) {
     const int q = q0 + qi;
     if (q >= N) continue;
-    const int L = (int)min(s_bcnt[qi], (unsigned)CAP);
-    int need = s_need[qi];
-    if (need > L) need = L;  // degenerate overflow: ties beyond CAP dropped
-    int base = (int)s_acc[qi];
-    // lanes own entries lane, lane+64, ... (CAP/WAVE = 4 slots max)
+    const int L = (int)min(s_cnt[qi], (unsigned)CAP);
+    int take = k < L ? k : L;
     float dv[CAP / WAVE];
     int iv[CAP / WAVE];
 #pragma unroll
@@ -226,7 +187,7 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
       iv[s] = p < L ? s_bi[qi][p] : 0x7fffffff;
     }
     int written = 0;
-    for (int r = 0; r < need; ++r) {
+    for (int r = 0; r < take; ++r) {
       float best = INFINITY;
       int bslot = 0;
 #pragma unroll
@@ -241,17 +202,15 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
       wave_argmin(bv, bidx);
       if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
         const int s = bidx / WAVE;
-        out_idx[((long)b * N + q) * k + base + r] = iv[s];
-#pragma unroll
-        for (int ss = 0; ss < CAP / WAVE; ++ss)
-          if (ss == s) dv[ss] = INFINITY;
+        out_idx[((long)b * N + q) * k + r] = iv[s];
+        dv[s] = INFINITY;
       }
       if (bidx != 0x7fffffff) ++written;
     }
-    // degenerate overflow (ties beyond CAP): pad remaining slots with the
-    // query itself (a valid neighbour; edge features become zero)
+    // degenerate clouds (ties beyond CAP / still-short buffer): pad with
+    // the query itself -- a valid neighbour with zero edge features
     if (lane == 0)
-      for (int r = base + written; r < k; ++r)
+      for (int r = written; r < k; ++r)
         out_idx[((long)b * N + q) * k + r] = q;
   }
 }
