@@ -642,6 +642,43 @@ torch::Tensor pv_corr_fused_bwd(torch::Tensor g_vox, torch::Tensor g_knn,
   return gcorr;
 }
 
+// Fused bf16 MFMA correlation GEMM + streaming top-K truncation
+// (reference corr.py:95-99 + corr.py:37): f1t (B,N,C), f2t (B,M,C) bf16
+// point-major -> top-K values (B,N,K) fp32 + indices (B,N,K) i32; the
+// (N, M) correlation matrix is never materialised.  Top-K SET is
+// unordered (all consumers are order-invariant).
+void launch_corr_topk(const void*, const void*, float*, float*, int*, float*,
+                      int*, int*, int, int, int, int, int, float,
+                      hipStream_t);
+
+std::vector<torch::Tensor> corr_topk(torch::Tensor f1t, torch::Tensor f2t,
+                                     int64_t K) {
+  TORCH_CHECK(f1t.is_cuda() && f1t.is_contiguous() && f1t.dim() == 3 &&
+              f1t.scalar_type() == torch::kBFloat16, "f1t must be (B,N,C) bf16");
+  TORCH_CHECK(f2t.is_cuda() && f2t.is_contiguous() && f2t.dim() == 3 &&
+              f2t.scalar_type() == torch::kBFloat16, "f2t must be (B,M,C) bf16");
+  const int B = f1t.size(0), N = f1t.size(1), C = f1t.size(2);
+  const int M = f2t.size(1);
+  TORCH_CHECK(f2t.size(0) == B && f2t.size(2) == C, "f1t/f2t mismatch");
+  TORCH_CHECK(C % 32 == 0 && C <= 256, "need C % 32 == 0, C <= 256");
+  TORCH_CHECK(K >= 1 && K <= M && K <= 1024, "need 1 <= K <= min(M, 1024)");
+  auto fopt = f1t.options().dtype(torch::kFloat32);
+  auto iopt = f1t.options().dtype(torch::kInt32);
+  const long R = (long)B * N;
+  auto thr = torch::empty({R, 2}, fopt);
+  auto out_v = torch::empty({B, (long)N, K}, fopt);
+  auto out_i = torch::empty({B, (long)N, K}, iopt);
+  auto band_v = torch::empty({R, 704}, fopt);
+  auto band_i = torch::empty({R, 704}, iopt);
+  auto cnt = zeros_fast({R, 2}, iopt);
+  launch_corr_topk(f1t.data_ptr(), f2t.data_ptr(), thr.data_ptr<float>(),
+                   out_v.data_ptr<float>(), out_i.data_ptr<int>(),
+                   band_v.data_ptr<float>(), band_i.data_ptr<int>(),
+                   cnt.data_ptr<int>(), B, N, M, C, (int)K,
+                   1.0f / sqrtf((float)C), stream());
+  return {out_v, out_i};
+}
+
 // vals (R, M) fp32 -> top-K largest per row: values (R, K), idx (R, K)
 std::vector<torch::Tensor> topk_rows(torch::Tensor vals, int64_t K) {
   check_f32(vals, "vals");
@@ -779,6 +816,7 @@ std::vector<torch::Tensor> seq_loss_bwd(std::vector<torch::Tensor> flows,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_rows", &topk_rows);
+  m.def("corr_topk", &corr_topk);
   m.def("seq_loss_fwd", &seq_loss_fwd);
   m.def("seq_loss_bwd", &seq_loss_bwd);
   m.def("gru_zr_fwd", &gru_zr_fwd);

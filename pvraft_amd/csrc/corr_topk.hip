@@ -1,0 +1,322 @@
+// K2+K3 fused: all-pair correlation GEMM with streaming top-K truncation
+// (reference model/corr.py:95-99 computes the full B x N x M matrix and
+// model/corr.py:37 topk's it; the round-1 GPU path still materialised
+// fp32 row-chunks for a separate selection kernel).
+//
+// Here the (N x M) matrix NEVER exists.  bf16 MFMA GEMM
+// (v_mfma_f32_16x16x32_bf16, fp32 accumulate, * 1/sqrt(C)) with a
+// sample-threshold streaming selection:
+//
+//   kernel A (sample): each query row computes its correlation against a
+//     1-in-ST column sample (<= 1024 columns) and extracts two exact
+//     sample order statistics by wave argmax rounds:
+//       P_hi = kp_hi-th largest sample value  (E[#full > P_hi] ~ K - 3sd)
+//       P_lo = kp_lo-th largest sample value  (E[#full > P_lo] ~ K + 3sd)
+//     For ST == 1 the sample is the full row and P_hi = P_lo = the exact
+//     K-th value (no estimation error).
+//   kernel B (sweep): full GEMM sweep; values > P_hi are emitted straight
+//     into the output slots (guaranteed top-K members, E ~ K - 3sd of
+//     them); values in [P_lo, P_hi] land in a per-row band workspace
+//     (E ~ 6sd entries).  Everything else is rejected with no atomics.
+//     Sample and sweep values are BITWISE equal (identical MFMA fragment
+//     and k-block order), so the threshold semantics are exact.
+//   kernel C (select): a wave per row extracts the remaining
+//     K - count(>P_hi) largest entries from the band.
+//
+//   Failure tails: count(>P_hi) > K needs a ~10-sigma sample deviation
+//     (never in practice; extra arrivals are dropped, top-K set then
+//     approximate).  Band overflow/underflow only happens on massive
+//     value ties (degenerate inputs), where any K of the tied values is a
+//     correct top-K SET -- remaining slots pad from the band head,
+//     matching the round-1 kernel's degenerate semantics.  The top-K SET
+//     is unordered (reference sorted=True is a torch.topk detail; every
+//     consumer is order-invariant).
+//
+// Inputs are point-major (B, N, C)/(B, M, C) bf16 (the LDS-tiled
+// transpose of the encoder's (B, C, N) maps); C % 32 == 0, C <= 256.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define CT_THREADS 256
+#define CT_SAMP 1024   // max sample columns per row
+#define CT_CAP 704     // band capacity per row
+#define CT_RWA 16      // rows per workgroup, sample kernel
+#define CT_RWB 64      // rows per workgroup, sweep kernel
+#define CT_CPAD 8      // bf16 padding per LDS f1 row
+
+// ------------------------------------------------------------- kernel A
+
+__global__ __launch_bounds__(CT_THREADS) void corr_sample_kernel(
+    const __hip_bfloat16 *__restrict__ f1t,  // (B, N, C)
+    const __hip_bfloat16 *__restrict__ f2t,  // (B, M, C)
+    float *__restrict__ thr,                 // (B, N, 2) {P_hi, P_lo}
+    int N, int M, int C, int st, int ns, int kp_hi, int kp_lo, float scale) {
+  __shared__ __hip_bfloat16 s_f1[CT_RWA][256 + CT_CPAD];
+  __shared__ float s_samp[CT_RWA][CT_SAMP + 1];
+
+  const int b = blockIdx.z;
+  const int n0 = blockIdx.x * CT_RWA;
+  const int lane = lane_id();
+  const int wv = wave_id();
+
+  // stage the 16 query rows (zero-padded past N)
+  for (int i = threadIdx.x; i < CT_RWA * C / 8; i += CT_THREADS) {
+    const int r = i / (C / 8);
+    const int c8 = (i % (C / 8)) * 8;
+    bf16x8 v = (bf16x8)(__bf16)0.0f;
+    if (n0 + r < N)
+      v = *(const bf16x8 *)(f1t + ((long)b * N + n0 + r) * C + c8);
+    *(bf16x8 *)&s_f1[r][c8] = v;
+  }
+  __syncthreads();
+
+  const int frow = lane & 15;
+  const int koff = (lane >> 4) * 8;
+  for (int cf = wv; cf * 16 < ns; cf += CT_THREADS / WAVE) {
+    const int j = cf * 16 + frow;            // sample index (B-frag row)
+    const long m = (long)j * st;
+    const bool ok = j < ns && m < M;
+    const __hip_bfloat16 *f2r = f2t + ((long)b * M + (ok ? m : 0)) * C;
+    f32x4 acc = (f32x4)(0.f);
+    for (int kb = 0; kb < C; kb += 32) {
+      const bf16x8 afrag = *(const bf16x8 *)&s_f1[frow][kb + koff];
+      bf16x8 bfrag = (bf16x8)(__bf16)0.0f;
+      if (ok) bfrag = *(const bf16x8 *)(f2r + kb + koff);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0, 0, 0);
+    }
+    // D map: col = lane&15 (sample j), row = (lane>>4)*4 + e (query)
+    const int jj = cf * 16 + (lane & 15);
+    if (jj < ns) {
+      const bool valid = (long)jj * st < M;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const int r = (lane >> 4) * 4 + e;
+        s_samp[r][jj] = valid ? acc[e] * scale : -INFINITY;
+      }
+    }
+  }
+  __syncthreads();
+
+  // per-row exact sample order statistics (argmax extraction rounds)
+  for (int r = wv * (CT_RWA / (CT_THREADS / WAVE));
+       r < (wv + 1) * (CT_RWA / (CT_THREADS / WAVE)); ++r) {
+    const int n = n0 + r;
+    if (n >= N) continue;
+    float dv[CT_SAMP / WAVE];
+#pragma unroll
+    for (int s = 0; s < CT_SAMP / WAVE; ++s) {
+      const int p = lane + s * WAVE;
+      dv[s] = p < ns ? -s_samp[r][p] : INFINITY;  // argmin of negated
+    }
+    float phi = INFINITY, plo = INFINITY;
+    for (int t = 0; t < kp_lo; ++t) {
+      float best = INFINITY;
+      int bslot = 0;
+#pragma unroll
+      for (int s = 0; s < CT_SAMP / WAVE; ++s)
+        if (dv[s] < best) {
+          best = dv[s];
+          bslot = s;
+        }
+      int bidx = lane + bslot * WAVE;
+      if (best == INFINITY) bidx = 0x7fffffff;
+      float bv = best;
+      wave_argmin(bv, bidx);
+      if (t == kp_hi - 1) phi = bv;
+      if (t == kp_lo - 1) plo = bv;
+      if (bidx != 0x7fffffff && (bidx % WAVE) == lane) dv[bidx / WAVE] = INFINITY;
+    }
+    if (lane == 0) {
+      thr[((long)b * N + n) * 2 + 0] = -phi;
+      thr[((long)b * N + n) * 2 + 1] = -plo;
+    }
+  }
+}
+
+// ------------------------------------------------------------- kernel B
+
+__global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
+    const __hip_bfloat16 *__restrict__ f1t, const __hip_bfloat16 *__restrict__ f2t,
+    const float *__restrict__ thr,
+    float *__restrict__ out_v,   // (B, N, K)
+    int *__restrict__ out_i,     // (B, N, K)
+    float *__restrict__ band_v,  // (B*N, CAP)
+    int *__restrict__ band_i,    // (B*N, CAP)
+    int *__restrict__ cnt,       // (B*N, 2) pre-zeroed {n_hi, n_band}
+    int N, int M, int C, int K, float scale) {
+  __shared__ __hip_bfloat16 s_f1[CT_RWB][256 + CT_CPAD];
+  __shared__ float s_phi[CT_RWB], s_plo[CT_RWB];
+  __shared__ unsigned s_hi[CT_RWB], s_bd[CT_RWB];
+
+  const int b = blockIdx.z;
+  const int n0 = blockIdx.x * CT_RWB;
+  const int lane = lane_id();
+  const int wv = wave_id();
+
+  for (int i = threadIdx.x; i < CT_RWB * C / 8; i += CT_THREADS) {
+    const int r = i / (C / 8);
+    const int c8 = (i % (C / 8)) * 8;
+    bf16x8 v = (bf16x8)(__bf16)0.0f;
+    if (n0 + r < N)
+      v = *(const bf16x8 *)(f1t + ((long)b * N + n0 + r) * C + c8);
+    *(bf16x8 *)&s_f1[r][c8] = v;
+  }
+  for (int r = threadIdx.x; r < CT_RWB; r += CT_THREADS) {
+    const int n = n0 + r;
+    s_phi[r] = n < N ? thr[((long)b * N + n) * 2 + 0] : INFINITY;
+    s_plo[r] = n < N ? thr[((long)b * N + n) * 2 + 1] : INFINITY;
+    s_hi[r] = 0;
+    s_bd[r] = 0;
+  }
+  __syncthreads();
+
+  const int frow = lane & 15;
+  const int koff = (lane >> 4) * 8;
+  const int nfrag = (M + 15) / 16;
+  for (int cf = wv; cf < nfrag; cf += CT_THREADS / WAVE) {
+    const int j = cf * 16 + frow;  // column (B-frag row)
+    const bool ok = j < M;
+    const __hip_bfloat16 *f2r = f2t + ((long)b * M + (ok ? j : 0)) * C;
+    f32x4 acc[CT_RWB / 16];
+#pragma unroll
+    for (int rf = 0; rf < CT_RWB / 16; ++rf) acc[rf] = (f32x4)(0.f);
+    for (int kb = 0; kb < C; kb += 32) {
+      bf16x8 bfrag = (bf16x8)(__bf16)0.0f;
+      if (ok) bfrag = *(const bf16x8 *)(f2r + kb + koff);
+#pragma unroll
+      for (int rf = 0; rf < CT_RWB / 16; ++rf) {
+        const bf16x8 afrag = *(const bf16x8 *)&s_f1[rf * 16 + frow][kb + koff];
+        acc[rf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[rf], 0, 0, 0);
+      }
+    }
+    const int jj = cf * 16 + (lane & 15);
+    if (jj < M) {
+#pragma unroll
+      for (int rf = 0; rf < CT_RWB / 16; ++rf)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const int r = rf * 16 + (lane >> 4) * 4 + e;
+          const long n = n0 + r;
+          if (n >= N) continue;
+          const float v = acc[rf][e] * scale;
+          if (v > s_phi[r]) {
+            const unsigned slot = atomicAdd(&s_hi[r], 1u);
+            if (slot < (unsigned)K) {
+              out_v[((long)b * N + n) * K + slot] = v;
+              out_i[((long)b * N + n) * K + slot] = jj;
+            }
+          } else if (v >= s_plo[r]) {
+            const unsigned p = atomicAdd(&s_bd[r], 1u);
+            if (p < CT_CAP) {
+              band_v[((long)b * N + n) * CT_CAP + p] = v;
+              band_i[((long)b * N + n) * CT_CAP + p] = jj;
+            }
+          }
+        }
+    }
+  }
+  __syncthreads();
+  for (int r = threadIdx.x; r < CT_RWB; r += CT_THREADS) {
+    const long n = n0 + r;
+    if (n < N) {
+      cnt[((long)b * N + n) * 2 + 0] = (int)min(s_hi[r], (unsigned)K);
+      cnt[((long)b * N + n) * 2 + 1] = (int)min(s_bd[r], (unsigned)CT_CAP);
+    }
+  }
+}
+
+// ------------------------------------------------------------- kernel C
+
+__global__ __launch_bounds__(CT_THREADS) void corr_band_select_kernel(
+    const float *__restrict__ band_v, const int *__restrict__ band_i,
+    const int *__restrict__ cnt, float *__restrict__ out_v,
+    int *__restrict__ out_i, long R, int K) {
+  const int lane = lane_id();
+  const long row = (long)blockIdx.x * (CT_THREADS / WAVE) + wave_id();
+  if (row >= R) return;
+  const int c_hi = cnt[row * 2 + 0];
+  const int L = cnt[row * 2 + 1];
+  int need = K - c_hi;
+  if (need <= 0) return;
+  int take = need < L ? need : L;
+
+  float dv[CT_CAP / WAVE];
+  int iv[CT_CAP / WAVE];
+#pragma unroll
+  for (int s = 0; s < CT_CAP / WAVE; ++s) {
+    const int p = lane + s * WAVE;
+    dv[s] = p < L ? -band_v[row * CT_CAP + p] : INFINITY;
+    iv[s] = p < L ? band_i[row * CT_CAP + p] : 0;
+  }
+  int written = 0;
+  for (int t = 0; t < take; ++t) {
+    float best = INFINITY;
+    int bslot = 0;
+#pragma unroll
+    for (int s = 0; s < CT_CAP / WAVE; ++s)
+      if (dv[s] < best) {
+        best = dv[s];
+        bslot = s;
+      }
+    int bidx = lane + bslot * WAVE;
+    if (best == INFINITY) bidx = 0x7fffffff;
+    float bv = best;
+    wave_argmin(bv, bidx);
+    if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
+      const int s = bidx / WAVE;
+      out_v[row * K + c_hi + t] = -dv[s];
+      out_i[row * K + c_hi + t] = iv[s];
+      dv[s] = INFINITY;
+    }
+    if (bidx != 0x7fffffff) ++written;
+  }
+  // degenerate underflow / tie-overflow: pad from the band head
+  if (lane == 0)
+    for (int t = c_hi + written; t < K; ++t) {
+      out_v[row * K + t] = L > 0 ? band_v[row * CT_CAP] : -INFINITY;
+      out_i[row * K + t] = L > 0 ? band_i[row * CT_CAP] : 0;
+    }
+}
+
+// ------------------------------------------------------------- launcher
+
+void launch_corr_topk(const void *f1t, const void *f2t, float *thr,
+                      float *out_v, int *out_i, float *band_v, int *band_i,
+                      int *cnt, int B, int N, int M, int C, int K,
+                      float scale, hipStream_t stream) {
+  const int st = M > CT_SAMP ? (M + CT_SAMP - 1) / CT_SAMP : 1;
+  const int ns = (M + st - 1) / st;
+  int kp_hi, kp_lo;
+  if (st == 1) {
+    kp_hi = kp_lo = K;
+  } else {
+    // st*r +- 3*st*sqrt(r) = K  (3-sigma margins on the full-count mean)
+    const double q = (double)K / st;
+    double x = (-3.0 + sqrt(9.0 + 4.0 * q)) / 2.0;
+    kp_hi = (int)(x * x);
+    x = (3.0 + sqrt(9.0 + 4.0 * q)) / 2.0;
+    kp_lo = (int)(x * x) + 1;
+    if (kp_hi < 1) kp_hi = 1;
+    if (kp_lo <= kp_hi) kp_lo = kp_hi + 1;
+    if (kp_lo > ns) kp_lo = ns;
+    if (kp_hi > kp_lo) kp_hi = kp_lo;
+  }
+  const dim3 ga((N + CT_RWA - 1) / CT_RWA, 1, B);
+  hipLaunchKernelGGL(corr_sample_kernel, ga, dim3(CT_THREADS), 0, stream,
+                     (const __hip_bfloat16 *)f1t, (const __hip_bfloat16 *)f2t,
+                     thr, N, M, C, st, ns, kp_hi, kp_lo, scale);
+  const dim3 gb((N + CT_RWB - 1) / CT_RWB, 1, B);
+  hipLaunchKernelGGL(corr_sweep_kernel, gb, dim3(CT_THREADS), 0, stream,
+                     (const __hip_bfloat16 *)f1t, (const __hip_bfloat16 *)f2t,
+                     thr, out_v, out_i, band_v, band_i, cnt, N, M, C, K,
+                     scale);
+  const long R = (long)B * N;
+  const int wpb = CT_THREADS / WAVE;
+  hipLaunchKernelGGL(corr_band_select_kernel,
+                     dim3((unsigned)((R + wpb - 1) / wpb)), dim3(CT_THREADS),
+                     0, stream, band_v, band_i, cnt, out_v, out_i, R, K);
+}

@@ -388,21 +388,47 @@ class _CorrTruncate(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, fmap1: Tensor, fmap2: Tensor, xyz2: Tensor, truncate_k: int):
-        # keep the GEMM + topk in fp32 even under bf16 autocast (selection
-        # quality and value precision of the correlation field follow the
-        # reference, corr.py:95-99); autocast would otherwise hijack the bmm
+        # accumulation stays fp32 everywhere (selection quality follows the
+        # reference, corr.py:95-99): the fused MFMA kernel accumulates fp32
+        # from bf16 operands; the fp32 fallback disables autocast so the
+        # bmm is a true fp32 GEMM
         with torch.no_grad(), torch.autocast("cuda", enabled=False):
             M = fmap2.shape[2]
-            if fmap1.is_cuda and M <= 8192 and _load_ext() is not None:
-                corr, idx, txyz = _CorrTruncate._gpu_forward(fmap1, fmap2, xyz2, truncate_k)
+            if (
+                fmap1.is_cuda
+                and fmap1.dtype == torch.bfloat16
+                and fmap2.dtype == torch.bfloat16
+                and fmap1.shape[1] % 32 == 0
+                and fmap1.shape[1] <= 256
+                and truncate_k <= 1024
+                and _load_ext() is not None
+            ):
+                # fused MFMA GEMM + streaming top-K: the (N, M) matrix is
+                # never materialised (csrc/corr_topk.hip)
+                K = min(truncate_k, M)
+                f1t = _EXT.batched_transpose(fmap1.contiguous())
+                f2t = _EXT.batched_transpose(fmap2.contiguous())
+                corr, idx32 = _EXT.corr_topk(f1t, f2t, K)
+                idx = idx32.long()
+                B, _, N = fmap1.shape
+                txyz = xyz2.gather(
+                    1, idx.reshape(B, N * K).unsqueeze(-1).expand(B, N * K, 3)
+                ).view(B, N, K, 3)
+            elif fmap1.is_cuda and M <= 8192 and _load_ext() is not None:
+                corr, idx, txyz = _CorrTruncate._gpu_forward(
+                    fmap1.float(), fmap2.float(), xyz2, truncate_k
+                )
             else:
                 corr, idx, txyz = reference.corr_truncate(
-                    fmap1, fmap2, xyz2, truncate_k, chunk=_CorrTruncate.CHUNK
+                    fmap1.float(), fmap2.float(), xyz2, truncate_k,
+                    chunk=_CorrTruncate.CHUNK,
                 )
         ctx.save_for_backward(fmap1, fmap2, idx)
         # backward GEMMs run bf16 when the surrounding step is bf16 autocast
         # (standard amp gradient precision; fp32 master weights untouched)
-        ctx.bf16_bwd = fmap1.is_cuda and torch.is_autocast_enabled()
+        ctx.bf16_bwd = fmap1.is_cuda and (
+            torch.is_autocast_enabled() or fmap1.dtype == torch.bfloat16
+        )
         return corr, idx, txyz
 
     @staticmethod
@@ -455,14 +481,13 @@ class _CorrTruncate(torch.autograd.Function):
 def corr_truncate(fmap1: Tensor, fmap2: Tensor, xyz2: Tensor, truncate_k: int):
     """(B,C,N),(B,C,M),(B,M,3) -> corr (B,N,K), idx (B,N,K), xyz (B,N,K,3).
 
-    GEMM runs on rocBLAS (chunked bmm + topk; the N x M matrix is tiled);
-    on GPU the custom-backward path avoids retaining it for autograd.
+    bf16 inputs (autocast) take the fused MFMA GEMM + streaming top-K
+    kernel; fp32 falls back to the chunked rocBLAS bmm + topk_rows path.
+    Values are always fp32 accumulate.
     """
-    fmap1 = fmap1.float()
-    fmap2 = fmap2.float()
     if fmap1.is_cuda and os.environ.get("PVRAFT_REF_OPS", "0") != "1":
         return _CorrTruncate.apply(fmap1, fmap2, xyz2, truncate_k)
-    return reference.corr_truncate(fmap1, fmap2, xyz2, truncate_k)
+    return reference.corr_truncate(fmap1.float(), fmap2.float(), xyz2, truncate_k)
 
 
 def voxel_corr(

@@ -25,6 +25,7 @@ SOURCES = [
     "pvraft_amd/csrc/transpose.hip",
     "pvraft_amd/csrc/pv_corr_fused.hip",
     "pvraft_amd/csrc/topk_rows.hip",
+    "pvraft_amd/csrc/corr_topk.hip",
     "pvraft_amd/csrc/gru_gates.hip",
     "pvraft_amd/csrc/seq_loss.hip",
 ]
