@@ -34,6 +34,43 @@ DEV_INLINE void wave_argmin(float &val, int &idx) {
   }
 }
 
+// One extraction round of a wave-cooperative k-smallest selection over
+// per-lane register slots.  CRITICAL: every slot access is statically
+// indexed (unrolled compare/select) -- a single dynamically-indexed
+// access would spill the whole array to scratch (global memory), which
+// measured ~40x slower on these kernels.  Returns the minimum value
+// across all lanes' live slots (INFINITY when exhausted), broadcasts its
+// payload int to every lane, and kills the winning slot.
+template <int S>
+DEV_INLINE float wave_extract_min(float (&dv)[S], const int (&iv)[S],
+                                  int &out_payload) {
+  float best = INFINITY;
+  int bslot = 0;
+#pragma unroll
+  for (int s = 0; s < S; ++s)
+    if (dv[s] < best) {
+      best = dv[s];
+      bslot = s;
+    }
+  int bi = lane_id() | (bslot << 6);
+  if (best == INFINITY) bi = 0x7fffffff;
+  float bv = best;
+  wave_argmin(bv, bi);
+  const int wl = bi & 63;
+  const int ws = bi >> 6;
+  int pay = 0;
+#pragma unroll
+  for (int s = 0; s < S; ++s)
+    if (s == ws) pay = iv[s];  // ws is wave-uniform; static select
+  out_payload = __shfl(pay, wl, WAVE);
+  if (bi != 0x7fffffff && wl == lane_id()) {
+#pragma unroll
+    for (int s = 0; s < S; ++s)
+      if (s == ws) dv[s] = INFINITY;
+  }
+  return bv;
+}
+
 #define HIP_CHECK_LAST()                                          \
   do {                                                            \
     hipError_t e = hipGetLastError();                             \

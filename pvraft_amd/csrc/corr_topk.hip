@@ -9,28 +9,30 @@
 //
 //   kernel A (sample): each query row computes its correlation against a
 //     1-in-ST column sample (<= 1024 columns) and extracts two exact
-//     sample order statistics by wave argmax rounds:
-//       P_hi = kp_hi-th largest sample value  (E[#full > P_hi] ~ K - 3sd)
-//       P_lo = kp_lo-th largest sample value  (E[#full > P_lo] ~ K + 3sd)
-//     For ST == 1 the sample is the full row and P_hi = P_lo = the exact
-//     K-th value (no estimation error).
+//     sample order statistics by wave extraction rounds:
+//       P_hi = kp_hi-th largest sample value, P_lo = kp_lo-th largest.
+//     An order statistic of rank r estimates the full count above it with
+//     sd ~ st*sqrt(r), so the ranks put ~4 sd between the expected direct
+//     count and K on one side and the expected coverage and K on the
+//     other.  For ST == 1 the sample is the full row and the thresholds
+//     are exact.
 //   kernel B (sweep): full GEMM sweep; values > P_hi are emitted straight
-//     into the output slots (guaranteed top-K members, E ~ K - 3sd of
-//     them); values in [P_lo, P_hi] land in a per-row band workspace
-//     (E ~ 6sd entries).  Everything else is rejected with no atomics.
+//     into the output slots; values in [P_lo, P_hi] land in a per-row
+//     band workspace; everything else is rejected with no atomics.
 //     Sample and sweep values are BITWISE equal (identical MFMA fragment
-//     and k-block order), so the threshold semantics are exact.
+//     and k-block order), so the threshold counts are exact -- and the
+//     kernel CHECKS them: rows whose direct count exceeded K, whose
+//     coverage missed K, or whose band overflowed get their thresholds
+//     adjusted from the EXACT counts and are re-swept (<= 4 rounds; only
+//     workgroups owning failed rows re-sweep).  After the loop only
+//     massive value ties (degenerate inputs) can remain short, and any K
+//     of tied values is a correct top-K SET -- remaining slots pad from
+//     the band head, like the round-1 kernel.
 //   kernel C (select): a wave per row extracts the remaining
 //     K - count(>P_hi) largest entries from the band.
 //
-//   Failure tails: count(>P_hi) > K needs a ~10-sigma sample deviation
-//     (never in practice; extra arrivals are dropped, top-K set then
-//     approximate).  Band overflow/underflow only happens on massive
-//     value ties (degenerate inputs), where any K of the tied values is a
-//     correct top-K SET -- remaining slots pad from the band head,
-//     matching the round-1 kernel's degenerate semantics.  The top-K SET
-//     is unordered (reference sorted=True is a torch.topk detail; every
-//     consumer is order-invariant).
+// The top-K SET is unordered (reference sorted=True is a torch.topk
+// detail; every consumer is order-invariant).
 //
 // Inputs are point-major (B, N, C)/(B, M, C) bf16 (the LDS-tiled
 // transpose of the encoder's (B, C, N) maps); C % 32 == 0, C <= 256.
@@ -45,17 +47,19 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define CT_SAMP 1024   // max sample columns per row
 #define CT_CAP 704     // band capacity per row
 #define CT_RWA 16      // rows per workgroup, sample kernel
-#define CT_RWB 64      // rows per workgroup, sweep kernel
+#define CT_RWB 32      // rows per workgroup, sweep kernel
 #define CT_CPAD 8      // bf16 padding per LDS f1 row
 
 // ------------------------------------------------------------- kernel A
 
+template <int KBN>  // C / 32
 __global__ __launch_bounds__(CT_THREADS) void corr_sample_kernel(
     const __hip_bfloat16 *__restrict__ f1t,  // (B, N, C)
     const __hip_bfloat16 *__restrict__ f2t,  // (B, M, C)
     float *__restrict__ thr,                 // (B, N, 2) {P_hi, P_lo}
-    int N, int M, int C, int st, int ns, int kp_hi, int kp_lo, float scale) {
-  __shared__ __hip_bfloat16 s_f1[CT_RWA][256 + CT_CPAD];
+    int N, int M, int st, int ns, int kp_hi, int kp_lo, float scale) {
+  constexpr int C = KBN * 32;
+  __shared__ __hip_bfloat16 s_f1[CT_RWA][C + CT_CPAD];
   __shared__ float s_samp[CT_RWA][CT_SAMP + 1];
 
   const int b = blockIdx.z;
@@ -63,7 +67,6 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sample_kernel(
   const int lane = lane_id();
   const int wv = wave_id();
 
-  // stage the 16 query rows (zero-padded past N)
   for (int i = threadIdx.x; i < CT_RWA * C / 8; i += CT_THREADS) {
     const int r = i / (C / 8);
     const int c8 = (i % (C / 8)) * 8;
@@ -77,58 +80,48 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sample_kernel(
   const int frow = lane & 15;
   const int koff = (lane >> 4) * 8;
   for (int cf = wv; cf * 16 < ns; cf += CT_THREADS / WAVE) {
-    const int j = cf * 16 + frow;            // sample index (B-frag row)
+    const int j = cf * 16 + frow;  // sample index (B-frag row)
     const long m = (long)j * st;
     const bool ok = j < ns && m < M;
     const __hip_bfloat16 *f2r = f2t + ((long)b * M + (ok ? m : 0)) * C;
+    bf16x8 bf[KBN];
+#pragma unroll
+    for (int kb = 0; kb < KBN; ++kb)
+      bf[kb] = ok ? *(const bf16x8 *)(f2r + kb * 32 + koff) : (bf16x8)(__bf16)0.0f;
     f32x4 acc = (f32x4)(0.f);
-    for (int kb = 0; kb < C; kb += 32) {
-      const bf16x8 afrag = *(const bf16x8 *)&s_f1[frow][kb + koff];
-      bf16x8 bfrag = (bf16x8)(__bf16)0.0f;
-      if (ok) bfrag = *(const bf16x8 *)(f2r + kb + koff);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0, 0, 0);
+#pragma unroll
+    for (int kb = 0; kb < KBN; ++kb) {
+      const bf16x8 afrag = *(const bf16x8 *)&s_f1[frow][kb * 32 + koff];
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bf[kb], acc, 0, 0, 0);
     }
-    // D map: col = lane&15 (sample j), row = (lane>>4)*4 + e (query)
     const int jj = cf * 16 + (lane & 15);
     if (jj < ns) {
       const bool valid = (long)jj * st < M;
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        const int r = (lane >> 4) * 4 + e;
-        s_samp[r][jj] = valid ? acc[e] * scale : -INFINITY;
-      }
+      for (int e = 0; e < 4; ++e)
+        s_samp[(lane >> 4) * 4 + e][jj] = valid ? acc[e] * scale : -INFINITY;
     }
   }
   __syncthreads();
 
-  // per-row exact sample order statistics (argmax extraction rounds)
-  for (int r = wv * (CT_RWA / (CT_THREADS / WAVE));
-       r < (wv + 1) * (CT_RWA / (CT_THREADS / WAVE)); ++r) {
+  // per-row exact sample order statistics via extraction rounds
+  for (int r = wv; r < CT_RWA; r += CT_THREADS / WAVE) {
     const int n = n0 + r;
     if (n >= N) continue;
     float dv[CT_SAMP / WAVE];
+    int iv[CT_SAMP / WAVE];
 #pragma unroll
     for (int s = 0; s < CT_SAMP / WAVE; ++s) {
       const int p = lane + s * WAVE;
-      dv[s] = p < ns ? -s_samp[r][p] : INFINITY;  // argmin of negated
+      dv[s] = p < ns ? -s_samp[r][p] : INFINITY;  // min of negated = max
+      iv[s] = 0;
     }
     float phi = INFINITY, plo = INFINITY;
     for (int t = 0; t < kp_lo; ++t) {
-      float best = INFINITY;
-      int bslot = 0;
-#pragma unroll
-      for (int s = 0; s < CT_SAMP / WAVE; ++s)
-        if (dv[s] < best) {
-          best = dv[s];
-          bslot = s;
-        }
-      int bidx = lane + bslot * WAVE;
-      if (best == INFINITY) bidx = 0x7fffffff;
-      float bv = best;
-      wave_argmin(bv, bidx);
+      int pay;
+      const float bv = wave_extract_min(dv, iv, pay);
       if (t == kp_hi - 1) phi = bv;
       if (t == kp_lo - 1) plo = bv;
-      if (bidx != 0x7fffffff && (bidx % WAVE) == lane) dv[bidx / WAVE] = INFINITY;
     }
     if (lane == 0) {
       thr[((long)b * N + n) * 2 + 0] = -phi;
@@ -139,6 +132,7 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sample_kernel(
 
 // ------------------------------------------------------------- kernel B
 
+template <int KBN>
 __global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
     const __hip_bfloat16 *__restrict__ f1t, const __hip_bfloat16 *__restrict__ f2t,
     const float *__restrict__ thr,
@@ -146,11 +140,13 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
     int *__restrict__ out_i,     // (B, N, K)
     float *__restrict__ band_v,  // (B*N, CAP)
     int *__restrict__ band_i,    // (B*N, CAP)
-    int *__restrict__ cnt,       // (B*N, 2) pre-zeroed {n_hi, n_band}
-    int N, int M, int C, int K, float scale) {
-  __shared__ __hip_bfloat16 s_f1[CT_RWB][256 + CT_CPAD];
+    int *__restrict__ cnt,       // (B*N, 2) {n_hi (<=K), n_band (<=CAP)}
+    int N, int M, int K, float scale) {
+  constexpr int C = KBN * 32;
+  __shared__ __hip_bfloat16 s_f1[CT_RWB][C + CT_CPAD];
   __shared__ float s_phi[CT_RWB], s_plo[CT_RWB];
   __shared__ unsigned s_hi[CT_RWB], s_bd[CT_RWB];
+  __shared__ unsigned s_active;
 
   const int b = blockIdx.z;
   const int n0 = blockIdx.x * CT_RWB;
@@ -172,54 +168,94 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
     s_hi[r] = 0;
     s_bd[r] = 0;
   }
+  if (threadIdx.x == 0) s_active = (unsigned)-1;
   __syncthreads();
 
   const int frow = lane & 15;
   const int koff = (lane >> 4) * 8;
   const int nfrag = (M + 15) / 16;
-  for (int cf = wv; cf < nfrag; cf += CT_THREADS / WAVE) {
-    const int j = cf * 16 + frow;  // column (B-frag row)
-    const bool ok = j < M;
-    const __hip_bfloat16 *f2r = f2t + ((long)b * M + (ok ? j : 0)) * C;
-    f32x4 acc[CT_RWB / 16];
+  for (int iter = 0; iter < 4 && s_active != 0u; ++iter) {
+    for (int cf = wv; cf < nfrag; cf += CT_THREADS / WAVE) {
+      const int j = cf * 16 + frow;  // column (B-frag row)
+      const bool ok = j < M;
+      const __hip_bfloat16 *f2r = f2t + ((long)b * M + (ok ? j : 0)) * C;
+      bf16x8 bf[KBN];
 #pragma unroll
-    for (int rf = 0; rf < CT_RWB / 16; ++rf) acc[rf] = (f32x4)(0.f);
-    for (int kb = 0; kb < C; kb += 32) {
-      bf16x8 bfrag = (bf16x8)(__bf16)0.0f;
-      if (ok) bfrag = *(const bf16x8 *)(f2r + kb + koff);
+      for (int kb = 0; kb < KBN; ++kb)
+        bf[kb] = ok ? *(const bf16x8 *)(f2r + kb * 32 + koff)
+                    : (bf16x8)(__bf16)0.0f;
+      f32x4 acc[CT_RWB / 16];
 #pragma unroll
-      for (int rf = 0; rf < CT_RWB / 16; ++rf) {
-        const bf16x8 afrag = *(const bf16x8 *)&s_f1[rf * 16 + frow][kb + koff];
-        acc[rf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[rf], 0, 0, 0);
-      }
-    }
-    const int jj = cf * 16 + (lane & 15);
-    if (jj < M) {
+      for (int rf = 0; rf < CT_RWB / 16; ++rf) acc[rf] = (f32x4)(0.f);
 #pragma unroll
-      for (int rf = 0; rf < CT_RWB / 16; ++rf)
+      for (int kb = 0; kb < KBN; ++kb)
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const int r = rf * 16 + (lane >> 4) * 4 + e;
-          const long n = n0 + r;
-          if (n >= N) continue;
-          const float v = acc[rf][e] * scale;
-          if (v > s_phi[r]) {
-            const unsigned slot = atomicAdd(&s_hi[r], 1u);
-            if (slot < (unsigned)K) {
-              out_v[((long)b * N + n) * K + slot] = v;
-              out_i[((long)b * N + n) * K + slot] = jj;
-            }
-          } else if (v >= s_plo[r]) {
-            const unsigned p = atomicAdd(&s_bd[r], 1u);
-            if (p < CT_CAP) {
-              band_v[((long)b * N + n) * CT_CAP + p] = v;
-              band_i[((long)b * N + n) * CT_CAP + p] = jj;
+        for (int rf = 0; rf < CT_RWB / 16; ++rf) {
+          const bf16x8 afrag =
+              *(const bf16x8 *)&s_f1[rf * 16 + frow][kb * 32 + koff];
+          acc[rf] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bf[kb], acc[rf], 0, 0, 0);
+        }
+      const int jj = cf * 16 + (lane & 15);
+      if (jj < M) {
+#pragma unroll
+        for (int rf = 0; rf < CT_RWB / 16; ++rf)
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const int r = rf * 16 + (lane >> 4) * 4 + e;
+            const long n = n0 + r;
+            if (n >= N || !(s_active >> r & 1u)) continue;
+            const float v = acc[rf][e] * scale;
+            if (v > s_phi[r]) {
+              const unsigned slot = atomicAdd(&s_hi[r], 1u);
+              if (slot < (unsigned)K) {
+                out_v[((long)b * N + n) * K + slot] = v;
+                out_i[((long)b * N + n) * K + slot] = jj;
+              }
+            } else if (v >= s_plo[r]) {
+              const unsigned p = atomicAdd(&s_bd[r], 1u);
+              if (p < CT_CAP) {
+                band_v[((long)b * N + n) * CT_CAP + p] = v;
+                band_i[((long)b * N + n) * CT_CAP + p] = jj;
+              }
             }
           }
-        }
+      }
     }
+    __syncthreads();
+    // exact-count check: adjust failed rows' thresholds and re-sweep them
+    for (int r = threadIdx.x; r < CT_RWB; r += CT_THREADS) {
+      if (!(s_active >> r & 1u)) continue;
+      if (n0 + r >= N) {
+        atomicAnd(&s_active, ~(1u << r));
+        continue;
+      }
+      const unsigned chi = s_hi[r], cbd = s_bd[r];
+      const bool over_hi = chi > (unsigned)K;
+      const bool under = chi + min(cbd, (unsigned)CT_CAP) < (unsigned)K;
+      const bool over_bd = cbd > (unsigned)CT_CAP && chi + cbd >= (unsigned)K;
+      if (!over_hi && !under && !over_bd) {
+        atomicAnd(&s_active, ~(1u << r));
+        continue;
+      }
+      const float span = s_phi[r] - s_plo[r] + 1e-6f * (1.f + fabsf(s_phi[r]));
+      if (over_hi) {
+        // too many strict accepts: push P_hi above them, band takes over
+        s_plo[r] = s_phi[r];
+        s_phi[r] = s_phi[r] + span * 4.f;
+      } else if (under) {
+        // coverage short of K: band region grows downward
+        s_phi[r] = s_plo[r];
+        s_plo[r] = s_plo[r] - span * 4.f;
+      } else {
+        // band overflowed its buffer: bisect the band from below
+        s_plo[r] = s_plo[r] + (s_phi[r] - s_plo[r]) * 0.5f;
+      }
+      s_hi[r] = 0;
+      s_bd[r] = 0;
+    }
+    __syncthreads();
   }
-  __syncthreads();
   for (int r = threadIdx.x; r < CT_RWB; r += CT_THREADS) {
     const long n = n0 + r;
     if (n < N) {
@@ -240,9 +276,9 @@ __global__ __launch_bounds__(CT_THREADS) void corr_band_select_kernel(
   if (row >= R) return;
   const int c_hi = cnt[row * 2 + 0];
   const int L = cnt[row * 2 + 1];
-  int need = K - c_hi;
+  const int need = K - c_hi;
   if (need <= 0) return;
-  int take = need < L ? need : L;
+  const int take = need < L ? need : L;
 
   float dv[CT_CAP / WAVE];
   int iv[CT_CAP / WAVE];
@@ -252,31 +288,17 @@ __global__ __launch_bounds__(CT_THREADS) void corr_band_select_kernel(
     dv[s] = p < L ? -band_v[row * CT_CAP + p] : INFINITY;
     iv[s] = p < L ? band_i[row * CT_CAP + p] : 0;
   }
-  int written = 0;
   for (int t = 0; t < take; ++t) {
-    float best = INFINITY;
-    int bslot = 0;
-#pragma unroll
-    for (int s = 0; s < CT_CAP / WAVE; ++s)
-      if (dv[s] < best) {
-        best = dv[s];
-        bslot = s;
-      }
-    int bidx = lane + bslot * WAVE;
-    if (best == INFINITY) bidx = 0x7fffffff;
-    float bv = best;
-    wave_argmin(bv, bidx);
-    if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
-      const int s = bidx / WAVE;
-      out_v[row * K + c_hi + t] = -dv[s];
-      out_i[row * K + c_hi + t] = iv[s];
-      dv[s] = INFINITY;
+    int pay;
+    const float bv = wave_extract_min(dv, iv, pay);
+    if (lane == 0) {
+      out_v[row * K + c_hi + t] = -bv;
+      out_i[row * K + c_hi + t] = pay;
     }
-    if (bidx != 0x7fffffff) ++written;
   }
-  // degenerate underflow / tie-overflow: pad from the band head
+  // degenerate tie overflow: pad from the band head
   if (lane == 0)
-    for (int t = c_hi + written; t < K; ++t) {
+    for (int t = c_hi + take; t < K; ++t) {
       out_v[row * K + t] = L > 0 ? band_v[row * CT_CAP] : -INFINITY;
       out_i[row * K + t] = L > 0 ? band_i[row * CT_CAP] : 0;
     }
@@ -294,11 +316,12 @@ void launch_corr_topk(const void *f1t, const void *f2t, float *thr,
   if (st == 1) {
     kp_hi = kp_lo = K;
   } else {
-    // st*r +- 3*st*sqrt(r) = K  (3-sigma margins on the full-count mean)
+    // rank-r sample order statistic -> full-count estimate sd ~ st*sqrt(r):
+    // put ~4 sd between E[direct] and K, and E[coverage] and K
     const double q = (double)K / st;
-    double x = (-3.0 + sqrt(9.0 + 4.0 * q)) / 2.0;
+    double x = (-4.0 + sqrt(16.0 + 4.0 * q)) / 2.0;
     kp_hi = (int)(x * x);
-    x = (3.0 + sqrt(9.0 + 4.0 * q)) / 2.0;
+    x = (4.0 + sqrt(16.0 + 4.0 * q)) / 2.0;
     kp_lo = (int)(x * x) + 1;
     if (kp_hi < 1) kp_hi = 1;
     if (kp_lo <= kp_hi) kp_lo = kp_hi + 1;
@@ -306,16 +329,31 @@ void launch_corr_topk(const void *f1t, const void *f2t, float *thr,
     if (kp_hi > kp_lo) kp_hi = kp_lo;
   }
   const dim3 ga((N + CT_RWA - 1) / CT_RWA, 1, B);
-  hipLaunchKernelGGL(corr_sample_kernel, ga, dim3(CT_THREADS), 0, stream,
-                     (const __hip_bfloat16 *)f1t, (const __hip_bfloat16 *)f2t,
-                     thr, N, M, C, st, ns, kp_hi, kp_lo, scale);
   const dim3 gb((N + CT_RWB - 1) / CT_RWB, 1, B);
-  hipLaunchKernelGGL(corr_sweep_kernel, gb, dim3(CT_THREADS), 0, stream,
-                     (const __hip_bfloat16 *)f1t, (const __hip_bfloat16 *)f2t,
-                     thr, out_v, out_i, band_v, band_i, cnt, N, M, C, K,
-                     scale);
   const long R = (long)B * N;
   const int wpb = CT_THREADS / WAVE;
+#define CT_DISPATCH(KBN)                                                       \
+  do {                                                                         \
+    hipLaunchKernelGGL((corr_sample_kernel<KBN>), ga, dim3(CT_THREADS), 0,     \
+                       stream, (const __hip_bfloat16 *)f1t,                    \
+                       (const __hip_bfloat16 *)f2t, thr, N, M, st, ns, kp_hi,  \
+                       kp_lo, scale);                                          \
+    hipLaunchKernelGGL((corr_sweep_kernel<KBN>), gb, dim3(CT_THREADS), 0,      \
+                       stream, (const __hip_bfloat16 *)f1t,                    \
+                       (const __hip_bfloat16 *)f2t, thr, out_v, out_i, band_v, \
+                       band_i, cnt, N, M, K, scale);                           \
+  } while (0)
+  switch (C / 32) {
+    case 1: CT_DISPATCH(1); break;
+    case 2: CT_DISPATCH(2); break;
+    case 3: CT_DISPATCH(3); break;
+    case 4: CT_DISPATCH(4); break;
+    case 5: CT_DISPATCH(5); break;
+    case 6: CT_DISPATCH(6); break;
+    case 7: CT_DISPATCH(7); break;
+    default: CT_DISPATCH(8); break;
+  }
+#undef CT_DISPATCH
   hipLaunchKernelGGL(corr_band_select_kernel,
                      dim3((unsigned)((R + wpb - 1) / wpb)), dim3(CT_THREADS),
                      0, stream, band_v, band_i, cnt, out_v, out_i, R, K);

@@ -57,8 +57,11 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
 
   const int st = N > SAMP ? (N + SAMP - 1) / SAMP : 1;
   const int ns = (N + st - 1) / st;
-  int kp = 2 * k;  // k' such that E[collected] ~ 2k
-  if (st > 1) kp = (2 * k + st - 1) / st;
+  // E[collected] target: 2k exact for st==1; 4k for sampled thresholds (a
+  // rank-r order statistic estimates the full count with sd ~ st*sqrt(r),
+  // so aim well clear of both k and CAP; the exact-count retry loop below
+  // repairs the tails)
+  int kp = st == 1 ? 2 * k : (4 * k + st - 1) / st;
   if (kp > ns) kp = ns;
   if (kp < 1) kp = 1;
 
@@ -93,40 +96,31 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
   }
   __syncthreads();
 
-  // ---- exact k'-th smallest of each query's sample: wave argmin rounds
+  // ---- exact k'-th smallest of each query's sample: wave extraction
   {
     const int lane = lane_id();
     for (int qi = wave_id(); qi < QB; qi += KNN_THREADS / WAVE) {
       float dv[SAMP / WAVE];
+      int iv[SAMP / WAVE];
 #pragma unroll
       for (int s = 0; s < SAMP / WAVE; ++s) {
         const int p = lane + s * WAVE;
         dv[s] = p < ns ? s_samp[qi][p] : INFINITY;
+        iv[s] = 0;
       }
       float tau = INFINITY;
       for (int r = 0; r < kp; ++r) {
-        float best = INFINITY;
-        int bslot = 0;
-#pragma unroll
-        for (int s = 0; s < SAMP / WAVE; ++s)
-          if (dv[s] < best) {
-            best = dv[s];
-            bslot = s;
-          }
-        int bidx = lane + bslot * WAVE;
-        if (best == INFINITY) bidx = 0x7fffffff;
-        float bv = best;
-        wave_argmin(bv, bidx);
-        tau = bv;
-        if (bidx != 0x7fffffff && (bidx % WAVE) == lane) dv[bidx / WAVE] = INFINITY;
+        int pay;
+        tau = wave_extract_min(dv, iv, pay);
       }
       if (lane == 0) s_tau[qi] = tau;
     }
   }
   __syncthreads();
 
-  // ---- pass B: collect d <= tau (re-sweep with widened tau on underflow)
-  for (int iter = 0; iter < 3 && s_active != 0; ++iter) {
+  // ---- pass B: collect d <= tau; retarget tau from the EXACT count on
+  // underflow (< k) or buffer overflow (> CAP) and re-sweep
+  for (int iter = 0; iter < 4 && s_active != 0; ++iter) {
     const unsigned active = s_active;
     for (int t0 = 0; t0 < N; t0 += TILE_PTS) {
       const int tn = min(TILE_PTS, N - t0);
@@ -159,11 +153,15 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
     if (threadIdx.x < QB) {
       const int qi = threadIdx.x;
       if (active >> qi & 1u) {
-        if (s_cnt[qi] >= (unsigned)k || q0 + qi >= N) {
+        const unsigned c = s_cnt[qi];
+        if ((c >= (unsigned)k && c <= (unsigned)CAP) || q0 + qi >= N ||
+            (iter == 3)) {
           atomicAnd(&s_active, ~(1u << qi));
         } else {
-          // underflow: widen (4x distance ~ 2x radius) and re-sweep
-          s_tau[qi] = s_tau[qi] * 4.f + 1e-30f;
+          // retarget E[count] to 4k using count ~ tau^1.5 (3-D volume);
+          // +epsilon escapes tau == 0 (coincident points)
+          const float ratio = (4.f * k) / (float)max(c, 2u);
+          s_tau[qi] = s_tau[qi] * __powf(ratio, 0.6667f) + 1e-30f;
           s_cnt[qi] = 0;
         }
       }
@@ -177,40 +175,24 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
     const int q = q0 + qi;
     if (q >= N) continue;
     const int L = (int)min(s_cnt[qi], (unsigned)CAP);
-    int take = k < L ? k : L;
+    const int take = k < L ? k : L;
     float dv[CAP / WAVE];
     int iv[CAP / WAVE];
 #pragma unroll
     for (int s = 0; s < CAP / WAVE; ++s) {
       const int p = lane + s * WAVE;
       dv[s] = p < L ? s_bd[qi][p] : INFINITY;
-      iv[s] = p < L ? s_bi[qi][p] : 0x7fffffff;
+      iv[s] = p < L ? s_bi[qi][p] : q;
     }
-    int written = 0;
     for (int r = 0; r < take; ++r) {
-      float best = INFINITY;
-      int bslot = 0;
-#pragma unroll
-      for (int s = 0; s < CAP / WAVE; ++s)
-        if (dv[s] < best) {
-          best = dv[s];
-          bslot = s;
-        }
-      int bidx = lane + bslot * WAVE;
-      if (best == INFINITY) bidx = 0x7fffffff;
-      float bv = best;
-      wave_argmin(bv, bidx);
-      if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
-        const int s = bidx / WAVE;
-        out_idx[((long)b * N + q) * k + r] = iv[s];
-        dv[s] = INFINITY;
-      }
-      if (bidx != 0x7fffffff) ++written;
+      int pay;
+      wave_extract_min(dv, iv, pay);
+      if (lane == 0) out_idx[((long)b * N + q) * k + r] = pay;
     }
     // degenerate clouds (ties beyond CAP / still-short buffer): pad with
     // the query itself -- a valid neighbour with zero edge features
     if (lane == 0)
-      for (int r = written; r < k; ++r)
+      for (int r = take; r < k; ++r)
         out_idx[((long)b * N + q) * k + r] = q;
   }
 }

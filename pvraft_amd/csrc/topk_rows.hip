@@ -125,6 +125,8 @@ __global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
   __syncthreads();
 
   // wave 0 selects the remaining `need` LARGEST from the boundary buffer
+  // (wave_extract_min keeps every slot access statically indexed -- a
+  // dynamic index would spill the arrays to scratch)
   if (wave_id() == 0) {
     const int lane = lane_id();
     const int L = (int)min(s_bcnt, (unsigned)TK_CAP);
@@ -137,35 +139,19 @@ __global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
     for (int s_ = 0; s_ < TK_CAP / WAVE; ++s_) {
       const int p = lane + s_ * WAVE;
       dv[s_] = p < L ? -s_bv[p] : INFINITY;  // argmin on negated = argmax
-      iv[s_] = p < L ? s_bi[p] : 0x7fffffff;
+      iv[s_] = p < L ? s_bi[p] : 0;
     }
-    int written = 0;
     for (int r = 0; r < take; ++r) {
-      float best = INFINITY;
-      int bslot = 0;
-#pragma unroll
-      for (int s_ = 0; s_ < TK_CAP / WAVE; ++s_)
-        if (dv[s_] < best) {
-          best = dv[s_];
-          bslot = s_;
-        }
-      int bidx = lane + bslot * WAVE;
-      if (best == INFINITY) bidx = 0x7fffffff;
-      float bv = best;
-      wave_argmin(bv, bidx);
-      if (bidx != 0x7fffffff && (bidx % WAVE) == lane) {
-        const int s_ = bidx / WAVE;
-        out_v[(long)row * K + base + r] = -dv[s_];
-        out_i[(long)row * K + base + r] = iv[s_];
-#pragma unroll
-        for (int ss = 0; ss < TK_CAP / WAVE; ++ss)
-          if (ss == s_) dv[ss] = INFINITY;
+      int pay;
+      const float bv = wave_extract_min(dv, iv, pay);
+      if (lane == 0) {
+        out_v[(long)row * K + base + r] = -bv;
+        out_i[(long)row * K + base + r] = pay;
       }
-      if (bidx != 0x7fffffff) ++written;
     }
     // degenerate ties beyond CAP: pad with the first boundary entry
     if (lane == 0)
-      for (int r = base + written; r < K; ++r) {
+      for (int r = base + take; r < K; ++r) {
         out_v[(long)row * K + r] = L > 0 ? s_bv[0] : -INFINITY;
         out_i[(long)row * K + r] = L > 0 ? s_bi[0] : 0;
       }
