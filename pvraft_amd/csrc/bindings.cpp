@@ -500,6 +500,63 @@ std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
   return {dw, dbias};
 }
 
+// Fused MFMA 1x1-conv forward: y = act(sum_i W_i @ x_i + bias + addend)
+// in one launch (csrc/pw_fwd.hip)
+void launch_pw_fwd(const void**, const void**, const int*, int, const float*,
+                   const void*, long, int, void*, int, int, long, int,
+                   hipStream_t);
+
+torch::Tensor pw_fwd(std::vector<torch::Tensor> ws,
+                     std::vector<torch::Tensor> xs,
+                     c10::optional<torch::Tensor> bias,
+                     c10::optional<torch::Tensor> addend, int64_t act) {
+  const int n = (int)ws.size();
+  TORCH_CHECK(n >= 1 && n <= 4 && (int)xs.size() == n, "1..4 parts");
+  const int Co = ws[0].size(0);
+  const int B = xs[0].size(0);
+  const long S = xs[0].size(2);
+  const void* wp[4];
+  const void* xp[4];
+  int cis[4];
+  for (int i = 0; i < n; ++i) {
+    TORCH_CHECK(ws[i].is_cuda() && ws[i].is_contiguous() && ws[i].dim() == 2 &&
+                ws[i].scalar_type() == torch::kBFloat16, "w must be (Co,Ci) bf16");
+    TORCH_CHECK(xs[i].is_cuda() && xs[i].is_contiguous() && xs[i].dim() == 3 &&
+                xs[i].scalar_type() == torch::kBFloat16, "x must be (B,Ci,S) bf16");
+    TORCH_CHECK(ws[i].size(0) == Co && xs[i].size(0) == B && xs[i].size(2) == S);
+    TORCH_CHECK(ws[i].size(1) == xs[i].size(1), "w/x channel mismatch");
+    TORCH_CHECK(ws[i].size(1) <= 224, "pw_fwd: Ci <= 224");
+    wp[i] = ws[i].data_ptr();
+    xp[i] = xs[i].data_ptr();
+    cis[i] = (int)ws[i].size(1);
+  }
+  TORCH_CHECK(Co <= 256, "pw_fwd: Co <= 256");
+  const float* bptr = nullptr;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->scalar_type() == torch::kFloat32 && bias->is_contiguous() &&
+                bias->numel() == Co, "bias must be fp32 (Co)");
+    bptr = bias->data_ptr<float>();
+  }
+  const void* aptr = nullptr;
+  long abstride = 0;
+  int afp32 = 0;
+  if (addend.has_value() && addend->defined()) {
+    auto& a = *addend;
+    TORCH_CHECK(a.is_cuda() && a.dim() == 3 && a.size(0) == B &&
+                a.size(1) == Co && a.size(2) == S, "addend must be (B,Co,S) view");
+    TORCH_CHECK(a.stride(2) == 1 && a.stride(1) == S,
+                "addend rows must be contiguous");
+    afp32 = a.scalar_type() == torch::kFloat32;
+    TORCH_CHECK(afp32 || a.scalar_type() == torch::kBFloat16, "addend fp32/bf16");
+    aptr = a.data_ptr();
+    abstride = a.stride(0);
+  }
+  auto y = torch::empty({B, (long)Co, S}, xs[0].options());
+  launch_pw_fwd(wp, xp, cis, n, bptr, aptr, abstride, afp32, y.data_ptr(), B,
+                Co, S, (int)act, stream());
+  return y;
+}
+
 // Batched deferred weight gradients: many (dy, x) -> dW problems in ONE
 // kernel launch, each ACCUMULATED into its weight's existing fp32 grad
 // buffer (no per-call zero-fill, no autograd accumulate-add).  ~180 such
@@ -827,6 +884,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pv_corr_fused_bwd", &pv_corr_fused_bwd);
   m.def("pw_wgrad", &pw_wgrad, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("schunks") = 0, pybind11::arg("with_bias") = false);
   m.def("pw_wgrad_batched", &pw_wgrad_batched);
+  m.def("pw_fwd", &pw_fwd);
   m.def("batched_transpose", &batched_transpose);
   m.def("group_norm_act_fwd", &group_norm_act_fwd);
   m.def("group_norm_act_bwd", &group_norm_act_bwd);

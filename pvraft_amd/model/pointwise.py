@@ -238,6 +238,128 @@ class _PwMatmul(torch.autograd.Function):
         return dw, dx, dbias, None, None
 
 
+class _PwFused(torch.autograd.Function):
+    """One MFMA launch for y = act(sum_i W_i @ x_i + bias + addend)
+    (csrc/pw_fwd.hip) -- the whole concat-free conv stack piece with its
+    epilogue, instead of per-part hipBLASLt GEMMs + add/bias/ReLU
+    elementwise launches.  Weight/bias grads ride the deferred batched
+    wgrad path when armed."""
+
+    @staticmethod
+    def forward(ctx, bias, addend, conf, *wx):
+        from pvraft_amd import _C
+
+        act_id, targets, p, compute_dt = conf
+        ws = [
+            w if compute_dt is None else _cast_cached(w, compute_dt)
+            for w in wx[:p]
+        ]
+        xs = list(wx[p:])
+        y = _C.pw_fwd(ws, xs, bias, addend, act_id)
+        ctx.save_for_backward(*ws, *xs, y)
+        ctx.conf = (act_id, targets, p)
+        ctx.has_bias = bias is not None
+        ctx.has_addend = addend is not None
+        ctx.w_dtypes = tuple(w.dtype for w in wx[:p])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: Tensor):
+        act_id, targets, p = ctx.conf
+        saved = ctx.saved_tensors
+        ws, xs, y = saved[:p], saved[p : 2 * p], saved[2 * p]
+        dy = dy.contiguous()
+        dpre = (
+            torch.ops.aten.threshold_backward(dy, y, 0) if act_id == 1 else dy
+        )
+        d_addend = dpre if ctx.has_addend else None
+        dxs = [
+            torch.bmm(w.t().unsqueeze(0).expand(dpre.shape[0], -1, -1), dpre)
+            if ctx.needs_input_grad[3 + p + i]
+            else None
+            for i, w in enumerate(ws)
+        ]
+        dws = [None] * p
+        d_bias = None
+        need_bias = ctx.has_bias and ctx.needs_input_grad[0]
+        if _DEFER.on and targets is not None:
+            for i in range(p):
+                if ctx.needs_input_grad[3 + i]:
+                    bt = None
+                    if i == 0 and need_bias and targets[p] is not None:
+                        bt = _grad_buffer(targets[p]).view(-1)
+                        need_bias = False
+                    _DEFER.jobs.append(
+                        (dpre, xs[i], _grad_buffer(targets[i]).view(-1), bt)
+                    )
+        else:
+            for i in range(p):
+                if ctx.needs_input_grad[3 + i]:
+                    try:
+                        from pvraft_amd import _C
+
+                        dw, db = _C.pw_wgrad(dpre, xs[i], 0, i == 0 and need_bias)
+                        if i == 0 and need_bias:
+                            d_bias = db
+                            need_bias = False
+                    except (ImportError, AttributeError):
+                        dw = torch.einsum("bos,bis->oi", dpre, xs[i])
+                    wd = ctx.w_dtypes[i]
+                    dws[i] = dw.to(wd) if dw.dtype != wd else dw
+        if need_bias:
+            d_bias = dpre.sum(dim=(0, 2)).float()
+        return (d_bias, d_addend, None, *dws, *dxs)
+
+
+def pw_fused(parts, bias=None, bias_target=None, addend=None, act="none") -> Tensor:
+    """act(sum_i w_i @ x_i + bias + addend) for 1x1-conv stacks.
+
+    parts: list of (w (Co, Ci), x (B, Ci, S), wgrad_target_or_None).
+    GPU + bf16 compute: one fused MFMA kernel.  Otherwise: composed
+    pw_matmul calls (+ add/activation), same semantics.
+    """
+    import torch.nn.functional as F
+
+    x0 = parts[0][1]
+    dt = None
+    if x0.is_cuda and torch.is_autocast_enabled():
+        dt = torch.get_autocast_dtype("cuda")
+    fused_ok = (
+        dt == torch.bfloat16
+        and len(parts) <= 4
+        and parts[0][0].shape[0] <= 256
+        and all(w.shape[1] <= 224 for w, _x, _t in parts)
+        and os.environ.get("PVRAFT_NO_PWFWD", "0") != "1"
+    )
+    if fused_ok:
+        try:
+            from pvraft_amd import _C  # noqa: F401
+        except ImportError:
+            fused_ok = False
+    if fused_ok:
+        act_id = {"none": 0, "relu": 1}[act]
+        p = len(parts)
+        targets = None
+        if all(t is not None for _w, _x, t in parts):
+            targets = tuple(t for _w, _x, t in parts) + (bias_target,)
+        wx = tuple(w for w, _x, _t in parts) + tuple(
+            x.to(dt).contiguous() for _w, x, _t in parts
+        )
+        b = bias.float() if bias is not None and bias.dtype != torch.float32 else bias
+        return _PwFused.apply(b, addend, (act_id, targets, p, dt), *wx)
+    out = None
+    for i, (w, x, tgt) in enumerate(parts):
+        b = bias if i == 0 else None
+        t = (tgt, bias_target) if tgt is not None else None
+        term = pw_matmul(w, x, b, targets=t)
+        out = term if out is None else out + term
+    if addend is not None:
+        out = out + addend
+    if act == "relu":
+        out = F.relu(out)
+    return out
+
+
 def pw_matmul(weight: Tensor, x: Tensor, bias: Tensor = None, targets=None) -> Tensor:
     """``targets`` = (w_target, bias_target): the fp32 accumulation targets
     for the DEFERRED wgrad path -- the underlying Parameters when ``weight``
@@ -258,20 +380,23 @@ def pw_matmul(weight: Tensor, x: Tensor, bias: Tensor = None, targets=None) -> T
 
 
 class PwConv1d(nn.Conv1d):
-    """nn.Conv1d(k=1) with a matmul forward."""
+    """nn.Conv1d(k=1): fused MFMA GEMM+bias on the bf16 GPU path, bmm
+    otherwise."""
 
     def forward(self, x: Tensor) -> Tensor:
-        # x (B, Cin, N) -> (B, Cout, N); bias fused into the GEMM Function
-        # (its gradient comes from the wgrad kernel, not a slow ATen reduce)
-        return pw_matmul(self.weight.squeeze(-1), x, self.bias,
-                         targets=(self.weight, self.bias))
+        return pw_fused(
+            [(self.weight.squeeze(-1), x, self.weight)],
+            bias=self.bias, bias_target=self.bias,
+        )
 
 
 class PwConv2d(nn.Conv2d):
-    """nn.Conv2d(k=1) with a matmul forward over flattened spatial dims."""
+    """nn.Conv2d(k=1) over flattened spatial dims (same GEMM)."""
 
     def forward(self, x: Tensor) -> Tensor:
         B, C, H, W = x.shape
-        y = pw_matmul(self.weight.view(self.out_channels, C), x.reshape(B, C, H * W),
-                      self.bias, targets=(self.weight, self.bias))
+        y = pw_fused(
+            [(self.weight.view(self.out_channels, C), x.reshape(B, C, H * W), self.weight)],
+            bias=self.bias, bias_target=self.bias,
+        )
         return y.view(B, self.out_channels, H, W)

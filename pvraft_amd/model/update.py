@@ -53,24 +53,25 @@ def _slice_weight(weight: Tensor, sizes):
     return out
 
 
-def _split_mm(weight: Tensor, bias, parts, wparts=None):
-    """sum_i W[:, lo_i:hi_i] @ parts[i] (+ bias): the concat-free 1x1 conv.
+def _split_mm(weight: Tensor, bias, parts, wparts=None, act="none"):
+    """act(sum_i W[:, lo_i:hi_i] @ parts[i] + bias): the concat-free 1x1
+    conv with its activation.
 
     weight: (Co, Ci_total, 1) conv weight; parts: list of (B, Ci_i, N)
     tensors whose channel sizes sum to Ci_total.  ``wparts`` supplies
     pre-sliced contiguous weights (cached once per forward -- the GRU loop
     reuses the same slices for all iterations, which also lets autocast's
-    weight-cast cache hit).
+    weight-cast cache hit).  On GPU the whole sum + bias + activation is
+    one fused MFMA kernel (pointwise.pw_fused).
     """
+    from .pointwise import pw_fused
+
     if wparts is None:
         wparts = _slice_weight(weight, [p.shape[1] for p in parts])
-    out = None
-    for i, ((w_i, tgt), p) in enumerate(zip(wparts, parts)):
-        # bias rides the first partial GEMM (fused fwd add + kernel dbias)
-        b = bias if i == 0 else None
-        term = pw_matmul(w_i, p, b, targets=(tgt, bias) if tgt is not None else None)
-        out = term if out is None else out + term
-    return out
+    return pw_fused(
+        [(w_i, p, tgt) for (w_i, tgt), p in zip(wparts, parts)],
+        bias=bias, bias_target=bias, act=act,
+    )
 
 
 class MotionEncoder(nn.Module):
@@ -87,12 +88,19 @@ class MotionEncoder(nn.Module):
         parts are kept separate here and consumed slice-wise downstream.
         """
         from pvraft_amd import ops
+        from .pointwise import pw_fused
 
         flow_t = ops.transpose_last2(flow)
-        cor = F.relu(self.conv_corr(corr))
-        flo = F.relu(self.conv_flow(flow_t))
+        cor = pw_fused(
+            [(self.conv_corr.weight.squeeze(-1), corr, self.conv_corr.weight)],
+            bias=self.conv_corr.bias, bias_target=self.conv_corr.bias, act="relu",
+        )
+        flo = pw_fused(
+            [(self.conv_flow.weight.squeeze(-1), flow_t, self.conv_flow.weight)],
+            bias=self.conv_flow.bias, bias_target=self.conv_flow.bias, act="relu",
+        )
         wp = wcache.get("motion") if wcache else None
-        out = F.relu(_split_mm(self.conv.weight, self.conv.bias, [cor, flo], wp))
+        out = _split_mm(self.conv.weight, self.conv.bias, [cor, flo], wp, act="relu")
         return out, flow_t
 
 
@@ -170,8 +178,9 @@ class ConvGRU(nn.Module):
     def precompute_inp(self, inp: Tensor, gw) -> Tensor:
         """Contribution of the iteration-constant context features to all
         three gate preactivations -- hoisted out of the GRU loop."""
-        t = gw["t_parts"][0]
-        return pw_matmul(gw["parts"][0], inp, targets=(t, None) if t is not None else None)
+        from .pointwise import pw_fused
+
+        return pw_fused([(gw["parts"][0], inp, gw["t_parts"][0])])
 
     def forward_parts(self, h: Tensor, x_parts, wcache=None, gw=None, pre=None) -> Tensor:
         """Gates from the concat parts (cross-gate fused GEMMs); the gate
@@ -179,24 +188,33 @@ class ConvGRU(nn.Module):
         fused HIP kernels per direction (ops.gru_zr / ops.gru_q)."""
         from pvraft_amd import ops
 
+        from .pointwise import pw_fused
+
         if gw is None:
             gw = self.gate_weights([p.shape[1] for p in x_parts])
         hd = self.convz.out_channels
-
-        def tgt(t):
-            return (t, None) if t is not None else None
-
-        # shared-input contribution for all gates: (B, 3*hd, N)
-        m = pre if pre is not None else pw_matmul(
-            gw["parts"][0], x_parts[0], targets=tgt(gw["t_parts"][0]))
-        for w_i, t, p in zip(gw["parts"][1:], gw["t_parts"][1:], x_parts[1:]):
-            m = m + pw_matmul(w_i, p, targets=tgt(t))
-        zr_t = (gw["t_zr_h"], gw["t_b_zr"]) if gw["t_zr_h"] is not None else None
-        q_t = (gw["t_q_h"], gw["b_q"]) if gw["t_q_h"] is not None else None
-        z, rh = ops.gru_zr(
-            pw_matmul(gw["zr_h"], h, gw["b_zr"], targets=zr_t) + m[:, : 2 * hd], h)
-        return ops.gru_q(
-            pw_matmul(gw["q_h"], rh, gw["b_q"], targets=q_t) + m[:, 2 * hd :], z, h)
+        # shared-input contribution for all gates: (B, 3*hd, N).  The
+        # iteration-varying parts (motion, flow) sum in ONE fused kernel
+        # with the hoisted context contribution as the carried addend.
+        if pre is not None:
+            m = pw_fused(
+                [(w_i, p, t) for w_i, t, p in
+                 zip(gw["parts"][1:], gw["t_parts"][1:], x_parts[1:])],
+                addend=pre,
+            )
+        else:
+            m = pw_fused(
+                [(w_i, p, t) for w_i, t, p in
+                 zip(gw["parts"], gw["t_parts"], x_parts)],
+            )
+        # gate preactivations: hidden-operand GEMM + the matching slice of
+        # m, again one kernel each
+        pre_zr = pw_fused([(gw["zr_h"], h, gw["t_zr_h"])], bias=gw["b_zr"],
+                          bias_target=gw["t_b_zr"], addend=m[:, : 2 * hd])
+        z, rh = ops.gru_zr(pre_zr, h)
+        pre_q = pw_fused([(gw["q_h"], rh, gw["t_q_h"])], bias=gw["b_q"],
+                         bias_target=gw["b_q"], addend=m[:, 2 * hd :])
+        return ops.gru_q(pre_q, z, h)
 
 
 class ConvRNN(nn.Module):
@@ -228,9 +246,10 @@ class FlowHead(nn.Module):
 
         out = self.conv1(x)
         out_set = ops.transpose_last2(self.setconv(ops.transpose_last2(x), graph))
-        # out_conv(cat([out_set, out])) without the cat
+        # out_conv(cat([out_set, out])) without the cat; ReLU fused
         wp = wcache.get("flowhead") if wcache else None
-        mid = F.relu(_split_mm(self.out_conv[0].weight, self.out_conv[0].bias, [out_set, out], wp))
+        mid = _split_mm(self.out_conv[0].weight, self.out_conv[0].bias,
+                        [out_set, out], wp, act="relu")
         return self.out_conv[2](mid)
 
 

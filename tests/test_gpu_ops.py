@@ -529,3 +529,75 @@ def test_corr_truncate_bf16_path_backward():
     corr.sum().backward()
     assert f1.grad is not None and f1.grad.abs().sum() > 0
     assert f2.grad is not None and f2.grad.abs().sum() > 0
+
+
+@pytest.mark.parametrize("parts,Co,S,act,with_bias,with_addend", [
+    (1, 64, 2048, 0, True, False),
+    (2, 61, 1000, 1, True, False),
+    (3, 192, 4096, 0, False, True),
+    (1, 128, 555, 1, False, False),
+])
+def test_pw_fwd_fused_matches_composed(parts, Co, S, act, with_bias, with_addend):
+    """One-launch MFMA conv stack vs composed fp32 GEMMs (bf16-rounded
+    inputs, fp32 accumulate both sides)."""
+    import pvraft_amd._C as _C
+
+    torch.manual_seed(7)
+    B = 2
+    cis = [64, 61, 37][:parts]
+    ws = [torch.randn(Co, ci, device=dev(), dtype=torch.bfloat16) for ci in cis]
+    xs = [torch.randn(B, ci, S, device=dev(), dtype=torch.bfloat16) for ci in cis]
+    bias = torch.randn(Co, device=dev()) if with_bias else None
+    add_full = torch.randn(B, Co + 32, S, device=dev(), dtype=torch.bfloat16)
+    addend = add_full[:, 16:16 + Co] if with_addend else None
+
+    y = _C.pw_fwd(ws, xs, bias, addend, act)
+
+    ref = sum(torch.bmm(w.float().unsqueeze(0).expand(B, -1, -1), x.float())
+              for w, x in zip(ws, xs))
+    if bias is not None:
+        ref = ref + bias.view(1, -1, 1)
+    if addend is not None:
+        ref = ref + addend.float()
+    if act == 1:
+        ref = torch.relu(ref)
+    assert y.shape == (B, Co, S) and y.dtype == torch.bfloat16
+    assert torch.allclose(y.float(), ref, atol=3e-2, rtol=2e-2), (
+        (y.float() - ref).abs().max()
+    )
+
+
+def test_pw_fused_autograd_matches_fallback():
+    """pw_fused under autocast (fused kernel) vs PVRAFT_NO_PWFWD=1
+    (composed bmm path): same forward and same gradients within bf16
+    tolerance."""
+    from pvraft_amd.model.pointwise import pw_fused
+
+    torch.manual_seed(11)
+    B, Co, S = 2, 64, 3000
+    w1 = torch.randn(Co, 64, device=dev(), requires_grad=True)
+    w2 = torch.randn(Co, 61, device=dev(), requires_grad=True)
+    x1 = torch.randn(B, 64, S, device=dev(), requires_grad=True)
+    x2 = torch.randn(B, 61, S, device=dev(), requires_grad=True)
+    bias = torch.randn(Co, device=dev(), requires_grad=True)
+    leaves = [w1, w2, x1, x2, bias]
+    clones = [t.detach().clone().requires_grad_(True) for t in leaves]
+
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = pw_fused([(w1, x1, None), (w2, x2, None)], bias=bias, act="relu")
+    y.float().square().mean().backward()
+
+    os.environ["PVRAFT_NO_PWFWD"] = "1"
+    try:
+        c1, c2, cx1, cx2, cb = clones
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            y2 = pw_fused([(c1, cx1, None), (c2, cx2, None)], bias=cb, act="relu")
+        y2.float().square().mean().backward()
+    finally:
+        os.environ.pop("PVRAFT_NO_PWFWD", None)
+
+    assert torch.allclose(y.float(), y2.float(), atol=5e-2, rtol=2e-2)
+    for a, b in zip(leaves, clones):
+        assert torch.allclose(a.grad, b.grad, atol=5e-2, rtol=5e-2), (
+            (a.grad - b.grad).abs().max()
+        )
