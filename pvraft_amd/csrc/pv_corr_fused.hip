@@ -5,27 +5,26 @@
 // per phase and the intermediate (B,N,K) distance / (B,N,27) scatter
 // tensors never exist.
 //
-// Geometry: one wave per point (block = 256 = 4 points).
-//
-// Voxel binning uses per-wave LDS histograms (ds_add_f32): a first cut
-// kept 27 (sum, count) accumulators per LANE in statically-indexed VGPRs
-// and reduced them with 54 wave butterflies per level -- ~2300 VALU ops
-// per wave per level; the LDS-atomic histogram replaces that with 2
-// LDS atomics per candidate (conflicts serialize inside the LDS pipe,
-// bounded by the cell occupancy) and no cross-lane reduction at all,
-// since the histogram is already wave-wide.
+// Geometry: one wave per point (block = 256 = 4 points).  Each lane owns
+// K/64 (<= 8) candidates in REGISTERS (relative xyz + corr + distance,
+// statically indexed): the round-1 version staged the whole field in LDS
+// and re-read it from every phase, which made the kernel LDS-op bound
+// (~9k LDS operations per point).  Register residency leaves LDS for the
+// things that need cross-lane communication only: the 27-cell voxel
+// accumulators and the per-level counts.
 //
 // forward outputs:
 //   voxel (B, L*27, N): per-level 3^3 mean of corr over quantised offsets;
 //   knn (B, 4, k, N): [corr; rel-xyz] of the k nearest candidates
-//     (wave argmin rounds, ties to the smallest index);
+//     (wave extraction rounds; tie order arbitrary -- all consumers are
+//     order-invariant);
 //   knn_idx (B, N, k) for backward.
 // backward (d corr only; quantisation/selection indices are constants to
 // autograd, reference corr.py:52-62 and coords detached per iteration):
 //   d corr[j] = sum_l [valid_l(j)] g_vox[l*27+cell_l(j)] / cnt_l(cell)
 //             + [j selected at slot t] g_knn[0, t]
 // with the kNN term resolved by an LDS scatter over the candidate row
-// (selected slots are unique within a row) instead of a k-way scan.
+// (selected slots are unique within a row).
 #include <hip/hip_runtime.h>
 #include "common.h"
 
@@ -60,19 +59,22 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
   const float *cand_xyz = xyz + p * K * 3;
   const float *cand_corr = corr + p * K;
 
-  // stage the wave's whole candidate field in LDS ONCE (xyz stored
-  // relative to the query point): the voxel levels, the distance pass and
-  // both emit paths re-read it, and the HBM re-reads (~6 x 8 KB per
-  // point) dominated the kernel (~100 us of a 187 us call)
-  __shared__ float s_cxyz[4][MAXK][3];
-  __shared__ float s_ccor[4][MAXK];
-  for (int j = lane; j < K; j += WAVE) {
-    s_ccor[w][j] = cand_corr[j];
-    s_cxyz[w][j][0] = cand_xyz[j * 3 + 0] - cx;
-    s_cxyz[w][j][1] = cand_xyz[j * 3 + 1] - cy;
-    s_cxyz[w][j][2] = cand_xyz[j * 3 + 2] - cz;
+  // this lane's candidates, register-resident (statically indexed)
+  float rx[MAXC], ry[MAXC], rz[MAXC], rc[MAXC], d[MAXC];
+#pragma unroll
+  for (int t = 0; t < MAXC; ++t) {
+    const int j = lane + t * WAVE;
+    if (j < K) {
+      rx[t] = cand_xyz[j * 3 + 0] - cx;
+      ry[t] = cand_xyz[j * 3 + 1] - cy;
+      rz[t] = cand_xyz[j * 3 + 2] - cz;
+      rc[t] = cand_corr[j];
+      d[t] = rx[t] * rx[t] + ry[t] * ry[t] + rz[t] * rz[t];
+    } else {
+      rx[t] = ry[t] = rz[t] = rc[t] = 0.f;
+      d[t] = INFINITY;
+    }
   }
-  __threadfence_block();
 
   // ---- voxel pyramid: per-wave LDS histogram per level
   for (int l = 0; l < L; ++l) {
@@ -82,15 +84,19 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
       s_cnt[w][lane] = 0.f;
     }
     __threadfence_block();
-    for (int j = lane; j < K; j += WAVE) {
-      const float dx = rintf(s_cxyz[w][j][0] * inv_r);
-      const float dy = rintf(s_cxyz[w][j][1] * inv_r);
-      const float dz = rintf(s_cxyz[w][j][2] * inv_r);
-      if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) && fabsf(dz) <= (R / 2)) {
-        const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
-                         ((int)dz + R / 2);
-        atomicAdd(&s_sum[w][cell], s_ccor[w][j]);
-        atomicAdd(&s_cnt[w][cell], 1.f);
+#pragma unroll
+    for (int t = 0; t < MAXC; ++t) {
+      if (lane + t * WAVE < K) {
+        const float dx = rintf(rx[t] * inv_r);
+        const float dy = rintf(ry[t] * inv_r);
+        const float dz = rintf(rz[t] * inv_r);
+        if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
+            fabsf(dz) <= (R / 2)) {
+          const int cell = ((int)dx + R / 2) * (R * R) +
+                           ((int)dy + R / 2) * R + ((int)dz + R / 2);
+          atomicAdd(&s_sum[w][cell], rc[t]);
+          atomicAdd(&s_cnt[w][cell], 1.f);
+        }
       }
     }
     __threadfence_block();
@@ -101,127 +107,35 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     // order within the wave (LDS ops issue in order from one wave)
   }
 
-  // ---- kNN branch: per-wave histogram select of the k smallest squared
-  // distances (non-negative floats order as unsigned ints).  MSB-first
-  // 8-bit refine rounds narrow a threshold prefix until the boundary bin
-  // is small; candidates strictly below the threshold are emitted
-  // directly, the remainder comes from wave-argmin rounds over the (<= 64
-  // entry) boundary buffer.  Replaces k sequential butterfly-argmin rounds
-  // (~2000 VALU ops per wave at k=32).  Output order is arbitrary -- every
-  // consumer (conv + max-pool over k, index-scatter backward) is
-  // order-invariant; selection matches exact k-NN except on > 64-way ties
-  // in the top 24 distance bits.
-  __shared__ unsigned s_hist[4][256];
-  __shared__ float s_bd[4][MAXKN];
-  __shared__ int s_bj[4][MAXKN];
-  __shared__ unsigned s_st[4][4];  // prefix, need, acc, bcnt
-
-  float d[MAXC];
-#pragma unroll
-  for (int t = 0; t < MAXC; ++t) {
-    const int j = lane + t * WAVE;
-    if (j < K) {
-      const float dx = s_cxyz[w][j][0];
-      const float dy = s_cxyz[w][j][1];
-      const float dz = s_cxyz[w][j][2];
-      d[t] = dx * dx + dy * dy + dz * dz;
-    } else {
-      d[t] = INFINITY;
-    }
-  }
-
-  unsigned prefix = 0;
-  int need = k;
-  int nbytes = 0;
-  for (int round = 0; round < 3; ++round) {
-    for (int e = lane; e < 256; e += WAVE) s_hist[w][e] = 0;
-    __threadfence_block();
-    const int shift = 24 - 8 * round;
-    const unsigned mask = round == 0 ? 0u : (0xFFFFFFFFu << (shift + 8));
-#pragma unroll
-    for (int t = 0; t < MAXC; ++t) {
-      const int j = lane + t * WAVE;
-      if (j < K) {
-        const unsigned key = __float_as_uint(d[t]);
-        if ((key & mask) == prefix)
-          atomicAdd(&s_hist[w][(key >> shift) & 0xFFu], 1u);
-      }
-    }
-    __threadfence_block();
-    if (lane == 0) {
-      unsigned cum = 0;
-      int T = 255;
-      for (int bin = 0; bin < 256; ++bin) {
-        const unsigned nxt = cum + s_hist[w][bin];
-        if (nxt >= (unsigned)need) {
-          T = bin;
-          break;
-        }
-        cum = nxt;
-      }
-      s_st[w][0] = prefix | ((unsigned)T << shift);
-      s_st[w][1] = (unsigned)(need - (int)cum);
-      s_st[w][2] = (s_hist[w][T] <= (unsigned)(MAXKN - 8) || round == 2) ? 1u : 0u;
-    }
-    __threadfence_block();
-    prefix = s_st[w][0];
-    need = (int)s_st[w][1];
-    nbytes = round + 1;
-    if (s_st[w][2]) break;
-  }
-
-  // collect: strictly-below -> direct emit; equal-prefix -> boundary
-  if (lane == 0) {
-    s_st[w][2] = 0;  // accepted count
-    s_st[w][3] = 0;  // boundary count
-  }
-  __threadfence_block();
+  // ---- kNN branch: k extraction rounds over the register distances.
+  // The winning lane owns the candidate's data in registers and emits it
+  // directly (static register selection -- dynamic indexing would spill).
   float *dst = knn + ((long)b * 4) * k * N + n;
   const long ch_stride = (long)k * N;
   int *idst = knn_idx + p * k;
-  const int shc = 32 - 8 * nbytes;
-  const unsigned tp = prefix >> shc;
+  int jd[MAXC];
 #pragma unroll
-  for (int t = 0; t < MAXC; ++t) {
-    const int j = lane + t * WAVE;
-    if (j < K) {
-      const unsigned kp = __float_as_uint(d[t]) >> shc;
-      if (kp < tp) {
-        const int slot = (int)atomicAdd(&s_st[w][2], 1u);
-        dst[(long)slot * N] = s_ccor[w][j];
-        dst[(long)slot * N + ch_stride] = s_cxyz[w][j][0];
-        dst[(long)slot * N + 2 * ch_stride] = s_cxyz[w][j][1];
-        dst[(long)slot * N + 3 * ch_stride] = s_cxyz[w][j][2];
-        idst[slot] = j;
-      } else if (kp == tp) {
-        const unsigned bp = atomicAdd(&s_st[w][3], 1u);
-        if (bp < MAXKN) {
-          s_bd[w][bp] = d[t];
-          s_bj[w][bp] = j;
+  for (int t = 0; t < MAXC; ++t) jd[t] = lane + t * WAVE;
+  for (int r = 0; r < k; ++r) {
+    int j;
+    const float dist = wave_extract_min(d, jd, j);
+    (void)dist;
+    if (lane == (j & 63)) {
+      const int ts = j >> 6;
+      float ex = 0.f, ey = 0.f, ez = 0.f, ec = 0.f;
+#pragma unroll
+      for (int t = 0; t < MAXC; ++t)
+        if (t == ts) {
+          ex = rx[t];
+          ey = ry[t];
+          ez = rz[t];
+          ec = rc[t];
         }
-      }
-    }
-  }
-  __threadfence_block();
-  const int base = (int)s_st[w][2];
-  const int bl = (int)min(s_st[w][3], (unsigned)MAXKN);
-  // need <= k <= MAXKN and the boundary bin holds >= need candidates, so
-  // base + need == k always; one boundary entry per lane
-  float bv = lane < bl ? s_bd[w][lane] : INFINITY;
-  int bj = lane < bl ? s_bj[w][lane] : 0x7fffffff;
-  const int take = need < bl ? need : bl;
-  for (int r = 0; r < take; ++r) {
-    float v = bv;
-    int j = bj;
-    wave_argmin(v, j);
-    if (j != 0x7fffffff && j == bj) {
-      dst[(long)(base + r) * N] = s_ccor[w][j];
-      dst[(long)(base + r) * N + ch_stride] = s_cxyz[w][j][0];
-      dst[(long)(base + r) * N + 2 * ch_stride] = s_cxyz[w][j][1];
-      dst[(long)(base + r) * N + 3 * ch_stride] = s_cxyz[w][j][2];
-      idst[base + r] = j;
-      bv = INFINITY;
-      bj = 0x7fffffff;
+      dst[(long)r * N] = ec;
+      dst[(long)r * N + ch_stride] = ex;
+      dst[(long)r * N + 2 * ch_stride] = ey;
+      dst[(long)r * N + 3 * ch_stride] = ez;
+      idst[r] = j;
     }
   }
 }
@@ -250,13 +164,18 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
   const float cz = coords[p * 3 + 2];
   const float *cand_xyz = xyz + p * K * 3;
 
-  // stage relative candidate xyz in LDS (read by the count histograms and
-  // again by the gradient pass)
-  __shared__ float s_cxyz[4][MAXK][3];
-  for (int j = lane; j < K; j += WAVE) {
-    s_cxyz[w][j][0] = cand_xyz[j * 3 + 0] - cx;
-    s_cxyz[w][j][1] = cand_xyz[j * 3 + 1] - cy;
-    s_cxyz[w][j][2] = cand_xyz[j * 3 + 2] - cz;
+  // register-resident relative candidate xyz (statically indexed)
+  float rx[MAXC], ry[MAXC], rz[MAXC];
+#pragma unroll
+  for (int t = 0; t < MAXC; ++t) {
+    const int j = lane + t * WAVE;
+    if (j < K) {
+      rx[t] = cand_xyz[j * 3 + 0] - cx;
+      ry[t] = cand_xyz[j * 3 + 1] - cy;
+      rz[t] = cand_xyz[j * 3 + 2] - cz;
+    } else {
+      rx[t] = ry[t] = rz[t] = 1e30f;  // lands outside every cell
+    }
   }
 
   // scatter the kNN channel-0 grads onto their candidate slots (unique)
@@ -277,30 +196,34 @@ __global__ __launch_bounds__(256) void pv_corr_fused_bwd_kernel(
           g_vox[((long)b * L * CELLS + l * CELLS + lane) * N + n];
     }
     __threadfence_block();
-    for (int j = lane; j < K; j += WAVE) {
-      const float dx = rintf(s_cxyz[w][j][0] * inv_r);
-      const float dy = rintf(s_cxyz[w][j][1] * inv_r);
-      const float dz = rintf(s_cxyz[w][j][2] * inv_r);
-      if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) && fabsf(dz) <= (R / 2))
-        atomicAdd(&s_cnt[w][l][(int)dx * (R * R) + (int)dy * R + (int)dz +
-                               (R / 2) * (R * R + R + 1)],
-                  1.f);
+#pragma unroll
+    for (int t = 0; t < MAXC; ++t) {
+      if (lane + t * WAVE < K) {
+        const float dx = rintf(rx[t] * inv_r);
+        const float dy = rintf(ry[t] * inv_r);
+        const float dz = rintf(rz[t] * inv_r);
+        if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
+            fabsf(dz) <= (R / 2))
+          atomicAdd(&s_cnt[w][l][((int)dx + R / 2) * (R * R) +
+                                 ((int)dy + R / 2) * R + ((int)dz + R / 2)],
+                    1.f);
+      }
     }
   }
   __threadfence_block();
 
-  for (int j = lane; j < K; j += WAVE) {
-    const float ox = s_cxyz[w][j][0];
-    const float oy = s_cxyz[w][j][1];
-    const float oz = s_cxyz[w][j][2];
+#pragma unroll
+  for (int t = 0; t < MAXC; ++t) {
+    const int j = lane + t * WAVE;
+    if (j >= K) continue;
     float g = s_kg[w][j];
 #pragma unroll
     for (int l = 0; l < MAXL; ++l) {
       if (l < L) {
         const float inv_r = 1.0f / (base_scale * (float)(1 << l));
-        const float dx = rintf(ox * inv_r);
-        const float dy = rintf(oy * inv_r);
-        const float dz = rintf(oz * inv_r);
+        const float dx = rintf(rx[t] * inv_r);
+        const float dy = rintf(ry[t] * inv_r);
+        const float dz = rintf(rz[t] * inv_r);
         if (fabsf(dx) <= (R / 2) && fabsf(dy) <= (R / 2) &&
             fabsf(dz) <= (R / 2)) {
           const int cell = ((int)dx + R / 2) * (R * R) + ((int)dy + R / 2) * R +
