@@ -234,7 +234,9 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
       const bool over_hi = chi > (unsigned)K;
       const bool under = chi + min(cbd, (unsigned)CT_CAP) < (unsigned)K;
       const bool over_bd = cbd > (unsigned)CT_CAP && chi + cbd >= (unsigned)K;
-      if (!over_hi && !under && !over_bd) {
+      if ((!over_hi && !under && !over_bd) || iter == 3) {
+        // done -- or out of retries (massive ties): KEEP the last sweep's
+        // counters/buffers; kernel C pads what is short
         atomicAnd(&s_active, ~(1u << r));
         continue;
       }

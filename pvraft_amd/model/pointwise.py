@@ -17,6 +17,8 @@ interchangeable with the reference.
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 from torch import Tensor
