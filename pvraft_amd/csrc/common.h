@@ -34,6 +34,40 @@ DEV_INLINE void wave_argmin(float &val, int &idx) {
   }
 }
 
+DEV_INLINE int wave_sum_int(int v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+// order-preserving float -> unsigned key: a < b  <=>  fkey(a) < fkey(b)
+DEV_INLINE unsigned fkey(float v) {
+  unsigned b = __float_as_uint(v);
+  return (b & 0x80000000u) ? ~b : (b | 0x80000000u);
+}
+DEV_INLINE float fkey_inv(unsigned k) {
+  return __uint_as_float((k & 0x80000000u) ? (k ^ 0x80000000u) : ~k);
+}
+
+// Exact r-th LARGEST key among the wave's register slots, by 32-step
+// radix descent with ballot counting: O(32 * S) wave-ops, vs the
+// O(r * S) of repeated argmax extraction (which measured ~50x slower at
+// r ~ 200).  Returns K* such that count(key >= K*) >= r and
+// count(key > K*) < r; K* is the key of an actual element.
+template <int S>
+DEV_INLINE unsigned wave_rank_key(const unsigned (&kv)[S], int r) {
+  unsigned cur = 0;
+#pragma unroll
+  for (int bit = 31; bit >= 0; --bit) {
+    const unsigned cand = cur | (1u << bit);
+    int c = 0;
+#pragma unroll
+    for (int s = 0; s < S; ++s) c += kv[s] >= cand;
+    if (wave_sum_int(c) >= r) cur = cand;
+  }
+  return cur;
+}
+
 // One extraction round of a wave-cooperative k-smallest selection over
 // per-lane register slots.  CRITICAL: every slot access is statically
 // indexed (unrolled compare/select) -- a single dynamically-indexed

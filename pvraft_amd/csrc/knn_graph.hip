@@ -96,24 +96,19 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
   }
   __syncthreads();
 
-  // ---- exact k'-th smallest of each query's sample: wave extraction
+  // ---- exact k'-th smallest of each query's sample via radix rank
+  // select on inverted keys (larger key = smaller distance)
   {
     const int lane = lane_id();
     for (int qi = wave_id(); qi < QB; qi += KNN_THREADS / WAVE) {
-      float dv[SAMP / WAVE];
-      int iv[SAMP / WAVE];
+      unsigned kv[SAMP / WAVE];
 #pragma unroll
       for (int s = 0; s < SAMP / WAVE; ++s) {
         const int p = lane + s * WAVE;
-        dv[s] = p < ns ? s_samp[qi][p] : INFINITY;
-        iv[s] = 0;
+        kv[s] = p < ns ? ~fkey(s_samp[qi][p]) : 0u;
       }
-      float tau = INFINITY;
-      for (int r = 0; r < kp; ++r) {
-        int pay;
-        tau = wave_extract_min(dv, iv, pay);
-      }
-      if (lane == 0) s_tau[qi] = tau;
+      const unsigned kt = wave_rank_key(kv, kp);
+      if (lane == 0) s_tau[qi] = fkey_inv(~kt);
     }
   }
   __syncthreads();

@@ -107,36 +107,39 @@ __global__ __launch_bounds__(256) void pv_corr_fused_fwd_kernel(
     // order within the wave (LDS ops issue in order from one wave)
   }
 
-  // ---- kNN branch: k extraction rounds over the register distances.
-  // The winning lane owns the candidate's data in registers and emits it
-  // directly (static register selection -- dynamic indexing would spill).
+  // ---- kNN branch: radix rank select of the k smallest distances
+  // (inverted keys: larger key = smaller distance), then ballot-prefix
+  // emission -- each lane owns its candidates' data in registers and
+  // writes its own winners.  Slot order is arbitrary; every consumer
+  // (conv + max-pool over k, index-scatter backward) is order-invariant.
   float *dst = knn + ((long)b * 4) * k * N + n;
   const long ch_stride = (long)k * N;
   int *idst = knn_idx + p * k;
-  int jd[MAXC];
+  unsigned kv[MAXC];
 #pragma unroll
-  for (int t = 0; t < MAXC; ++t) jd[t] = lane + t * WAVE;
-  for (int r = 0; r < k; ++r) {
-    int j;
-    const float dist = wave_extract_min(d, jd, j);
-    (void)dist;
-    if (lane == (j & 63)) {
-      const int ts = j >> 6;
-      float ex = 0.f, ey = 0.f, ez = 0.f, ec = 0.f;
+  for (int t = 0; t < MAXC; ++t)
+    kv[t] = (lane + t * WAVE) < K ? ~fkey(d[t]) : 0u;
+  const unsigned kt = wave_rank_key(kv, k);
+  int base = 0;
+  for (int pass = 0; pass < 2; ++pass) {
 #pragma unroll
-      for (int t = 0; t < MAXC; ++t)
-        if (t == ts) {
-          ex = rx[t];
-          ey = ry[t];
-          ez = rz[t];
-          ec = rc[t];
+    for (int t = 0; t < MAXC; ++t) {
+      const bool elig = (lane + t * WAVE) < K &&
+                        (pass == 0 ? kv[t] > kt : kv[t] == kt);
+      const unsigned long long m = __ballot(elig);
+      if (elig) {
+        const int slot = base + __popcll(m & ((1ull << lane) - 1ull));
+        if (slot < k) {
+          dst[(long)slot * N] = rc[t];
+          dst[(long)slot * N + ch_stride] = rx[t];
+          dst[(long)slot * N + 2 * ch_stride] = ry[t];
+          dst[(long)slot * N + 3 * ch_stride] = rz[t];
+          idst[slot] = lane + t * WAVE;
         }
-      dst[(long)r * N] = ec;
-      dst[(long)r * N + ch_stride] = ex;
-      dst[(long)r * N + 2 * ch_stride] = ey;
-      dst[(long)r * N + 3 * ch_stride] = ez;
-      idst[r] = j;
+      }
+      base += __popcll(m);
     }
+    if (base >= k) break;
   }
 }
 

@@ -19,11 +19,7 @@
 #define TK_MAXM 8192  // LDS row cache (32 KB)
 #define TK_CAP 128    // boundary-buffer capacity
 
-// order-preserving map: a < b (as floats, incl. negatives) <=> key(a) < key(b)
-DEV_INLINE unsigned fkey(float v) {
-  unsigned b = __float_as_uint(v);
-  return (b & 0x80000000u) ? ~b : (b | 0x80000000u);
-}
+// order-preserving float->unsigned map fkey() comes from common.h
 
 __global__ __launch_bounds__(TK_THREADS) void topk_rows_kernel(
     const float *__restrict__ vals,  // (R, M)
