@@ -244,8 +244,11 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
         s_phi[r] = s_plo[r];
         s_plo[r] = s_plo[r] - span * 4.f;
       } else {
-        // band overflowed its buffer: bisect the band from below
-        s_plo[r] = s_plo[r] + (s_phi[r] - s_plo[r]) * 0.5f;
+        // band overflowed its buffer: shrink the band span by the COUNT
+        // ratio (cbd is the true count), with slack for the next sweep
+        const float f =
+            fminf(1.f, (float)(K - (int)chi + 64) / (float)cbd * 1.25f);
+        s_plo[r] = s_phi[r] - (s_phi[r] - s_plo[r]) * f;
       }
       s_hi[r] = 0;
       s_bd[r] = 0;
@@ -333,7 +336,9 @@ void launch_corr_topk(const void *f1t, const void *f2t, float *thr,
     const double q = (double)K / st;
     double x = (-4.0 + sqrt(16.0 + 4.0 * q)) / 2.0;
     kp_hi = (int)(x * x);
-    x = (4.0 + sqrt(16.0 + 4.0 * q)) / 2.0;
+    // coverage margin 3 sd (4 sd made E[band] ~ CAP-sized and the band-
+    // overflow retry path hot); the exact-count retry covers the tail
+    x = (3.0 + sqrt(9.0 + 4.0 * q)) / 2.0;
     kp_lo = (int)(x * x) + 1;
     if (kp_hi < 1) kp_hi = 1;
     if (kp_lo <= kp_hi) kp_lo = kp_hi + 1;

@@ -596,8 +596,17 @@ def test_pw_fused_autograd_matches_fallback():
     finally:
         os.environ.pop("PVRAFT_NO_PWFWD", None)
 
-    assert torch.allclose(y.float(), y2.float(), atol=5e-2, rtol=2e-2)
+    # the fallback rounds to bf16 after EACH part GEMM and again after the
+    # bias add; the fused kernel keeps fp32 until the final store -- element
+    # diffs up to a few bf16 ulps of the ACCUMULATED magnitude are expected
+    # (the strict numerics test against fp32 is test_pw_fwd_fused_matches_
+    # composed above)
+    scale = y2.float().abs().max().item()
+    assert torch.allclose(y.float(), y2.float(), atol=3e-2 * scale, rtol=5e-2), (
+        (y.float() - y2.float()).abs().max(), scale
+    )
     for a, b in zip(leaves, clones):
-        assert torch.allclose(a.grad, b.grad, atol=5e-2, rtol=5e-2), (
+        gs = b.grad.abs().max().item() + 1e-6
+        assert torch.allclose(a.grad, b.grad, atol=3e-2 * gs, rtol=5e-2), (
             (a.grad - b.grad).abs().max()
         )
