@@ -169,6 +169,20 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
   const int koff = (lane >> 4) * 8;
   const int nfrag = (M + 15) / 16;
   for (int iter = 0; iter < 4 && s_active != 0u; ++iter) {
+    // register copies of this lane's row thresholds/active bits: the LDS
+    // counter atomics below alias LDS for the compiler, which would
+    // otherwise re-read s_phi/s_plo/s_active per emitted element
+    float rphi[CT_RWB / 16][4], rplo[CT_RWB / 16][4];
+    bool ract[CT_RWB / 16][4];
+#pragma unroll
+    for (int rf = 0; rf < CT_RWB / 16; ++rf)
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const int r = rf * 16 + (lane >> 4) * 4 + e;
+        rphi[rf][e] = s_phi[r];
+        rplo[rf][e] = s_plo[r];
+        ract[rf][e] = (s_active >> r & 1u) != 0u;
+      }
     for (int cf = wv; cf < nfrag; cf += CT_THREADS / WAVE) {
       const int j = cf * 16 + frow;  // column (B-frag row)
       const bool ok = j < M;
@@ -198,15 +212,15 @@ __global__ __launch_bounds__(CT_THREADS) void corr_sweep_kernel(
           for (int e = 0; e < 4; ++e) {
             const int r = rf * 16 + (lane >> 4) * 4 + e;
             const long n = n0 + r;
-            if (n >= N || !(s_active >> r & 1u)) continue;
+            if (n >= N || !ract[rf][e]) continue;
             const float v = acc[rf][e] * scale;
-            if (v > s_phi[r]) {
+            if (v > rphi[rf][e]) {
               const unsigned slot = atomicAdd(&s_hi[r], 1u);
               if (slot < (unsigned)K) {
                 out_v[((long)b * N + n) * K + slot] = v;
                 out_i[((long)b * N + n) * K + slot] = jj;
               }
-            } else if (v >= s_plo[r]) {
+            } else if (v >= rplo[rf][e]) {
               const unsigned p = atomicAdd(&s_bd[r], 1u);
               if (p < CT_CAP) {
                 band_v[((long)b * N + n) * CT_CAP + p] = v;

@@ -76,6 +76,18 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
   if (threadIdx.x == 0) s_active = (1u << QB) - 1;
   __syncthreads();
 
+  // register copies of the query coordinates: the LDS atomics below alias
+  // LDS in the compiler's view, so s_q reads inside the hot loops would
+  // otherwise be re-issued per (candidate, query) -- measured ~4x VALU/LDS
+  // instruction bloat
+  float qx[QB], qy[QB], qz[QB];
+#pragma unroll
+  for (int qi = 0; qi < QB; ++qi) {
+    qx[qi] = s_q[qi][0];
+    qy[qi] = s_q[qi][1];
+    qz[qi] = s_q[qi][2];
+  }
+
   // ---- pass A: sample distances (cloud is L2-resident; strided gather)
   for (int si = threadIdx.x; si < ns; si += KNN_THREADS) {
     const long c = (long)si * st;
@@ -88,9 +100,9 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
     }
 #pragma unroll
     for (int qi = 0; qi < QB; ++qi) {
-      const float dx = cx - s_q[qi][0];
-      const float dy = cy - s_q[qi][1];
-      const float dz = cz - s_q[qi][2];
+      const float dx = cx - qx[qi];
+      const float dy = cy - qy[qi];
+      const float dz = cz - qz[qi];
       s_samp[qi][si] = ok ? dx * dx + dy * dy + dz * dz : INFINITY;
     }
   }
@@ -117,6 +129,10 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
   // underflow (< k) or buffer overflow (> CAP) and re-sweep
   for (int iter = 0; iter < 4 && s_active != 0; ++iter) {
     const unsigned active = s_active;
+    float tau[QB];
+#pragma unroll
+    for (int qi = 0; qi < QB; ++qi)
+      tau[qi] = (active >> qi & 1u) ? s_tau[qi] : -1.f;  // -1: accept none
     for (int t0 = 0; t0 < N; t0 += TILE_PTS) {
       const int tn = min(TILE_PTS, N - t0);
       __syncthreads();
@@ -129,12 +145,11 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
         const float cz = s_tile[c * 3 + 2];
 #pragma unroll
         for (int qi = 0; qi < QB; ++qi) {
-          if (!(active >> qi & 1u)) continue;
-          const float dx = cx - s_q[qi][0];
-          const float dy = cy - s_q[qi][1];
-          const float dz = cz - s_q[qi][2];
+          const float dx = cx - qx[qi];
+          const float dy = cy - qy[qi];
+          const float dz = cz - qz[qi];
           const float d = dx * dx + dy * dy + dz * dz;
-          if (d <= s_tau[qi]) {
+          if (d <= tau[qi]) {
             const unsigned p = atomicAdd(&s_cnt[qi], 1u);
             if (p < CAP) {
               s_bd[qi][p] = d;
