@@ -57,7 +57,7 @@ void launch_transpose(const void*, void*, long, int, long, long, bool,
                       hipStream_t);
 void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
                      long, int, hipStream_t);
-void launch_gn_bwd_extract(float*, float*, float*, float*, int, int,
+void launch_gn_bwd_extract(float*, float*, float*, float*, int, int, int,
                            hipStream_t);
 void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
                       float*, const float*, const float*, void*,
@@ -274,13 +274,46 @@ std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
   return {y, mean, rstd};
 }
 
+
+// Finish a GN-family backward: either materialise fresh dweight/dbias/
+// dslope tensors (plain autograd) or ACCUMULATE into the parameters' own
+// fp32 grad buffers (deferred mode, returns an empty vector tail).
+static std::vector<torch::Tensor> gn_grads_finish(
+    torch::Tensor& ws, torch::Tensor dx, int rows, int C,
+    const torch::TensorOptions& fopt,
+    c10::optional<torch::Tensor> wt, c10::optional<torch::Tensor> bt,
+    c10::optional<torch::Tensor> st) {
+  if (wt.has_value()) {
+    TORCH_CHECK(wt->is_cuda() && wt->is_contiguous() &&
+                wt->scalar_type() == torch::kFloat32 && wt->numel() == C &&
+                bt.has_value() && bt->numel() == C,
+                "gn accumulate targets must be fp32 (C)");
+    float* sp = nullptr;
+    if (st.has_value() && st->defined() && st->numel() == 1)
+      sp = st->data_ptr<float>();
+    launch_gn_bwd_extract(ws.data_ptr<float>(), wt->data_ptr<float>(),
+                          bt->data_ptr<float>(), sp, rows, C, 1, stream());
+    return {dx};
+  }
+  auto dweight = torch::empty({C}, fopt);
+  auto dbias = torch::empty({C}, fopt);
+  auto dslope = torch::empty({1}, fopt);
+  launch_gn_bwd_extract(ws.data_ptr<float>(), dweight.data_ptr<float>(),
+                        dbias.data_ptr<float>(), dslope.data_ptr<float>(),
+                        rows, C, 0, stream());
+  return {dx, dweight, dbias, dslope};
+}
+
 std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
                                               torch::Tensor mean,
                                               torch::Tensor rstd, int64_t G,
                                               torch::Tensor weight,
                                               torch::Tensor bias, int64_t act,
                                               double slope,
-                                              c10::optional<torch::Tensor> slope_t) {
+                                              c10::optional<torch::Tensor> slope_t,
+                                              c10::optional<torch::Tensor> wtarget = c10::nullopt,
+                                              c10::optional<torch::Tensor> btarget = c10::nullopt,
+                                              c10::optional<torch::Tensor> starget = c10::nullopt) {
   const float* slope_ptr = nullptr;
   if (act == 2) {
     TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
@@ -307,13 +340,7 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
                 bias.data_ptr<float>(), row_ws, chan_ws, slope_ws,
                 dx.data_ptr(), rows, row_len, S, C, (int)G, (int)act,
                 (float)slope, slope_ptr, bf16, stream());
-  auto dweight = torch::empty({C}, fopt);
-  auto dbias = torch::empty({C}, fopt);
-  auto dslope = torch::empty({1}, fopt);
-  launch_gn_bwd_extract(ws.data_ptr<float>(), dweight.data_ptr<float>(),
-                        dbias.data_ptr<float>(), dslope.data_ptr<float>(),
-                        rows, C, stream());
-  return {dx, dweight, dbias, dslope};
+  return gn_grads_finish(ws, dx, rows, C, fopt, wtarget, btarget, starget);
 }
 
 // x (B, C, K, N); GN stats over full (K, N); returns pooled
@@ -354,7 +381,10 @@ std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
 std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
     torch::Tensor dy, torch::Tensor x, torch::Tensor am, torch::Tensor mean,
     torch::Tensor rstd, int64_t G, torch::Tensor weight, torch::Tensor bias,
-    int64_t act, double slope, c10::optional<torch::Tensor> slope_t) {
+    int64_t act, double slope, c10::optional<torch::Tensor> slope_t,
+    c10::optional<torch::Tensor> wtarget = c10::nullopt,
+    c10::optional<torch::Tensor> btarget = c10::nullopt,
+    c10::optional<torch::Tensor> starget = c10::nullopt) {
   const float* slope_ptr = nullptr;
   if (act == 2) {
     TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
@@ -377,13 +407,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
                   weight.data_ptr<float>(), bias.data_ptr<float>(), row_ws,
                   chan_ws, slope_ws, dx.data_ptr(), rows, row_len, N, K, C,
                   (int)G, (int)act, (float)slope, slope_ptr, bf16, stream());
-  auto dweight = torch::empty({C}, fopt);
-  auto dbias = torch::empty({C}, fopt);
-  auto dslope = torch::empty({1}, fopt);
-  launch_gn_bwd_extract(ws.data_ptr<float>(), dweight.data_ptr<float>(),
-                        dbias.data_ptr<float>(), dslope.data_ptr<float>(),
-                        rows, C, stream());
-  return {dx, dweight, dbias, dslope};
+  return gn_grads_finish(ws, dx, rows, C, fopt, wtarget, btarget, starget);
 }
 
 // SetConv stage 1 on the linearly-restructured operands: wg (B, N, M)
@@ -435,7 +459,10 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
     torch::Tensor dy, torch::Tensor wg, torch::Tensor idx, torch::Tensor am,
     torch::Tensor order, torch::Tensor offsets, torch::Tensor mean,
     torch::Tensor rstd, int64_t G, torch::Tensor weight, torch::Tensor bias,
-    int64_t act, double slope, c10::optional<torch::Tensor> slope_t) {
+    int64_t act, double slope, c10::optional<torch::Tensor> slope_t,
+    c10::optional<torch::Tensor> wtarget = c10::nullopt,
+    c10::optional<torch::Tensor> btarget = c10::nullopt,
+    c10::optional<torch::Tensor> starget = c10::nullopt) {
   const float* slope_ptr = nullptr;
   if (act == 2) {
     TORCH_CHECK(slope_t.has_value() && slope_t->numel() == 1, "act=2 needs a scalar slope tensor");
@@ -463,13 +490,7 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
                    ws.data_ptr<float>(), dwg.data_ptr(), B, N, K, M, (int)G,
                    (int)act, (float)slope, slope_ptr, bf16, rchunks,
                    stream());
-  auto dweight = torch::empty({M}, fopt);
-  auto dbias = torch::empty({M}, fopt);
-  auto dslope = torch::empty({1}, fopt);
-  launch_gn_bwd_extract(ws.data_ptr<float>(), dweight.data_ptr<float>(),
-                        dbias.data_ptr<float>(), dslope.data_ptr<float>(),
-                        rows, M, stream());
-  return {dwg, dweight, dbias, dslope};
+  return gn_grads_finish(ws, dwg, rows, M, fopt, wtarget, btarget, starget);
 }
 
 // dy (B, Co, S) bf16, x (B, Ci, S) bf16 -> dW (Co, Ci) fp32 (split-K MFMA)

@@ -40,6 +40,30 @@ def _load_ext():
     return _EXT
 
 
+def _gn_defer_targets(weight, bias, slope_t, act):
+    """fp32 grad-accumulation targets for a GN-family backward when the
+    deferred-wgrad step is armed (GradReducer flat views), else None.
+    Accumulating inside the extract kernel removes the per-call grad
+    tensors and their AccumulateGrad add launches (~190/step)."""
+    from pvraft_amd.model import pointwise
+
+    if not (pointwise._DEFER.on and torch.is_grad_enabled()):
+        return None
+    for t in (weight, bias):
+        if not (t.is_leaf and t.requires_grad and t.dtype == torch.float32):
+            return None
+    st = None
+    if act == 2:
+        if not (slope_t is not None and slope_t.is_leaf and slope_t.requires_grad):
+            return None
+        st = pointwise._grad_buffer(slope_t).view(-1)
+    return (
+        pointwise._grad_buffer(weight).view(-1),
+        pointwise._grad_buffer(bias).view(-1),
+        st,
+    )
+
+
 def _use_hip(*tensors: Tensor) -> bool:
     if os.environ.get("PVRAFT_REF_OPS", "0") == "1":
         return False
@@ -166,14 +190,23 @@ class _GroupNormAct(torch.autograd.Function):
         y, mean, rstd = _EXT.group_norm_act_fwd(x, num_groups, w, b, eps, act, slope, st_)
         ctx.save_for_backward(x, mean, rstd, w, b, st_)
         ctx.conf = (num_groups, act, slope, weight.dtype)
+        ctx.params = (weight, bias, slope_t)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, mean, rstd, w, b, st_ = ctx.saved_tensors
         num_groups, act, slope, wdtype = ctx.conf
+        tgt = _gn_defer_targets(*ctx.params, act)
+        if tgt is not None:
+            (dx,) = _EXT.group_norm_act_bwd(
+                dy.contiguous(), x, mean, rstd, num_groups, w, b, act, slope,
+                st_, tgt[0], tgt[1], tgt[2],
+            )
+            return dx, None, None, None, None, None, None, None
         dx, dw, db, dsl = _EXT.group_norm_act_bwd(
-            dy.contiguous(), x, mean, rstd, num_groups, w, b, act, slope, st_
+            dy.contiguous(), x, mean, rstd, num_groups, w, b, act, slope, st_,
+            None, None, None,
         )
         dslope = dsl.to(wdtype) if act == 2 else None
         return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None, dslope
@@ -234,14 +267,23 @@ class _GroupNormActMaxpool(torch.autograd.Function):
         y, am, mean, rstd = _EXT.group_norm_act_maxpool_fwd(x, num_groups, w, b, eps, act, slope, st_)
         ctx.save_for_backward(x, am, mean, rstd, w, b, st_)
         ctx.conf = (num_groups, act, slope, weight.dtype)
+        ctx.params = (weight, bias, slope_t)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         x, am, mean, rstd, w, b, st_ = ctx.saved_tensors
         num_groups, act, slope, wdtype = ctx.conf
+        tgt = _gn_defer_targets(*ctx.params, act)
+        if tgt is not None:
+            (dx,) = _EXT.group_norm_act_maxpool_bwd(
+                dy.contiguous(), x, am, mean, rstd, num_groups, w, b, act,
+                slope, st_, tgt[0], tgt[1], tgt[2],
+            )
+            return dx, None, None, None, None, None, None, None
         dx, dw, db, dsl = _EXT.group_norm_act_maxpool_bwd(
-            dy.contiguous(), x, am, mean, rstd, num_groups, w, b, act, slope, st_
+            dy.contiguous(), x, am, mean, rstd, num_groups, w, b, act, slope,
+            st_, None, None, None,
         )
         dslope = dsl.to(wdtype) if act == 2 else None
         return dx, None, dw.to(wdtype), db.to(wdtype), None, None, None, dslope
@@ -287,15 +329,24 @@ class _EdgeGNMP(torch.autograd.Function):
         y, am, mean, rstd = _EXT.edge_gnmp_fwd(wg_t, idx, num_groups, w, b, eps, act, slope, st_)
         ctx.save_for_backward(wg_t, idx, am, order, offsets, mean, rstd, w, b, st_)
         ctx.conf = (num_groups, act, slope, weight.dtype)
+        ctx.params = (weight, bias, slope_t)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         wg_t, idx, am, order, offsets, mean, rstd, w, b, st_ = ctx.saved_tensors
         num_groups, act, slope, wdtype = ctx.conf
+        tgt = _gn_defer_targets(*ctx.params, act)
+        if tgt is not None:
+            (dwg,) = _EXT.edge_gnmp_bwd(
+                dy.contiguous(), wg_t, idx, am, order, offsets, mean, rstd,
+                num_groups, w, b, act, slope, st_, tgt[0], tgt[1], tgt[2],
+            )
+            return (dwg, None, None, None, None, None, None, None, None,
+                    None, None)
         dwg, dw, db, dsl = _EXT.edge_gnmp_bwd(
             dy.contiguous(), wg_t, idx, am, order, offsets, mean, rstd,
-            num_groups, w, b, act, slope, st_,
+            num_groups, w, b, act, slope, st_, None, None, None,
         )
         dslope = dsl.to(wdtype) if act == 2 else None
         return (dwg, None, None, None, None, dw.to(wdtype), db.to(wdtype),
