@@ -23,12 +23,15 @@ class Predictor:
         self.iters = iters
         self.amp = amp
         self.device = next(model.parameters()).device
-        # KNOWN ISSUE (under investigation, profiles/README.md): graph
-        # capture+replay of the inference forward intermittently raises an
-        # illegal memory access for batch >= 5 (eager is clean at every
-        # batch size, all per-op GPU tests pass, and batch <= 4 graphs are
-        # exercised throughout training/eval).  Serve large batches eagerly
-        # until root-caused.
+        # ROOT-CAUSED TOOLCHAIN BUG (round 2, profiles/README.md): hipGraph
+        # replay of the inference forward at batch >= 5 faults in the ROCm
+        # graph-pool page mapping -- the PURE-ATEN forward (PVRAFT_REF_OPS=1,
+        # zero custom kernels) reproduces it identically ("write access to a
+        # read-only page" / aperture violation on first replay), eager is
+        # clean at every batch size under strict per-tensor allocation, and
+        # batch <= 4 graphs are exercised throughout training/eval.
+        # Repros: scripts/stress_graph_bs.py, scripts/graph_region_bisect.py.
+        # Permanent guard: serve batch >= 5 eagerly on this toolchain.
         self.use_graph = use_graph and self.device.type == "cuda" and batch <= 4
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self._in1 = torch.zeros(batch, points, 3, device=self.device)
