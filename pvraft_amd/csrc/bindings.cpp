@@ -66,8 +66,8 @@ void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
 void launch_egnmp_bwd(const void*, const void*, const int*,
                       const unsigned char*, const int*, const int*,
                       const float*, const float*, const float*, const float*,
-                      float*, float*, float*, void*, int, long, int, int, int,
-                      int, float, const float*, bool, int, hipStream_t);
+                      float*, float*, void*, int, long, int, int, int, int,
+                      float, const float*, bool, int, hipStream_t);
 int egnmp_reduce_chunks(long, int, int);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
@@ -481,17 +481,15 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
   auto& ws = persistent_ws(ws_len, fopt);
   const int rchunks = egnmp_reduce_chunks(N, M, B);
   auto scratch = torch::empty({ws_len, (long)rchunks * B}, fopt);
-  // argmax-dxhat scatter target for the apply pass (see edge_gnmp.hip)
-  auto dxh = zeros_fast({(long)B * N, (long)M}, fopt);
   auto dwg = torch::empty_like(wg);
   launch_egnmp_bwd(dy.data_ptr(), wg.data_ptr(), idx.data_ptr<int>(),
                    am.data_ptr<unsigned char>(), order.data_ptr<int>(),
                    offsets.data_ptr<int>(), mean.data_ptr<float>(),
                    rstd.data_ptr<float>(), weight.data_ptr<float>(),
                    bias.data_ptr<float>(), scratch.data_ptr<float>(),
-                   ws.data_ptr<float>(), dxh.data_ptr<float>(),
-                   dwg.data_ptr(), B, N, K, M, (int)G, (int)act,
-                   (float)slope, slope_ptr, bf16, rchunks, stream());
+                   ws.data_ptr<float>(), dwg.data_ptr(), B, N, K, M, (int)G,
+                   (int)act, (float)slope, slope_ptr, bf16, rchunks,
+                   stream());
   return gn_grads_finish(ws, dwg, rows, M, fopt, wtarget, btarget, starget);
 }
 

@@ -33,16 +33,15 @@
 // kernel folds the partials into the workspace the shared finalize /
 // extract kernels consume.
 //
-// Backward: the gradient w.r.t. WgT at point p is
+// Backward data path is deterministic too: the gradient w.r.t. WgT at
+// point p is
 //   dWg[m, p] = sum_{edges e=(j,n): idx[n,j]=p} dx1[m, j, n]   (incoming)
 //             - sum_j dx1[m, j, p]                             (centre)
 // where dx1 is the standard GroupNorm+act+maxpool backward element
-// (exact same formulas as gnmp_bwd_* in group_norm.hip).  The incoming
-// sum walks the same inverse-adjacency CSR (order/offsets, edge id =
-// j*N + n) as round 1, but the per-edge terms are grouped algebraically
-// so the walk only gathers wg, with the argmax term pre-scattered by the
-// reduce pass (fp32 atomics -- see the apply kernel's comment on
-// determinism); every (p, m) output is written exactly once.
+// (exact same formulas as gnmp_bwd_* in group_norm.hip), recomputed on
+// the fly from WgT / argmax / saved stats.  The incoming sum walks the
+// same inverse-adjacency CSR (order/offsets, edge id = j*N + n) the
+// round-1 CSR backward used; every (p, m) output is written exactly once.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include "common.h"
@@ -231,21 +230,14 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_apply_kernel(
 // the argmax element of each (n, c) carries dy).  Partials land in
 // scratch exactly like the forward reduce; out layout matches the ws
 // layout [rows*2 | M*2 | 1] so one fold kernel serves both.
-//
-// The pass ALSO scatters each (n, c)'s single dxhat value onto its argmax
-// neighbour (dxh[nb][c] += gv*gamma, fp32 atomic): the apply pass then
-// reads dxh instead of re-deriving the argmax term per incoming CSR edge,
-// which cuts its gather loads ~3x (it no longer touches am/dy per edge).
 template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_reduce_kernel(
     const T *__restrict__ dy,  // (B, N, M) pooled grad (point-major)
     const T *__restrict__ wg, const int *__restrict__ idx,
     const unsigned char *__restrict__ am, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
-    const float *__restrict__ beta, float *__restrict__ scratch,
-    float *__restrict__ dxh,   // (B, N, M) fp32, pre-zeroed
-    long N, int K, int M, int G, float slope,
-    const float *__restrict__ slope_ptr) {
+    const float *__restrict__ beta, float *__restrict__ scratch, long N,
+    int K, int M, int G, float slope, const float *__restrict__ slope_ptr) {
   const int b = blockIdx.z;
   const int B = gridDim.z;
   const int tpc = M / 4;
@@ -300,7 +292,6 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_reduce_kernel(
         const float dxhat = gv * ga[e];
         sum_dx[e] += dxhat;
         sum_dxx[e] += dxhat * xhat;
-        atomicAdd(&dxh[((long)b * N + nb) * M + c4 * 4 + e], dxhat);
       }
     }
 #pragma unroll
@@ -321,17 +312,9 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_reduce_kernel(
     scratch[i * stride + col] = sbins[i];
 }
 
-// pass 2: dWgT (B, N, M).  dx1 = (dxhat - (s1 + xhat*s2)/n) * r summed
-// over the centre edges (minus) and the incoming CSR edges (plus), with
-// the per-edge sums GROUPED ALGEBRAICALLY so the loops only gather wg:
-//   sum dx1 = r*sum(dxhat) - r*(cnt*s1 + s2*sum(xhat))/n
-//   sum_in xhat  = r*(indeg*(wg_p - m) - sum_in wg_n)
-//   sum_ctr xhat = r*(sum_j wg_nb - K*(wg_p + m))
-// sum_in(dxhat) comes pre-scattered from the reduce pass (dxh); the
-// centre's single argmax dxhat is recomputed from values the centre loop
-// reads anyway.  (The dxh scatter uses fp32 atomics, so this path is
-// reduction-order deterministic only per run -- same as the GN statistic
-// reductions elsewhere in the step.)
+// pass 2: dWgT (B, N, M), deterministic.  dx1 elements are recomputed on
+// the fly with the exact gnmp_bwd_apply formula; the incoming sum walks
+// the inverse-adjacency CSR (edge id = j*N + n, sorted by target).
 template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ wg,
@@ -340,9 +323,9 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     const int *__restrict__ offsets,  // (B, N+1)
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
-    const float *__restrict__ row_ws, const float *__restrict__ dxh,
-    T *__restrict__ dwg, long N, int K, int M, int G, long row_len,
-    float slope, const float *__restrict__ slope_ptr) {
+    const float *__restrict__ row_ws, T *__restrict__ dwg, long N, int K,
+    int M, int G, long row_len, float slope,
+    const float *__restrict__ slope_ptr) {
   const int b = blockIdx.z;
   const int tpc = M / 4;
   const int ppb = EG_THREADS / tpc;
@@ -376,54 +359,61 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
   for (long p = (long)blockIdx.x * ppb + p_l; p < N;
        p += (long)gridDim.x * ppb) {
     const Quad<T> pq = *(const Quad<T> *)(wgb + p * M + c4 * 4);
-    const uchar4 aq = *(const uchar4 *)(amb + p * M + c4 * 4);
-    const Quad<T> gq = *(const Quad<T> *)(dyb + p * M + c4 * 4);
-    const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
-    // centre: sum_j wg_nb and the argmax channel's neighbour value
-    float sum_nb[4] = {0.f, 0.f, 0.f, 0.f};
-    float nb_sel[4] = {0.f, 0.f, 0.f, 0.f};
-    const int *irow = idxb + p * K;
-    for (int j = 0; j < K; ++j) {
-      const Quad<T> nq = *(const Quad<T> *)(wgb + (long)irow[j] * M + c4 * 4);
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    // centre term: -sum_j dx1[c, j, p]
+    {
+      const uchar4 aq = *(const uchar4 *)(amb + p * M + c4 * 4);
+      const Quad<T> gq = *(const Quad<T> *)(dyb + p * M + c4 * 4);
+      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
+      const int *irow = idxb + p * K;
+      for (int j = 0; j < K; ++j) {
+        const Quad<T> nq =
+            *(const Quad<T> *)(wgb + (long)irow[j] * M + c4 * 4);
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        const float v = (float)nq.v[e];
-        sum_nb[e] += v;
-        if (j == ks[e]) nb_sel[e] = v;
+        for (int e = 0; e < 4; ++e) {
+          const float v = (float)nq.v[e] - (float)pq.v[e];
+          const float xhat = (v - m[e]) * r[e];
+          float dxhat = 0.f;
+          if (j == ks[e]) {
+            float gs = (float)gq.v[e];
+            if (ACT >= 1) {
+              const float pre = xhat * ga[e] + be[e];
+              gs = pre > 0.f ? gs : gs * slope;
+            }
+            dxhat = gs * ga[e];
+          }
+          acc[e] -= (dxhat - (s1[e] + xhat * s2[e]) * inv_n) * r[e];
+        }
       }
     }
-    // incoming: sum_t wg_n only
+    // incoming term: edges whose neighbour is p
     const int lo = offb[p], hi = offb[p + 1];
-    const float indeg = (float)(hi - lo);
-    float sum_in[4] = {0.f, 0.f, 0.f, 0.f};
     for (int t = lo; t < hi; ++t) {
       int j, n;
       split_edge(ordb[t], (int)N, invN, j, n);
       const Quad<T> nq = *(const Quad<T> *)(wgb + (long)n * M + c4 * 4);
+      const uchar4 aq = *(const uchar4 *)(amb + (long)n * M + c4 * 4);
+      const Quad<T> gq = *(const Quad<T> *)(dyb + (long)n * M + c4 * 4);
+      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
 #pragma unroll
-      for (int e = 0; e < 4; ++e) sum_in[e] += (float)nq.v[e];
+      for (int e = 0; e < 4; ++e) {
+        const float v = (float)pq.v[e] - (float)nq.v[e];
+        const float xhat = (v - m[e]) * r[e];
+        float dxhat = 0.f;
+        if (j == ks[e]) {
+          float gs = (float)gq.v[e];
+          if (ACT >= 1) {
+            const float pre = xhat * ga[e] + be[e];
+            gs = pre > 0.f ? gs : gs * slope;
+          }
+          dxhat = gs * ga[e];
+        }
+        acc[e] += (dxhat - (s1[e] + xhat * s2[e]) * inv_n) * r[e];
+      }
     }
-    const float *dxhp = dxh + ((long)b * N + p) * M + c4 * 4;
     Quad<T> oq;
 #pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      const float wp = (float)pq.v[e];
-      // centre argmax dxhat (recomputed from the gathered value)
-      const float xh_sel = (nb_sel[e] - wp - m[e]) * r[e];
-      float gs = (float)gq.v[e];
-      if (ACT >= 1) {
-        const float pre = xh_sel * ga[e] + be[e];
-        gs = pre > 0.f ? gs : gs * slope;
-      }
-      const float cen_dxhat = gs * ga[e];
-      const float sum_xh_ctr = r[e] * (sum_nb[e] - (float)K * (wp + m[e]));
-      const float sum_xh_in = r[e] * (indeg * (wp - m[e]) - sum_in[e]);
-      const float cen = r[e] * cen_dxhat -
-                        r[e] * inv_n * ((float)K * s1[e] + s2[e] * sum_xh_ctr);
-      const float inc = r[e] * dxhp[e] -
-                        r[e] * inv_n * (indeg * s1[e] + s2[e] * sum_xh_in);
-      oq.v[e] = (T)(inc - cen);
-    }
+    for (int e = 0; e < 4; ++e) oq.v[e] = (T)acc[e];
     *(Quad<T> *)(dwg + ((long)b * N + p) * M + c4 * 4) = oq;
   }
 }
@@ -474,9 +464,8 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
                     const unsigned char *am, const int *order,
                     const int *offsets, const float *mean, const float *rstd,
                     const float *gamma, const float *beta, float *scratch,
-                    float *ws, float *dxh, T *dwg, int B, long N, int K,
-                    int M, int G, int act, float slope,
-                    const float *slope_ptr, int rchunks,
+                    float *ws, T *dwg, int B, long N, int K, int M, int G,
+                    int act, float slope, const float *slope_ptr, int rchunks,
                     hipStream_t stream) {
   const int tpc = M / 4;
   const int ppb = EG_THREADS / tpc;
@@ -490,7 +479,7 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
   do {                                                                         \
     hipLaunchKernelGGL((egnmp_bwd_reduce_kernel<T, A>), rgrid,                 \
                        dim3(EG_THREADS), shmem, stream, dy, wg, idx, am, mean, \
-                       rstd, gamma, beta, scratch, dxh, N, K, M, G, slope,     \
+                       rstd, gamma, beta, scratch, N, K, M, G, slope,          \
                        slope_ptr);                                             \
     hipLaunchKernelGGL(egnmp_sum_partials_kernel,                              \
                        dim3((n_out + waves_per_block - 1) / waves_per_block),  \
@@ -498,8 +487,8 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
                        (long)rchunks * B, n_out);                              \
     hipLaunchKernelGGL((egnmp_bwd_apply_kernel<T, A>), agrid,                  \
                        dim3(EG_THREADS), 0, stream, dy, wg, idx, am, order,    \
-                       offsets, mean, rstd, gamma, beta, ws, dxh, dwg, N, K,   \
-                       M, G, row_len, slope, slope_ptr);                       \
+                       offsets, mean, rstd, gamma, beta, ws, dwg, N, K, M, G,  \
+                       row_len, slope, slope_ptr);                             \
   } while (0)
   if (act == 2) EG_BWD(2);
   else if (act == 1) EG_BWD(1);
@@ -538,19 +527,17 @@ void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
                       const unsigned char *am, const int *order,
                       const int *offsets, const float *mean, const float *rstd,
                       const float *gamma, const float *beta, float *scratch,
-                      float *ws, float *dxh, void *dwg, int B, long N, int K,
-                      int M, int G, int act, float slope,
-                      const float *slope_ptr, bool bf16, int rchunks,
-                      hipStream_t stream) {
+                      float *ws, void *dwg, int B, long N, int K, int M, int G,
+                      int act, float slope, const float *slope_ptr, bool bf16,
+                      int rchunks, hipStream_t stream) {
   if (bf16)
     egnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)wg, idx, am, order,
-        offsets, mean, rstd, gamma, beta, scratch, ws, dxh,
-        (__hip_bfloat16 *)dwg, B, N, K, M, G, act, slope, slope_ptr, rchunks,
-        stream);
+        offsets, mean, rstd, gamma, beta, scratch, ws, (__hip_bfloat16 *)dwg,
+        B, N, K, M, G, act, slope, slope_ptr, rchunks, stream);
   else
     egnmp_bwd_impl<float>((const float *)dy, (const float *)wg, idx, am, order,
-                          offsets, mean, rstd, gamma, beta, scratch, ws, dxh,
+                          offsets, mean, rstd, gamma, beta, scratch, ws,
                           (float *)dwg, B, N, K, M, G, act, slope, slope_ptr,
                           rchunks, stream);
 }
