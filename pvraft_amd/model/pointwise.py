@@ -331,6 +331,10 @@ def pw_fused(parts, bias=None, bias_target=None, addend=None, act="none") -> Ten
         and len(parts) <= 4
         and parts[0][0].shape[0] <= 256
         and all(w.shape[1] <= 224 for w, _x, _t in parts)
+        # degenerate skinny-k giant-S GEMMs (the kNN-branch 4->64 conv over
+        # K*N) pad k to 32 and waste ~7/8 of the MFMA work; hipBLASLt wins
+        # there (30 vs ~60 us measured) -- keep those on the bmm path
+        and not (x0.shape[2] > 100_000 and min(w.shape[1] for w, _x, _t in parts) < 16)
         and os.environ.get("PVRAFT_NO_PWFWD", "0") != "1"
     )
     if fused_ok:

@@ -610,3 +610,18 @@ def test_pw_fused_autograd_matches_fallback():
         assert torch.allclose(a.grad, b.grad, atol=3e-2 * gs, rtol=5e-2), (
             (a.grad - b.grad).abs().max()
         )
+
+
+@pytest.mark.parametrize("B,R,C,dtype", [
+    (2, 8192, 96, torch.bfloat16),   # narrow fast path
+    (2, 8191, 48, torch.bfloat16),   # narrow, ragged R
+    (1, 513, 16, torch.float32),
+    (2, 500, 35, torch.float32),     # narrow-scalar (C % W != 0)
+    (2, 300, 128, torch.bfloat16),   # general tiled path
+])
+def test_batched_transpose_shapes(B, R, C, dtype):
+    import pvraft_amd._C as _C
+
+    x = torch.randn(B, R, C, device=dev()).to(dtype)
+    y = _C.batched_transpose(x.contiguous())
+    assert torch.equal(y, x.transpose(1, 2).contiguous())
