@@ -105,25 +105,6 @@ DEV_INLINE float wave_extract_min(float (&dv)[S], const int (&iv)[S],
   return bv;
 }
 
-// Last-block-arrival detection for fused kernel epilogues: every block
-// increments the counter AFTER its own reads of the shared workspace
-// (fence + barrier first); exactly one block sees the final count, and it
-// self-resets the counter so a persistent slot serves every call/replay.
-// Arrivals are spread over the kernel's runtime, so the single-address
-// atomics do not serialize at one instant.
-DEV_INLINE bool last_block_arrival(unsigned *counter, unsigned total) {
-  __shared__ bool last;
-  __threadfence();
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    const unsigned prev = atomicAdd(counter, 1u);
-    last = (prev + 1 == total);
-    if (last) *counter = 0u;
-  }
-  __syncthreads();
-  return last;
-}
-
 #define HIP_CHECK_LAST()                                          \
   do {                                                            \
     hipError_t e = hipGetLastError();                             \

@@ -128,40 +128,12 @@ __global__ void gn_fwd_finalize_kernel(float *__restrict__ ws,
   rstd[row] = rsqrtf(fmaxf(var, 0.f) + eps);
 }
 
-// finalize-in-apply: every block derives mean/rstd for its row straight
-// from the summed workspace (two extra loads); the LAST block to finish
-// also writes the mean/rstd tensors for backward and re-zeroes the
-// workspace -- the separate 4.4 us finalize launch (one per GN call, ~67
-// per train step) disappears.
-DEV_INLINE void gn_stats_from_ws(const float *ws, int row, long row_len,
-                                 float eps, float &m, float &r) {
-  m = ws[row * 2 + 0] / (float)row_len;
-  const float var = ws[row * 2 + 1] / (float)row_len - m * m;
-  r = rsqrtf(fmaxf(var, 0.f) + eps);
-}
-
-DEV_INLINE void gn_fwd_epilogue(float *ws, float *mean, float *rstd,
-                                long row_len, int rows, float eps,
-                                unsigned total_blocks) {
-  if (last_block_arrival((unsigned *)(ws + (long)rows * 2), total_blocks)) {
-    for (int i = threadIdx.x; i < rows; i += blockDim.x) {
-      float m, r;
-      gn_stats_from_ws(ws, i, row_len, eps, m, r);
-      mean[i] = m;
-      rstd[i] = r;
-      ws[i * 2 + 0] = 0.f;
-      ws[i * 2 + 1] = 0.f;
-    }
-  }
-}
-
-// pass 2 (fused into the applies): y = act((x-mean)*rstd*gamma + beta)
+// pass 3: y = act((x - mean) * rstd * gamma + beta), per channel slice
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
-    const T *__restrict__ x, T *__restrict__ y, float *__restrict__ ws,
-    float *__restrict__ mean, float *__restrict__ rstd,
-    const float *__restrict__ gamma, const float *__restrict__ beta, long S,
-    int C, int G, long row_len, float eps, unsigned total_blocks, float slope,
+    const T *__restrict__ x, T *__restrict__ y, const float *__restrict__ mean,
+    const float *__restrict__ rstd, const float *__restrict__ gamma,
+    const float *__restrict__ beta, long S, int C, int G, float slope,
     const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
@@ -169,8 +141,8 @@ __global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
   const int b = blockIdx.z;
   const int Cg = C / G;
   const int row = b * G + c / Cg;
-  float m, r;
-  gn_stats_from_ws(ws, row, row_len, eps, m, r);
+  const float m = mean[row];
+  const float r = rstd[row];
   const float ga = gamma[c], be = beta[c];
   if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const T *xb = x + ((long)b * C + c) * S;
@@ -198,8 +170,6 @@ __global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
       st(yb + i, v);
     }
   }
-  gn_fwd_epilogue(ws, mean, rstd, row_len, G * (int)gridDim.z, eps,
-                  total_blocks);
 }
 
 // ---------------------------------------------------------------- backward
@@ -276,54 +246,13 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
   }
 }
 
-// extract-in-apply: the LAST apply block drains the channel/slope sums
-// into the gradient outputs (or ACCUMULATES into the parameters' grad
-// buffers, deferred mode) and re-zeroes the whole workspace -- the
-// separate extract launch (~67 per step) disappears.  ws layout:
-// [rows*2 row sums | C*2 channel sums | 1 d slope | 1 arrival counter].
-DEV_INLINE void gn_bwd_epilogue(float *ws, float *dweight, float *dbias,
-                                float *dslope, int rows, int C,
-                                int accumulate, unsigned total_blocks) {
-  unsigned *counter = (unsigned *)(ws + (long)rows * 2 + C * 2 + 1);
-  if (!last_block_arrival(counter, total_blocks)) return;
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    const float db = ws[rows * 2 + 2 * i];
-    const float dw = ws[rows * 2 + 2 * i + 1];
-    if (accumulate) {
-      dbias[i] += db;
-      dweight[i] += dw;
-    } else {
-      dbias[i] = db;
-      dweight[i] = dw;
-    }
-    ws[rows * 2 + 2 * i] = 0.f;
-    ws[rows * 2 + 2 * i + 1] = 0.f;
-  }
-  for (int i = threadIdx.x; i < rows; i += blockDim.x) {
-    ws[2 * i] = 0.f;
-    ws[2 * i + 1] = 0.f;
-  }
-  if (threadIdx.x == 0) {
-    const float dsl = ws[rows * 2 + C * 2];
-    if (dslope != nullptr) {
-      if (accumulate)
-        dslope[0] += dsl;
-      else
-        dslope[0] = dsl;
-    }
-    ws[rows * 2 + C * 2] = 0.f;
-  }
-}
-
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
-    float *__restrict__ row_ws, float *__restrict__ dweight,
-    float *__restrict__ dbias, float *__restrict__ dslope, int accumulate,
-    unsigned total_blocks, T *__restrict__ dx, long S, int C, int G,
-    long row_len, float slope, const float *__restrict__ slope_ptr) {
+    const float *__restrict__ row_ws, T *__restrict__ dx, long S, int C,
+    int G, long row_len, float slope, const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
   const int c = blockIdx.y;
@@ -378,8 +307,6 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
     }
   }
 #undef GN_BWD_APPLY_BODY
-  gn_bwd_epilogue(row_ws, dweight, dbias, dslope, G * (int)gridDim.z, C,
-                  accumulate, total_blocks);
 }
 
 // ------------------------------------------------- GN + act + max-pool(K)
@@ -394,18 +321,17 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gnmp_fwd_apply_kernel(
     const T *__restrict__ x, T *__restrict__ y, unsigned char *__restrict__ am,
-    float *__restrict__ ws, float *__restrict__ mean, float *__restrict__ rstd,
+    const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta, long N,
-    int K, int C, int G, long row_len, float eps, unsigned total_blocks,
-    float slope, const float *__restrict__ slope_ptr) {
+    int K, int C, int G, float slope, const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
   const int c = blockIdx.y;
   const int b = blockIdx.z;
   const int Cg = C / G;
   const int row = b * G + c / Cg;
-  float m, r;
-  gn_stats_from_ws(ws, row, row_len, eps, m, r);
+  const float m = mean[row];
+  const float r = rstd[row];
   const float ga = gamma[c], be = beta[c];
   if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const T *xb = x + ((long)b * C + c) * K * N;
@@ -458,8 +384,6 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_fwd_apply_kernel(
       amb[n] = (unsigned char)bk;
     }
   }
-  gn_fwd_epilogue(ws, mean, rstd, row_len, G * (int)gridDim.z, eps,
-                  total_blocks);
 }
 
 // backward pass 1: row/channel(/slope) sums over the POOLED domain
@@ -521,9 +445,7 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
     const unsigned char *__restrict__ am, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
-    const float *__restrict__ beta, float *__restrict__ row_ws,
-    float *__restrict__ dweight, float *__restrict__ dbias,
-    float *__restrict__ dslope, int accumulate, unsigned total_blocks,
+    const float *__restrict__ beta, const float *__restrict__ row_ws,
     T *__restrict__ dx, long N, int K, int C, int G, long row_len,
     int ksplit, float slope, const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
@@ -599,8 +521,6 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
       }
     }
   }
-  gn_bwd_epilogue(row_ws, dweight, dbias, dslope, G * (int)gridDim.z, C,
-                  accumulate, total_blocks);
 }
 
 // --------------------------------------------------------------- launchers
@@ -636,11 +556,12 @@ void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
   const dim3 grid(pick_chunks(S, (long)B * C), C, B);
   hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, rgrid, dim3(GN_THREADS), 0,
                      stream, x, ws, S, C, G);
-  const unsigned tb = grid.x * grid.y * grid.z;
+  hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
+                     0, stream, ws, mean, rstd, row_len, rows, eps);
 #define GN_FWD_APPLY(A)                                                        \
   hipLaunchKernelGGL((gn_fwd_apply_kernel<T, A>), grid, dim3(GN_THREADS), 0,   \
-                     stream, x, y, ws, mean, rstd, gamma, beta, S, C, G,       \
-                     row_len, eps, tb, slope, slope_ptr)
+                     stream, x, y, mean, rstd, gamma, beta, S, C, G, slope,    \
+                     slope_ptr)
   if (act == 2) GN_FWD_APPLY(2);
   else if (act == 1) GN_FWD_APPLY(1);
   else GN_FWD_APPLY(0);
@@ -650,23 +571,20 @@ void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
 template <typename T>
 void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
                  const float *gamma, const float *beta, float *row_ws,
-                 float *chan_ws, float *slope_ws, float *dweight, float *dbias,
-                 float *dslope, int accumulate, T *dx, int rows, long row_len,
+                 float *chan_ws, float *slope_ws, T *dx, int rows, long row_len,
                  long S, int C, int G, int act, float slope,
                  const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
   const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
   const dim3 grid(pick_chunks(S, (long)B * C), C, B);
-  const unsigned tb = grid.x * grid.y * grid.z;
 #define GN_BWD(A)                                                              \
   do {                                                                         \
     hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, A>), rgrid, dim3(GN_THREADS),  \
                        0, stream, dy, x, mean, rstd, gamma, beta, row_ws,      \
                        chan_ws, slope_ws, S, C, G, slope, slope_ptr);          \
     hipLaunchKernelGGL((gn_bwd_apply_kernel<T, A>), grid, dim3(GN_THREADS),    \
-                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws,      \
-                       dweight, dbias, dslope, accumulate, tb, dx, S, C, G,    \
-                       row_len, slope, slope_ptr);                             \
+                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws, dx,  \
+                       S, C, G, row_len, slope, slope_ptr);                    \
   } while (0)
   if (act == 2) GN_BWD(2);
   else if (act == 1) GN_BWD(1);
@@ -685,6 +603,8 @@ void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
   const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
   hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, rgrid, dim3(GN_THREADS), 0,
                      stream, x, ws, S, C, G);
+  hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
+                     0, stream, ws, mean, rstd, row_len, rows, eps);
   // the apply loops all K per thread (argmax), so a thread's work is K x
   // its vector count: allow chunking down to ~1 vector per thread instead
   // of pick_chunks' 16-iteration floor (C*B is small -> it underfilled)
@@ -693,11 +613,10 @@ void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
   if (pch > pv / GN_THREADS) pch = pv / GN_THREADS;
   if (pch < 1) pch = 1;
   const dim3 pgrid((unsigned)pch, C, B);
-  const unsigned tb = pgrid.x * pgrid.y * pgrid.z;
 #define GNMP_FWD(A)                                                            \
   hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, A>), pgrid, dim3(GN_THREADS),   \
-                     0, stream, x, y, am, ws, mean, rstd, gamma, beta, N, K,   \
-                     C, G, row_len, eps, tb, slope, slope_ptr)
+                     0, stream, x, y, am, mean, rstd, gamma, beta, N, K, C, G, \
+                     slope, slope_ptr)
   if (act == 2) GNMP_FWD(2);
   else if (act == 1) GNMP_FWD(1);
   else GNMP_FWD(0);
@@ -708,10 +627,9 @@ template <typename T>
 void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                    const float *mean, const float *rstd, const float *gamma,
                    const float *beta, float *row_ws, float *chan_ws,
-                   float *slope_ws, float *dweight, float *dbias,
-                   float *dslope, int accumulate, T *dx, int rows,
-                   long row_len, long N, int K, int C, int G, int act,
-                   float slope, const float *slope_ptr, hipStream_t stream) {
+                   float *slope_ws, T *dx, int rows, long row_len, long N,
+                   int K, int C, int G, int act, float slope,
+                   const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
   const dim3 rgrid(pick_chunks(N, (long)B * C, C / G), C, B);
   const int nchunks = pick_chunks(N, (long)B * C);
@@ -719,7 +637,6 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
   if (ksplit > K) ksplit = K;
   if (ksplit < 1) ksplit = 1;
   const dim3 grid(nchunks * ksplit, C, B);
-  const unsigned tb = grid.x * grid.y * grid.z;
 #define GNMP_BWD(A)                                                            \
   do {                                                                         \
     hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, A>), rgrid,                  \
@@ -728,8 +645,7 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                        slope, slope_ptr);                                      \
     hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, A>), grid,                    \
                        dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,     \
-                       gamma, beta, row_ws, dweight, dbias, dslope,            \
-                       accumulate, tb, dx, N, K, C, G, row_len, ksplit,        \
+                       gamma, beta, row_ws, dx, N, K, C, G, row_len, ksplit,   \
                        slope, slope_ptr);                                      \
   } while (0)
   if (act == 2) GNMP_BWD(2);
@@ -756,23 +672,20 @@ void launch_gn_fwd(const void *x, void *y, float *ws, float *mean, float *rstd,
 
 void launch_gn_bwd(const void *dy, const void *x, const float *mean,
                    const float *rstd, const float *gamma, const float *beta,
-                   float *row_ws, float *chan_ws, float *slope_ws,
-                   float *dweight, float *dbias, float *dslope,
-                   int accumulate, void *dx, int rows, long row_len, long S,
-                   int C, int G, int act, float slope, const float *slope_ptr,
-                   bool bf16, hipStream_t stream) {
+                   float *row_ws, float *chan_ws, float *slope_ws, void *dx,
+                   int rows, long row_len, long S, int C, int G, int act,
+                   float slope, const float *slope_ptr, bool bf16,
+                   hipStream_t stream) {
   if (bf16)
     gn_bwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)dy,
                                 (const __hip_bfloat16 *)x, mean, rstd, gamma,
-                                beta, row_ws, chan_ws, slope_ws, dweight,
-                                dbias, dslope, accumulate,
+                                beta, row_ws, chan_ws, slope_ws,
                                 (__hip_bfloat16 *)dx, rows, row_len, S, C, G,
                                 act, slope, slope_ptr, stream);
   else
     gn_bwd_impl<float>((const float *)dy, (const float *)x, mean, rstd, gamma,
-                       beta, row_ws, chan_ws, slope_ws, dweight, dbias,
-                       dslope, accumulate, (float *)dx, rows, row_len, S, C,
-                       G, act, slope, slope_ptr, stream);
+                       beta, row_ws, chan_ws, slope_ws, (float *)dx, rows,
+                       row_len, S, C, G, act, slope, slope_ptr, stream);
 }
 
 void launch_gnmp_fwd(const void *x, void *y, unsigned char *am, float *ws,
@@ -794,22 +707,19 @@ void launch_gnmp_fwd(const void *x, void *y, unsigned char *am, float *ws,
 void launch_gnmp_bwd(const void *dy, const void *x, const unsigned char *am,
                      const float *mean, const float *rstd, const float *gamma,
                      const float *beta, float *row_ws, float *chan_ws,
-                     float *slope_ws, float *dweight, float *dbias,
-                     float *dslope, int accumulate, void *dx, int rows,
-                     long row_len, long N, int K, int C, int G, int act,
-                     float slope, const float *slope_ptr, bool bf16,
-                     hipStream_t stream) {
+                     float *slope_ws, void *dx, int rows, long row_len, long N,
+                     int K, int C, int G, int act, float slope,
+                     const float *slope_ptr, bool bf16, hipStream_t stream) {
   if (bf16)
     gnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, am, mean, rstd,
-        gamma, beta, row_ws, chan_ws, slope_ws, dweight, dbias, dslope,
-        accumulate, (__hip_bfloat16 *)dx, rows, row_len, N, K, C, G, act,
-        slope, slope_ptr, stream);
+        gamma, beta, row_ws, chan_ws, slope_ws, (__hip_bfloat16 *)dx, rows,
+        row_len, N, K, C, G, act, slope, slope_ptr, stream);
   else
     gnmp_bwd_impl<float>((const float *)dy, (const float *)x, am, mean, rstd,
-                         gamma, beta, row_ws, chan_ws, slope_ws, dweight,
-                         dbias, dslope, accumulate, (float *)dx, rows,
-                         row_len, N, K, C, G, act, slope, slope_ptr, stream);
+                         gamma, beta, row_ws, chan_ws, slope_ws, (float *)dx,
+                         rows, row_len, N, K, C, G, act, slope, slope_ptr,
+                         stream);
 }
 
 // extract dbias/dweight/dslope from the backward workspace and re-zero the
