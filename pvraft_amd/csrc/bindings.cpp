@@ -23,9 +23,12 @@ void launch_gn_fwd(const void*, void*, float*, float*, float*, const float*,
                    const float*, int, long, long, int, int, float, int, float,
                    const float*, bool, hipStream_t);
 void launch_gn_bwd(const void*, const void*, const float*, const float*,
-                   const float*, const float*, float*, float*, float*, void*,
-                   int, long, long, int, int, int, float, const float*, bool,
-                   hipStream_t);
+                   const float*, const float*, float*, float*, float*,
+                   float*, int, void*, int, long, long, int, int, int, float,
+                   const float*, bool, hipStream_t);
+int gn_reduce_chunks(long, int, int, int);
+int gnmp_reduce_chunks(long, int, int, int, int);
+int gnmp_bwd_reduce_chunks(long, int, int, int);
 void launch_gnmp_fwd(const void*, void*, unsigned char*, float*, float*,
                      float*, const float*, const float*, int, long, long, int,
                      int, int, float, int, float, const float*, bool,
@@ -71,8 +74,9 @@ void launch_egnmp_bwd(const void*, const void*, const int*,
 int egnmp_reduce_chunks(long, int, int);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
-                     float*, float*, float*, void*, int, long, long, int, int,
-                     int, int, float, const float*, bool, hipStream_t);
+                     float*, float*, float*, float*, int, void*, int, long,
+                     long, int, int, int, int, float, const float*, bool,
+                     hipStream_t);
 
 namespace {
 
@@ -259,14 +263,17 @@ std::vector<torch::Tensor> group_norm_act_fwd(torch::Tensor x, int64_t G,
   const int rows = B * G;
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto& ws = persistent_ws((long)rows * 2, fopt);
+  // per-call deterministic partial-sum scratch (one slot pair per reduce
+  // block) -- no persistent zeroed workspace, no finalize launch
+  const int rchunks = gn_reduce_chunks(S, B, C, (int)G);
+  auto scratch = torch::empty({(long)B * C * 2 * rchunks}, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty_like(x);
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32,
               "group_norm_act: dtype must be float32 or bfloat16");
-  launch_gn_fwd(x.data_ptr(), y.data_ptr(), ws.data_ptr<float>(),
+  launch_gn_fwd(x.data_ptr(), y.data_ptr(), scratch.data_ptr<float>(),
                 mean.data_ptr<float>(), rstd.data_ptr<float>(),
                 weight.data_ptr<float>(), bias.data_ptr<float>(), rows,
                 row_len, S, C, (int)G, (float)eps, (int)act, (float)slope,
@@ -325,22 +332,39 @@ std::vector<torch::Tensor> group_norm_act_bwd(torch::Tensor dy, torch::Tensor x,
   const int rows = B * G;
   const long row_len = (C / G) * S;
   auto fopt = x.options().dtype(torch::kFloat32);
-  // one zeroed workspace: [rows*2 row sums | C*2 channel sums | 1 d slope]
-  auto& ws = persistent_ws((long)rows * 2 + C * 2 + 1, fopt);
-  float* row_ws = ws.data_ptr<float>();
-  float* chan_ws = row_ws + rows * 2;
-  float* slope_ws = chan_ws + C * 2;
+  const int rchunks = gn_reduce_chunks(S, B, C, (int)G);
+  auto scratch = torch::empty({(long)B * C * 5 * rchunks}, fopt);
   auto dx = torch::empty_like(x);
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32,
               "group_norm_act: dtype must be float32 or bfloat16");
   TORCH_CHECK(dy.scalar_type() == x.scalar_type(), "dy/x dtype mismatch");
+  // channel grads drain inside the apply pass: into fresh tensors, or
+  // ACCUMULATED into the parameters' grad buffers (deferred mode)
+  torch::Tensor dweight, dbias, dslope;
+  float *dwp, *dbp, *dsp;
+  int accumulate = 0;
+  if (wtarget.has_value()) {
+    accumulate = 1;
+    dwp = wtarget->data_ptr<float>();
+    dbp = btarget->data_ptr<float>();
+    dsp = (starget.has_value() && starget->defined() && starget->numel() == 1)
+              ? starget->data_ptr<float>() : nullptr;
+  } else {
+    dweight = torch::empty({C}, fopt);
+    dbias = torch::empty({C}, fopt);
+    dslope = torch::empty({1}, fopt);
+    dwp = dweight.data_ptr<float>();
+    dbp = dbias.data_ptr<float>();
+    dsp = dslope.data_ptr<float>();
+  }
   launch_gn_bwd(dy.data_ptr(), x.data_ptr(), mean.data_ptr<float>(),
                 rstd.data_ptr<float>(), weight.data_ptr<float>(),
-                bias.data_ptr<float>(), row_ws, chan_ws, slope_ws,
-                dx.data_ptr(), rows, row_len, S, C, (int)G, (int)act,
-                (float)slope, slope_ptr, bf16, stream());
-  return gn_grads_finish(ws, dx, rows, C, fopt, wtarget, btarget, starget);
+                bias.data_ptr<float>(), scratch.data_ptr<float>(), dwp, dbp,
+                dsp, accumulate, dx.data_ptr(), rows, row_len, S, C, (int)G,
+                (int)act, (float)slope, slope_ptr, bf16, stream());
+  if (accumulate) return {dx};
+  return {dx, dweight, dbias, dslope};
 }
 
 // x (B, C, K, N); GN stats over full (K, N); returns pooled
@@ -363,7 +387,8 @@ std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto& ws = persistent_ws((long)rows * 2, fopt);
+  const int rchunks = gnmp_reduce_chunks(N, K, B, C, (int)G);
+  auto scratch = torch::empty({(long)B * C * 2 * rchunks}, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty({B, C, N}, x.options());
@@ -371,7 +396,7 @@ std::vector<torch::Tensor> group_norm_act_maxpool_fwd(
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32, "fp32/bf16 only");
   launch_gnmp_fwd(x.data_ptr(), y.data_ptr(), am.data_ptr<unsigned char>(),
-                  ws.data_ptr<float>(), mean.data_ptr<float>(),
+                  scratch.data_ptr<float>(), mean.data_ptr<float>(),
                   rstd.data_ptr<float>(), weight.data_ptr<float>(),
                   bias.data_ptr<float>(), rows, row_len, N, K, C, (int)G,
                   (float)eps, (int)act, (float)slope, slope_ptr, bf16, stream());
@@ -396,18 +421,35 @@ std::vector<torch::Tensor> group_norm_act_maxpool_bwd(
   const int rows = B * G;
   const long row_len = (C / G) * K * N;
   auto fopt = x.options().dtype(torch::kFloat32);
-  auto& ws = persistent_ws((long)rows * 2 + C * 2 + 1, fopt);
-  float* row_ws = ws.data_ptr<float>();
-  float* chan_ws = row_ws + rows * 2;
-  float* slope_ws = chan_ws + C * 2;
+  const int rchunks = gnmp_bwd_reduce_chunks(N, B, C, (int)G);
+  auto scratch = torch::empty({(long)B * C * 5 * rchunks}, fopt);
   auto dx = torch::empty_like(x);
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
+  torch::Tensor dweight, dbias, dslope;
+  float *dwp, *dbp, *dsp;
+  int accumulate = 0;
+  if (wtarget.has_value()) {
+    accumulate = 1;
+    dwp = wtarget->data_ptr<float>();
+    dbp = btarget->data_ptr<float>();
+    dsp = (starget.has_value() && starget->defined() && starget->numel() == 1)
+              ? starget->data_ptr<float>() : nullptr;
+  } else {
+    dweight = torch::empty({C}, fopt);
+    dbias = torch::empty({C}, fopt);
+    dslope = torch::empty({1}, fopt);
+    dwp = dweight.data_ptr<float>();
+    dbp = dbias.data_ptr<float>();
+    dsp = dslope.data_ptr<float>();
+  }
   launch_gnmp_bwd(dy.data_ptr(), x.data_ptr(), am.data_ptr<unsigned char>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                  weight.data_ptr<float>(), bias.data_ptr<float>(), row_ws,
-                  chan_ws, slope_ws, dx.data_ptr(), rows, row_len, N, K, C,
-                  (int)G, (int)act, (float)slope, slope_ptr, bf16, stream());
-  return gn_grads_finish(ws, dx, rows, C, fopt, wtarget, btarget, starget);
+                  weight.data_ptr<float>(), bias.data_ptr<float>(),
+                  scratch.data_ptr<float>(), dwp, dbp, dsp, accumulate,
+                  dx.data_ptr(), rows, row_len, N, K, C, (int)G, (int)act,
+                  (float)slope, slope_ptr, bf16, stream());
+  if (accumulate) return {dx};
+  return {dx, dweight, dbias, dslope};
 }
 
 // SetConv stage 1 on the linearly-restructured operands: wg (B, N, M)

@@ -69,10 +69,16 @@ DEV_INLINE float block_sum(float v) {
 
 // ---------------------------------------------------------------- forward
 
-// pass 1: per-channel-slice partial sum/sumsq -> ws[row] = {sum, sumsq}
+// pass 1: per-channel-slice partial sum/sumsq, one DETERMINISTIC slot per
+// block: scratch[((b*C + c)*2 + {0,1})*chunks + chunk].  (Round-1 used
+// 2 block-reduced atomics into a tiny workspace, which needed grid caps
+// to bound same-address serialization, a zeroed persistent buffer, and a
+// separate finalize kernel; the partial-slot scheme needs none of those
+// -- the apply pass folds the handful of partials inline.)
 template <typename T>
 __global__ __launch_bounds__(GN_THREADS) void gn_fwd_reduce_kernel(
-    const T *__restrict__ x, float *__restrict__ ws, long S, int C, int G) {
+    const T *__restrict__ x, float *__restrict__ scratch, long S, int C,
+    int G) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
   const int c = blockIdx.y;
@@ -104,13 +110,34 @@ __global__ __launch_bounds__(GN_THREADS) void gn_fwd_reduce_kernel(
   }
   s = block_sum(s);
   ss = block_sum(ss);
+  (void)row;
   if (threadIdx.x == 0) {
-    atomicAdd(&ws[row * 2 + 0], s);
-    atomicAdd(&ws[row * 2 + 1], ss);
+    const long base = ((long)(b * C + c) * 2) * gridDim.x + blockIdx.x;
+    scratch[base] = s;
+    scratch[base + gridDim.x] = ss;
   }
 }
 
-// pass 2: rows threads -> mean/rstd.  The workspace is self-cleaning:
+// fold this row's per-block partials (Cg channels x chunks slots each)
+DEV_INLINE void gn_fold_stats(const float *scratch, int b, int C, int c,
+                              int Cg, int chunks, long row_len, float eps,
+                              float &m, float &r) {
+  const int c0 = (c / Cg) * Cg;
+  float s = 0.f, ss = 0.f;
+  for (int cc = c0; cc < c0 + Cg; ++cc) {
+    const float *p = scratch + ((long)(b * C + cc) * 2) * chunks;
+    for (int u = 0; u < chunks; ++u) {
+      s += p[u];
+      ss += p[u + chunks];
+    }
+  }
+  m = s / (float)row_len;
+  const float var = ss / (float)row_len - m * m;
+  r = rsqrtf(fmaxf(var, 0.f) + eps);
+}
+
+// pass 2 (legacy, used by the edge_gnmp path): rows threads -> mean/rstd.
+// The workspace is self-cleaning:
 // finalize is its only consumer and writes it back to zero, so the SAME
 // persistent buffer serves every call (and every hipGraph replay) with no
 // per-call allocation or fill kernel -- calls are ordered on the stream.
@@ -128,21 +155,28 @@ __global__ void gn_fwd_finalize_kernel(float *__restrict__ ws,
   rstd[row] = rsqrtf(fmaxf(var, 0.f) + eps);
 }
 
-// pass 3: y = act((x - mean) * rstd * gamma + beta), per channel slice
+// pass 3: y = act((x - mean) * rstd * gamma + beta), per channel slice.
+// Stats are folded inline from the reduce partials; the (blockIdx.x==0,
+// first channel of the group) block also writes mean/rstd for backward.
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
-    const T *__restrict__ x, T *__restrict__ y, const float *__restrict__ mean,
-    const float *__restrict__ rstd, const float *__restrict__ gamma,
-    const float *__restrict__ beta, long S, int C, int G, float slope,
-    const float *__restrict__ slope_ptr) {
+    const T *__restrict__ x, T *__restrict__ y,
+    const float *__restrict__ scratch, int chunks, float *__restrict__ mean,
+    float *__restrict__ rstd, const float *__restrict__ gamma,
+    const float *__restrict__ beta, long S, int C, int G, long row_len,
+    float eps, float slope, const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
   const int c = blockIdx.y;
   const int b = blockIdx.z;
   const int Cg = C / G;
   const int row = b * G + c / Cg;
-  const float m = mean[row];
-  const float r = rstd[row];
+  float m, r;
+  gn_fold_stats(scratch, b, C, c, Cg, chunks, row_len, eps, m, r);
+  if (blockIdx.x == 0 && c % Cg == 0 && threadIdx.x == 0) {
+    mean[row] = m;
+    rstd[row] = r;
+  }
   const float ga = gamma[c], be = beta[c];
   if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const T *xb = x + ((long)b * C + c) * S;
@@ -175,15 +209,15 @@ __global__ __launch_bounds__(GN_THREADS) void gn_fwd_apply_kernel(
 // ---------------------------------------------------------------- backward
 
 // partial sums: per-row {sum dxhat, sum dxhat*xhat}, per-channel
-// {sum dy_norm, sum dy_norm*xhat}, optional d slope; one channel slice per
-// block, four (five) block-reduced atomics.
+// {sum dy_norm, sum dy_norm*xhat}, optional d slope; one channel slice
+// per block, FIVE deterministic partial slots per block at
+// scratch[((b*C + c)*5 + slot)*chunks + chunk].
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
-    float *__restrict__ row_ws, float *__restrict__ chan_ws,
-    float *__restrict__ slope_ws, long S, int C, int G, float slope,
+    float *__restrict__ scratch, long S, int C, int G, float slope,
     const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
@@ -237,12 +271,70 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_reduce_kernel(
   c_db = block_sum(c_db);
   c_dg = block_sum(c_dg);
   if (ACT == 2) d_sl = block_sum(d_sl);
+  (void)row;
   if (threadIdx.x == 0) {
-    atomicAdd(&row_ws[row * 2 + 0], sum_dx);
-    atomicAdd(&row_ws[row * 2 + 1], sum_dxx);
-    atomicAdd(&chan_ws[c * 2 + 0], c_db);
-    atomicAdd(&chan_ws[c * 2 + 1], c_dg);
-    if (ACT == 2) atomicAdd(slope_ws, d_sl);
+    const long base = ((long)(b * C + c) * 5) * gridDim.x + blockIdx.x;
+    scratch[base + 0 * gridDim.x] = sum_dx;
+    scratch[base + 1 * gridDim.x] = sum_dxx;
+    scratch[base + 2 * gridDim.x] = c_db;
+    scratch[base + 3 * gridDim.x] = c_dg;
+    scratch[base + 4 * gridDim.x] = ACT == 2 ? d_sl : 0.f;
+  }
+}
+
+// fold the row's {sum dxhat, sum dxhat*xhat} from the 5-slot partials
+DEV_INLINE void gn_fold_row(const float *scratch, int b, int C, int c,
+                            int Cg, int chunks, float &s1, float &s2) {
+  const int c0 = (c / Cg) * Cg;
+  s1 = 0.f;
+  s2 = 0.f;
+  for (int cc = c0; cc < c0 + Cg; ++cc) {
+    const float *p = scratch + ((long)(b * C + cc) * 5) * chunks;
+    for (int u = 0; u < chunks; ++u) {
+      s1 += p[u];
+      s2 += p[u + chunks];
+    }
+  }
+}
+
+// channel-gradient drain: the (blockIdx.x==0, b==0) block of channel c
+// folds its dgamma/dbeta partials over (b', chunk) and writes (or, in
+// deferred mode, ACCUMULATES into the parameters' grad buffers); block
+// (0, 0, 0) additionally folds d slope over everything.  The partials are
+// complete before any apply block launches, so no cross-block
+// synchronization is needed.
+DEV_INLINE void gn_drain_channel_grads(const float *scratch, int chunks,
+                                       int B, int C, int c, int act,
+                                       float *dweight, float *dbias,
+                                       float *dslope, int accumulate) {
+  if (blockIdx.x != 0 || blockIdx.z != 0) return;
+  if (threadIdx.x == 0) {
+    float db = 0.f, dg = 0.f;
+    for (int b = 0; b < B; ++b) {
+      const float *p = scratch + ((long)(b * C + c) * 5) * chunks;
+      for (int u = 0; u < chunks; ++u) {
+        db += p[u + 2 * chunks];
+        dg += p[u + 3 * chunks];
+      }
+    }
+    if (accumulate) {
+      dbias[c] += db;
+      dweight[c] += dg;
+    } else {
+      dbias[c] = db;
+      dweight[c] = dg;
+    }
+  }
+  if (act == 2 && c == 0 && threadIdx.x == 1 && dslope != nullptr) {
+    float dsl = 0.f;
+    for (long i = 0; i < (long)B * C; ++i) {
+      const float *p = scratch + i * 5 * chunks;
+      for (int u = 0; u < chunks; ++u) dsl += p[u + 4 * chunks];
+    }
+    if (accumulate)
+      dslope[0] += dsl;
+    else
+      dslope[0] = dsl;
   }
 }
 
@@ -251,8 +343,10 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
-    const float *__restrict__ row_ws, T *__restrict__ dx, long S, int C,
-    int G, long row_len, float slope, const float *__restrict__ slope_ptr) {
+    const float *__restrict__ scratch, int chunks, float *__restrict__ dweight,
+    float *__restrict__ dbias, float *__restrict__ dslope, int accumulate,
+    T *__restrict__ dx, long S, int C, int G, long row_len, float slope,
+    const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
   const int c = blockIdx.y;
@@ -264,8 +358,10 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
   const float ga = gamma[c], be = beta[c];
   if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const float inv_n = 1.0f / (float)row_len;
-  const float s1 = row_ws[row * 2 + 0];
-  const float s2 = row_ws[row * 2 + 1];
+  float s1, s2;
+  gn_fold_row(scratch, b, C, c, Cg, chunks, s1, s2);
+  gn_drain_channel_grads(scratch, chunks, (int)gridDim.z, C, c, ACT, dweight,
+                         dbias, dslope, accumulate);
   const T *xb = x + ((long)b * C + c) * S;
   const T *dyb = dy + ((long)b * C + c) * S;
   T *dxb = dx + ((long)b * C + c) * S;
@@ -321,17 +417,23 @@ __global__ __launch_bounds__(GN_THREADS) void gn_bwd_apply_kernel(
 template <typename T, int ACT>
 __global__ __launch_bounds__(GN_THREADS) void gnmp_fwd_apply_kernel(
     const T *__restrict__ x, T *__restrict__ y, unsigned char *__restrict__ am,
-    const float *__restrict__ mean, const float *__restrict__ rstd,
-    const float *__restrict__ gamma, const float *__restrict__ beta, long N,
-    int K, int C, int G, float slope, const float *__restrict__ slope_ptr) {
+    const float *__restrict__ scratch, int chunks, float *__restrict__ mean,
+    float *__restrict__ rstd, const float *__restrict__ gamma,
+    const float *__restrict__ beta, long N, int K, int C, int G,
+    long row_len, float eps, float slope,
+    const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
   const int c = blockIdx.y;
   const int b = blockIdx.z;
   const int Cg = C / G;
   const int row = b * G + c / Cg;
-  const float m = mean[row];
-  const float r = rstd[row];
+  float m, r;
+  gn_fold_stats(scratch, b, C, c, Cg, chunks, row_len, eps, m, r);
+  if (blockIdx.x == 0 && c % Cg == 0 && threadIdx.x == 0) {
+    mean[row] = m;
+    rstd[row] = r;
+  }
   const float ga = gamma[c], be = beta[c];
   if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const T *xb = x + ((long)b * C + c) * K * N;
@@ -392,9 +494,8 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_reduce_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
     const unsigned char *__restrict__ am, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
-    const float *__restrict__ beta, float *__restrict__ row_ws,
-    float *__restrict__ chan_ws, float *__restrict__ slope_ws, long N, int K,
-    int C, int G, float slope, const float *__restrict__ slope_ptr) {
+    const float *__restrict__ beta, float *__restrict__ scratch, long N,
+    int K, int C, int G, float slope, const float *__restrict__ slope_ptr) {
   const int c = blockIdx.y;
   const int b = blockIdx.z;
   const int Cg = C / G;
@@ -429,12 +530,14 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_reduce_kernel(
   c_db = block_sum(c_db);
   c_dg = block_sum(c_dg);
   if (ACT == 2) d_sl = block_sum(d_sl);
+  (void)row;
   if (threadIdx.x == 0) {
-    atomicAdd(&row_ws[row * 2 + 0], sum_dx);
-    atomicAdd(&row_ws[row * 2 + 1], sum_dxx);
-    atomicAdd(&chan_ws[c * 2 + 0], c_db);
-    atomicAdd(&chan_ws[c * 2 + 1], c_dg);
-    if (ACT == 2) atomicAdd(slope_ws, d_sl);
+    const long base = ((long)(b * C + c) * 5) * gridDim.x + blockIdx.x;
+    scratch[base + 0 * gridDim.x] = sum_dx;
+    scratch[base + 1 * gridDim.x] = sum_dxx;
+    scratch[base + 2 * gridDim.x] = c_db;
+    scratch[base + 3 * gridDim.x] = c_dg;
+    scratch[base + 4 * gridDim.x] = ACT == 2 ? d_sl : 0.f;
   }
 }
 
@@ -445,9 +548,11 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ x,
     const unsigned char *__restrict__ am, const float *__restrict__ mean,
     const float *__restrict__ rstd, const float *__restrict__ gamma,
-    const float *__restrict__ beta, const float *__restrict__ row_ws,
-    T *__restrict__ dx, long N, int K, int C, int G, long row_len,
-    int ksplit, float slope, const float *__restrict__ slope_ptr) {
+    const float *__restrict__ beta, const float *__restrict__ scratch,
+    int chunks, float *__restrict__ dweight, float *__restrict__ dbias,
+    float *__restrict__ dslope, int accumulate, T *__restrict__ dx, long N,
+    int K, int C, int G, long row_len, int ksplit, float slope,
+    const float *__restrict__ slope_ptr) {
   constexpr int W = VecT<T>::W;
   using V = typename VecT<T>::type;
   // blockIdx.x jointly spans spatial chunks and a strided K split (the
@@ -464,8 +569,11 @@ __global__ __launch_bounds__(GN_THREADS) void gnmp_bwd_apply_kernel(
   const float ga = gamma[c], be = beta[c];
   if (ACT >= 1) slope = act_slope<ACT>(slope, slope_ptr);
   const float inv_n = 1.0f / (float)row_len;
-  const float s1 = row_ws[row * 2 + 0];
-  const float s2 = row_ws[row * 2 + 1];
+  float s1, s2;
+  gn_fold_row(scratch, b, C, c, Cg, chunks, s1, s2);
+  if (kc == 0)
+    gn_drain_channel_grads(scratch, chunks, (int)gridDim.z, C, c, ACT,
+                           dweight, dbias, dslope, accumulate);
   const long base = ((long)b * C + c) * K * N;
   const long pooled = ((long)b * C + c) * N;
   if (N % W == 0) {
@@ -545,23 +653,28 @@ static int pick_chunks(long spatial, long bc, int cg = 0) {
   return (int)chunks;
 }
 
+int gn_reduce_chunks(long S, int B, int C, int G) {
+  return pick_chunks(S, (long)B * C, C / G);
+}
+int gnmp_reduce_chunks(long N, int K, int B, int C, int G) {
+  return pick_chunks((long)K * N, (long)B * C, C / G);
+}
+
 template <typename T>
-void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
+void gn_fwd_impl(const T *x, T *y, float *scratch, float *mean, float *rstd,
                  const float *gamma, const float *beta, int rows, long row_len,
                  long S, int C, int G, float eps, int act, float slope,
                  const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
-  // reduce grid bounds atomics per row; apply has no atomics -> fill freely
-  const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
+  const int rchunks = pick_chunks(S, (long)B * C, C / G);
+  const dim3 rgrid(rchunks, C, B);
   const dim3 grid(pick_chunks(S, (long)B * C), C, B);
   hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, rgrid, dim3(GN_THREADS), 0,
-                     stream, x, ws, S, C, G);
-  hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
-                     0, stream, ws, mean, rstd, row_len, rows, eps);
+                     stream, x, scratch, S, C, G);
 #define GN_FWD_APPLY(A)                                                        \
   hipLaunchKernelGGL((gn_fwd_apply_kernel<T, A>), grid, dim3(GN_THREADS), 0,   \
-                     stream, x, y, mean, rstd, gamma, beta, S, C, G, slope,    \
-                     slope_ptr)
+                     stream, x, y, scratch, rchunks, mean, rstd, gamma, beta,  \
+                     S, C, G, row_len, eps, slope, slope_ptr)
   if (act == 2) GN_FWD_APPLY(2);
   else if (act == 1) GN_FWD_APPLY(1);
   else GN_FWD_APPLY(0);
@@ -570,21 +683,23 @@ void gn_fwd_impl(const T *x, T *y, float *ws, float *mean, float *rstd,
 
 template <typename T>
 void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
-                 const float *gamma, const float *beta, float *row_ws,
-                 float *chan_ws, float *slope_ws, T *dx, int rows, long row_len,
-                 long S, int C, int G, int act, float slope,
-                 const float *slope_ptr, hipStream_t stream) {
+                 const float *gamma, const float *beta, float *scratch,
+                 float *dweight, float *dbias, float *dslope, int accumulate,
+                 T *dx, int rows, long row_len, long S, int C, int G, int act,
+                 float slope, const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
-  const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
+  const int rchunks = pick_chunks(S, (long)B * C, C / G);
+  const dim3 rgrid(rchunks, C, B);
   const dim3 grid(pick_chunks(S, (long)B * C), C, B);
 #define GN_BWD(A)                                                              \
   do {                                                                         \
     hipLaunchKernelGGL((gn_bwd_reduce_kernel<T, A>), rgrid, dim3(GN_THREADS),  \
-                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws,      \
-                       chan_ws, slope_ws, S, C, G, slope, slope_ptr);          \
+                       0, stream, dy, x, mean, rstd, gamma, beta, scratch, S,  \
+                       C, G, slope, slope_ptr);                                \
     hipLaunchKernelGGL((gn_bwd_apply_kernel<T, A>), grid, dim3(GN_THREADS),    \
-                       0, stream, dy, x, mean, rstd, gamma, beta, row_ws, dx,  \
-                       S, C, G, row_len, slope, slope_ptr);                    \
+                       0, stream, dy, x, mean, rstd, gamma, beta, scratch,     \
+                       rchunks, dweight, dbias, dslope, accumulate, dx, S, C,  \
+                       G, row_len, slope, slope_ptr);                          \
   } while (0)
   if (act == 2) GN_BWD(2);
   else if (act == 1) GN_BWD(1);
@@ -593,18 +708,17 @@ void gn_bwd_impl(const T *dy, const T *x, const float *mean, const float *rstd,
 }
 
 template <typename T>
-void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
-                   float *rstd, const float *gamma, const float *beta, int rows,
-                   long row_len, long N, int K, int C, int G, float eps,
-                   int act, float slope, const float *slope_ptr,
-                   hipStream_t stream) {
+void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *scratch,
+                   float *mean, float *rstd, const float *gamma,
+                   const float *beta, int rows, long row_len, long N, int K,
+                   int C, int G, float eps, int act, float slope,
+                   const float *slope_ptr, hipStream_t stream) {
   const int B = rows / G;
   const long S = (long)K * N;
-  const dim3 rgrid(pick_chunks(S, (long)B * C, C / G), C, B);
+  const int rchunks = pick_chunks(S, (long)B * C, C / G);
+  const dim3 rgrid(rchunks, C, B);
   hipLaunchKernelGGL(gn_fwd_reduce_kernel<T>, rgrid, dim3(GN_THREADS), 0,
-                     stream, x, ws, S, C, G);
-  hipLaunchKernelGGL(gn_fwd_finalize_kernel, dim3((rows + 255) / 256), dim3(256),
-                     0, stream, ws, mean, rstd, row_len, rows, eps);
+                     stream, x, scratch, S, C, G);
   // the apply loops all K per thread (argmax), so a thread's work is K x
   // its vector count: allow chunking down to ~1 vector per thread instead
   // of pick_chunks' 16-iteration floor (C*B is small -> it underfilled)
@@ -615,23 +729,31 @@ void gnmp_fwd_impl(const T *x, T *y, unsigned char *am, float *ws, float *mean,
   const dim3 pgrid((unsigned)pch, C, B);
 #define GNMP_FWD(A)                                                            \
   hipLaunchKernelGGL((gnmp_fwd_apply_kernel<T, A>), pgrid, dim3(GN_THREADS),   \
-                     0, stream, x, y, am, mean, rstd, gamma, beta, N, K, C, G, \
-                     slope, slope_ptr)
+                     0, stream, x, y, am, scratch, rchunks, mean, rstd,        \
+                     gamma, beta, N, K, C, G, row_len, eps, slope, slope_ptr)
   if (act == 2) GNMP_FWD(2);
   else if (act == 1) GNMP_FWD(1);
   else GNMP_FWD(0);
 #undef GNMP_FWD
 }
 
+// NOTE: the gnmp backward REDUCE runs over the pooled domain (N), so its
+// chunk count must match what the binding sized the scratch for
+int gnmp_bwd_reduce_chunks(long N, int B, int C, int G) {
+  return pick_chunks(N, (long)B * C, C / G);
+}
+
 template <typename T>
 void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
                    const float *mean, const float *rstd, const float *gamma,
-                   const float *beta, float *row_ws, float *chan_ws,
-                   float *slope_ws, T *dx, int rows, long row_len, long N,
-                   int K, int C, int G, int act, float slope,
-                   const float *slope_ptr, hipStream_t stream) {
+                   const float *beta, float *scratch, float *dweight,
+                   float *dbias, float *dslope, int accumulate, T *dx,
+                   int rows, long row_len, long N, int K, int C, int G,
+                   int act, float slope, const float *slope_ptr,
+                   hipStream_t stream) {
   const int B = rows / G;
-  const dim3 rgrid(pick_chunks(N, (long)B * C, C / G), C, B);
+  const int rchunks = pick_chunks(N, (long)B * C, C / G);
+  const dim3 rgrid(rchunks, C, B);
   const int nchunks = pick_chunks(N, (long)B * C);
   int ksplit = (int)(1024 / ((long)nchunks * B * C));
   if (ksplit > K) ksplit = K;
@@ -641,12 +763,12 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
   do {                                                                         \
     hipLaunchKernelGGL((gnmp_bwd_reduce_kernel<T, A>), rgrid,                  \
                        dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,     \
-                       gamma, beta, row_ws, chan_ws, slope_ws, N, K, C, G,     \
-                       slope, slope_ptr);                                      \
+                       gamma, beta, scratch, N, K, C, G, slope, slope_ptr);    \
     hipLaunchKernelGGL((gnmp_bwd_apply_kernel<T, A>), grid,                    \
                        dim3(GN_THREADS), 0, stream, dy, x, am, mean, rstd,     \
-                       gamma, beta, row_ws, dx, N, K, C, G, row_len, ksplit,   \
-                       slope, slope_ptr);                                      \
+                       gamma, beta, scratch, rchunks, dweight, dbias, dslope,  \
+                       accumulate, dx, N, K, C, G, row_len, ksplit, slope,     \
+                       slope_ptr);                                             \
   } while (0)
   if (act == 2) GNMP_BWD(2);
   else if (act == 1) GNMP_BWD(1);
@@ -655,71 +777,77 @@ void gnmp_bwd_impl(const T *dy, const T *x, const unsigned char *am,
 }
 
 // type-erased entry points (bindings.cpp is host-compiled, no HIP types)
-void launch_gn_fwd(const void *x, void *y, float *ws, float *mean, float *rstd,
-                   const float *gamma, const float *beta, int rows,
-                   long row_len, long S, int C, int G, float eps, int act,
-                   float slope, const float *slope_ptr, bool bf16,
+void launch_gn_fwd(const void *x, void *y, float *scratch, float *mean,
+                   float *rstd, const float *gamma, const float *beta,
+                   int rows, long row_len, long S, int C, int G, float eps,
+                   int act, float slope, const float *slope_ptr, bool bf16,
                    hipStream_t stream) {
   if (bf16)
     gn_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)x, (__hip_bfloat16 *)y,
-                                ws, mean, rstd, gamma, beta, rows, row_len, S,
-                                C, G, eps, act, slope, slope_ptr, stream);
+                                scratch, mean, rstd, gamma, beta, rows,
+                                row_len, S, C, G, eps, act, slope, slope_ptr,
+                                stream);
   else
-    gn_fwd_impl<float>((const float *)x, (float *)y, ws, mean, rstd, gamma,
-                       beta, rows, row_len, S, C, G, eps, act, slope,
+    gn_fwd_impl<float>((const float *)x, (float *)y, scratch, mean, rstd,
+                       gamma, beta, rows, row_len, S, C, G, eps, act, slope,
                        slope_ptr, stream);
 }
 
 void launch_gn_bwd(const void *dy, const void *x, const float *mean,
                    const float *rstd, const float *gamma, const float *beta,
-                   float *row_ws, float *chan_ws, float *slope_ws, void *dx,
-                   int rows, long row_len, long S, int C, int G, int act,
-                   float slope, const float *slope_ptr, bool bf16,
-                   hipStream_t stream) {
+                   float *scratch, float *dweight, float *dbias,
+                   float *dslope, int accumulate, void *dx, int rows,
+                   long row_len, long S, int C, int G, int act, float slope,
+                   const float *slope_ptr, bool bf16, hipStream_t stream) {
   if (bf16)
     gn_bwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)dy,
                                 (const __hip_bfloat16 *)x, mean, rstd, gamma,
-                                beta, row_ws, chan_ws, slope_ws,
-                                (__hip_bfloat16 *)dx, rows, row_len, S, C, G,
-                                act, slope, slope_ptr, stream);
+                                beta, scratch, dweight, dbias, dslope,
+                                accumulate, (__hip_bfloat16 *)dx, rows,
+                                row_len, S, C, G, act, slope, slope_ptr,
+                                stream);
   else
     gn_bwd_impl<float>((const float *)dy, (const float *)x, mean, rstd, gamma,
-                       beta, row_ws, chan_ws, slope_ws, (float *)dx, rows,
-                       row_len, S, C, G, act, slope, slope_ptr, stream);
+                       beta, scratch, dweight, dbias, dslope, accumulate,
+                       (float *)dx, rows, row_len, S, C, G, act, slope,
+                       slope_ptr, stream);
 }
 
-void launch_gnmp_fwd(const void *x, void *y, unsigned char *am, float *ws,
-                     float *mean, float *rstd, const float *gamma,
-                     const float *beta, int rows, long row_len, long N, int K,
-                     int C, int G, float eps, int act, float slope,
-                     const float *slope_ptr, bool bf16, hipStream_t stream) {
+void launch_gnmp_fwd(const void *x, void *y, unsigned char *am,
+                     float *scratch, float *mean, float *rstd,
+                     const float *gamma, const float *beta, int rows,
+                     long row_len, long N, int K, int C, int G, float eps,
+                     int act, float slope, const float *slope_ptr, bool bf16,
+                     hipStream_t stream) {
   if (bf16)
     gnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)x,
-                                  (__hip_bfloat16 *)y, am, ws, mean, rstd,
-                                  gamma, beta, rows, row_len, N, K, C, G, eps,
-                                  act, slope, slope_ptr, stream);
+                                  (__hip_bfloat16 *)y, am, scratch, mean,
+                                  rstd, gamma, beta, rows, row_len, N, K, C,
+                                  G, eps, act, slope, slope_ptr, stream);
   else
-    gnmp_fwd_impl<float>((const float *)x, (float *)y, am, ws, mean, rstd,
-                         gamma, beta, rows, row_len, N, K, C, G, eps, act,
-                         slope, slope_ptr, stream);
+    gnmp_fwd_impl<float>((const float *)x, (float *)y, am, scratch, mean,
+                         rstd, gamma, beta, rows, row_len, N, K, C, G, eps,
+                         act, slope, slope_ptr, stream);
 }
 
 void launch_gnmp_bwd(const void *dy, const void *x, const unsigned char *am,
                      const float *mean, const float *rstd, const float *gamma,
-                     const float *beta, float *row_ws, float *chan_ws,
-                     float *slope_ws, void *dx, int rows, long row_len, long N,
-                     int K, int C, int G, int act, float slope,
-                     const float *slope_ptr, bool bf16, hipStream_t stream) {
+                     const float *beta, float *scratch, float *dweight,
+                     float *dbias, float *dslope, int accumulate, void *dx,
+                     int rows, long row_len, long N, int K, int C, int G,
+                     int act, float slope, const float *slope_ptr, bool bf16,
+                     hipStream_t stream) {
   if (bf16)
     gnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)x, am, mean, rstd,
-        gamma, beta, row_ws, chan_ws, slope_ws, (__hip_bfloat16 *)dx, rows,
-        row_len, N, K, C, G, act, slope, slope_ptr, stream);
+        gamma, beta, scratch, dweight, dbias, dslope, accumulate,
+        (__hip_bfloat16 *)dx, rows, row_len, N, K, C, G, act, slope,
+        slope_ptr, stream);
   else
     gnmp_bwd_impl<float>((const float *)dy, (const float *)x, am, mean, rstd,
-                         gamma, beta, row_ws, chan_ws, slope_ws, (float *)dx,
-                         rows, row_len, N, K, C, G, act, slope, slope_ptr,
-                         stream);
+                         gamma, beta, scratch, dweight, dbias, dslope,
+                         accumulate, (float *)dx, rows, row_len, N, K, C, G,
+                         act, slope, slope_ptr, stream);
 }
 
 // extract dbias/dweight/dslope from the backward workspace and re-zero the
