@@ -63,9 +63,10 @@ void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
 void launch_gn_bwd_extract(float*, float*, float*, float*, int, int, int,
                            hipStream_t);
 void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
-                      float*, const float*, const float*, void*,
-                      unsigned char*, int, long, int, int, int, float, int,
-                      float, const float*, bool, int, hipStream_t);
+                      float*, const float*, const float*, void*, void*,
+                      unsigned char*, unsigned char*, void*, unsigned char*,
+                      int, long, int, int, int, float, int, float,
+                      const float*, bool, int, hipStream_t);
 void launch_egnmp_bwd(const void*, const void*, const int*,
                       const unsigned char*, const int*, const int*,
                       const float*, const float*, const float*, const float*,
@@ -487,13 +488,22 @@ std::vector<torch::Tensor> edge_gnmp_fwd(torch::Tensor wg, torch::Tensor idx,
   auto rstd = torch::empty({rows}, fopt);
   auto y = torch::empty_like(wg);
   auto am = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
+  // per-(n, c) gather extremes tracked by the reduce pass (the pick pass
+  // is elementwise; the second full gather sweep is gone)
+  auto vmax = torch::empty_like(wg);
+  auto vmin = torch::empty_like(wg);
+  auto amax = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
+  auto amin = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
   launch_egnmp_fwd(wg.data_ptr(), idx.data_ptr<int>(),
                    scratch.data_ptr<float>(), ws.data_ptr<float>(),
                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
                    weight.data_ptr<float>(), bias.data_ptr<float>(),
-                   y.data_ptr(), am.data_ptr<unsigned char>(), B, N, K, M,
-                   (int)G, (float)eps, (int)act, (float)slope, slope_ptr,
-                   bf16, rchunks, stream());
+                   vmax.data_ptr(), vmin.data_ptr(),
+                   amax.data_ptr<unsigned char>(),
+                   amin.data_ptr<unsigned char>(), y.data_ptr(),
+                   am.data_ptr<unsigned char>(), B, N, K, M, (int)G,
+                   (float)eps, (int)act, (float)slope, slope_ptr, bf16,
+                   rchunks, stream());
   return {y, am, mean, rstd};
 }
 

@@ -87,11 +87,19 @@ DEV_INLINE void split_edge(int e, int N, float invN, int &j, int &n) {
 // pass 1: per-block partial sum/sumsq per group over all (c, j, n)
 // gather-diff elements -> scratch[(out)*(gridX*B) + gx*B + b], out in
 // [0, G*2).  Deterministic (no atomics past the LDS bins).
+// ... and, per (n, c), the EXTREMES of v over j with their arg j: the
+// activation(GN(v)) is piecewise monotone in v (pre = v*(rstd*gamma) +
+// const; LeakyReLU/PReLU monotone per sign region), so the K-max-pool
+// resolves later from just (vmax, vmin) -- the apply pass needs NO gather
+// sweep at all (it was a second full pass over the 50M gathered values).
 template <typename T>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
     const T *__restrict__ wg,      // (B, N, M)
     const int *__restrict__ idx,   // (B, N, K)
     float *__restrict__ scratch,   // (B*G*2, gridX*B)
+    T *__restrict__ vmax, T *__restrict__ vmin,        // (B, N, M)
+    unsigned char *__restrict__ amax,
+    unsigned char *__restrict__ amin,                  // (B, N, M)
     long N, int K, int M, int G) {
   const int b = blockIdx.z;
   const int B = gridDim.z;
@@ -116,6 +124,13 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
          n += (long)gridDim.x * ppb) {
       const Quad<T> cq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
       const int *row = idxb + n * K;
+      float vmx[4], vmn[4];
+      int jmx[4] = {0, 0, 0, 0}, jmn[4] = {0, 0, 0, 0};
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        vmx[e] = -INFINITY;
+        vmn[e] = INFINITY;
+      }
       for (int j = 0; j < K; ++j) {
         const Quad<T> nq = *(const Quad<T> *)(wgb + (long)row[j] * M + c4 * 4);
 #pragma unroll
@@ -123,8 +138,32 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
           const float v = (float)nq.v[e] - (float)cq.v[e];
           s[e] += v;
           ss[e] += v * v;
+          if (v > vmx[e]) {
+            vmx[e] = v;
+            jmx[e] = j;
+          }
+          if (v < vmn[e]) {
+            vmn[e] = v;
+            jmn[e] = j;
+          }
         }
       }
+      Quad<T> qx, qn;
+      uchar4 ax, an;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        qx.v[e] = (T)vmx[e];
+        qn.v[e] = (T)vmn[e];
+      }
+      ax.x = (unsigned char)jmx[0]; ax.y = (unsigned char)jmx[1];
+      ax.z = (unsigned char)jmx[2]; ax.w = (unsigned char)jmx[3];
+      an.x = (unsigned char)jmn[0]; an.y = (unsigned char)jmn[1];
+      an.z = (unsigned char)jmn[2]; an.w = (unsigned char)jmn[3];
+      const long pi = ((long)b * N + n) * M + c4 * 4;
+      *(Quad<T> *)(vmax + pi) = qx;
+      *(Quad<T> *)(vmin + pi) = qn;
+      *(uchar4 *)(amax + pi) = ax;
+      *(uchar4 *)(amin + pi) = an;
     }
 #pragma unroll
     for (int e = 0; e < 4; ++e) {
@@ -155,71 +194,58 @@ __global__ void egnmp_sum_partials_kernel(const float *__restrict__ scratch,
   if (lane_id() == 0) ws[out] = acc;
 }
 
-// pass 3: normalized+act gather-diff values, max over j -> pooled yT
-// (B, N, M) + u8 argmax
+// pass 3: ELEMENTWISE pick from the tracked extremes -- for each (n, c)
+// the pooled value is max(act(gn(vmax)), act(gn(vmin))) with the matching
+// arg j (act(gn(v)) is piecewise monotone in v, so the max over all K
+// values is attained at one of the two extremes, for either sign of
+// gamma and any activation slope).
 template <typename T, int ACT>
-__global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_apply_kernel(
-    const T *__restrict__ wg, const int *__restrict__ idx,
-    const float *__restrict__ mean, const float *__restrict__ rstd,
-    const float *__restrict__ gamma, const float *__restrict__ beta,
+__global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
+    const T *__restrict__ vmax, const T *__restrict__ vmin,
+    const unsigned char *__restrict__ amax,
+    const unsigned char *__restrict__ amin, const float *__restrict__ mean,
+    const float *__restrict__ rstd, const float *__restrict__ gamma,
+    const float *__restrict__ beta,
     T *__restrict__ y,               // (B, N, M)
     unsigned char *__restrict__ am,  // (B, N, M)
-    long N, int K, int M, int G, float slope,
+    long total, long NM, int M, int G, float slope,
     const float *__restrict__ slope_ptr) {
-  const int b = blockIdx.z;
-  const int tpc = M / 4;
-  const int ppb = EG_THREADS / tpc;
-  const int p_l = (int)threadIdx.x / tpc;
-  const int c4 = (int)threadIdx.x % tpc;
-  if (p_l >= ppb) return;
-  const int Cg = M / G;
-  float m[4], r[4], ga[4], be[4];
-#pragma unroll
-  for (int e = 0; e < 4; ++e) {
-    const int c = c4 * 4 + e;
-    const int row = b * G + c / Cg;
-    m[e] = mean[row];
-    r[e] = rstd[row];
-    ga[e] = gamma[c];
-    be[e] = beta[c];
-  }
   if (ACT == 2) slope = *slope_ptr;
-
-  const T *wgb = wg + (long)b * N * M;
-  const int *idxb = idx + (long)b * N * K;
-  for (long n = (long)blockIdx.x * ppb + p_l; n < N;
-       n += (long)gridDim.x * ppb) {
-    const Quad<T> cq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
-    const int *irow = idxb + n * K;
-    float best[4];
-    int bk[4];
-#pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      best[e] = -INFINITY;
-      bk[e] = 0;
-    }
-    for (int j = 0; j < K; ++j) {
-      const Quad<T> nq = *(const Quad<T> *)(wgb + (long)irow[j] * M + c4 * 4);
-#pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        float v = ((float)nq.v[e] - (float)cq.v[e] - m[e]) * r[e] * ga[e] + be[e];
-        if (ACT >= 1) v = v > 0.f ? v : v * slope;
-        if (v > best[e]) {
-          best[e] = v;
-          bk[e] = j;
-        }
-      }
-    }
+  const int Cg = M / G;
+  for (long i4 = (long)blockIdx.x * EG_THREADS + threadIdx.x; i4 * 4 < total;
+       i4 += (long)gridDim.x * EG_THREADS) {
+    const long i = i4 * 4;
+    const long b = i / NM;
+    const int c0 = (int)(i % M);
+    const Quad<T> qx = *(const Quad<T> *)(vmax + i);
+    const Quad<T> qn = *(const Quad<T> *)(vmin + i);
+    const uchar4 ax = *(const uchar4 *)(amax + i);
+    const uchar4 an = *(const uchar4 *)(amin + i);
+    const int axs[4] = {ax.x, ax.y, ax.z, ax.w};
+    const int ans[4] = {an.x, an.y, an.z, an.w};
     Quad<T> oq;
     uchar4 aq;
+    unsigned char out_j[4];
 #pragma unroll
-    for (int e = 0; e < 4; ++e) oq.v[e] = (T)best[e];
-    aq.x = (unsigned char)bk[0];
-    aq.y = (unsigned char)bk[1];
-    aq.z = (unsigned char)bk[2];
-    aq.w = (unsigned char)bk[3];
-    *(Quad<T> *)(y + ((long)b * N + n) * M + c4 * 4) = oq;
-    *(uchar4 *)(am + ((long)b * N + n) * M + c4 * 4) = aq;
+    for (int e = 0; e < 4; ++e) {
+      const int c = c0 + e;
+      const int row = (int)b * G + c / Cg;
+      const float m = mean[row];
+      const float r = rstd[row];
+      const float ga = gamma[c], be = beta[c];
+      float vhi = ((float)qx.v[e] - m) * r * ga + be;
+      float vlo = ((float)qn.v[e] - m) * r * ga + be;
+      if (ACT >= 1) {
+        vhi = vhi > 0.f ? vhi : vhi * slope;
+        vlo = vlo > 0.f ? vlo : vlo * slope;
+      }
+      const bool hi = vhi >= vlo;
+      oq.v[e] = (T)(hi ? vhi : vlo);
+      out_j[e] = (unsigned char)(hi ? axs[e] : ans[e]);
+    }
+    aq.x = out_j[0]; aq.y = out_j[1]; aq.z = out_j[2]; aq.w = out_j[3];
+    *(Quad<T> *)(y + i) = oq;
+    *(uchar4 *)(am + i) = aq;
   }
 }
 
@@ -431,16 +457,16 @@ static int eg_chunks(long N, int ppb, int B, int cap) {
 template <typename T>
 void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
                     float *mean, float *rstd, const float *gamma,
-                    const float *beta, T *y, unsigned char *am, int B, long N,
-                    int K, int M, int G, float eps, int act, float slope,
-                    const float *slope_ptr, int rchunks, hipStream_t stream) {
-  const int tpc = M / 4;
-  const int ppb = EG_THREADS / tpc;
+                    const float *beta, T *vmax, T *vmin, unsigned char *amax,
+                    unsigned char *amin, T *y, unsigned char *am, int B,
+                    long N, int K, int M, int G, float eps, int act,
+                    float slope, const float *slope_ptr, int rchunks,
+                    hipStream_t stream) {
   const dim3 rgrid(rchunks, 1, B);
   const int n_out_f = B * G * 2;
   hipLaunchKernelGGL(egnmp_fwd_reduce_kernel<T>, rgrid, dim3(EG_THREADS),
                      (size_t)n_out_f * sizeof(float), stream, wg, idx,
-                     scratch, N, K, M, G);
+                     scratch, vmax, vmin, amax, amin, N, K, M, G);
   const int waves_per_block = EG_THREADS / WAVE;
   hipLaunchKernelGGL(egnmp_sum_partials_kernel,
                      dim3((n_out_f + waves_per_block - 1) / waves_per_block),
@@ -448,11 +474,15 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
                      (long)rchunks * B, n_out_f);
   launch_gn_finalize(ws, mean, rstd, (long)(M / G) * K * N, B * G, eps,
                      stream);
-  const dim3 agrid(eg_chunks(N, ppb, B, 2048), 1, B);
+  const long total = (long)B * N * M;
+  long pb = (total / 4 + EG_THREADS - 1) / EG_THREADS;
+  if (pb > 2048) pb = 2048;
+  if (pb < 1) pb = 1;
 #define EG_FWD(A)                                                             \
-  hipLaunchKernelGGL((egnmp_fwd_apply_kernel<T, A>), agrid, dim3(EG_THREADS), \
-                     0, stream, wg, idx, mean, rstd, gamma, beta, y, am, N,   \
-                     K, M, G, slope, slope_ptr)
+  hipLaunchKernelGGL((egnmp_fwd_pick_kernel<T, A>), dim3((unsigned)pb),       \
+                     dim3(EG_THREADS), 0, stream, vmax, vmin, amax, amin,     \
+                     mean, rstd, gamma, beta, y, am, total, N * M, M, G,      \
+                     slope, slope_ptr)
   if (act == 2) EG_FWD(2);
   else if (act == 1) EG_FWD(1);
   else EG_FWD(0);
@@ -507,20 +537,24 @@ int egnmp_reduce_chunks(long N, int M, int B) {
 
 void launch_egnmp_fwd(const void *wg, const int *idx, float *scratch,
                       float *ws, float *mean, float *rstd, const float *gamma,
-                      const float *beta, void *y, unsigned char *am, int B,
-                      long N, int K, int M, int G, float eps, int act,
-                      float slope, const float *slope_ptr, bool bf16,
-                      int rchunks, hipStream_t stream) {
+                      const float *beta, void *vmax, void *vmin,
+                      unsigned char *amax, unsigned char *amin, void *y,
+                      unsigned char *am, int B, long N, int K, int M, int G,
+                      float eps, int act, float slope, const float *slope_ptr,
+                      bool bf16, int rchunks, hipStream_t stream) {
   if (bf16)
     egnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)wg, idx, scratch,
                                    ws, mean, rstd, gamma, beta,
+                                   (__hip_bfloat16 *)vmax,
+                                   (__hip_bfloat16 *)vmin, amax, amin,
                                    (__hip_bfloat16 *)y, am, B, N, K, M, G,
                                    eps, act, slope, slope_ptr, rchunks,
                                    stream);
   else
     egnmp_fwd_impl<float>((const float *)wg, idx, scratch, ws, mean, rstd,
-                          gamma, beta, (float *)y, am, B, N, K, M, G, eps,
-                          act, slope, slope_ptr, rchunks, stream);
+                          gamma, beta, (float *)vmax, (float *)vmin, amax,
+                          amin, (float *)y, am, B, N, K, M, G, eps, act,
+                          slope, slope_ptr, rchunks, stream);
 }
 
 void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
