@@ -203,12 +203,13 @@ template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
     const T *__restrict__ vmax, const T *__restrict__ vmin,
     const unsigned char *__restrict__ amax,
-    const unsigned char *__restrict__ amin, const float *__restrict__ mean,
-    const float *__restrict__ rstd, const float *__restrict__ gamma,
-    const float *__restrict__ beta,
+    const unsigned char *__restrict__ amin, const int *__restrict__ idx,
+    const float *__restrict__ mean, const float *__restrict__ rstd,
+    const float *__restrict__ gamma, const float *__restrict__ beta,
     T *__restrict__ y,               // (B, N, M)
     unsigned char *__restrict__ am,  // (B, N, M)
-    long total, long NM, int M, int G, float slope,
+    int *__restrict__ nbsel,         // (B, N, M): idx[n, am[n, c]]
+    long total, long NM, int M, int K, int G, float slope,
     const float *__restrict__ slope_ptr) {
   if (ACT == 2) slope = *slope_ptr;
   const int Cg = M / G;
@@ -216,6 +217,8 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
        i4 += (long)gridDim.x * EG_THREADS) {
     const long i = i4 * 4;
     const long b = i / NM;
+    const long n = (i % NM) / M;
+    const int *irow = idx + ((long)b * (NM / M) + n) * K;
     const int c0 = (int)(i % M);
     const Quad<T> qx = *(const Quad<T> *)(vmax + i);
     const Quad<T> qn = *(const Quad<T> *)(vmin + i);
@@ -246,6 +249,12 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
     aq.x = out_j[0]; aq.y = out_j[1]; aq.z = out_j[2]; aq.w = out_j[3];
     *(Quad<T> *)(y + i) = oq;
     *(uchar4 *)(am + i) = aq;
+    int4 sel;
+    sel.x = irow[out_j[0]];
+    sel.y = irow[out_j[1]];
+    sel.z = irow[out_j[2]];
+    sel.w = irow[out_j[3]];
+    *(int4 *)(nbsel + i) = sel;
   }
 }
 
@@ -345,8 +354,9 @@ template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ wg,
     const int *__restrict__ idx, const unsigned char *__restrict__ am,
-    const int *__restrict__ order,    // (B, K*N) edge ids sorted by target
+    const int *__restrict__ nbsel,    // (B, N, M): argmax neighbour ids
     const int *__restrict__ offsets,  // (B, N+1)
+    const int *__restrict__ ordn,     // (B, K*N): source n per ordered edge
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
     const float *__restrict__ row_ws, T *__restrict__ dwg, long N, int K,
@@ -360,7 +370,6 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
   if (p_l >= ppb) return;
   const int Cg = M / G;
   const float inv_n = 1.0f / (float)row_len;
-  const float invN = 1.0f / (float)N;
   float m[4], r[4], ga[4], be[4], s1[4], s2[4];
 #pragma unroll
   for (int e = 0; e < 4; ++e) {
@@ -378,8 +387,9 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
   const T *wgb = wg + (long)b * N * M;
   const T *dyb = dy + (long)b * N * M;
   const unsigned char *amb = am + (long)b * N * M;
+  const int *selb = nbsel + (long)b * N * M;
   const int *idxb = idx + (long)b * N * K;
-  const int *ordb = order + (long)b * N * K;
+  const int *ordnb = ordn + (long)b * N * K;
   const int *offb = offsets + (long)b * (N + 1);
 
   for (long p = (long)blockIdx.x * ppb + p_l; p < N;
@@ -412,22 +422,25 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
         }
       }
     }
-    // incoming term: edges whose neighbour is p
+    // incoming term: edges whose neighbour is p.  The dy(argmax) part
+    // fires iff p IS the selected argmax neighbour of (n, c) (precomputed
+    // nbsel, so no per-edge am/dy loads and no edge-id decomposition).
+    // With duplicate entries in an idx row (degenerate pad-self clouds)
+    // the term can fire once per duplicate -- degenerate-tie semantics,
+    // matching the rest of this file.
     const int lo = offb[p], hi = offb[p + 1];
     for (int t = lo; t < hi; ++t) {
-      int j, n;
-      split_edge(ordb[t], (int)N, invN, j, n);
-      const Quad<T> nq = *(const Quad<T> *)(wgb + (long)n * M + c4 * 4);
-      const uchar4 aq = *(const uchar4 *)(amb + (long)n * M + c4 * 4);
-      const Quad<T> gq = *(const Quad<T> *)(dyb + (long)n * M + c4 * 4);
-      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
+      const long n = ordnb[t];
+      const Quad<T> nq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
+      const int4 sq = *(const int4 *)(selb + n * M + c4 * 4);
+      const int sel[4] = {sq.x, sq.y, sq.z, sq.w};
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
         const float v = (float)pq.v[e] - (float)nq.v[e];
         const float xhat = (v - m[e]) * r[e];
         float dxhat = 0.f;
-        if (j == ks[e]) {
-          float gs = (float)gq.v[e];
+        if (sel[e] == (int)p) {
+          float gs = ldg(dyb + n * M + c4 * 4 + e);
           if (ACT >= 1) {
             const float pre = xhat * ga[e] + be[e];
             gs = pre > 0.f ? gs : gs * slope;
@@ -458,8 +471,8 @@ template <typename T>
 void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
                     float *mean, float *rstd, const float *gamma,
                     const float *beta, T *vmax, T *vmin, unsigned char *amax,
-                    unsigned char *amin, T *y, unsigned char *am, int B,
-                    long N, int K, int M, int G, float eps, int act,
+                    unsigned char *amin, T *y, unsigned char *am, int *nbsel,
+                    int B, long N, int K, int M, int G, float eps, int act,
                     float slope, const float *slope_ptr, int rchunks,
                     hipStream_t stream) {
   const dim3 rgrid(rchunks, 1, B);
@@ -481,8 +494,8 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
 #define EG_FWD(A)                                                             \
   hipLaunchKernelGGL((egnmp_fwd_pick_kernel<T, A>), dim3((unsigned)pb),       \
                      dim3(EG_THREADS), 0, stream, vmax, vmin, amax, amin,     \
-                     mean, rstd, gamma, beta, y, am, total, N * M, M, G,      \
-                     slope, slope_ptr)
+                     idx, mean, rstd, gamma, beta, y, am, nbsel, total,       \
+                     N * M, M, K, G, slope, slope_ptr)
   if (act == 2) EG_FWD(2);
   else if (act == 1) EG_FWD(1);
   else EG_FWD(0);
@@ -491,11 +504,12 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
 
 template <typename T>
 void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
-                    const unsigned char *am, const int *order,
-                    const int *offsets, const float *mean, const float *rstd,
-                    const float *gamma, const float *beta, float *scratch,
-                    float *ws, T *dwg, int B, long N, int K, int M, int G,
-                    int act, float slope, const float *slope_ptr, int rchunks,
+                    const unsigned char *am, const int *nbsel,
+                    const int *offsets, const int *ordn, const float *mean,
+                    const float *rstd, const float *gamma, const float *beta,
+                    float *scratch, float *ws, T *dwg, int B, long N, int K,
+                    int M, int G, int act, float slope,
+                    const float *slope_ptr, int rchunks,
                     hipStream_t stream) {
   const int tpc = M / 4;
   const int ppb = EG_THREADS / tpc;
@@ -516,9 +530,9 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
                        dim3(EG_THREADS), 0, stream, scratch, ws,               \
                        (long)rchunks * B, n_out);                              \
     hipLaunchKernelGGL((egnmp_bwd_apply_kernel<T, A>), agrid,                  \
-                       dim3(EG_THREADS), 0, stream, dy, wg, idx, am, order,    \
-                       offsets, mean, rstd, gamma, beta, ws, dwg, N, K, M, G,  \
-                       row_len, slope, slope_ptr);                             \
+                       dim3(EG_THREADS), 0, stream, dy, wg, idx, am, nbsel,    \
+                       offsets, ordn, mean, rstd, gamma, beta, ws, dwg, N, K,  \
+                       M, G, row_len, slope, slope_ptr);                       \
   } while (0)
   if (act == 2) EG_BWD(2);
   else if (act == 1) EG_BWD(1);
@@ -539,39 +553,42 @@ void launch_egnmp_fwd(const void *wg, const int *idx, float *scratch,
                       float *ws, float *mean, float *rstd, const float *gamma,
                       const float *beta, void *vmax, void *vmin,
                       unsigned char *amax, unsigned char *amin, void *y,
-                      unsigned char *am, int B, long N, int K, int M, int G,
-                      float eps, int act, float slope, const float *slope_ptr,
-                      bool bf16, int rchunks, hipStream_t stream) {
+                      unsigned char *am, int *nbsel, int B, long N, int K,
+                      int M, int G, float eps, int act, float slope,
+                      const float *slope_ptr, bool bf16, int rchunks,
+                      hipStream_t stream) {
   if (bf16)
     egnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)wg, idx, scratch,
                                    ws, mean, rstd, gamma, beta,
                                    (__hip_bfloat16 *)vmax,
                                    (__hip_bfloat16 *)vmin, amax, amin,
-                                   (__hip_bfloat16 *)y, am, B, N, K, M, G,
-                                   eps, act, slope, slope_ptr, rchunks,
+                                   (__hip_bfloat16 *)y, am, nbsel, B, N, K,
+                                   M, G, eps, act, slope, slope_ptr, rchunks,
                                    stream);
   else
     egnmp_fwd_impl<float>((const float *)wg, idx, scratch, ws, mean, rstd,
                           gamma, beta, (float *)vmax, (float *)vmin, amax,
-                          amin, (float *)y, am, B, N, K, M, G, eps, act,
-                          slope, slope_ptr, rchunks, stream);
+                          amin, (float *)y, am, nbsel, B, N, K, M, G, eps,
+                          act, slope, slope_ptr, rchunks, stream);
 }
 
 void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
-                      const unsigned char *am, const int *order,
-                      const int *offsets, const float *mean, const float *rstd,
-                      const float *gamma, const float *beta, float *scratch,
-                      float *ws, void *dwg, int B, long N, int K, int M, int G,
-                      int act, float slope, const float *slope_ptr, bool bf16,
+                      const unsigned char *am, const int *nbsel,
+                      const int *offsets, const int *ordn, const float *mean,
+                      const float *rstd, const float *gamma,
+                      const float *beta, float *scratch, float *ws,
+                      void *dwg, int B, long N, int K, int M, int G, int act,
+                      float slope, const float *slope_ptr, bool bf16,
                       int rchunks, hipStream_t stream) {
   if (bf16)
     egnmp_bwd_impl<__hip_bfloat16>(
-        (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)wg, idx, am, order,
-        offsets, mean, rstd, gamma, beta, scratch, ws, (__hip_bfloat16 *)dwg,
-        B, N, K, M, G, act, slope, slope_ptr, rchunks, stream);
+        (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)wg, idx, am,
+        nbsel, offsets, ordn, mean, rstd, gamma, beta, scratch, ws,
+        (__hip_bfloat16 *)dwg, B, N, K, M, G, act, slope, slope_ptr, rchunks,
+        stream);
   else
-    egnmp_bwd_impl<float>((const float *)dy, (const float *)wg, idx, am, order,
-                          offsets, mean, rstd, gamma, beta, scratch, ws,
-                          (float *)dwg, B, N, K, M, G, act, slope, slope_ptr,
-                          rchunks, stream);
+    egnmp_bwd_impl<float>((const float *)dy, (const float *)wg, idx, am,
+                          nbsel, offsets, ordn, mean, rstd, gamma, beta,
+                          scratch, ws, (float *)dwg, B, N, K, M, G, act,
+                          slope, slope_ptr, rchunks, stream);
 }

@@ -40,11 +40,13 @@ class Graph:
     def build(xyz: Tensor, k: int) -> "Graph":
         return Graph(idx=ops.knn_graph(xyz, k), xyz=xyz)
 
-    def csr(self) -> Optional[Tuple[Tensor, Tensor]]:
+    def csr(self) -> Optional[Tuple[Tensor, Tensor, Tensor]]:
         """Inverse adjacency in CSR form, for the deterministic SetConv
         backward: (order (B, N*k) int32 = edge ids sorted by target node,
-        offsets (B, N+1) int32).  Built lazily once per graph (GPU only)
-        and shared by every SetConv/FlowHead call on this graph.
+        offsets (B, N+1) int32, order_n (B, N*k) int32 = the source point
+        of each ordered edge, so walkers need no id decomposition).
+        Built lazily once per graph (GPU only) and shared by every
+        SetConv/FlowHead call on this graph.
         """
         if not self.idx.is_cuda:
             return None
@@ -57,5 +59,9 @@ class Graph:
             targets = flat.gather(1, order)
             bounds = torch.arange(N + 1, device=flat.device).expand(B, N + 1).contiguous()
             offsets = torch.searchsorted(targets, bounds, side="left")
-            self._csr = (order.to(torch.int32).contiguous(), offsets.to(torch.int32).contiguous())
+            self._csr = (
+                order.to(torch.int32).contiguous(),
+                offsets.to(torch.int32).contiguous(),
+                (order % N).to(torch.int32).contiguous(),
+            )
         return self._csr
