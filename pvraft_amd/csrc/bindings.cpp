@@ -65,14 +65,14 @@ void launch_gn_bwd_extract(float*, float*, float*, float*, int, int, int,
 void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
                       float*, const float*, const float*, void*, void*,
                       unsigned char*, unsigned char*, void*, unsigned char*,
-                      int*, int, long, int, int, int, float, int, float,
+                      int, long, int, int, int, float, int, float,
                       const float*, bool, int, hipStream_t);
 void launch_egnmp_bwd(const void*, const void*, const int*,
                       const unsigned char*, const int*, const int*,
-                      const int*, const float*, const float*, const float*,
-                      const float*, float*, float*, void*, int, long, int,
-                      int, int, int, float, const float*, bool, int,
-                      hipStream_t);
+                      const unsigned char*, const float*, const float*,
+                      const float*, const float*, float*, float*, void*, int,
+                      long, int, int, int, int, float, const float*, bool,
+                      int, hipStream_t);
 int egnmp_reduce_chunks(long, int, int);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
@@ -495,9 +495,6 @@ std::vector<torch::Tensor> edge_gnmp_fwd(torch::Tensor wg, torch::Tensor idx,
   auto vmin = torch::empty_like(wg);
   auto amax = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
   auto amin = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
-  // selected argmax-neighbour ids: lets the CSR backward test membership
-  // without per-edge am/dy loads
-  auto nbsel = torch::empty({B, N, M}, wg.options().dtype(torch::kInt32));
   launch_egnmp_fwd(wg.data_ptr(), idx.data_ptr<int>(),
                    scratch.data_ptr<float>(), ws.data_ptr<float>(),
                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
@@ -505,16 +502,16 @@ std::vector<torch::Tensor> edge_gnmp_fwd(torch::Tensor wg, torch::Tensor idx,
                    vmax.data_ptr(), vmin.data_ptr(),
                    amax.data_ptr<unsigned char>(),
                    amin.data_ptr<unsigned char>(), y.data_ptr(),
-                   am.data_ptr<unsigned char>(), nbsel.data_ptr<int>(), B, N,
+                   am.data_ptr<unsigned char>(), B, N,
                    K, M, (int)G, (float)eps, (int)act, (float)slope,
                    slope_ptr, bf16, rchunks, stream());
-  return {y, am, nbsel, mean, rstd};
+  return {y, am, mean, rstd};
 }
 
 std::vector<torch::Tensor> edge_gnmp_bwd(
     torch::Tensor dy, torch::Tensor wg, torch::Tensor idx, torch::Tensor am,
-    torch::Tensor nbsel, torch::Tensor order, torch::Tensor offsets,
-    torch::Tensor order_n, torch::Tensor mean, torch::Tensor rstd, int64_t G,
+    torch::Tensor offsets, torch::Tensor order_n, torch::Tensor order_j,
+    torch::Tensor mean, torch::Tensor rstd, int64_t G,
     torch::Tensor weight, torch::Tensor bias,
     int64_t act, double slope, c10::optional<torch::Tensor> slope_t,
     c10::optional<torch::Tensor> wtarget = c10::nullopt,
@@ -527,8 +524,12 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
   }
   TORCH_CHECK(dy.is_contiguous() && wg.is_contiguous(), "dy/wg must be contiguous");
   TORCH_CHECK(dy.scalar_type() == wg.scalar_type(), "dy/wg dtype mismatch");
-  TORCH_CHECK(order.scalar_type() == torch::kInt32 && order.is_contiguous());
   TORCH_CHECK(offsets.scalar_type() == torch::kInt32 && offsets.is_contiguous());
+  TORCH_CHECK(order_n.scalar_type() == torch::kInt32 && order_n.is_contiguous());
+  TORCH_CHECK(order_j.scalar_type() == torch::kUInt8 && order_j.is_contiguous());
+  TORCH_CHECK(offsets.scalar_type() == torch::kInt32 && offsets.is_contiguous());
+  TORCH_CHECK(order_n.scalar_type() == torch::kInt32 && order_n.is_contiguous());
+  TORCH_CHECK(order_j.scalar_type() == torch::kUInt8 && order_j.is_contiguous());
   const int B = wg.size(0), M = wg.size(2), K = idx.size(2);
   const long N = wg.size(1);
   const bool bf16 = wg.scalar_type() == torch::kBFloat16;
@@ -539,10 +540,10 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
   const int rchunks = egnmp_reduce_chunks(N, M, B);
   auto scratch = torch::empty({ws_len, (long)rchunks * B}, fopt);
   auto dwg = torch::empty_like(wg);
-  (void)order;
   launch_egnmp_bwd(dy.data_ptr(), wg.data_ptr(), idx.data_ptr<int>(),
-                   am.data_ptr<unsigned char>(), nbsel.data_ptr<int>(),
+                   am.data_ptr<unsigned char>(),
                    offsets.data_ptr<int>(), order_n.data_ptr<int>(),
+                   order_j.data_ptr<unsigned char>(),
                    mean.data_ptr<float>(),
                    rstd.data_ptr<float>(), weight.data_ptr<float>(),
                    bias.data_ptr<float>(), scratch.data_ptr<float>(),

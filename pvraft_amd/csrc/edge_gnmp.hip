@@ -203,13 +203,12 @@ template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
     const T *__restrict__ vmax, const T *__restrict__ vmin,
     const unsigned char *__restrict__ amax,
-    const unsigned char *__restrict__ amin, const int *__restrict__ idx,
+    const unsigned char *__restrict__ amin,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
     T *__restrict__ y,               // (B, N, M)
     unsigned char *__restrict__ am,  // (B, N, M)
-    int *__restrict__ nbsel,         // (B, N, M): idx[n, am[n, c]]
-    long total, long NM, int M, int K, int G, float slope,
+    long total, long NM, int M, int G, float slope,
     const float *__restrict__ slope_ptr) {
   if (ACT == 2) slope = *slope_ptr;
   const int Cg = M / G;
@@ -217,8 +216,6 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
        i4 += (long)gridDim.x * EG_THREADS) {
     const long i = i4 * 4;
     const long b = i / NM;
-    const long n = (i % NM) / M;
-    const int *irow = idx + ((long)b * (NM / M) + n) * K;
     const int c0 = (int)(i % M);
     const Quad<T> qx = *(const Quad<T> *)(vmax + i);
     const Quad<T> qn = *(const Quad<T> *)(vmin + i);
@@ -249,12 +246,6 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
     aq.x = out_j[0]; aq.y = out_j[1]; aq.z = out_j[2]; aq.w = out_j[3];
     *(Quad<T> *)(y + i) = oq;
     *(uchar4 *)(am + i) = aq;
-    int4 sel;
-    sel.x = irow[out_j[0]];
-    sel.y = irow[out_j[1]];
-    sel.z = irow[out_j[2]];
-    sel.w = irow[out_j[3]];
-    *(int4 *)(nbsel + i) = sel;
   }
 }
 
@@ -354,9 +345,9 @@ template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ wg,
     const int *__restrict__ idx, const unsigned char *__restrict__ am,
-    const int *__restrict__ nbsel,    // (B, N, M): argmax neighbour ids
     const int *__restrict__ offsets,  // (B, N+1)
     const int *__restrict__ ordn,     // (B, K*N): source n per ordered edge
+    const unsigned char *__restrict__ ordj,  // (B, K*N): its neighbour slot
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
     const float *__restrict__ row_ws, T *__restrict__ dwg, long N, int K,
@@ -387,9 +378,9 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
   const T *wgb = wg + (long)b * N * M;
   const T *dyb = dy + (long)b * N * M;
   const unsigned char *amb = am + (long)b * N * M;
-  const int *selb = nbsel + (long)b * N * M;
   const int *idxb = idx + (long)b * N * K;
   const int *ordnb = ordn + (long)b * N * K;
+  const unsigned char *ordjb = ordj + (long)b * N * K;
   const int *offb = offsets + (long)b * (N + 1);
 
   for (long p = (long)blockIdx.x * ppb + p_l; p < N;
@@ -423,23 +414,23 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
       }
     }
     // incoming term: edges whose neighbour is p.  The dy(argmax) part
-    // fires iff p IS the selected argmax neighbour of (n, c) (precomputed
-    // nbsel, so no per-edge am/dy loads and no edge-id decomposition).
-    // With duplicate entries in an idx row (degenerate pad-self clouds)
-    // the term can fire once per duplicate -- degenerate-tie semantics,
-    // matching the rest of this file.
+    // fires iff this edge's slot j is the pooled argmax of (n, c) -- the
+    // CSR side arrays (ordn, ordj) give (n, j) with no id decomposition,
+    // and dy is a scalar load only on the (rare) argmax hit.  Exact under
+    // duplicate idx entries: only the recorded slot fires.
     const int lo = offb[p], hi = offb[p + 1];
     for (int t = lo; t < hi; ++t) {
       const long n = ordnb[t];
+      const int j = ordjb[t];
       const Quad<T> nq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
-      const int4 sq = *(const int4 *)(selb + n * M + c4 * 4);
-      const int sel[4] = {sq.x, sq.y, sq.z, sq.w};
+      const uchar4 aq = *(const uchar4 *)(amb + n * M + c4 * 4);
+      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
         const float v = (float)pq.v[e] - (float)nq.v[e];
         const float xhat = (v - m[e]) * r[e];
         float dxhat = 0.f;
-        if (sel[e] == (int)p) {
+        if (j == ks[e]) {
           float gs = ldg(dyb + n * M + c4 * 4 + e);
           if (ACT >= 1) {
             const float pre = xhat * ga[e] + be[e];
@@ -471,7 +462,7 @@ template <typename T>
 void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
                     float *mean, float *rstd, const float *gamma,
                     const float *beta, T *vmax, T *vmin, unsigned char *amax,
-                    unsigned char *amin, T *y, unsigned char *am, int *nbsel,
+                    unsigned char *amin, T *y, unsigned char *am,
                     int B, long N, int K, int M, int G, float eps, int act,
                     float slope, const float *slope_ptr, int rchunks,
                     hipStream_t stream) {
@@ -494,8 +485,8 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
 #define EG_FWD(A)                                                             \
   hipLaunchKernelGGL((egnmp_fwd_pick_kernel<T, A>), dim3((unsigned)pb),       \
                      dim3(EG_THREADS), 0, stream, vmax, vmin, amax, amin,     \
-                     idx, mean, rstd, gamma, beta, y, am, nbsel, total,       \
-                     N * M, M, K, G, slope, slope_ptr)
+                     mean, rstd, gamma, beta, y, am, total,                   \
+                     N * M, M, G, slope, slope_ptr)
   if (act == 2) EG_FWD(2);
   else if (act == 1) EG_FWD(1);
   else EG_FWD(0);
@@ -504,8 +495,9 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
 
 template <typename T>
 void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
-                    const unsigned char *am, const int *nbsel,
-                    const int *offsets, const int *ordn, const float *mean,
+                    const unsigned char *am, const int *offsets,
+                    const int *ordn, const unsigned char *ordj,
+                    const float *mean,
                     const float *rstd, const float *gamma, const float *beta,
                     float *scratch, float *ws, T *dwg, int B, long N, int K,
                     int M, int G, int act, float slope,
@@ -530,9 +522,9 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
                        dim3(EG_THREADS), 0, stream, scratch, ws,               \
                        (long)rchunks * B, n_out);                              \
     hipLaunchKernelGGL((egnmp_bwd_apply_kernel<T, A>), agrid,                  \
-                       dim3(EG_THREADS), 0, stream, dy, wg, idx, am, nbsel,    \
-                       offsets, ordn, mean, rstd, gamma, beta, ws, dwg, N, K,  \
-                       M, G, row_len, slope, slope_ptr);                       \
+                       dim3(EG_THREADS), 0, stream, dy, wg, idx, am,           \
+                       offsets, ordn, ordj, mean, rstd, gamma, beta, ws, dwg,  \
+                       N, K, M, G, row_len, slope, slope_ptr);                 \
   } while (0)
   if (act == 2) EG_BWD(2);
   else if (act == 1) EG_BWD(1);
@@ -553,7 +545,7 @@ void launch_egnmp_fwd(const void *wg, const int *idx, float *scratch,
                       float *ws, float *mean, float *rstd, const float *gamma,
                       const float *beta, void *vmax, void *vmin,
                       unsigned char *amax, unsigned char *amin, void *y,
-                      unsigned char *am, int *nbsel, int B, long N, int K,
+                      unsigned char *am, int B, long N, int K,
                       int M, int G, float eps, int act, float slope,
                       const float *slope_ptr, bool bf16, int rchunks,
                       hipStream_t stream) {
@@ -562,19 +554,20 @@ void launch_egnmp_fwd(const void *wg, const int *idx, float *scratch,
                                    ws, mean, rstd, gamma, beta,
                                    (__hip_bfloat16 *)vmax,
                                    (__hip_bfloat16 *)vmin, amax, amin,
-                                   (__hip_bfloat16 *)y, am, nbsel, B, N, K,
+                                   (__hip_bfloat16 *)y, am, B, N, K,
                                    M, G, eps, act, slope, slope_ptr, rchunks,
                                    stream);
   else
     egnmp_fwd_impl<float>((const float *)wg, idx, scratch, ws, mean, rstd,
                           gamma, beta, (float *)vmax, (float *)vmin, amax,
-                          amin, (float *)y, am, nbsel, B, N, K, M, G, eps,
+                          amin, (float *)y, am, B, N, K, M, G, eps,
                           act, slope, slope_ptr, rchunks, stream);
 }
 
 void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
-                      const unsigned char *am, const int *nbsel,
-                      const int *offsets, const int *ordn, const float *mean,
+                      const unsigned char *am, const int *offsets,
+                      const int *ordn, const unsigned char *ordj,
+                      const float *mean,
                       const float *rstd, const float *gamma,
                       const float *beta, float *scratch, float *ws,
                       void *dwg, int B, long N, int K, int M, int G, int act,
@@ -583,12 +576,12 @@ void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
   if (bf16)
     egnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)wg, idx, am,
-        nbsel, offsets, ordn, mean, rstd, gamma, beta, scratch, ws,
+        offsets, ordn, ordj, mean, rstd, gamma, beta, scratch, ws,
         (__hip_bfloat16 *)dwg, B, N, K, M, G, act, slope, slope_ptr, rchunks,
         stream);
   else
     egnmp_bwd_impl<float>((const float *)dy, (const float *)wg, idx, am,
-                          nbsel, offsets, ordn, mean, rstd, gamma, beta,
+                          offsets, ordn, ordj, mean, rstd, gamma, beta,
                           scratch, ws, (float *)dwg, B, N, K, M, G, act,
                           slope, slope_ptr, rchunks, stream);
 }

@@ -43,8 +43,9 @@ class Graph:
     def csr(self) -> Optional[Tuple[Tensor, Tensor, Tensor]]:
         """Inverse adjacency in CSR form, for the deterministic SetConv
         backward: (order (B, N*k) int32 = edge ids sorted by target node,
-        offsets (B, N+1) int32, order_n (B, N*k) int32 = the source point
-        of each ordered edge, so walkers need no id decomposition).
+        offsets (B, N+1) int32, order_n (B, N*k) int32 = source point of
+        each ordered edge, order_j (B, N*k) uint8 = its neighbour slot, so
+        walkers need no id decomposition).
         Built lazily once per graph (GPU only) and shared by every
         SetConv/FlowHead call on this graph.
         """
@@ -63,5 +64,6 @@ class Graph:
                 order.to(torch.int32).contiguous(),
                 offsets.to(torch.int32).contiguous(),
                 (order % N).to(torch.int32).contiguous(),
+                (order // N).to(torch.uint8).contiguous(),
             )
         return self._csr

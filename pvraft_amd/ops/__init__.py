@@ -322,32 +322,32 @@ class _EdgeGNMP(torch.autograd.Function):
     deterministic CSR walk."""
 
     @staticmethod
-    def forward(ctx, wg_t, idx, order, offsets, order_n, num_groups, weight, bias, eps, act, slope, slope_t):
+    def forward(ctx, wg_t, idx, offsets, order_n, order_j, num_groups, weight, bias, eps, act, slope, slope_t):
         w = weight.float().contiguous()
         b = bias.float().contiguous()
         st_ = slope_t.float().reshape(1).contiguous() if slope_t is not None else None
-        y, am, nbsel, mean, rstd = _EXT.edge_gnmp_fwd(wg_t, idx, num_groups, w, b, eps, act, slope, st_)
-        ctx.save_for_backward(wg_t, idx, am, nbsel, order, offsets, order_n, mean, rstd, w, b, st_)
+        y, am, mean, rstd = _EXT.edge_gnmp_fwd(wg_t, idx, num_groups, w, b, eps, act, slope, st_)
+        ctx.save_for_backward(wg_t, idx, am, offsets, order_n, order_j, mean, rstd, w, b, st_)
         ctx.conf = (num_groups, act, slope, weight.dtype)
         ctx.params = (weight, bias, slope_t)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        (wg_t, idx, am, nbsel, order, offsets, order_n, mean, rstd, w, b,
+        (wg_t, idx, am, offsets, order_n, order_j, mean, rstd, w, b,
          st_) = ctx.saved_tensors
         num_groups, act, slope, wdtype = ctx.conf
         tgt = _gn_defer_targets(*ctx.params, act)
         if tgt is not None:
             (dwg,) = _EXT.edge_gnmp_bwd(
-                dy.contiguous(), wg_t, idx, am, nbsel, order, offsets,
-                order_n, mean, rstd, num_groups, w, b, act, slope, st_,
+                dy.contiguous(), wg_t, idx, am, offsets, order_n, order_j,
+                mean, rstd, num_groups, w, b, act, slope, st_,
                 tgt[0], tgt[1], tgt[2],
             )
             return (dwg, None, None, None, None, None, None, None, None,
                     None, None, None)
         dwg, dw, db, dsl = _EXT.edge_gnmp_bwd(
-            dy.contiguous(), wg_t, idx, am, nbsel, order, offsets, order_n,
+            dy.contiguous(), wg_t, idx, am, offsets, order_n, order_j,
             mean, rstd, num_groups, w, b, act, slope, st_, None, None, None,
         )
         dslope = dsl.to(wdtype) if act == 2 else None
@@ -372,10 +372,10 @@ def edge_gnmp(
     if not _use_hip(wg_t):
         raise RuntimeError("edge_gnmp is a GPU-only fused op")
     act_id = {"none": 0, "lrelu": 1, "prelu": 2}[act]
-    order, offsets, order_n = csr
+    _order, offsets, order_n, order_j = csr
     return _EdgeGNMP.apply(
-        wg_t.contiguous(), idx, order, offsets, order_n, num_groups, weight,
-        bias, eps, act_id, slope, slope_t,
+        wg_t.contiguous(), idx, offsets, order_n, order_j, num_groups,
+        weight, bias, eps, act_id, slope, slope_t,
     )
 
 
