@@ -27,15 +27,20 @@
 //
 // Candidate tiles (TILE_PTS x 3 fp32) are staged in LDS and shared by
 // QB=8 queries per workgroup; grid (ceil(N/QB), B) fills the chip.
+//
+// LDS is sized for 4 workgroups per CU (~38 KB of the 160 KB): the first
+// cut staged the QB x SAMP sample-distance matrix in LDS too (33 KB),
+// which capped residency at 2 WGs/CU and left the kernel latency-bound
+// (PMC: SQ_WAIT_ANY ~8.6x SQ_BUSY_CYCLES) -- sample distances now go
+// straight into the ranking wave's registers.
 #include <hip/hip_runtime.h>
 #include "common.h"
 
 #define KNN_THREADS 256
-#define TILE_PTS 2048
+#define TILE_PTS 1792
 #define QB 8            // queries per workgroup
 #define CAP 256         // collect-buffer capacity per query
 #define SAMP 1024       // max sample size per query
-#define SPAD 4          // LDS row padding (bank spread across qi)
 #define KNN_MAXK 48     // model uses 32 (reference extractor.py:10)
 
 __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
@@ -43,7 +48,6 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
     int *__restrict__ out_idx,      // (B, N, k)
     int N, int k) {
   __shared__ float s_tile[TILE_PTS * 3];
-  __shared__ float s_samp[QB][SAMP + SPAD];
   __shared__ float s_bd[QB][CAP];
   __shared__ int s_bi[QB][CAP];
   __shared__ unsigned s_cnt[QB];  // collect counters
@@ -88,36 +92,29 @@ __global__ __launch_bounds__(KNN_THREADS) void knn_select_kernel(
     qz[qi] = s_q[qi][2];
   }
 
-  // ---- pass A: sample distances (cloud is L2-resident; strided gather)
-  for (int si = threadIdx.x; si < ns; si += KNN_THREADS) {
-    const long c = (long)si * st;
-    float cx = 0.f, cy = 0.f, cz = 0.f;
-    const bool ok = c < N;
-    if (ok) {
-      cx = cloud[c * 3 + 0];
-      cy = cloud[c * 3 + 1];
-      cz = cloud[c * 3 + 2];
-    }
-#pragma unroll
-    for (int qi = 0; qi < QB; ++qi) {
-      const float dx = cx - qx[qi];
-      const float dy = cy - qy[qi];
-      const float dz = cz - qz[qi];
-      s_samp[qi][si] = ok ? dx * dx + dy * dy + dz * dz : INFINITY;
-    }
-  }
-  __syncthreads();
-
-  // ---- exact k'-th smallest of each query's sample via radix rank
-  // select on inverted keys (larger key = smaller distance)
+  // ---- pass A + exact k'-th smallest, fused per wave: the ranking
+  // wave computes its queries' sample distances straight into the radix
+  // key registers (p < ns implies p*st < N).  Query coords come from
+  // s_q, not the qx registers: qi is a runtime index here and a
+  // dynamically-indexed per-lane array would spill ALL of qx/qy/qz to
+  // scratch.
   {
     const int lane = lane_id();
     for (int qi = wave_id(); qi < QB; qi += KNN_THREADS / WAVE) {
+      const float sx = s_q[qi][0], sy = s_q[qi][1], sz = s_q[qi][2];
       unsigned kv[SAMP / WAVE];
 #pragma unroll
       for (int s = 0; s < SAMP / WAVE; ++s) {
         const int p = lane + s * WAVE;
-        kv[s] = p < ns ? ~fkey(s_samp[qi][p]) : 0u;
+        unsigned key = 0u;
+        if (p < ns) {
+          const long c = (long)p * st;
+          const float dx = cloud[c * 3 + 0] - sx;
+          const float dy = cloud[c * 3 + 1] - sy;
+          const float dz = cloud[c * 3 + 2] - sz;
+          key = ~fkey(dx * dx + dy * dy + dz * dz);
+        }
+        kv[s] = key;
       }
       const unsigned kt = wave_rank_key(kv, kp);
       if (lane == 0) s_tau[qi] = fkey_inv(~kt);

@@ -60,9 +60,9 @@ class PVRaft(nn.Module):
         )
 
     def forward(self, p, num_iters: int = 12) -> List[Tensor]:
-        from .pointwise import clear_step_cache
+        from .pointwise import refresh_casts
 
-        clear_step_cache()  # fresh autocast weight casts each forward
+        refresh_casts()  # re-fill the bf16 weight mirrors (one foreach)
         xyz1, xyz2 = p
         graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
         fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
@@ -137,9 +137,9 @@ class PVRaftRefine(nn.Module):
                 pmt.requires_grad_(False)
 
     def forward(self, p, num_iters: int = 32) -> Tensor:
-        from .pointwise import clear_step_cache
+        from .pointwise import refresh_casts
 
-        clear_step_cache()
+        refresh_casts()
         with torch.no_grad():
             xyz1, xyz2 = p
             graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)

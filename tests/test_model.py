@@ -180,9 +180,14 @@ def test_predictor_cpu_fallback_exposes_flows():
 
 
 def test_cast_cache_invalidated_by_inplace_update():
-    """_STEP_CACHE must not serve stale bf16 copies after an
-    optimizer step mutates the fp32 weight in place (version-keyed)."""
-    from pvraft_amd.model.pointwise import _cast_cached, clear_step_cache
+    """the cast mirror must not serve stale bf16 values after an
+    optimizer step mutates the fp32 weight in place (version-checked on
+    hit, batch-refreshed by refresh_casts)."""
+    from pvraft_amd.model.pointwise import (
+        _cast_cached,
+        clear_step_cache,
+        refresh_casts,
+    )
 
     clear_step_cache()
     w = torch.randn(4, 4)
@@ -191,5 +196,8 @@ def test_cast_cache_invalidated_by_inplace_update():
     w.add_(1.0)  # in-place update bumps w._version
     c2 = _cast_cached(w, torch.bfloat16)
     assert torch.equal(c2, w.to(torch.bfloat16))
-    assert not torch.equal(c1, c2)
+    # the batched per-step refresh path re-fills the mirror too
+    w.mul_(2.0)
+    refresh_casts()
+    assert torch.equal(_cast_cached(w, torch.bfloat16), w.to(torch.bfloat16))
     clear_step_cache()
