@@ -137,6 +137,51 @@ __global__ __launch_bounds__(256) void transpose_narrow_kernel(
   }
 }
 
+// Narrow-R fast path: the opposite boundary, (B, R, C) with R the CHANNEL
+// count (<= 96, multiple of the vector width) and C the long point dim --
+// e.g. wg (B, mid, N) -> (B, N, mid).  The general kernel's 64x64 tile
+// never fits R, so everything took the scalar edge path (~8 us per call,
+// ~96 calls/step).  Tile = full R x 64 columns: vector loads along C,
+// transposed vector stores along R.
+template <typename T>
+__global__ __launch_bounds__(256) void transpose_narrow_r_kernel(
+    const T *__restrict__ in, T *__restrict__ out, long in_bstride, int R,
+    long C) {
+  constexpr int W = TVec<T>::W;
+  using V = typename TVec<T>::type;
+  __shared__ T tile[96][TP + W];
+  const long c0 = (long)blockIdx.x * TP;
+  const int b = blockIdx.z;
+  const T *src = in + (long)b * in_bstride;
+  T *dst = out + (long)b * C * R;
+  constexpr int CV = TP / W;
+  if ((C % W == 0) && (c0 + TP <= C)) {
+    for (int i = threadIdx.x; i < R * CV; i += 256) {
+      const int r = i / CV;
+      const int cc = (i % CV) * W;
+      *(V *)&tile[r][cc] = *(const V *)&src[(long)r * C + c0 + cc];
+    }
+  } else {
+    for (int i = threadIdx.x; i < R * TP; i += 256) {
+      const int r = i / TP;
+      const int cc = i % TP;
+      tile[r][cc] = (c0 + cc < C) ? src[(long)r * C + c0 + cc] : (T)0.0f;
+    }
+  }
+  __syncthreads();
+  const int RV = R / W;  // launcher guarantees R % W == 0
+  for (int i = threadIdx.x; i < TP * RV; i += 256) {
+    const int cc = i / RV;
+    const int rr = (i % RV) * W;
+    if (c0 + cc < C) {
+      V v;
+#pragma unroll
+      for (int e = 0; e < W; ++e) v.v[e] = tile[rr + e][cc];
+      *(V *)&dst[(c0 + cc) * (long)R + rr] = v;
+    }
+  }
+}
+
 void launch_transpose(const void *in, void *out, long in_bstride, int B,
                       long R, long C, bool bf16, hipStream_t stream) {
   const int W = bf16 ? 8 : 4;
@@ -150,6 +195,18 @@ void launch_transpose(const void *in, void *out, long in_bstride, int B,
       hipLaunchKernelGGL(transpose_narrow_kernel<float>, grid, dim3(256), 0,
                          stream, (const float *)in, (float *)out, in_bstride,
                          R, (int)C);
+    return;
+  }
+  if (R <= 96 && R % W == 0) {
+    dim3 grid((unsigned)((C + TP - 1) / TP), 1, B);
+    if (bf16)
+      hipLaunchKernelGGL(transpose_narrow_r_kernel<__hip_bfloat16>, grid,
+                         dim3(256), 0, stream, (const __hip_bfloat16 *)in,
+                         (__hip_bfloat16 *)out, in_bstride, (int)R, C);
+    else
+      hipLaunchKernelGGL(transpose_narrow_r_kernel<float>, grid, dim3(256),
+                         0, stream, (const float *)in, (float *)out,
+                         in_bstride, (int)R, C);
     return;
   }
   dim3 grid((R + TP - 1) / TP, (C + TP - 1) / TP, B);

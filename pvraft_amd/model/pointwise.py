@@ -44,11 +44,18 @@ def refresh_casts() -> None:
         return
     srcs = []
     dsts = []
-    for ent in _CASTS.values():
-        srcs.append(ent[0])
+    for key, ent in _CASTS.items():
+        if isinstance(key[0], tuple):
+            continue  # derived stack: re-filled when its fresh tensor arrives
+        # detach FRESH each refresh: a stored detached alias of a view
+        # (e.g. weight.view(o, -1)) trips autograd's stale-view check
+        # once the optimizer updates the base in place
+        srcs.append(ent[0].detach())
         dsts.append(ent[1])
         ent[2] = ent[0]._version
-    torch._foreach_copy_(dsts, srcs)
+    if srcs:
+        with torch.no_grad():
+            torch._foreach_copy_(dsts, srcs)
 
 
 def clear_step_cache() -> None:
@@ -160,7 +167,10 @@ def _cast_cached(t: Tensor, dt) -> Tensor:
 
     if os.environ.get("PVRAFT_NO_CAST_CACHE", "0") == "1":
         return t.to(dt)
-    key = (id(t), dt)
+    # derived weights (per-forward cat/slice stacks) carry a stable
+    # _cast_key so each step's fresh tensor REPLACES its entry instead of
+    # growing the registry forever (eager-mode leak otherwise)
+    key = (getattr(t, "_cast_key", None) or id(t), dt)
     ent = _CASTS.get(key)
     if ent is None:
         _CASTS[key] = [t, t.detach().to(dt), t._version]
@@ -169,7 +179,11 @@ def _cast_cached(t: Tensor, dt) -> Tensor:
     # sub-block calls outside PVRaft.forward) bump _version: re-fill the
     # mirror in place.  Inside hipGraph capture _version is stable and
     # the captured refresh re-reads the fp32 source on every replay.
-    if ent[2] != t._version:
+    if ent[0] is not t:
+        ent[0] = t
+        ent[1].copy_(t.detach())
+        ent[2] = t._version
+    elif ent[2] != t._version:
         ent[1].copy_(t.detach())
         ent[2] = t._version
     return ent[1]

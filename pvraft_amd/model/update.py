@@ -39,6 +39,9 @@ def _slice_weight(weight: Tensor, sizes):
     out, lo = [], 0
     for s_ in sizes:
         sl = w[:, lo : lo + s_].contiguous()
+        # stable registry key: each step's fresh slice replaces (not
+        # grows) its autocast-mirror entry (pointwise._cast_cached)
+        sl._cast_key = (id(weight), "slice", lo, s_)
         tgt = None
         if defer:
             lo_c, sz = lo, s_
@@ -139,11 +142,14 @@ class ConvGRU(nn.Module):
                "b_q": self.convq.bias,
                "parts": [],
                "t_zr_h": None, "t_q_h": None, "t_b_zr": None, "t_parts": []}
+        out["zr_h"]._cast_key = (id(self), "zr_h")
+        out["q_h"]._cast_key = (id(self), "q_h")
+        out["b_zr"]._cast_key = (id(self), "b_zr")
         lo = hd
         for sz in part_sizes:
-            out["parts"].append(
-                torch.cat([wz[:, lo : lo + sz], wr[:, lo : lo + sz], wq[:, lo : lo + sz]], dim=0).contiguous()
-            )
+            part = torch.cat([wz[:, lo : lo + sz], wr[:, lo : lo + sz], wq[:, lo : lo + sz]], dim=0).contiguous()
+            part._cast_key = (id(self), "part", lo, sz)
+            out["parts"].append(part)
             lo += sz
         if pointwise.wgrad_defer_active():
             dev = wz.device

@@ -58,7 +58,30 @@ def main():
         if any(t in name.lower() for t in interesting) and ev.self_device_time_total > 0:
             print(
                 f"{name[:70]:70s} calls={ev.count:5d} "
-                f"self_gpu={ev.self_device_time_total/1000:8.1f}us"
+                f"self_gpu={ev.self_device_time_total:10.0f}us"
+            )
+
+    # second pass WITH stacks: where do the add_/copy_ calls come from?
+    with profile(
+        activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+        with_stack=True,
+    ) as prof2:
+        step()
+        torch.cuda.synchronize()
+    print("\n=== add_/copy_ by stack ===")
+    shown = 0
+    for ev in sorted(
+        prof2.key_averages(group_by_stack_n=6),
+        key=lambda e: -e.self_device_time_total,
+    ):
+        if ev.key in ("aten::add_", "aten::copy_", "aten::add") and shown < 14:
+            shown += 1
+            stk = " <- ".join(
+                s.split("/")[-1] for s in (ev.stack or [])[:6]
+            )
+            print(
+                f"{ev.key:14s} calls={ev.count:4d} "
+                f"gpu={ev.self_device_time_total:8.0f}us  {stk[:200]}"
             )
 
 
