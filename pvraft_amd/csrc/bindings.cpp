@@ -64,11 +64,13 @@ void launch_gn_bwd_extract(float*, float*, float*, float*, int, int, int,
                            hipStream_t);
 void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
                       float*, const float*, const float*, void*, void*,
-                      unsigned char*, unsigned char*, void*, unsigned char*,
+                      unsigned char*, unsigned char*, float*, void*,
+                      unsigned char*, void*,
                       int, long, int, int, int, float, int, float,
                       const float*, bool, int, hipStream_t);
 void launch_egnmp_bwd(const void*, const void*, const int*,
-                      const unsigned char*, const int*, const int*,
+                      const unsigned char*, const void*, const float*,
+                      const int*, const int*,
                       const unsigned char*, const float*, const float*,
                       const float*, const float*, float*, float*, void*, int,
                       long, int, int, int, int, float, const float*, bool,
@@ -495,21 +497,27 @@ std::vector<torch::Tensor> edge_gnmp_fwd(torch::Tensor wg, torch::Tensor idx,
   auto vmin = torch::empty_like(wg);
   auto amax = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
   auto amin = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
+  // saved for the closed-form backward centre term: per-point gather sum
+  // (fp32) and the selected pre-GN extreme value
+  auto vsum = torch::empty({B, N, M}, fopt);
+  auto vsel = torch::empty_like(wg);
   launch_egnmp_fwd(wg.data_ptr(), idx.data_ptr<int>(),
                    scratch.data_ptr<float>(), ws.data_ptr<float>(),
                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
                    weight.data_ptr<float>(), bias.data_ptr<float>(),
                    vmax.data_ptr(), vmin.data_ptr(),
                    amax.data_ptr<unsigned char>(),
-                   amin.data_ptr<unsigned char>(), y.data_ptr(),
-                   am.data_ptr<unsigned char>(), B, N,
+                   amin.data_ptr<unsigned char>(), vsum.data_ptr<float>(),
+                   y.data_ptr(),
+                   am.data_ptr<unsigned char>(), vsel.data_ptr(), B, N,
                    K, M, (int)G, (float)eps, (int)act, (float)slope,
                    slope_ptr, bf16, rchunks, stream());
-  return {y, am, mean, rstd};
+  return {y, am, vsel, vsum, mean, rstd};
 }
 
 std::vector<torch::Tensor> edge_gnmp_bwd(
     torch::Tensor dy, torch::Tensor wg, torch::Tensor idx, torch::Tensor am,
+    torch::Tensor vsel, torch::Tensor vsum,
     torch::Tensor offsets, torch::Tensor order_n, torch::Tensor order_j,
     torch::Tensor mean, torch::Tensor rstd, int64_t G,
     torch::Tensor weight, torch::Tensor bias,
@@ -541,7 +549,8 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
   auto scratch = torch::empty({ws_len, (long)rchunks * B}, fopt);
   auto dwg = torch::empty_like(wg);
   launch_egnmp_bwd(dy.data_ptr(), wg.data_ptr(), idx.data_ptr<int>(),
-                   am.data_ptr<unsigned char>(),
+                   am.data_ptr<unsigned char>(), vsel.data_ptr(),
+                   vsum.data_ptr<float>(),
                    offsets.data_ptr<int>(), order_n.data_ptr<int>(),
                    order_j.data_ptr<unsigned char>(),
                    mean.data_ptr<float>(),

@@ -100,6 +100,7 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
     T *__restrict__ vmax, T *__restrict__ vmin,        // (B, N, M)
     unsigned char *__restrict__ amax,
     unsigned char *__restrict__ amin,                  // (B, N, M)
+    float *__restrict__ vsum,      // (B, N, M): sum_j v per point (fp32)
     long N, int K, int M, int G) {
   const int b = blockIdx.z;
   const int B = gridDim.z;
@@ -124,7 +125,7 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
          n += (long)gridDim.x * ppb) {
       const Quad<T> cq = *(const Quad<T> *)(wgb + n * M + c4 * 4);
       const int *row = idxb + n * K;
-      float vmx[4], vmn[4];
+      float vmx[4], vmn[4], ps[4] = {0.f, 0.f, 0.f, 0.f};
       int jmx[4] = {0, 0, 0, 0}, jmn[4] = {0, 0, 0, 0};
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
@@ -136,7 +137,7 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
           const float v = (float)nq.v[e] - (float)cq.v[e];
-          s[e] += v;
+          ps[e] += v;
           ss[e] += v * v;
           if (v > vmx[e]) {
             vmx[e] = v;
@@ -148,6 +149,8 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
           }
         }
       }
+#pragma unroll
+      for (int e = 0; e < 4; ++e) s[e] += ps[e];
       Quad<T> qx, qn;
       uchar4 ax, an;
 #pragma unroll
@@ -164,6 +167,10 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
       *(Quad<T> *)(vmin + pi) = qn;
       *(uchar4 *)(amax + pi) = ax;
       *(uchar4 *)(amin + pi) = an;
+      // per-point gather sum: lets the backward centre term close over j
+      // (sum_j xhat_j derives from it) with no K-loop at all
+      *(float4 *)(vsum + pi) =
+          make_float4(ps[0], ps[1], ps[2], ps[3]);
     }
 #pragma unroll
     for (int e = 0; e < 4; ++e) {
@@ -208,6 +215,7 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
     const float *__restrict__ gamma, const float *__restrict__ beta,
     T *__restrict__ y,               // (B, N, M)
     unsigned char *__restrict__ am,  // (B, N, M)
+    T *__restrict__ vsel,            // (B, N, M): chosen pre-GN extreme
     long total, long NM, int M, int G, float slope,
     const float *__restrict__ slope_ptr) {
   if (ACT == 2) slope = *slope_ptr;
@@ -223,7 +231,7 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
     const uchar4 an = *(const uchar4 *)(amin + i);
     const int axs[4] = {ax.x, ax.y, ax.z, ax.w};
     const int ans[4] = {an.x, an.y, an.z, an.w};
-    Quad<T> oq;
+    Quad<T> oq, sq;
     uchar4 aq;
     unsigned char out_j[4];
 #pragma unroll
@@ -241,11 +249,13 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
       }
       const bool hi = vhi >= vlo;
       oq.v[e] = (T)(hi ? vhi : vlo);
+      sq.v[e] = hi ? qx.v[e] : qn.v[e];
       out_j[e] = (unsigned char)(hi ? axs[e] : ans[e]);
     }
     aq.x = out_j[0]; aq.y = out_j[1]; aq.z = out_j[2]; aq.w = out_j[3];
     *(Quad<T> *)(y + i) = oq;
     *(uchar4 *)(am + i) = aq;
+    *(Quad<T> *)(vsel + i) = sq;
   }
 }
 
@@ -338,13 +348,19 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_reduce_kernel(
     scratch[i * stride + col] = sbins[i];
 }
 
-// pass 2: dWgT (B, N, M), deterministic.  dx1 elements are recomputed on
-// the fly with the exact gnmp_bwd_apply formula; the incoming sum walks
-// the inverse-adjacency CSR (edge id = j*N + n, sorted by target).
+// pass 2: dWgT (B, N, M), deterministic.  The centre term
+// -sum_j dx1[c, j, p] closes over j analytically: only j = am carries dy
+// (recovered from the saved pre-GN extreme vsel), and sum_j xhat_j
+// derives from the per-point gather sum the forward reduce stored
+// (vsum) -- so no K-loop and no idx reads here.  The incoming term walks
+// the inverse-adjacency CSR (ordn/ordj side arrays), with dy a scalar
+// load only on the (rare) argmax hit.
 template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ wg,
-    const int *__restrict__ idx, const unsigned char *__restrict__ am,
+    const unsigned char *__restrict__ am,
+    const T *__restrict__ vsel,       // (B, N, M): chosen pre-GN extreme
+    const float *__restrict__ vsum,   // (B, N, M): sum_j v per point
     const int *__restrict__ offsets,  // (B, N+1)
     const int *__restrict__ ordn,     // (B, K*N): source n per ordered edge
     const unsigned char *__restrict__ ordj,  // (B, K*N): its neighbour slot
@@ -378,7 +394,8 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
   const T *wgb = wg + (long)b * N * M;
   const T *dyb = dy + (long)b * N * M;
   const unsigned char *amb = am + (long)b * N * M;
-  const int *idxb = idx + (long)b * N * K;
+  const T *vselb = vsel + (long)b * N * M;
+  const float *vsumb = vsum + (long)b * N * M;
   const int *ordnb = ordn + (long)b * N * K;
   const unsigned char *ordjb = ordj + (long)b * N * K;
   const int *offb = offsets + (long)b * (N + 1);
@@ -387,30 +404,26 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
        p += (long)gridDim.x * ppb) {
     const Quad<T> pq = *(const Quad<T> *)(wgb + p * M + c4 * 4);
     float acc[4] = {0.f, 0.f, 0.f, 0.f};
-    // centre term: -sum_j dx1[c, j, p]
+    // centre term, closed over j:
+    //   -sum_j (dxhat_j - (s1 + xhat_j*s2)*inv_n)*r
+    //   = -dxhat_am*r + (K*s1 + s2*sum_j xhat_j)*inv_n*r,
+    //   sum_j xhat_j = (vsum - K*mean)*rstd
     {
-      const uchar4 aq = *(const uchar4 *)(amb + p * M + c4 * 4);
       const Quad<T> gq = *(const Quad<T> *)(dyb + p * M + c4 * 4);
-      const int ks[4] = {aq.x, aq.y, aq.z, aq.w};
-      const int *irow = idxb + p * K;
-      for (int j = 0; j < K; ++j) {
-        const Quad<T> nq =
-            *(const Quad<T> *)(wgb + (long)irow[j] * M + c4 * 4);
+      const Quad<T> sq = *(const Quad<T> *)(vselb + p * M + c4 * 4);
+      const float4 vs = *(const float4 *)(vsumb + p * M + c4 * 4);
+      const float vss[4] = {vs.x, vs.y, vs.z, vs.w};
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const float v = (float)nq.v[e] - (float)pq.v[e];
-          const float xhat = (v - m[e]) * r[e];
-          float dxhat = 0.f;
-          if (j == ks[e]) {
-            float gs = (float)gq.v[e];
-            if (ACT >= 1) {
-              const float pre = xhat * ga[e] + be[e];
-              gs = pre > 0.f ? gs : gs * slope;
-            }
-            dxhat = gs * ga[e];
-          }
-          acc[e] -= (dxhat - (s1[e] + xhat * s2[e]) * inv_n) * r[e];
+      for (int e = 0; e < 4; ++e) {
+        const float xhat_am = ((float)sq.v[e] - m[e]) * r[e];
+        float gs = (float)gq.v[e];
+        if (ACT >= 1) {
+          const float pre = xhat_am * ga[e] + be[e];
+          gs = pre > 0.f ? gs : gs * slope;
         }
+        const float sum_xhat = (vss[e] - (float)K * m[e]) * r[e];
+        acc[e] = -gs * ga[e] * r[e] +
+                 ((float)K * s1[e] + sum_xhat * s2[e]) * inv_n * r[e];
       }
     }
     // incoming term: edges whose neighbour is p.  The dy(argmax) part
@@ -462,7 +475,8 @@ template <typename T>
 void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
                     float *mean, float *rstd, const float *gamma,
                     const float *beta, T *vmax, T *vmin, unsigned char *amax,
-                    unsigned char *amin, T *y, unsigned char *am,
+                    unsigned char *amin, float *vsum, T *y,
+                    unsigned char *am, T *vsel,
                     int B, long N, int K, int M, int G, float eps, int act,
                     float slope, const float *slope_ptr, int rchunks,
                     hipStream_t stream) {
@@ -470,7 +484,7 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
   const int n_out_f = B * G * 2;
   hipLaunchKernelGGL(egnmp_fwd_reduce_kernel<T>, rgrid, dim3(EG_THREADS),
                      (size_t)n_out_f * sizeof(float), stream, wg, idx,
-                     scratch, vmax, vmin, amax, amin, N, K, M, G);
+                     scratch, vmax, vmin, amax, amin, vsum, N, K, M, G);
   const int waves_per_block = EG_THREADS / WAVE;
   hipLaunchKernelGGL(egnmp_sum_partials_kernel,
                      dim3((n_out_f + waves_per_block - 1) / waves_per_block),
@@ -485,7 +499,7 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
 #define EG_FWD(A)                                                             \
   hipLaunchKernelGGL((egnmp_fwd_pick_kernel<T, A>), dim3((unsigned)pb),       \
                      dim3(EG_THREADS), 0, stream, vmax, vmin, amax, amin,     \
-                     mean, rstd, gamma, beta, y, am, total,                   \
+                     mean, rstd, gamma, beta, y, am, vsel, total,             \
                      N * M, M, G, slope, slope_ptr)
   if (act == 2) EG_FWD(2);
   else if (act == 1) EG_FWD(1);
@@ -495,7 +509,8 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
 
 template <typename T>
 void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
-                    const unsigned char *am, const int *offsets,
+                    const unsigned char *am, const T *vsel,
+                    const float *vsum, const int *offsets,
                     const int *ordn, const unsigned char *ordj,
                     const float *mean,
                     const float *rstd, const float *gamma, const float *beta,
@@ -522,7 +537,7 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
                        dim3(EG_THREADS), 0, stream, scratch, ws,               \
                        (long)rchunks * B, n_out);                              \
     hipLaunchKernelGGL((egnmp_bwd_apply_kernel<T, A>), agrid,                  \
-                       dim3(EG_THREADS), 0, stream, dy, wg, idx, am,           \
+                       dim3(EG_THREADS), 0, stream, dy, wg, am, vsel, vsum,    \
                        offsets, ordn, ordj, mean, rstd, gamma, beta, ws, dwg,  \
                        N, K, M, G, row_len, slope, slope_ptr);                 \
   } while (0)
@@ -544,28 +559,30 @@ int egnmp_reduce_chunks(long N, int M, int B) {
 void launch_egnmp_fwd(const void *wg, const int *idx, float *scratch,
                       float *ws, float *mean, float *rstd, const float *gamma,
                       const float *beta, void *vmax, void *vmin,
-                      unsigned char *amax, unsigned char *amin, void *y,
-                      unsigned char *am, int B, long N, int K,
-                      int M, int G, float eps, int act, float slope,
+                      unsigned char *amax, unsigned char *amin, float *vsum,
+                      void *y, unsigned char *am, void *vsel, int B, long N,
+                      int K, int M, int G, float eps, int act, float slope,
                       const float *slope_ptr, bool bf16, int rchunks,
                       hipStream_t stream) {
   if (bf16)
     egnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)wg, idx, scratch,
                                    ws, mean, rstd, gamma, beta,
                                    (__hip_bfloat16 *)vmax,
-                                   (__hip_bfloat16 *)vmin, amax, amin,
-                                   (__hip_bfloat16 *)y, am, B, N, K,
+                                   (__hip_bfloat16 *)vmin, amax, amin, vsum,
+                                   (__hip_bfloat16 *)y, am,
+                                   (__hip_bfloat16 *)vsel, B, N, K,
                                    M, G, eps, act, slope, slope_ptr, rchunks,
                                    stream);
   else
     egnmp_fwd_impl<float>((const float *)wg, idx, scratch, ws, mean, rstd,
                           gamma, beta, (float *)vmax, (float *)vmin, amax,
-                          amin, (float *)y, am, B, N, K, M, G, eps,
-                          act, slope, slope_ptr, rchunks, stream);
+                          amin, vsum, (float *)y, am, (float *)vsel, B, N, K,
+                          M, G, eps, act, slope, slope_ptr, rchunks, stream);
 }
 
 void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
-                      const unsigned char *am, const int *offsets,
+                      const unsigned char *am, const void *vsel,
+                      const float *vsum, const int *offsets,
                       const int *ordn, const unsigned char *ordj,
                       const float *mean,
                       const float *rstd, const float *gamma,
@@ -576,11 +593,13 @@ void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
   if (bf16)
     egnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)wg, idx, am,
+        (const __hip_bfloat16 *)vsel, vsum,
         offsets, ordn, ordj, mean, rstd, gamma, beta, scratch, ws,
         (__hip_bfloat16 *)dwg, B, N, K, M, G, act, slope, slope_ptr, rchunks,
         stream);
   else
     egnmp_bwd_impl<float>((const float *)dy, (const float *)wg, idx, am,
+                          (const float *)vsel, vsum,
                           offsets, ordn, ordj, mean, rstd, gamma, beta,
                           scratch, ws, (float *)dwg, B, N, K, M, G, act,
                           slope, slope_ptr, rchunks, stream);

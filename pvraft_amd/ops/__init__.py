@@ -326,28 +326,30 @@ class _EdgeGNMP(torch.autograd.Function):
         w = weight.float().contiguous()
         b = bias.float().contiguous()
         st_ = slope_t.float().reshape(1).contiguous() if slope_t is not None else None
-        y, am, mean, rstd = _EXT.edge_gnmp_fwd(wg_t, idx, num_groups, w, b, eps, act, slope, st_)
-        ctx.save_for_backward(wg_t, idx, am, offsets, order_n, order_j, mean, rstd, w, b, st_)
+        y, am, vsel, vsum, mean, rstd = _EXT.edge_gnmp_fwd(wg_t, idx, num_groups, w, b, eps, act, slope, st_)
+        ctx.save_for_backward(wg_t, idx, am, vsel, vsum, offsets, order_n, order_j, mean, rstd, w, b, st_)
         ctx.conf = (num_groups, act, slope, weight.dtype)
         ctx.params = (weight, bias, slope_t)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        (wg_t, idx, am, offsets, order_n, order_j, mean, rstd, w, b,
-         st_) = ctx.saved_tensors
+        (wg_t, idx, am, vsel, vsum, offsets, order_n, order_j, mean, rstd,
+         w, b, st_) = ctx.saved_tensors
         num_groups, act, slope, wdtype = ctx.conf
         tgt = _gn_defer_targets(*ctx.params, act)
         if tgt is not None:
             (dwg,) = _EXT.edge_gnmp_bwd(
-                dy.contiguous(), wg_t, idx, am, offsets, order_n, order_j,
+                dy.contiguous(), wg_t, idx, am, vsel, vsum, offsets,
+                order_n, order_j,
                 mean, rstd, num_groups, w, b, act, slope, st_,
                 tgt[0], tgt[1], tgt[2],
             )
             return (dwg, None, None, None, None, None, None, None, None,
                     None, None, None)
         dwg, dw, db, dsl = _EXT.edge_gnmp_bwd(
-            dy.contiguous(), wg_t, idx, am, offsets, order_n, order_j,
+            dy.contiguous(), wg_t, idx, am, vsel, vsum, offsets, order_n,
+            order_j,
             mean, rstd, num_groups, w, b, act, slope, st_, None, None, None,
         )
         dslope = dsl.to(wdtype) if act == 2 else None
