@@ -63,13 +63,13 @@ void launch_pw_wgrad(const void*, const void*, float*, float*, int, int, int,
 void launch_gn_bwd_extract(float*, float*, float*, float*, int, int, int,
                            hipStream_t);
 void launch_egnmp_fwd(const void*, const int*, float*, float*, float*,
-                      float*, const float*, const float*, void*, void*,
+                      float*, const float*, const float*, float*, float*,
                       unsigned char*, unsigned char*, float*, void*,
-                      unsigned char*, void*,
+                      unsigned char*, float*,
                       int, long, int, int, int, float, int, float,
                       const float*, bool, int, hipStream_t);
 void launch_egnmp_bwd(const void*, const void*, const int*,
-                      const unsigned char*, const void*, const float*,
+                      const unsigned char*, const float*, const float*,
                       const int*, const int*,
                       const unsigned char*, const float*, const float*,
                       const float*, const float*, float*, float*, void*, int,
@@ -492,24 +492,25 @@ std::vector<torch::Tensor> edge_gnmp_fwd(torch::Tensor wg, torch::Tensor idx,
   auto y = torch::empty_like(wg);
   auto am = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
   // per-(n, c) gather extremes tracked by the reduce pass (the pick pass
-  // is elementwise; the second full gather sweep is gone)
-  auto vmax = torch::empty_like(wg);
-  auto vmin = torch::empty_like(wg);
+  // is elementwise; the second full gather sweep is gone).  fp32: the
+  // backward derives activation-branch signs from them.
+  auto vmax = torch::empty({B, N, M}, fopt);
+  auto vmin = torch::empty({B, N, M}, fopt);
   auto amax = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
   auto amin = torch::empty({B, N, M}, wg.options().dtype(torch::kUInt8));
   // saved for the closed-form backward centre term: per-point gather sum
   // (fp32) and the selected pre-GN extreme value
   auto vsum = torch::empty({B, N, M}, fopt);
-  auto vsel = torch::empty_like(wg);
+  auto vsel = torch::empty({B, N, M}, fopt);
   launch_egnmp_fwd(wg.data_ptr(), idx.data_ptr<int>(),
                    scratch.data_ptr<float>(), ws.data_ptr<float>(),
                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
                    weight.data_ptr<float>(), bias.data_ptr<float>(),
-                   vmax.data_ptr(), vmin.data_ptr(),
+                   vmax.data_ptr<float>(), vmin.data_ptr<float>(),
                    amax.data_ptr<unsigned char>(),
                    amin.data_ptr<unsigned char>(), vsum.data_ptr<float>(),
                    y.data_ptr(),
-                   am.data_ptr<unsigned char>(), vsel.data_ptr(), B, N,
+                   am.data_ptr<unsigned char>(), vsel.data_ptr<float>(), B, N,
                    K, M, (int)G, (float)eps, (int)act, (float)slope,
                    slope_ptr, bf16, rchunks, stream());
   return {y, am, vsel, vsum, mean, rstd};
@@ -549,7 +550,7 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
   auto scratch = torch::empty({ws_len, (long)rchunks * B}, fopt);
   auto dwg = torch::empty_like(wg);
   launch_egnmp_bwd(dy.data_ptr(), wg.data_ptr(), idx.data_ptr<int>(),
-                   am.data_ptr<unsigned char>(), vsel.data_ptr(),
+                   am.data_ptr<unsigned char>(), vsel.data_ptr<float>(),
                    vsum.data_ptr<float>(),
                    offsets.data_ptr<int>(), order_n.data_ptr<int>(),
                    order_j.data_ptr<unsigned char>(),

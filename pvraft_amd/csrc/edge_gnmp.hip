@@ -97,7 +97,9 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
     const T *__restrict__ wg,      // (B, N, M)
     const int *__restrict__ idx,   // (B, N, K)
     float *__restrict__ scratch,   // (B*G*2, gridX*B)
-    T *__restrict__ vmax, T *__restrict__ vmin,        // (B, N, M)
+    float *__restrict__ vmax, float *__restrict__ vmin,  // (B, N, M) fp32:
+    // the backward recovers activation-branch signs from these; a bf16
+    // round here flips the slope branch near zero (0.9 max grad error)
     unsigned char *__restrict__ amax,
     unsigned char *__restrict__ amin,                  // (B, N, M)
     float *__restrict__ vsum,      // (B, N, M): sum_j v per point (fp32)
@@ -151,20 +153,14 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_reduce_kernel(
       }
 #pragma unroll
       for (int e = 0; e < 4; ++e) s[e] += ps[e];
-      Quad<T> qx, qn;
       uchar4 ax, an;
-#pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        qx.v[e] = (T)vmx[e];
-        qn.v[e] = (T)vmn[e];
-      }
       ax.x = (unsigned char)jmx[0]; ax.y = (unsigned char)jmx[1];
       ax.z = (unsigned char)jmx[2]; ax.w = (unsigned char)jmx[3];
       an.x = (unsigned char)jmn[0]; an.y = (unsigned char)jmn[1];
       an.z = (unsigned char)jmn[2]; an.w = (unsigned char)jmn[3];
       const long pi = ((long)b * N + n) * M + c4 * 4;
-      *(Quad<T> *)(vmax + pi) = qx;
-      *(Quad<T> *)(vmin + pi) = qn;
+      *(float4 *)(vmax + pi) = make_float4(vmx[0], vmx[1], vmx[2], vmx[3]);
+      *(float4 *)(vmin + pi) = make_float4(vmn[0], vmn[1], vmn[2], vmn[3]);
       *(uchar4 *)(amax + pi) = ax;
       *(uchar4 *)(amin + pi) = an;
       // per-point gather sum: lets the backward centre term close over j
@@ -208,14 +204,14 @@ __global__ void egnmp_sum_partials_kernel(const float *__restrict__ scratch,
 // gamma and any activation slope).
 template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
-    const T *__restrict__ vmax, const T *__restrict__ vmin,
+    const float *__restrict__ vmax, const float *__restrict__ vmin,
     const unsigned char *__restrict__ amax,
     const unsigned char *__restrict__ amin,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ gamma, const float *__restrict__ beta,
     T *__restrict__ y,               // (B, N, M)
     unsigned char *__restrict__ am,  // (B, N, M)
-    T *__restrict__ vsel,            // (B, N, M): chosen pre-GN extreme
+    float *__restrict__ vsel,        // (B, N, M): chosen pre-GN extreme
     long total, long NM, int M, int G, float slope,
     const float *__restrict__ slope_ptr) {
   if (ACT == 2) slope = *slope_ptr;
@@ -225,13 +221,16 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
     const long i = i4 * 4;
     const long b = i / NM;
     const int c0 = (int)(i % M);
-    const Quad<T> qx = *(const Quad<T> *)(vmax + i);
-    const Quad<T> qn = *(const Quad<T> *)(vmin + i);
+    const float4 qx = *(const float4 *)(vmax + i);
+    const float4 qn = *(const float4 *)(vmin + i);
     const uchar4 ax = *(const uchar4 *)(amax + i);
     const uchar4 an = *(const uchar4 *)(amin + i);
+    const float qxs[4] = {qx.x, qx.y, qx.z, qx.w};
+    const float qns[4] = {qn.x, qn.y, qn.z, qn.w};
     const int axs[4] = {ax.x, ax.y, ax.z, ax.w};
     const int ans[4] = {an.x, an.y, an.z, an.w};
-    Quad<T> oq, sq;
+    Quad<T> oq;
+    float sq[4];
     uchar4 aq;
     unsigned char out_j[4];
 #pragma unroll
@@ -241,21 +240,21 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_fwd_pick_kernel(
       const float m = mean[row];
       const float r = rstd[row];
       const float ga = gamma[c], be = beta[c];
-      float vhi = ((float)qx.v[e] - m) * r * ga + be;
-      float vlo = ((float)qn.v[e] - m) * r * ga + be;
+      float vhi = (qxs[e] - m) * r * ga + be;
+      float vlo = (qns[e] - m) * r * ga + be;
       if (ACT >= 1) {
         vhi = vhi > 0.f ? vhi : vhi * slope;
         vlo = vlo > 0.f ? vlo : vlo * slope;
       }
       const bool hi = vhi >= vlo;
       oq.v[e] = (T)(hi ? vhi : vlo);
-      sq.v[e] = hi ? qx.v[e] : qn.v[e];
+      sq[e] = hi ? qxs[e] : qns[e];
       out_j[e] = (unsigned char)(hi ? axs[e] : ans[e]);
     }
     aq.x = out_j[0]; aq.y = out_j[1]; aq.z = out_j[2]; aq.w = out_j[3];
     *(Quad<T> *)(y + i) = oq;
     *(uchar4 *)(am + i) = aq;
-    *(Quad<T> *)(vsel + i) = sq;
+    *(float4 *)(vsel + i) = make_float4(sq[0], sq[1], sq[2], sq[3]);
   }
 }
 
@@ -359,7 +358,7 @@ template <typename T, int ACT>
 __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     const T *__restrict__ dy, const T *__restrict__ wg,
     const unsigned char *__restrict__ am,
-    const T *__restrict__ vsel,       // (B, N, M): chosen pre-GN extreme
+    const float *__restrict__ vsel,   // (B, N, M): chosen pre-GN extreme
     const float *__restrict__ vsum,   // (B, N, M): sum_j v per point
     const int *__restrict__ offsets,  // (B, N+1)
     const int *__restrict__ ordn,     // (B, K*N): source n per ordered edge
@@ -394,7 +393,7 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
   const T *wgb = wg + (long)b * N * M;
   const T *dyb = dy + (long)b * N * M;
   const unsigned char *amb = am + (long)b * N * M;
-  const T *vselb = vsel + (long)b * N * M;
+  const float *vselb = vsel + (long)b * N * M;
   const float *vsumb = vsum + (long)b * N * M;
   const int *ordnb = ordn + (long)b * N * K;
   const unsigned char *ordjb = ordj + (long)b * N * K;
@@ -410,12 +409,13 @@ __global__ __launch_bounds__(EG_THREADS) void egnmp_bwd_apply_kernel(
     //   sum_j xhat_j = (vsum - K*mean)*rstd
     {
       const Quad<T> gq = *(const Quad<T> *)(dyb + p * M + c4 * 4);
-      const Quad<T> sq = *(const Quad<T> *)(vselb + p * M + c4 * 4);
+      const float4 sv = *(const float4 *)(vselb + p * M + c4 * 4);
       const float4 vs = *(const float4 *)(vsumb + p * M + c4 * 4);
+      const float sqs[4] = {sv.x, sv.y, sv.z, sv.w};
       const float vss[4] = {vs.x, vs.y, vs.z, vs.w};
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
-        const float xhat_am = ((float)sq.v[e] - m[e]) * r[e];
+        const float xhat_am = (sqs[e] - m[e]) * r[e];
         float gs = (float)gq.v[e];
         if (ACT >= 1) {
           const float pre = xhat_am * ga[e] + be[e];
@@ -474,9 +474,10 @@ static int eg_chunks(long N, int ppb, int B, int cap) {
 template <typename T>
 void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
                     float *mean, float *rstd, const float *gamma,
-                    const float *beta, T *vmax, T *vmin, unsigned char *amax,
+                    const float *beta, float *vmax, float *vmin,
+                    unsigned char *amax,
                     unsigned char *amin, float *vsum, T *y,
-                    unsigned char *am, T *vsel,
+                    unsigned char *am, float *vsel,
                     int B, long N, int K, int M, int G, float eps, int act,
                     float slope, const float *slope_ptr, int rchunks,
                     hipStream_t stream) {
@@ -509,7 +510,7 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
 
 template <typename T>
 void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
-                    const unsigned char *am, const T *vsel,
+                    const unsigned char *am, const float *vsel,
                     const float *vsum, const int *offsets,
                     const int *ordn, const unsigned char *ordj,
                     const float *mean,
@@ -558,30 +559,29 @@ int egnmp_reduce_chunks(long N, int M, int B) {
 
 void launch_egnmp_fwd(const void *wg, const int *idx, float *scratch,
                       float *ws, float *mean, float *rstd, const float *gamma,
-                      const float *beta, void *vmax, void *vmin,
+                      const float *beta, float *vmax, float *vmin,
                       unsigned char *amax, unsigned char *amin, float *vsum,
-                      void *y, unsigned char *am, void *vsel, int B, long N,
+                      void *y, unsigned char *am, float *vsel, int B, long N,
                       int K, int M, int G, float eps, int act, float slope,
                       const float *slope_ptr, bool bf16, int rchunks,
                       hipStream_t stream) {
   if (bf16)
     egnmp_fwd_impl<__hip_bfloat16>((const __hip_bfloat16 *)wg, idx, scratch,
                                    ws, mean, rstd, gamma, beta,
-                                   (__hip_bfloat16 *)vmax,
-                                   (__hip_bfloat16 *)vmin, amax, amin, vsum,
+                                   vmax, vmin, amax, amin, vsum,
                                    (__hip_bfloat16 *)y, am,
-                                   (__hip_bfloat16 *)vsel, B, N, K,
+                                   vsel, B, N, K,
                                    M, G, eps, act, slope, slope_ptr, rchunks,
                                    stream);
   else
     egnmp_fwd_impl<float>((const float *)wg, idx, scratch, ws, mean, rstd,
-                          gamma, beta, (float *)vmax, (float *)vmin, amax,
-                          amin, vsum, (float *)y, am, (float *)vsel, B, N, K,
+                          gamma, beta, vmax, vmin, amax,
+                          amin, vsum, (float *)y, am, vsel, B, N, K,
                           M, G, eps, act, slope, slope_ptr, rchunks, stream);
 }
 
 void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
-                      const unsigned char *am, const void *vsel,
+                      const unsigned char *am, const float *vsel,
                       const float *vsum, const int *offsets,
                       const int *ordn, const unsigned char *ordj,
                       const float *mean,
@@ -593,13 +593,13 @@ void launch_egnmp_bwd(const void *dy, const void *wg, const int *idx,
   if (bf16)
     egnmp_bwd_impl<__hip_bfloat16>(
         (const __hip_bfloat16 *)dy, (const __hip_bfloat16 *)wg, idx, am,
-        (const __hip_bfloat16 *)vsel, vsum,
+        vsel, vsum,
         offsets, ordn, ordj, mean, rstd, gamma, beta, scratch, ws,
         (__hip_bfloat16 *)dwg, B, N, K, M, G, act, slope, slope_ptr, rchunks,
         stream);
   else
     egnmp_bwd_impl<float>((const float *)dy, (const float *)wg, idx, am,
-                          (const float *)vsel, vsum,
+                          vsel, vsum,
                           offsets, ordn, ordj, mean, rstd, gamma, beta,
                           scratch, ws, (float *)dwg, B, N, K, M, G, act,
                           slope, slope_ptr, rchunks, stream);

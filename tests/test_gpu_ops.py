@@ -613,11 +613,15 @@ def test_pw_fused_autograd_matches_fallback():
 
 
 @pytest.mark.parametrize("B,R,C,dtype", [
-    (2, 8192, 96, torch.bfloat16),   # narrow fast path
-    (2, 8191, 48, torch.bfloat16),   # narrow, ragged R
+    (2, 8192, 96, torch.bfloat16),   # narrow-C fast path
+    (2, 8191, 48, torch.bfloat16),   # narrow-C, ragged R
     (1, 513, 16, torch.float32),
     (2, 500, 35, torch.float32),     # narrow-scalar (C % W != 0)
     (2, 300, 128, torch.bfloat16),   # general tiled path
+    (2, 48, 8192, torch.bfloat16),   # narrow-R fast path (wg direction)
+    (2, 96, 8191, torch.bfloat16),   # narrow-R, ragged C
+    (1, 16, 513, torch.float32),     # narrow-R fp32 (W=4)
+    (2, 61, 8192, torch.bfloat16),   # R % W != 0: general kernel
 ])
 def test_batched_transpose_shapes(B, R, C, dtype):
     import pvraft_amd._C as _C
