@@ -169,8 +169,16 @@ def _cast_cached(t: Tensor, dt) -> Tensor:
         return t.to(dt)
     # derived weights (per-forward cat/slice stacks) carry a stable
     # _cast_key so each step's fresh tensor REPLACES its entry instead of
-    # growing the registry forever (eager-mode leak otherwise)
-    key = (getattr(t, "_cast_key", None) or id(t), dt)
+    # growing the registry forever (eager-mode leak otherwise).  Plain
+    # VIEWS (weight.squeeze(-1) / .view(o, -1), fresh objects every call)
+    # key by their base parameter + view geometry, so the 8-iteration GRU
+    # loop hits instead of re-casting ~40 weights per step.
+    ck = getattr(t, "_cast_key", None)
+    if ck is None:
+        base = t._base if t._base is not None else t
+        ck = (id(base), tuple(t.shape), tuple(t.stride()),
+              t.storage_offset())
+    key = (ck, dt)
     ent = _CASTS.get(key)
     if ent is None:
         _CASTS[key] = [t, t.detach().to(dt), t._version]
@@ -179,11 +187,10 @@ def _cast_cached(t: Tensor, dt) -> Tensor:
     # sub-block calls outside PVRaft.forward) bump _version: re-fill the
     # mirror in place.  Inside hipGraph capture _version is stable and
     # the captured refresh re-reads the fp32 source on every replay.
-    if ent[0] is not t:
+    # a fresh view object over the SAME storage is still a hit; only new
+    # storage (rebuilt cat stacks) or an in-place base update re-fills
+    if ent[0].data_ptr() != t.data_ptr() or ent[2] != t._version:
         ent[0] = t
-        ent[1].copy_(t.detach())
-        ent[2] = t._version
-    elif ent[2] != t._version:
         ent[1].copy_(t.detach())
         ent[2] = t._version
     return ent[1]
