@@ -76,6 +76,16 @@ void launch_egnmp_bwd(const void*, const void*, const int*,
                       long, int, int, int, int, float, const float*, bool,
                       int, hipStream_t);
 int egnmp_reduce_chunks(long, int, int);
+void launch_kg_fwd(const float*, const float*, const float*, float*, float*,
+                   float*, float*, const float*, const float*, float*,
+                   float*, unsigned char*, unsigned char*, void*,
+                   unsigned char*, float*, int, long, int, int, int, float,
+                   const float*, bool, int, hipStream_t);
+void launch_kg_bwd(const void*, const float*, const float*, const float*,
+                   const unsigned char*, const float*, const float*,
+                   const float*, const float*, const float*, float*, float*,
+                   float*, float*, float*, float*, int, long, int, int, int,
+                   const float*, bool, int, hipStream_t);
 void launch_gnmp_bwd(const void*, const void*, const unsigned char*,
                      const float*, const float*, const float*, const float*,
                      float*, float*, float*, float*, int, void*, int, long,
@@ -563,6 +573,102 @@ std::vector<torch::Tensor> edge_gnmp_bwd(
   return gn_grads_finish(ws, dwg, rows, M, fopt, wtarget, btarget, starget);
 }
 
+// Fused kNN-correlation branch head (csrc/knn_gnmp.hip): conv(4->C) ->
+// GroupNorm(G) -> PReLU -> max over K, without the (B, C, K, N)
+// intermediate.  raw (B, 4, K, N) fp32; y is (B, N, C) point-major (the
+// caller transposes with the narrow-R fast path).
+std::vector<torch::Tensor> knn_gnmp_fwd(torch::Tensor raw, torch::Tensor W,
+                                        torch::Tensor cb, int64_t G,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta, double eps,
+                                        torch::Tensor slope_t,
+                                        bool out_bf16) {
+  check_f32(raw, "raw");
+  check_f32(W, "W");
+  check_f32(cb, "cb");
+  check_f32(gamma, "gamma");
+  check_f32(beta, "beta");
+  TORCH_CHECK(raw.dim() == 4 && raw.size(1) == 4, "raw must be (B,4,K,N)");
+  TORCH_CHECK(slope_t.numel() == 1 && slope_t.scalar_type() == torch::kFloat32);
+  const int B = raw.size(0), K = raw.size(2);
+  const long N = raw.size(3);
+  const int C = W.size(0);
+  TORCH_CHECK(W.dim() == 2 && W.size(1) == 4, "W must be (C,4)");
+  TORCH_CHECK(C % 16 == 0 && C % G == 0 && C / G >= 4 && K <= 255,
+              "knn_gnmp: need C % 16 == 0, C/G >= 4, K <= 255");
+  auto fopt = raw.options();
+  const int rows = B * (int)G;
+  auto& ws = persistent_ws((long)rows * 2, fopt);
+  int nblk = (int)((N + 255) / 256);
+  if (nblk < 1) nblk = 1;
+  const int chunks_f = C / 8;
+  auto scratch = torch::empty({(long)rows * 2, (long)nblk * chunks_f * B}, fopt);
+  auto mean = torch::empty({rows}, fopt);
+  auto rstd = torch::empty({rows}, fopt);
+  auto vmax = torch::empty({B, N, C}, fopt);
+  auto vmin = torch::empty({B, N, C}, fopt);
+  auto vsel = torch::empty({B, N, C}, fopt);
+  auto amax = torch::empty({B, N, C}, raw.options().dtype(torch::kUInt8));
+  auto amin = torch::empty({B, N, C}, raw.options().dtype(torch::kUInt8));
+  auto am = torch::empty({B, N, C}, raw.options().dtype(torch::kUInt8));
+  auto y = torch::empty({B, N, C}, raw.options().dtype(
+                                       out_bf16 ? torch::kBFloat16
+                                                : torch::kFloat32));
+  launch_kg_fwd(raw.data_ptr<float>(), W.data_ptr<float>(),
+                cb.data_ptr<float>(), scratch.data_ptr<float>(),
+                ws.data_ptr<float>(), mean.data_ptr<float>(),
+                rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                beta.data_ptr<float>(), vmax.data_ptr<float>(),
+                vmin.data_ptr<float>(), amax.data_ptr<unsigned char>(),
+                amin.data_ptr<unsigned char>(), y.data_ptr(),
+                am.data_ptr<unsigned char>(), vsel.data_ptr<float>(), B, N,
+                K, C, (int)G, (float)eps, slope_t.data_ptr<float>(),
+                out_bf16, nblk, stream());
+  return {y, am, vsel, mean, rstd};
+}
+
+std::vector<torch::Tensor> knn_gnmp_bwd(
+    torch::Tensor dyT, torch::Tensor raw, torch::Tensor W, torch::Tensor cb,
+    torch::Tensor am, torch::Tensor vsel, torch::Tensor mean,
+    torch::Tensor rstd, int64_t G, torch::Tensor gamma, torch::Tensor beta,
+    torch::Tensor slope_t) {
+  TORCH_CHECK(dyT.is_cuda() && dyT.is_contiguous() && dyT.dim() == 3,
+              "dyT must be contiguous (B,N,C)");
+  check_f32(raw, "raw");
+  const int B = raw.size(0), K = raw.size(2);
+  const long N = raw.size(3);
+  const int C = W.size(0);
+  const bool bf16 = dyT.scalar_type() == torch::kBFloat16;
+  auto fopt = raw.options();
+  const int rows = B * (int)G;
+  int nblk = (int)((N + 255) / 256);
+  if (nblk < 1) nblk = 1;
+  const long n_out = (long)rows * 2 + C * 2 + 1;
+  auto scratch = torch::empty({n_out, (long)nblk * B}, fopt);
+  auto ws = torch::empty({n_out}, fopt);
+  const int chunks_b = C / 16;
+  auto wscratch =
+      torch::empty({(long)C * 5, (long)nblk * chunks_b * B}, fopt);
+  auto ws2 = torch::empty({(long)C * 5}, fopt);
+  auto draw_part = torch::empty({(long)chunks_b, B, 4, K, N}, fopt);
+  auto draw = torch::empty({B, 4, K, N}, fopt);
+  launch_kg_bwd(dyT.data_ptr(), raw.data_ptr<float>(), W.data_ptr<float>(),
+                cb.data_ptr<float>(), am.data_ptr<unsigned char>(),
+                vsel.data_ptr<float>(), mean.data_ptr<float>(),
+                rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                beta.data_ptr<float>(), scratch.data_ptr<float>(),
+                ws.data_ptr<float>(), wscratch.data_ptr<float>(),
+                ws2.data_ptr<float>(), draw_part.data_ptr<float>(),
+                draw.data_ptr<float>(), B, N, K, C, (int)G,
+                slope_t.data_ptr<float>(), bf16, nblk, stream());
+  auto tail = gn_grads_finish(ws, draw, rows, C, fopt, c10::nullopt,
+                              c10::nullopt, c10::nullopt);
+  auto dW = ws2.narrow(0, 0, (long)C * 4).view({C, 4});
+  auto dcb = ws2.narrow(0, (long)C * 4, C);
+  // tail = {draw, dgamma, dbeta, dslope}
+  return {draw, dW, dcb, tail[1], tail[2], tail[3]};
+}
+
 // dy (B, Co, S) bf16, x (B, Ci, S) bf16 -> dW (Co, Ci) fp32 (split-K MFMA)
 std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
                                     int64_t schunks = 0, bool with_bias = false) {
@@ -983,6 +1089,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_act_maxpool_bwd", &group_norm_act_maxpool_bwd);
   m.def("edge_gnmp_fwd", &edge_gnmp_fwd);
   m.def("edge_gnmp_bwd", &edge_gnmp_bwd);
+  m.def("knn_gnmp_fwd", &knn_gnmp_fwd);
+  m.def("knn_gnmp_bwd", &knn_gnmp_bwd);
   m.def("knn_graph", &knn_graph, "fused kNN graph (CDNA4)");
   m.def("gather_edge_concat_fwd", &gather_edge_concat_fwd);
   m.def("gather_edge_concat_bwd", &gather_edge_concat_bwd);

@@ -548,6 +548,37 @@ void egnmp_bwd_impl(const T *dy, const T *wg, const int *idx,
 #undef EG_BWD
 }
 
+// Standalone pick launcher for sibling fused ops (csrc/knn_gnmp.hip):
+// elementwise max(act(gn(vmax)), act(gn(vmin))) + argmax + selected
+// pre-GN value, over any (B, N, M) extremes tensors.
+void launch_gnmp_pick(const float *vmax, const float *vmin,
+                      const unsigned char *amax, const unsigned char *amin,
+                      const float *mean, const float *rstd,
+                      const float *gamma, const float *beta, void *y,
+                      unsigned char *am, float *vsel, long total, long NM,
+                      int M, int G, int act, float slope,
+                      const float *slope_ptr, bool bf16,
+                      hipStream_t stream) {
+  long pb = (total / 4 + EG_THREADS - 1) / EG_THREADS;
+  if (pb > 2048) pb = 2048;
+  if (pb < 1) pb = 1;
+#define EG_PICK(T, A)                                                         \
+  hipLaunchKernelGGL((egnmp_fwd_pick_kernel<T, A>), dim3((unsigned)pb),       \
+                     dim3(EG_THREADS), 0, stream, vmax, vmin, amax, amin,     \
+                     mean, rstd, gamma, beta, (T *)y, am, vsel, total, NM,    \
+                     M, G, slope, slope_ptr)
+  if (bf16) {
+    if (act == 2) EG_PICK(__hip_bfloat16, 2);
+    else if (act == 1) EG_PICK(__hip_bfloat16, 1);
+    else EG_PICK(__hip_bfloat16, 0);
+  } else {
+    if (act == 2) EG_PICK(float, 2);
+    else if (act == 1) EG_PICK(float, 1);
+    else EG_PICK(float, 0);
+  }
+#undef EG_PICK
+}
+
 int egnmp_reduce_chunks(long N, int M, int B) {
   const int ppb = EG_THREADS / (M / 4);
   long want = 1024 / (B > 0 ? B : 1);

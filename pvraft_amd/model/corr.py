@@ -109,12 +109,28 @@ class CorrBlock(nn.Module):
         return self._knn_conv(feat)
 
     def _knn_conv(self, feat: Tensor) -> Tensor:
-        # knn_conv = Sequential(conv, GN, PReLU) + max over the k axis;
-        # executed as conv -> fused GN+PReLU+maxpool (one kernel pipeline,
-        # the (B, 64, k, N) activation never materialises post-GN)
+        # knn_conv = Sequential(conv, GN, PReLU) + max over the k axis
         conv, gn, prelu = self.knn_conv[0], self.knn_conv[1], self.knn_conv[2]
-        feat = conv(feat)
-        feat = ops.group_norm_act_maxpool(
-            feat, gn.num_groups, gn.weight, gn.bias, gn.eps, act="prelu", slope_t=prelu.weight
-        )  # B, 64, N
+        import os
+
+        if (
+            feat.is_cuda
+            and ops.hip_available()
+            and feat.dtype == torch.float32
+            and conv.out_channels % 16 == 0
+            and os.environ.get("PVRAFT_REF_OPS", "0") != "1"
+        ):
+            # one fused pipeline: the conv contraction is only 4-wide, so
+            # it is evaluated inline with GN+PReLU+maxpool and the
+            # (B, 64, k, N) activation never exists at all
+            feat = ops.knn_gnmp(
+                feat, conv.weight, conv.bias, gn.num_groups, gn.weight,
+                gn.bias, gn.eps, prelu.weight,
+            )  # B, 64, N
+        else:
+            feat = conv(feat)
+            feat = ops.group_norm_act_maxpool(
+                feat, gn.num_groups, gn.weight, gn.bias, gn.eps, act="prelu",
+                slope_t=prelu.weight,
+            )  # B, 64, N
         return self.knn_out(feat)

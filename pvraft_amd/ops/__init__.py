@@ -381,6 +381,49 @@ def edge_gnmp(
     )
 
 
+class _KnnGNMP(torch.autograd.Function):
+    """Fused kNN-corr branch head: conv(4->C) + GN + PReLU + max over K
+    (csrc/knn_gnmp.hip) -- the (B, C, K, N) conv activation never exists.
+    fp32 compute (raw is fp32; the contraction is 4-wide), output in the
+    autocast dtype."""
+
+    @staticmethod
+    def forward(ctx, raw, weight, cbias, num_groups, gamma, beta, eps, slope_t):
+        w = weight.view(weight.shape[0], 4).float().contiguous()
+        cb = cbias.float().contiguous()
+        ga = gamma.float().contiguous()
+        be = beta.float().contiguous()
+        st = slope_t.float().reshape(1).contiguous()
+        out_bf16 = raw.is_cuda and torch.is_autocast_enabled() and (
+            torch.get_autocast_dtype("cuda") == torch.bfloat16)
+        y, am, vsel, mean, rstd = _EXT.knn_gnmp_fwd(
+            raw, w, cb, num_groups, ga, be, eps, st, out_bf16)
+        ctx.save_for_backward(raw, w, cb, am, vsel, mean, rstd, ga, be, st)
+        ctx.conf = (num_groups, weight.shape, weight.dtype)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        raw, w, cb, am, vsel, mean, rstd, ga, be, st = ctx.saved_tensors
+        num_groups, wshape, wdtype = ctx.conf
+        draw, dW, dcb, dgamma, dbeta, dslope = _EXT.knn_gnmp_bwd(
+            dy.contiguous(), raw, w, cb, am, vsel, mean, rstd, num_groups,
+            ga, be, st)
+        return (draw, dW.view(wshape).to(wdtype), dcb, None, dgamma, dbeta,
+                None, dslope.reshape(1))
+
+
+def knn_gnmp(raw: Tensor, weight: Tensor, cbias: Tensor, num_groups: int,
+             gamma: Tensor, beta: Tensor, eps: float, slope_t: Tensor) -> Tensor:
+    """raw (B, 4, K, N) fp32 -> (B, C, N): conv -> GN -> PReLU -> max over
+    K in one fused pipeline, GPU only."""
+    if not _use_hip(raw):
+        raise RuntimeError("knn_gnmp is a GPU-only fused op")
+    y = _KnnGNMP.apply(raw.contiguous(), weight, cbias, num_groups, gamma,
+                       beta, eps, slope_t)
+    return transpose_last2(y)  # (B, N, C) -> (B, C, N), narrow-R fast path
+
+
 # ---------------------------------------------------------------------------
 # public functional API (model code calls these)
 # ---------------------------------------------------------------------------

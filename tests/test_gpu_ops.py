@@ -629,3 +629,45 @@ def test_batched_transpose_shapes(B, R, C, dtype):
     x = torch.randn(B, R, C, device=dev()).to(dtype)
     y = _C.batched_transpose(x.contiguous())
     assert torch.equal(y, x.transpose(1, 2).contiguous())
+
+@pytest.mark.parametrize("B,N,K,C,G,autocast_on", [
+    (2, 311, 32, 64, 8, False),
+    (1, 1024, 16, 64, 8, True),
+    (2, 500, 32, 32, 8, False),
+])
+def test_knn_gnmp_matches_reference(B, N, K, C, G, autocast_on):
+    """Fused conv(4->C)+GN+PReLU+maxpool vs the explicit composition
+    (reference model/corr.py:70-76 semantics)."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(3)
+    raw = torch.randn(B, 4, K, N, device=dev()).requires_grad_(True)
+    w = torch.randn(C, 4, 1, 1, device=dev()).mul(0.2).requires_grad_(True)
+    cb = torch.randn(C, device=dev()).mul(0.1).requires_grad_(True)
+    ga = torch.rand(C, device=dev()).add(0.5).requires_grad_(True)
+    be = torch.randn(C, device=dev()).mul(0.1).requires_grad_(True)
+    sl = torch.tensor([0.25], device=dev()).requires_grad_(True)
+
+    leaves = [raw, w, cb, ga, be, sl]
+    clones = [t.detach().clone().requires_grad_(True) for t in leaves]
+    raw2, w2, cb2, ga2, be2, sl2 = clones
+
+    with torch.autocast("cuda", dtype=torch.bfloat16, enabled=autocast_on):
+        y = ops.knn_gnmp(raw, w, cb, G, ga, be, 1e-5, sl)
+    y.float().square().mean().backward()
+
+    x = F.conv2d(raw2, w2, cb2)
+    x = F.group_norm(x, G, ga2, be2, 1e-5)
+    x = F.prelu(x, sl2)
+    y_ref = x.max(dim=2).values  # (B, C, N)
+    y_ref.square().mean().backward()
+
+    tol = 3e-2 if autocast_on else 2e-3
+    assert torch.allclose(y.float(), y_ref, atol=tol, rtol=1e-2), (
+        (y.float() - y_ref).abs().max()
+    )
+    for a, b in zip(leaves, clones):
+        scale = b.grad.abs().max().item() + 1e-6
+        assert torch.allclose(a.grad.float(), b.grad, atol=0.03 * scale, rtol=5e-2), (
+            (a.grad.float() - b.grad).abs().max(), scale
+        )
