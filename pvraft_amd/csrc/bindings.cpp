@@ -601,7 +601,7 @@ std::vector<torch::Tensor> knn_gnmp_fwd(torch::Tensor raw, torch::Tensor W,
   auto& ws = persistent_ws((long)rows * 2, fopt);
   int nblk = (int)((N + 255) / 256);
   if (nblk < 1) nblk = 1;
-  const int chunks_f = C / 8;
+  const int chunks_f = C / 4;
   auto scratch = torch::empty({(long)rows * 2, (long)nblk * chunks_f * B}, fopt);
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
@@ -646,9 +646,9 @@ std::vector<torch::Tensor> knn_gnmp_bwd(
   const long n_out = (long)rows * 2 + C * 2 + 1;
   auto scratch = torch::empty({n_out, (long)nblk * B}, fopt);
   auto ws = torch::empty({n_out}, fopt);
-  const int chunks_b = C / 16;
+  const int chunks_b = C / 8;
   auto wscratch =
-      torch::empty({(long)C * 5, (long)nblk * chunks_b * B}, fopt);
+      torch::empty({(long)C * 5, (long)nblk * chunks_b * 2 * B}, fopt);
   auto ws2 = torch::empty({(long)C * 5}, fopt);
   auto draw_part = torch::empty({(long)chunks_b, B, 4, K, N}, fopt);
   auto draw = torch::empty({B, 4, K, N}, fopt);

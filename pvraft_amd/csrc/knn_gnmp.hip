@@ -248,7 +248,7 @@ __global__ __launch_bounds__(KG_THREADS) void kg_bwd_reduce_kernel(
 // dv[c, j, n] = (dxhat - (s1 + xhat*s2)/len) * rstd, dxhat only at the
 // pooled argmax j; d_raw[:, j, n] = sum_c W[c, :] * dv[c]; dW[c, :] +=
 // dv[c] * raw[:, j, n]; dcb[c] += dv[c].
-template <typename T, int KG_CH>
+template <typename T, int KG_CH, int JCH>
 __global__ __launch_bounds__(KG_THREADS) void kg_bwd_apply_kernel(
     const T *__restrict__ dyT,               // (B, N, C)
     const float *__restrict__ raw,           // (B, 4, K, N)
@@ -263,8 +263,14 @@ __global__ __launch_bounds__(KG_THREADS) void kg_bwd_apply_kernel(
     const float *__restrict__ slope_ptr) {
   const int b = blockIdx.z;
   const int B = gridDim.z;
-  const int chunk = blockIdx.y;
+  const int chunk = blockIdx.y % ((C + KG_CH - 1) / KG_CH);
+  const int jc = blockIdx.y / ((C + KG_CH - 1) / KG_CH);
   const int c0 = chunk * KG_CH;
+  // j-range owned by this block: the per-edge outputs partition cleanly
+  // over j (draw writes disjoint slices; dW partials fold per block)
+  const int jn = (K + JCH - 1) / JCH;
+  const int j_lo = jc * jn;
+  const int j_hi = min(K, j_lo + jn);
   const int Cg = C / G;
   const float slope = *slope_ptr;
   const float inv_n = 1.0f / (float)row_len;
@@ -319,7 +325,7 @@ __global__ __launch_bounds__(KG_THREADS) void kg_bwd_apply_kernel(
         ksel[c] = ks[e];
       }
     }
-    for (int j = 0; j < K; ++j) {
+    for (int j = j_lo; j < j_hi; ++j) {
       const float r0 = rawb[((long)0 * K + j) * N + n];
       const float r1 = rawb[((long)1 * K + j) * N + n];
       const float r2 = rawb[((long)2 * K + j) * N + n];
@@ -397,9 +403,9 @@ void launch_kg_fwd(const float *raw, const float *W, const float *cb,
                    int B, long N, int K, int C, int G, float eps,
                    const float *slope_ptr, bool bf16, int nblk,
                    hipStream_t stream) {
-  const int chunks = C / 8;
+  const int chunks = C / 4;
   const dim3 grid(nblk, chunks, B);
-  hipLaunchKernelGGL(kg_fwd_reduce_kernel<8>, grid, dim3(KG_THREADS), 0,
+  hipLaunchKernelGGL(kg_fwd_reduce_kernel<4>, grid, dim3(KG_THREADS), 0,
                      stream, raw, W, cb, scratch, vmax, vmin, amax, amin, N,
                      K, C, G);
   const int n_out = B * G * 2;
@@ -422,7 +428,7 @@ void launch_kg_bwd(const void *dyT, const float *raw, const float *W,
                    float *draw, int B, long N, int K, int C, int G,
                    const float *slope_ptr, bool bf16, int nblk,
                    hipStream_t stream) {
-  const int chunks = C / 16;
+  const int chunks = C / 8;
   const long row_len = (long)(C / G) * K * N;
   {
     const dim3 rgrid(nblk, 1, B);
@@ -443,15 +449,15 @@ void launch_kg_bwd(const void *dyT, const float *raw, const float *W,
                        stream, scratch, ws, (long)nblk * B, n_out);
   }
   {
-    const dim3 agrid(nblk, chunks, B);
+    const dim3 agrid(nblk, chunks * 2, B);
     if (bf16)
-      hipLaunchKernelGGL((kg_bwd_apply_kernel<__hip_bfloat16, 16>), agrid,
+      hipLaunchKernelGGL((kg_bwd_apply_kernel<__hip_bfloat16, 8, 2>), agrid,
                          dim3(KG_THREADS), 0, stream,
                          (const __hip_bfloat16 *)dyT, raw, W, cb, am, vsel,
                          mean, rstd, gamma, beta, ws, draw_part, wscratch, N,
                          K, C, G, row_len, slope_ptr);
     else
-      hipLaunchKernelGGL((kg_bwd_apply_kernel<float, 16>), agrid,
+      hipLaunchKernelGGL((kg_bwd_apply_kernel<float, 8, 2>), agrid,
                          dim3(KG_THREADS), 0, stream, (const float *)dyT,
                          raw, W, cb, am, vsel, mean, rstd, gamma, beta, ws,
                          draw_part, wscratch, N, K, C, G, row_len,
@@ -460,7 +466,7 @@ void launch_kg_bwd(const void *dyT, const float *raw, const float *W,
     const int wpb = KG_THREADS / WAVE;
     hipLaunchKernelGGL(egnmp_sum_partials_kernel,
                        dim3((n_out + wpb - 1) / wpb), dim3(KG_THREADS), 0,
-                       stream, wscratch, ws2, (long)nblk * chunks * B,
+                       stream, wscratch, ws2, (long)nblk * chunks * 2 * B,
                        n_out);
     const long per_b = (long)4 * K * N;
     long blocks = (per_b * B + KG_THREADS - 1) / KG_THREADS;
