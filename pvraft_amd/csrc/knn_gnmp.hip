@@ -363,7 +363,7 @@ __global__ __launch_bounds__(KG_THREADS) void kg_bwd_apply_kernel(
   __syncthreads();
   // wscratch rows: [C*4 dW | C dcb]; this chunk owns rows c0..c0+CH
   const long cols = (long)gridDim.x * gridDim.y * B;
-  const long col = ((long)blockIdx.x * gridDim.y + chunk) * B + b;
+  const long col = ((long)blockIdx.x * gridDim.y + blockIdx.y) * B + b;
   for (unsigned i = threadIdx.x; i < (unsigned)n_out; i += KG_THREADS) {
     float v = 0.f;
     if (i < (unsigned)(C * 4)) {
