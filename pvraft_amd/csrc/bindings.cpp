@@ -4,6 +4,8 @@
 #include <hip/hip_runtime.h>
 
 void launch_knn_graph(const float*, int*, int, int, int, hipStream_t);
+void launch_morton_keys(const float*, const float*, const float*, long*,
+                        int, long, hipStream_t);
 void launch_gather_edge_fwd(const void*, const int*, const float*, void*,
                             int, int, int, int, bool, hipStream_t);
 void launch_gather_edge_bwd(const void*, const int*, float*, int, int, int,
@@ -144,6 +146,21 @@ torch::Tensor& persistent_ws(long len, const torch::TensorOptions& opt) {
 }
 
 }  // namespace
+
+torch::Tensor morton_keys(torch::Tensor xyz, torch::Tensor mn,
+                          torch::Tensor inv_ext) {
+  check_f32(xyz, "xyz");
+  check_f32(mn, "mn");
+  check_f32(inv_ext, "inv_ext");
+  TORCH_CHECK(xyz.dim() == 3 && xyz.size(2) == 3, "xyz must be (B,N,3)");
+  const int B = xyz.size(0);
+  const long N = xyz.size(1);
+  auto keys = torch::empty({B, N}, xyz.options().dtype(torch::kInt64));
+  launch_morton_keys(xyz.data_ptr<float>(), mn.data_ptr<float>(),
+                     inv_ext.data_ptr<float>(), keys.data_ptr<long>(), B, N,
+                     stream());
+  return keys;
+}
 
 torch::Tensor knn_graph(torch::Tensor xyz, int64_t k) {
   check_f32(xyz, "xyz");
@@ -1089,6 +1106,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_norm_act_maxpool_bwd", &group_norm_act_maxpool_bwd);
   m.def("edge_gnmp_fwd", &edge_gnmp_fwd);
   m.def("edge_gnmp_bwd", &edge_gnmp_bwd);
+  m.def("morton_keys", &morton_keys);
   m.def("knn_gnmp_fwd", &knn_gnmp_fwd);
   m.def("knn_gnmp_bwd", &knn_gnmp_bwd);
   m.def("knn_graph", &knn_graph, "fused kNN graph (CDNA4)");

@@ -210,3 +210,45 @@ void launch_knn_graph(const float *xyz, int *out_idx, int B, int N, int k,
   hipLaunchKernelGGL(knn_select_kernel, grid, dim3(KNN_THREADS), 0, stream,
                      xyz, out_idx, N, k);
 }
+
+// 30-bit Morton (Z-order) keys for spatial point relabeling: gathers all
+// over the model (SetConv neighbour rows, correlation lookups) touch
+// RANDOM point ids on unordered clouds, so every 8 B quad pulls its own
+// cacheline.  Sorting points along a space-filling curve once per pair
+// makes kNN neighbourhoods id-local, and the gather kernels L2/L1-hit.
+__device__ __forceinline__ unsigned mg_part(unsigned v) {
+  v &= 0x3ffu;
+  v = (v | (v << 16)) & 0x030000FFu;
+  v = (v | (v << 8)) & 0x0300F00Fu;
+  v = (v | (v << 4)) & 0x030C30C3u;
+  v = (v | (v << 2)) & 0x09249249u;
+  return v;
+}
+
+__global__ void morton_keys_kernel(const float *__restrict__ xyz,
+                                   const float *__restrict__ mn,   // (B, 3)
+                                   const float *__restrict__ inv_ext,
+                                   long *__restrict__ keys, long N) {
+  const int b = blockIdx.y;
+  const float mx0 = mn[b * 3 + 0], mx1 = mn[b * 3 + 1], mx2 = mn[b * 3 + 2];
+  const float e0 = inv_ext[b * 3 + 0], e1 = inv_ext[b * 3 + 1],
+              e2 = inv_ext[b * 3 + 2];
+  for (long n = (long)blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += (long)gridDim.x * blockDim.x) {
+    const float *p = xyz + ((long)b * N + n) * 3;
+    unsigned q0 = (unsigned)fminf(fmaxf((p[0] - mx0) * e0, 0.f), 1023.f);
+    unsigned q1 = (unsigned)fminf(fmaxf((p[1] - mx1) * e1, 0.f), 1023.f);
+    unsigned q2 = (unsigned)fminf(fmaxf((p[2] - mx2) * e2, 0.f), 1023.f);
+    keys[(long)b * N + n] =
+        (long)((mg_part(q2) << 2) | (mg_part(q1) << 1) | mg_part(q0));
+  }
+}
+
+void launch_morton_keys(const float *xyz, const float *mn,
+                        const float *inv_ext, long *keys, int B, long N,
+                        hipStream_t stream) {
+  long blocks = (N + 255) / 256;
+  if (blocks > 1024) blocks = 1024;
+  hipLaunchKernelGGL(morton_keys_kernel, dim3((unsigned)blocks, B),
+                     dim3(256), 0, stream, xyz, mn, inv_ext, keys, N);
+}

@@ -15,11 +15,14 @@ checkpoints are interchangeable both ways.
 
 from __future__ import annotations
 
+import os
 from typing import List
 
 import torch
 import torch.nn as nn
 from torch import Tensor
+
+from pvraft_amd import ops
 
 from .corr import CorrBlock
 from .encoder import PointEncoder
@@ -64,6 +67,7 @@ class PVRaft(nn.Module):
 
         refresh_casts()  # re-fill the bf16 weight mirrors (one foreach)
         xyz1, xyz2 = p
+        xyz1, xyz2, inv1 = _morton_relabel(xyz1, xyz2)
         graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
         fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
         fmap2, _ = self.feature_extractor(xyz2)
@@ -90,7 +94,29 @@ class PVRaft(nn.Module):
             )
             coords2 = coords2 + delta_flow
             flow_predictions.append(coords2 - coords1)
+        if inv1 is not None:
+            g = inv1.unsqueeze(-1).expand(-1, -1, 3)
+            flow_predictions = [f.gather(1, g) for f in flow_predictions]
         return flow_predictions
+
+
+def _morton_relabel(xyz1, xyz2):
+    """Sort both clouds along the Morton curve (GPU path): kNN
+    neighbourhoods become id-local, so every gather kernel (SetConv
+    rows, correlation lookups, CSR walks) hits L2/L1 instead of pulling
+    one cacheline per 8 B quad.  Returns the relabeled clouds plus the
+    inverse permutation that maps pc1-aligned outputs back to the
+    caller's original point order (flow row i must describe input point
+    i).  Identity on CPU / reference mode."""
+    if not (xyz1.is_cuda and ops.hip_available()) or (
+        os.environ.get("PVRAFT_REF_OPS", "0") == "1"
+    ):
+        return xyz1, xyz2, None
+    perm1, inv1 = ops.morton_order(xyz1)
+    perm2, _ = ops.morton_order(xyz2)
+    g1 = perm1.unsqueeze(-1).expand(-1, -1, 3)
+    g2 = perm2.unsqueeze(-1).expand(-1, -1, 3)
+    return xyz1.gather(1, g1), xyz2.gather(1, g2), inv1
 
 
 class PVRaftRefine(nn.Module):
@@ -142,6 +168,7 @@ class PVRaftRefine(nn.Module):
         refresh_casts()
         with torch.no_grad():
             xyz1, xyz2 = p
+            xyz1, xyz2, inv1 = _morton_relabel(xyz1, xyz2)
             graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
             fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
             fmap2, _ = self.feature_extractor(xyz2)
@@ -163,7 +190,10 @@ class PVRaftRefine(nn.Module):
                     net, inp, corr, flow, graph_context, wcache, inp_pre=inp_pre
                 )
                 coords2 = coords2 + delta_flow
-        return self.refine_block(coords2 - coords1, graph1)
+        out = self.refine_block(coords2 - coords1, graph1)
+        if inv1 is not None:
+            out = out.gather(1, inv1.unsqueeze(-1).expand(-1, -1, 3))
+        return out
 
 
 def build_model(args):

@@ -429,6 +429,22 @@ def knn_gnmp(raw: Tensor, weight: Tensor, cbias: Tensor, num_groups: int,
 # ---------------------------------------------------------------------------
 
 
+def morton_order(xyz: Tensor):
+    """(B, N, 3) -> (perm, inv) int64 (B, N): point relabeling along a
+    30-bit Morton curve.  Gathers all over the model (SetConv neighbour
+    rows, correlation lookups) touch random point ids on unordered
+    clouds; Z-order relabeling makes kNN neighbourhoods id-local so those
+    kernels L2/L1-hit.  ``xyz.gather(1, perm...)`` sorts; ``gather(1,
+    inv...)`` restores the original order."""
+    mn = xyz.amin(dim=1).contiguous()
+    ext = (xyz.amax(dim=1) - mn).clamp_min(1e-9)
+    inv_ext = (1023.0 / ext).contiguous()
+    keys = _EXT.morton_keys(xyz.contiguous(), mn, inv_ext)
+    perm = keys.argsort(dim=1)
+    inv = perm.argsort(dim=1)
+    return perm, inv
+
+
 def knn_graph(xyz: Tensor, k: int) -> Tensor:
     """(B,N,3) -> (B,N,k) int64 neighbour indices (self included)."""
     xyz = xyz.contiguous().float()
