@@ -108,8 +108,10 @@ def _morton_relabel(xyz1, xyz2):
     inverse permutation that maps pc1-aligned outputs back to the
     caller's original point order (flow row i must describe input point
     i).  Identity on CPU / reference mode."""
-    if not (xyz1.is_cuda and ops.hip_available()) or (
-        os.environ.get("PVRAFT_REF_OPS", "0") == "1"
+    if (
+        not (xyz1.is_cuda and ops.hip_available())
+        or os.environ.get("PVRAFT_REF_OPS", "0") == "1"
+        or os.environ.get("PVRAFT_NO_MORTON", "0") == "1"
     ):
         return xyz1, xyz2, None
     perm1, inv1 = ops.morton_order(xyz1)
