@@ -436,6 +436,8 @@ def morton_order(xyz: Tensor):
     clouds; Z-order relabeling makes kNN neighbourhoods id-local so those
     kernels L2/L1-hit.  ``xyz.gather(1, perm...)`` sorts; ``gather(1,
     inv...)`` restores the original order."""
+    if not _use_hip(xyz):
+        raise RuntimeError("morton_order is a GPU-only op")
     mn = xyz.amin(dim=1).contiguous()
     ext = (xyz.amax(dim=1) - mn).clamp_min(1e-9)
     inv_ext = (1023.0 / ext).contiguous()
