@@ -62,12 +62,12 @@ class PVRaft(nn.Module):
             truncate_k=args.truncate_k,
         )
 
-    def forward(self, p, num_iters: int = 12) -> List[Tensor]:
+    def forward(self, p, num_iters: int = 12, morton: bool = True) -> List[Tensor]:
         from .pointwise import refresh_casts
 
         refresh_casts()  # re-fill the bf16 weight mirrors (one foreach)
         xyz1, xyz2 = p
-        xyz1, xyz2, inv1 = _morton_relabel(xyz1, xyz2)
+        xyz1, xyz2, inv1 = _morton_relabel(xyz1, xyz2, morton)
         graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
         fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
         fmap2, _ = self.feature_extractor(xyz2)
@@ -100,16 +100,24 @@ class PVRaft(nn.Module):
         return flow_predictions
 
 
-def _morton_relabel(xyz1, xyz2):
+def _morton_relabel(xyz1, xyz2, enabled: bool = True):
     """Sort both clouds along the Morton curve (GPU path): kNN
     neighbourhoods become id-local, so every gather kernel (SetConv
     rows, correlation lookups, CSR walks) hits L2/L1 instead of pulling
     one cacheline per 8 B quad.  Returns the relabeled clouds plus the
     inverse permutation that maps pc1-aligned outputs back to the
     caller's original point order (flow row i must describe input point
-    i).  Identity on CPU / reference mode."""
+    i).  Identity on CPU / reference mode.
+
+    ``enabled=False`` is for callers that pre-permute OUTSIDE a hipGraph
+    capture (engine/graphed.py): the in-capture argsort itself replays
+    fine, but its temp allocations shift the graph pool layout enough to
+    re-trigger the ROCm pool page-mapping fault (same toolchain bug as
+    the bs>=5 replay fault; reproduced bench-only, allocator-history
+    dependent)."""
     if (
-        not (xyz1.is_cuda and ops.hip_available())
+        not enabled
+        or not (xyz1.is_cuda and ops.hip_available())
         or os.environ.get("PVRAFT_REF_OPS", "0") == "1"
         or os.environ.get("PVRAFT_NO_MORTON", "0") == "1"
     ):
@@ -164,13 +172,13 @@ class PVRaftRefine(nn.Module):
             for pmt in m.parameters():
                 pmt.requires_grad_(False)
 
-    def forward(self, p, num_iters: int = 32) -> Tensor:
+    def forward(self, p, num_iters: int = 32, morton: bool = True) -> Tensor:
         from .pointwise import refresh_casts
 
         refresh_casts()
         with torch.no_grad():
             xyz1, xyz2 = p
-            xyz1, xyz2, inv1 = _morton_relabel(xyz1, xyz2)
+            xyz1, xyz2, inv1 = _morton_relabel(xyz1, xyz2, morton)
             graph1 = Graph.build(xyz1, self.feature_extractor.num_neighbors)
             fmap1, _ = self.feature_extractor(xyz1, graph=graph1)
             fmap2, _ = self.feature_extractor(xyz2)
