@@ -116,16 +116,17 @@ def build_graphed_step(model, batch, num_iters: int, gamma: float,
         def pre():
             with torch.no_grad():
                 perm1, inv1 = ops.morton_order(seq[0])
-                perm2, _ = ops.morton_order(seq[1])
+                perm2, _ = ops.morton_order(seq[1], need_inv=False)
                 inv_s.copy_(inv1)
                 g1 = perm1.unsqueeze(-1)
-                data_p["sequence"][0].copy_(
-                    seq[0].gather(1, g1.expand(-1, -1, seq[0].shape[-1])))
-                data_p["sequence"][1].copy_(seq[1].gather(
-                    1, perm2.unsqueeze(-1).expand(-1, -1, seq[1].shape[-1])))
+                torch.gather(seq[0], 1, g1.expand(-1, -1, seq[0].shape[-1]),
+                             out=data_p["sequence"][0])
+                torch.gather(seq[1], 1,
+                             perm2.unsqueeze(-1).expand(-1, -1, seq[1].shape[-1]),
+                             out=data_p["sequence"][1])
                 for dst, src in zip(data_p["ground_truth"], gt):
-                    dst.copy_(src.gather(
-                        1, g1.expand(-1, -1, src.shape[-1])))
+                    torch.gather(src, 1, g1.expand(-1, -1, src.shape[-1]),
+                                 out=dst)
 
     else:
         batch_use = batch

@@ -123,7 +123,7 @@ def _morton_relabel(xyz1, xyz2, enabled: bool = True):
     ):
         return xyz1, xyz2, None
     perm1, inv1 = ops.morton_order(xyz1)
-    perm2, _ = ops.morton_order(xyz2)
+    perm2, _ = ops.morton_order(xyz2, need_inv=False)
     g1 = perm1.unsqueeze(-1).expand(-1, -1, 3)
     g2 = perm2.unsqueeze(-1).expand(-1, -1, 3)
     return xyz1.gather(1, g1), xyz2.gather(1, g2), inv1

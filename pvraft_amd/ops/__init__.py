@@ -429,21 +429,34 @@ def knn_gnmp(raw: Tensor, weight: Tensor, cbias: Tensor, num_groups: int,
 # ---------------------------------------------------------------------------
 
 
-def morton_order(xyz: Tensor):
+_ARANGE_CACHE: dict = {}
+
+
+def morton_order(xyz: Tensor, need_inv: bool = True):
     """(B, N, 3) -> (perm, inv) int64 (B, N): point relabeling along a
     30-bit Morton curve.  Gathers all over the model (SetConv neighbour
     rows, correlation lookups) touch random point ids on unordered
     clouds; Z-order relabeling makes kNN neighbourhoods id-local so those
     kernels L2/L1-hit.  ``xyz.gather(1, perm...)`` sorts; ``gather(1,
-    inv...)`` restores the original order."""
+    inv...)`` restores the original order.  ``need_inv=False`` skips the
+    inverse (pc2 never needs one)."""
     if not _use_hip(xyz):
         raise RuntimeError("morton_order is a GPU-only op")
-    mn = xyz.amin(dim=1).contiguous()
-    ext = (xyz.amax(dim=1) - mn).clamp_min(1e-9)
-    inv_ext = (1023.0 / ext).contiguous()
+    mn, mx = torch.aminmax(xyz, dim=1)
+    mn = mn.contiguous()
+    inv_ext = (1023.0 / (mx - mn).clamp_min(1e-9)).contiguous()
     keys = _EXT.morton_keys(xyz.contiguous(), mn, inv_ext)
     perm = keys.argsort(dim=1)
-    inv = perm.argsort(dim=1)
+    inv = None
+    if need_inv:
+        B, N = perm.shape
+        ck = (B, N, xyz.device.index)
+        ar = _ARANGE_CACHE.get(ck)
+        if ar is None:
+            ar = torch.arange(N, device=xyz.device).expand(B, N).contiguous()
+            _ARANGE_CACHE[ck] = ar
+        inv = torch.empty_like(perm)
+        inv.scatter_(1, perm, ar)  # one scatter instead of a second sort
     return perm, inv
 
 
