@@ -93,6 +93,7 @@ def build_graphed_step(model, batch, num_iters: int, gamma: float,
     use_morton = (
         seq[0].is_cuda
         and ops.hip_available()
+        and num_iters >= 12  # pre-phase sorts amortise past ~12 iterations
         and os.environ.get("PVRAFT_REF_OPS", "0") != "1"
         and os.environ.get("PVRAFT_NO_MORTON", "0") != "1"
     )
