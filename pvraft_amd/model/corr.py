@@ -119,6 +119,7 @@ class CorrBlock(nn.Module):
             and feat.dtype == torch.float32
             and conv.out_channels % 16 == 0
             and os.environ.get("PVRAFT_REF_OPS", "0") != "1"
+            and os.environ.get("PVRAFT_NO_KNN_GNMP", "0") != "1"
         ):
             # one fused pipeline: the conv contraction is only 4-wide, so
             # it is evaluated inline with GN+PReLU+maxpool and the
