@@ -78,6 +78,9 @@ void launch_egnmp_bwd(const void*, const void*, const int*,
                       long, int, int, int, int, float, const float*, bool,
                       int, hipStream_t);
 int egnmp_reduce_chunks(long, int, int);
+struct CastDesc { const float* src; void* dst; int n; };
+struct CastChunk { CastDesc d[128]; int count; };
+void launch_multi_cast(const CastChunk*, int, hipStream_t);
 void launch_kg_fwd(const float*, const float*, const float*, float*, float*,
                    float*, float*, const float*, const float*, float*,
                    float*, unsigned char*, unsigned char*, void*,
@@ -686,6 +689,37 @@ std::vector<torch::Tensor> knn_gnmp_bwd(
   return {draw, dW, dcb, tail[1], tail[2], tail[3]};
 }
 
+// Capture-safe batched fp32->bf16 mirror refresh (csrc/cast_pack.hip):
+// pointer tables ride kernel arguments, so a hipGraph replay re-reads the
+// live fp32 weights (ATen _foreach_copy_'s staging upload does not).
+void multi_cast_bf16(std::vector<torch::Tensor> srcs,
+                     std::vector<torch::Tensor> dsts) {
+  const int n = (int)srcs.size();
+  TORCH_CHECK((int)dsts.size() == n, "src/dst count mismatch");
+  std::vector<CastChunk> chunks;
+  chunks.reserve((n + 127) / 128);
+  CastChunk cur;
+  cur.count = 0;
+  for (int i = 0; i < n; ++i) {
+    auto& s = srcs[i];
+    auto& d = dsts[i];
+    TORCH_CHECK(s.is_cuda() && s.is_contiguous() &&
+                s.scalar_type() == torch::kFloat32, "src must be fp32 contig");
+    TORCH_CHECK(d.is_cuda() && d.is_contiguous() &&
+                d.scalar_type() == torch::kBFloat16 && d.numel() == s.numel(),
+                "dst must be matching bf16");
+    cur.d[cur.count++] = CastDesc{s.data_ptr<float>(), d.data_ptr(),
+                                  (int)s.numel()};
+    if (cur.count == 128) {
+      chunks.push_back(cur);
+      cur.count = 0;
+    }
+  }
+  if (cur.count) chunks.push_back(cur);
+  if (!chunks.empty())
+    launch_multi_cast(chunks.data(), (int)chunks.size(), stream());
+}
+
 // dy (B, Co, S) bf16, x (B, Ci, S) bf16 -> dW (Co, Ci) fp32 (split-K MFMA)
 std::vector<torch::Tensor> pw_wgrad(torch::Tensor dy, torch::Tensor x,
                                     int64_t schunks = 0, bool with_bias = false) {
@@ -1107,6 +1141,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("edge_gnmp_fwd", &edge_gnmp_fwd);
   m.def("edge_gnmp_bwd", &edge_gnmp_bwd);
   m.def("morton_keys", &morton_keys);
+  m.def("multi_cast_bf16", &multi_cast_bf16);
   m.def("knn_gnmp_fwd", &knn_gnmp_fwd);
   m.def("knn_gnmp_bwd", &knn_gnmp_bwd);
   m.def("knn_graph", &knn_graph, "fused kNN graph (CDNA4)");

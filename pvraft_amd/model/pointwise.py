@@ -44,18 +44,40 @@ def refresh_casts() -> None:
         return
     srcs = []
     dsts = []
+    slow = []
     for key, ent in _CASTS.items():
-        if isinstance(key[0], tuple):
-            continue  # derived stack: re-filled when its fresh tensor arrives
+        if isinstance(key[0], tuple) and not isinstance(key[0][1], tuple):
+            # _cast_key-tagged derived stack (fresh storage each forward):
+            # re-filled when its fresh tensor arrives, skip here
+            continue
         # detach FRESH each refresh: a stored detached alias of a view
         # (e.g. weight.view(o, -1)) trips autograd's stale-view check
         # once the optimizer updates the base in place
-        srcs.append(ent[0].detach())
-        dsts.append(ent[1])
+        src = ent[0].detach()
         ent[2] = ent[0]._version
-    if srcs:
-        with torch.no_grad():
-            torch._foreach_copy_(dsts, srcs)
+        if (
+            src.is_cuda
+            and src.is_contiguous()
+            and src.dtype == torch.float32
+            and ent[1].dtype == torch.bfloat16
+        ):
+            srcs.append(src)
+            dsts.append(ent[1])
+        else:
+            slow.append((src, ent[1]))
+    with torch.no_grad():
+        if srcs:
+            # capture-safe batched refresh: pointer tables ride kernel
+            # arguments (csrc/cast_pack.hip), so a hipGraph replay
+            # re-reads the live fp32 weights.  ATen _foreach_copy_ was
+            # measured to FREEZE mirrors at capture values under replay
+            # (its device pointer staging is not re-uploaded) -- graphed
+            # training silently stopped learning.
+            from pvraft_amd import _C
+
+            _C.multi_cast_bf16(srcs, dsts)
+        for src, dst in slow:
+            dst.copy_(src)
 
 
 def clear_step_cache() -> None:

@@ -22,6 +22,7 @@ SOURCES = [
     "pvraft_amd/csrc/group_norm.hip",
     "pvraft_amd/csrc/edge_gnmp.hip",
     "pvraft_amd/csrc/knn_gnmp.hip",
+    "pvraft_amd/csrc/cast_pack.hip",
     "pvraft_amd/csrc/pw_wgrad.hip",
     "pvraft_amd/csrc/pw_fwd.hip",
     "pvraft_amd/csrc/transpose.hip",
