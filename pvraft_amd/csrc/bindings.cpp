@@ -622,7 +622,13 @@ std::vector<torch::Tensor> knn_gnmp_fwd(torch::Tensor raw, torch::Tensor W,
   int nblk = (int)((N + 255) / 256);
   if (nblk < 1) nblk = 1;
   const int chunks_f = C / 4;
-  auto scratch = torch::empty({(long)rows * 2, (long)nblk * chunks_f * B}, fopt);
+  // transient (write-then-read within this enqueued call): persistent
+  // buffers, NOT per-call pool allocations -- inside a captured training
+  // step the per-call workspaces (~85 MB x 12 calls) push the graph pool
+  // into the ROCm page-mapping-bug regime (silent garbage at replay;
+  // the op alone replays clean, scripts/graph_kg_repro.py)
+  auto scratch = persistent_ws((long)rows * 2 * nblk * chunks_f * B, fopt)
+                     .view({(long)rows * 2, (long)nblk * chunks_f * B});
   auto mean = torch::empty({rows}, fopt);
   auto rstd = torch::empty({rows}, fopt);
   auto vmax = torch::empty({B, N, C}, fopt);
@@ -664,13 +670,21 @@ std::vector<torch::Tensor> knn_gnmp_bwd(
   int nblk = (int)((N + 255) / 256);
   if (nblk < 1) nblk = 1;
   const long n_out = (long)rows * 2 + C * 2 + 1;
-  auto scratch = torch::empty({n_out, (long)nblk * B}, fopt);
-  auto ws = torch::empty({n_out}, fopt);
+  // transient call-internal workspaces: persistent (see knn_gnmp_fwd).
+  // draw (the returned gradient) and ws2 (dW/dcb grad views) stay
+  // per-call: autograd's execution order gives no cross-node liveness
+  // guarantee for returned gradients.
+  auto scratch = persistent_ws(n_out * nblk * B, fopt)
+                     .view({n_out, (long)nblk * B});
+  auto ws = persistent_ws(n_out, fopt);
   const int chunks_b = C / 8;
   auto wscratch =
-      torch::empty({(long)C * 5, (long)nblk * chunks_b * 2 * B}, fopt);
+      persistent_ws((long)C * 5 * nblk * chunks_b * 2 * B, fopt)
+          .view({(long)C * 5, (long)nblk * chunks_b * 2 * B});
   auto ws2 = torch::empty({(long)C * 5}, fopt);
-  auto draw_part = torch::empty({(long)chunks_b, B, 4, K, N}, fopt);
+  auto draw_part =
+      persistent_ws((long)chunks_b * B * 4 * K * N, fopt)
+          .view({(long)chunks_b, B, 4, K, N});
   auto draw = torch::empty({B, 4, K, N}, fopt);
   launch_kg_bwd(dyT.data_ptr(), raw.data_ptr<float>(), W.data_ptr<float>(),
                 cb.data_ptr<float>(), am.data_ptr<unsigned char>(),
