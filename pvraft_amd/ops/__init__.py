@@ -400,15 +400,36 @@ class _KnnGNMP(torch.autograd.Function):
             raw, w, cb, num_groups, ga, be, eps, st, out_bf16)
         ctx.save_for_backward(raw, w, cb, am, vsel, mean, rstd, ga, be, st)
         ctx.conf = (num_groups, weight.shape, weight.dtype)
+        ctx.param_refs = (weight, cbias, gamma, beta, slope_t)
         return y
 
     @staticmethod
     def backward(ctx, dy):
+        from pvraft_amd.model import pointwise
+
         raw, w, cb, am, vsel, mean, rstd, ga, be, st = ctx.saved_tensors
         num_groups, wshape, wdtype = ctx.conf
         draw, dW, dcb, dgamma, dbeta, dslope = _EXT.knn_gnmp_bwd(
             dy.contiguous(), raw, w, cb, am, vsel, mean, rstd, num_groups,
             ga, be, st)
+        params = ctx.param_refs
+        if pointwise.wgrad_defer_active() and all(
+            p.is_leaf and p.requires_grad for p in params
+        ):
+            # accumulate in-place on the CURRENT stream and return no
+            # grads: an AccumulateGrad node pinned to the warmup side
+            # stream mis-orders against the producing kernels under
+            # hipGraph capture (replays read the grad workspaces before
+            # they are written -- measured 1e13-garbage conv grads,
+            # scripts/graph_step_gradcheck.py); every other weight grad
+            # already rides the deferred/drain path for the same reason.
+            wt, cbt, gat, bet, slt = params
+            pointwise._grad_buffer(wt).view(-1).add_(dW.view(-1))
+            pointwise._grad_buffer(cbt).add_(dcb)
+            pointwise._grad_buffer(gat).add_(dgamma)
+            pointwise._grad_buffer(bet).add_(dbeta)
+            pointwise._grad_buffer(slt).view(-1).add_(dslope)
+            return (draw, None, None, None, None, None, None, None)
         return (draw, dW.view(wshape).to(wdtype), dcb, None, dgamma, dbeta,
                 None, dslope.reshape(1))
 
