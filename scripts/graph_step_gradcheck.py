@@ -47,6 +47,16 @@ def main():
         torch.cuda.synchronize()
         ggrads = {n: p.grad.clone() for n, p in model.named_parameters()
                   if p.grad is not None}
+        # determinism probe: replay AGAIN on the same data -- a race gives
+        # different garbage, a deterministic bug gives identical values
+        graphed.replay()
+        torch.cuda.synchronize()
+        kk = "corr_block.knn_conv.0.weight"
+        g2 = dict(model.named_parameters())[kk].grad
+        same = torch.equal(ggrads[kk], g2)
+        print(f"  replay-determinism({kk}): {'identical' if same else 'DIFFERS'}"
+              f" max1={ggrads[kk].abs().max().item():.3e}"
+              f" max2={g2.abs().max().item():.3e}")
 
         # eager step, same weights, same data (weights were never updated)
         model.load_state_dict(sd0)
