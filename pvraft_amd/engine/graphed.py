@@ -58,8 +58,13 @@ class GraphedTrainStep:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
 
+        # capture on the SAME stream the warmup ran on: per-leaf
+        # AccumulateGrad nodes are pinned to the stream of the first
+        # backward that touched them (the warmup stream) and a capture on
+        # a different stream records their writes as cross-stream work
+        # that RACES at replay
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        with torch.cuda.graph(self.graph, stream=s):
             self.static_loss, self.static_final_flow = self.fn()
 
     def replay(self) -> torch.Tensor:

@@ -55,7 +55,10 @@ class Predictor:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
+        # capture on the warmup stream (see engine/graphed.py: nodes
+        # pinned to another stream record cross-stream work that races
+        # at replay)
+        with torch.cuda.graph(self._graph, stream=s):
             self._flows = self._forward()
 
     @torch.no_grad()

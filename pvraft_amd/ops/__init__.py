@@ -47,7 +47,13 @@ def _gn_defer_targets(weight, bias, slope_t, act):
     tensors and their AccumulateGrad add launches (~190/step)."""
     from pvraft_amd.model import pointwise
 
-    if not (pointwise._DEFER.on and torch.is_grad_enabled()):
+    # NOTE: checked in BACKWARD, where the engine runs custom Functions
+    # with grad mode DISABLED -- an is_grad_enabled() condition here is
+    # always False and silently reroutes every GN weight grad through
+    # AccumulateGrad (whose nodes are pinned to the FIRST-backward
+    # stream; inside a hipGraph capture that cross-stream write races at
+    # replay -- measured 1e13 garbage grads, scripts/graph_step_gradcheck.py)
+    if not pointwise._DEFER.on:
         return None
     for t in (weight, bias):
         if not (t.is_leaf and t.requires_grad and t.dtype == torch.float32):
@@ -413,7 +419,7 @@ class _KnnGNMP(torch.autograd.Function):
             dy.contiguous(), raw, w, cb, am, vsel, mean, rstd, num_groups,
             ga, be, st)
         params = ctx.param_refs
-        if pointwise.wgrad_defer_active() and all(
+        if pointwise._DEFER.on and all(
             p.is_leaf and p.requires_grad for p in params
         ):
             # accumulate in-place on the CURRENT stream and return no
