@@ -37,9 +37,9 @@
 #include "common.h"
 
 #define KNN_THREADS 256
-#define TILE_PTS 1280
+#define TILE_PTS 1792
 #define QB 8            // queries per workgroup
-#define CAP 192         // collect-buffer capacity per query
+#define CAP 256         // collect-buffer capacity per query
 #define SAMP 1024       // max sample size per query
 #define KNN_MAXK 48     // model uses 32 (reference extractor.py:10)
 
