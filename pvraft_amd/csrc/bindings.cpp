@@ -679,8 +679,8 @@ std::vector<torch::Tensor> knn_gnmp_bwd(
   auto ws = persistent_ws(n_out, fopt);
   const int chunks_b = C / 8;
   auto wscratch =
-      persistent_ws((long)C * 5 * nblk * chunks_b * 2 * B, fopt)
-          .view({(long)C * 5, (long)nblk * chunks_b * 2 * B});
+      persistent_ws((long)C * 5 * nblk * chunks_b * 4 * B, fopt)
+          .view({(long)C * 5, (long)nblk * chunks_b * 4 * B});
   auto ws2 = torch::empty({(long)C * 5}, fopt);
   auto draw_part =
       persistent_ws((long)chunks_b * B * 4 * K * N, fopt)

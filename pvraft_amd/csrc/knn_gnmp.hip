@@ -449,15 +449,15 @@ void launch_kg_bwd(const void *dyT, const float *raw, const float *W,
                        stream, scratch, ws, (long)nblk * B, n_out);
   }
   {
-    const dim3 agrid(nblk, chunks * 2, B);
+    const dim3 agrid(nblk, chunks * 4, B);
     if (bf16)
-      hipLaunchKernelGGL((kg_bwd_apply_kernel<__hip_bfloat16, 8, 2>), agrid,
+      hipLaunchKernelGGL((kg_bwd_apply_kernel<__hip_bfloat16, 8, 4>), agrid,
                          dim3(KG_THREADS), 0, stream,
                          (const __hip_bfloat16 *)dyT, raw, W, cb, am, vsel,
                          mean, rstd, gamma, beta, ws, draw_part, wscratch, N,
                          K, C, G, row_len, slope_ptr);
     else
-      hipLaunchKernelGGL((kg_bwd_apply_kernel<float, 8, 2>), agrid,
+      hipLaunchKernelGGL((kg_bwd_apply_kernel<float, 8, 4>), agrid,
                          dim3(KG_THREADS), 0, stream, (const float *)dyT,
                          raw, W, cb, am, vsel, mean, rstd, gamma, beta, ws,
                          draw_part, wscratch, N, K, C, G, row_len,
@@ -466,7 +466,7 @@ void launch_kg_bwd(const void *dyT, const float *raw, const float *W,
     const int wpb = KG_THREADS / WAVE;
     hipLaunchKernelGGL(egnmp_sum_partials_kernel,
                        dim3((n_out + wpb - 1) / wpb), dim3(KG_THREADS), 0,
-                       stream, wscratch, ws2, (long)nblk * chunks * 2 * B,
+                       stream, wscratch, ws2, (long)nblk * chunks * 4 * B,
                        n_out);
     const long per_b = (long)4 * K * N;
     long blocks = (per_b * B + KG_THREADS - 1) / KG_THREADS;
