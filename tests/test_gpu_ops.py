@@ -671,3 +671,60 @@ def test_knn_gnmp_matches_reference(B, N, K, C, G, autocast_on):
         assert torch.allclose(a.grad.float(), b.grad, atol=0.03 * scale, rtol=5e-2), (
             (a.grad.float() - b.grad).abs().max(), scale
         )
+
+def test_morton_order_valid_permutation():
+    torch.manual_seed(5)
+    xyz = torch.randn(3, 2048, 3, device=dev())
+    perm, inv = ops.morton_order(xyz)
+    B, N = perm.shape
+    ar = torch.arange(N, device=dev())
+    # perm and inv are permutations and mutual inverses
+    assert torch.equal(perm.sort(1).values, ar.expand(B, N))
+    assert torch.equal(inv.sort(1).values, ar.expand(B, N))
+    assert torch.equal(perm.gather(1, inv), ar.expand(B, N))
+    # gathering by perm sorts the Morton keys
+    mn, mx = torch.aminmax(xyz, dim=1)
+    inv_ext = (1023.0 / (mx - mn).clamp_min(1e-9)).contiguous()
+    import pvraft_amd._C as _C
+    keys = _C.morton_keys(xyz.contiguous(), mn.contiguous(), inv_ext)
+    ks = keys.gather(1, perm)
+    assert torch.all(ks[:, 1:] >= ks[:, :-1])
+
+
+def test_multi_cast_bf16_matches_to():
+    import pvraft_amd._C as _C
+
+    torch.manual_seed(6)
+    srcs = [torch.randn((i % 7) + 1, (i % 13) + 1, device=dev())
+            for i in range(150)]
+    dsts = [torch.empty_like(t, dtype=torch.bfloat16) for t in srcs]
+    _C.multi_cast_bf16(srcs, dsts)
+    torch.cuda.synchronize()
+    for s_, d in zip(srcs, dsts):
+        assert torch.equal(d, s_.to(torch.bfloat16))
+
+
+def test_model_output_invariant_to_morton_relabel():
+    """Morton relabeling permutes internals but must return flows in the
+    caller's point order: compare a 16-iter forward (relabeling active)
+    against PVRAFT_NO_MORTON on the same weights/inputs.  Not bitwise:
+    kNN ties and fp32-atomic GN stats differ between labelings."""
+    from pvraft_amd.model import PVRaft
+
+    torch.manual_seed(11)
+    model = PVRaft(truncate_k=256).to(dev())
+    x1 = torch.randn(2, 2048, 3, device=dev())
+    x2 = x1 + 0.05 * torch.randn_like(x1)
+    with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16):
+        f_m = model([x1, x2], num_iters=16)[-1].float()
+    os.environ["PVRAFT_NO_MORTON"] = "1"
+    try:
+        with torch.no_grad(), torch.autocast("cuda", dtype=torch.bfloat16):
+            f_p = model([x1, x2], num_iters=16)[-1].float()
+    finally:
+        os.environ.pop("PVRAFT_NO_MORTON", None)
+    cos = torch.nn.functional.cosine_similarity(
+        f_m.flatten(), f_p.flatten(), dim=0).item()
+    assert cos > 0.98, cos
+    scale = f_p.abs().max().item() + 1e-6
+    assert (f_m - f_p).abs().median().item() < 0.05 * scale
