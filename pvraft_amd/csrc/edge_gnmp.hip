@@ -68,20 +68,6 @@ struct alignas(4 * sizeof(T)) Quad {
   T v[4];
 };
 
-// e = j*N + n  ->  (j, n) without a hardware integer division
-DEV_INLINE void split_edge(int e, int N, float invN, int &j, int &n) {
-  j = (int)((float)e * invN);
-  n = e - j * N;
-  while (n < 0) {
-    --j;
-    n += N;
-  }
-  while (n >= N) {
-    ++j;
-    n -= N;
-  }
-}
-
 // ---------------------------------------------------------------- forward
 
 // pass 1: per-block partial sum/sumsq per group over all (c, j, n)
