@@ -1,10 +1,10 @@
 // K8: fused kNN-correlation branch head: Conv2d(4->C) -> GroupNorm ->
 // PReLU -> max over k, in one kernel pipeline.
 //
-// Reference semantics (model/corr.py:70-76 + 44-56): the kNN branch feeds
-// raw per-candidate features [corr; rel-xyz] (B, 4, K, N) through a 1x1
-// conv to C=64 channels, GroupNorm(8), PReLU, then max-pools over the K
-// candidates.  Run as a GEMM the conv materialises a (B, C, K, N)
+// Reference semantics (model/corr.py:75-98 kNN branch; knn_conv defined
+// at corr.py:23-29): raw per-candidate features [corr; rel-xyz]
+// (B, 4, K, N) run through a 1x1 conv to C=64 channels, GroupNorm(8),
+// PReLU, then max-pool over the K candidates.  Run as a GEMM the conv materialises a (B, C, K, N)
 // activation (~33 M elements at the flagship shape) that GroupNorm and
 // the pool each re-read -- ~1.5 ms/step of traffic + launches for a
 // 4-wide contraction.
