@@ -31,8 +31,14 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define PF_THREADS 256
-#define PF_TC 64        // output columns per workgroup
 #define PF_MAXP 4       // max summed parts
+// Output columns per workgroup is a TEMPLATE parameter: at the flagship
+// S=8192, 64-column tiles give only ceil(S/64)*B = 256 workgroups -- one
+// per CU, 4 waves resident, latency-bound (~10 us for a ~2 us problem).
+// Narrower tiles split the FOUR waves between column tiles and
+// row-fragment halves (same LDS traffic, more workgroups): TC=32 doubles
+// the grid, TC=16 quadruples it.  The launcher picks the widest TC that
+// still yields >= 2 workgroups per CU.
 
 struct PwFwdPart {
   const void *w;  // (Co, Ci) bf16 row-major
@@ -40,7 +46,7 @@ struct PwFwdPart {
   int ci;
 };
 
-template <int NRF>
+template <int NRF, int TC>
 __global__ __launch_bounds__(PF_THREADS) void pw_fwd_kernel(
     PwFwdPart p0, PwFwdPart p1, PwFwdPart p2, PwFwdPart p3, int nparts,
     const float *__restrict__ bias,   // (Co) fp32 or null
@@ -48,18 +54,23 @@ __global__ __launch_bounds__(PF_THREADS) void pw_fwd_kernel(
     long addend_bstride, int addend_fp32,
     __hip_bfloat16 *__restrict__ y,   // (B, Co, S)
     int Co, long S, int act, int pitch) {
+  constexpr int COL16 = TC / 16;        // column tiles per WG
+  constexpr int RFG = 4 / COL16;        // wave groups along rows
+  constexpr int NRF_PER = (NRF + RFG - 1) / RFG;  // row frags per wave
   extern __shared__ __hip_bfloat16 smem[];
   __hip_bfloat16 *s_w = smem;                       // (NRF*16, pitch)
-  __hip_bfloat16 *s_x = smem + (long)NRF * 16 * pitch;  // (PF_TC, pitch)
+  __hip_bfloat16 *s_x = smem + (long)NRF * 16 * pitch;  // (TC, pitch)
 
   const int b = blockIdx.z;
-  const long s0 = (long)blockIdx.x * PF_TC;
+  const long s0 = (long)blockIdx.x * TC;
   const int lane = lane_id();
   const int wv = wave_id();
+  const int ct = wv % COL16;            // this wave's column tile
+  const int rf0 = (wv / COL16) * NRF_PER;  // first row fragment
 
-  f32x4 acc[NRF];
+  f32x4 acc[NRF_PER];
 #pragma unroll
-  for (int rf = 0; rf < NRF; ++rf) acc[rf] = (f32x4)(0.f);
+  for (int rf = 0; rf < NRF_PER; ++rf) acc[rf] = (f32x4)(0.f);
 
 #pragma unroll
   for (int pi = 0; pi < PF_MAXP; ++pi) {
@@ -92,9 +103,9 @@ __global__ __launch_bounds__(PF_THREADS) void pw_fwd_kernel(
     }
     // stage the activation tile TRANSPOSED: read x[r][s0+c] coalesced,
     // write s_x[c][r]
-    for (int i = threadIdx.x; i < cip * (PF_TC / 4); i += PF_THREADS) {
-      const int r = i / (PF_TC / 4);         // channel (k)
-      const int c4 = (i % (PF_TC / 4)) * 4;  // column group
+    for (int i = threadIdx.x; i < cip * (TC / 4); i += PF_THREADS) {
+      const int r = i / (TC / 4);         // channel (k)
+      const int c4 = (i % (TC / 4)) * 4;  // column group
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
         const long s = s0 + c4 + e;
@@ -107,27 +118,29 @@ __global__ __launch_bounds__(PF_THREADS) void pw_fwd_kernel(
     const int koff = (lane >> 4) * 8;
     for (int kb = 0; kb < cip; kb += 32) {
       const bf16x8 bfrag =
-          *(const bf16x8 *)&s_x[(long)(wv * 16 + frow) * pitch + kb + koff];
+          *(const bf16x8 *)&s_x[(long)(ct * 16 + frow) * pitch + kb + koff];
 #pragma unroll
-      for (int rf = 0; rf < NRF; ++rf) {
-        const bf16x8 afrag =
-            *(const bf16x8 *)&s_w[(long)(rf * 16 + frow) * pitch + kb + koff];
-        acc[rf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                          acc[rf], 0, 0, 0);
+      for (int rf = 0; rf < NRF_PER; ++rf) {
+        if (rf0 + rf < NRF) {  // wave-uniform guard (keeps the unroll)
+          const bf16x8 afrag = *(const bf16x8 *)
+              &s_w[(long)((rf0 + rf) * 16 + frow) * pitch + kb + koff];
+          acc[rf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                            acc[rf], 0, 0, 0);
+        }
       }
     }
   }
 
   // epilogue: bias + addend + activation, bf16 store
-  const long sc = s0 + wv * 16 + (lane & 15);
+  const long sc = s0 + ct * 16 + (lane & 15);
   if (sc >= S) return;
   const int crow = (lane >> 4) * 4;
 #pragma unroll
-  for (int rf = 0; rf < NRF; ++rf)
+  for (int rf = 0; rf < NRF_PER; ++rf)
 #pragma unroll
     for (int e = 0; e < 4; ++e) {
-      const int o = rf * 16 + crow + e;
-      if (o < Co) {
+      const int o = (rf0 + rf) * 16 + crow + e;
+      if (rf0 + rf < NRF && o < Co) {
         float v = acc[rf][e];
         if (bias != nullptr) v += bias[o];
         if (addend != nullptr) {
@@ -154,13 +167,26 @@ void launch_pw_fwd(const void **ws, const void **xs, const int *cis,
   }
   const int pitch = cip_max + 8;
   const int nrf = (Co + 15) / 16;
-  const dim3 grid((unsigned)((S + PF_TC - 1) / PF_TC), 1, B);
-  const size_t shmem = (size_t)(nrf * 16 + PF_TC) * pitch * sizeof(__hip_bfloat16);
+  // widest tile that still fills >= 2 workgroups per CU
+  int tc = 64;
+  if ((long)B * ((S + 63) / 64) < 512) tc = 32;
+  if ((long)B * ((S + 31) / 32) < 512) tc = 16;
+#define PF_LAUNCH_T(NRF, TC)                                                  \
+  do {                                                                        \
+    const dim3 grid((unsigned)((S + TC - 1) / TC), 1, B);                     \
+    const size_t shmem =                                                      \
+        (size_t)(nrf * 16 + TC) * pitch * sizeof(__hip_bfloat16);             \
+    hipLaunchKernelGGL((pw_fwd_kernel<NRF, TC>), grid, dim3(PF_THREADS),      \
+                       shmem, stream, p[0], p[1], p[2], p[3], nparts, bias,   \
+                       addend, addend_bstride, addend_fp32,                   \
+                       (__hip_bfloat16 *)y, Co, S, act, pitch);               \
+  } while (0)
 #define PF_LAUNCH(NRF)                                                        \
-  hipLaunchKernelGGL((pw_fwd_kernel<NRF>), grid, dim3(PF_THREADS), shmem,     \
-                     stream, p[0], p[1], p[2], p[3], nparts, bias, addend,    \
-                     addend_bstride, addend_fp32, (__hip_bfloat16 *)y, Co, S, \
-                     act, pitch)
+  do {                                                                        \
+    if (tc == 64) PF_LAUNCH_T(NRF, 64);                                       \
+    else if (tc == 32) PF_LAUNCH_T(NRF, 32);                                  \
+    else PF_LAUNCH_T(NRF, 16);                                                \
+  } while (0)
   switch (nrf) {
     case 1: PF_LAUNCH(1); break;
     case 2: PF_LAUNCH(2); break;
@@ -174,4 +200,5 @@ void launch_pw_fwd(const void **ws, const void **xs, const int *cis,
     default: PF_LAUNCH(16); break;
   }
 #undef PF_LAUNCH
+#undef PF_LAUNCH_T
 }
