@@ -116,6 +116,12 @@ _DEFER = _DeferState()
 
 
 def wgrad_defer_active() -> bool:
+    """FORWARD-side check (wcache construction): grad mode reflects the
+    caller there, so a no_grad inference forward skips the defer buffers.
+    BACKWARD-side code must test ``_DEFER.on`` directly instead -- the
+    engine runs Function.backward with grad mode DISABLED, and gating on
+    is_grad_enabled() there silently reroutes weight grads through
+    AccumulateGrad (see ops._gn_defer_targets)."""
     return _DEFER.on and torch.is_grad_enabled()
 
 
