@@ -48,10 +48,10 @@
 
 #define EG_THREADS 256
 
-// shared finalize pass (mean/rstd from the summed workspace, re-zeroing
-// it) lives in group_norm.hip
-void launch_gn_finalize(float *, float *, float *, long, int, float,
-                        hipStream_t);
+// shared finalize pass (mean/rstd folded straight from the per-block
+// partial scratch) lives in group_norm.hip
+void launch_gn_finalize_scratch(const float *, long, float *, float *, long,
+                                int, float, hipStream_t);
 
 template <typename T>
 DEV_INLINE float ldg(const T *p) {
@@ -472,13 +472,9 @@ void egnmp_fwd_impl(const T *wg, const int *idx, float *scratch, float *ws,
   hipLaunchKernelGGL(egnmp_fwd_reduce_kernel<T>, rgrid, dim3(EG_THREADS),
                      (size_t)n_out_f * sizeof(float), stream, wg, idx,
                      scratch, vmax, vmin, amax, amin, vsum, N, K, M, G);
-  const int waves_per_block = EG_THREADS / WAVE;
-  hipLaunchKernelGGL(egnmp_sum_partials_kernel,
-                     dim3((n_out_f + waves_per_block - 1) / waves_per_block),
-                     dim3(EG_THREADS), 0, stream, scratch, ws,
-                     (long)rchunks * B, n_out_f);
-  launch_gn_finalize(ws, mean, rstd, (long)(M / G) * K * N, B * G, eps,
-                     stream);
+  (void)ws;
+  launch_gn_finalize_scratch(scratch, (long)rchunks * B, mean, rstd,
+                             (long)(M / G) * K * N, B * G, eps, stream);
   const long total = (long)B * N * M;
   long pb = (total / 4 + EG_THREADS - 1) / EG_THREADS;
   if (pb > 2048) pb = 2048;

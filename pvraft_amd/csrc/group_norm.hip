@@ -155,6 +155,44 @@ __global__ void gn_fwd_finalize_kernel(float *__restrict__ ws,
   rstd[row] = rsqrtf(fmaxf(var, 0.f) + eps);
 }
 
+// Finalize DIRECTLY from the per-block partial scratch (rows*2, cols):
+// one wave per row folds both halves inline, so the egnmp/knn_gnmp
+// forwards need no separate sum-partials launch and no persistent ws at
+// all (~26 extra 6-us launches/step removed).
+__global__ void gn_finalize_scratch_kernel(const float *__restrict__ scratch,
+                                           long cols,
+                                           float *__restrict__ mean,
+                                           float *__restrict__ rstd,
+                                           long row_len, int rows,
+                                           float eps) {
+  const int row = blockIdx.x * (blockDim.x / WAVE) + wave_id();
+  if (row >= rows) return;
+  const float *s0 = scratch + (long)(row * 2 + 0) * cols;
+  const float *s1 = scratch + (long)(row * 2 + 1) * cols;
+  float a = 0.f, b = 0.f;
+  for (long i = lane_id(); i < cols; i += WAVE) {
+    a += s0[i];
+    b += s1[i];
+  }
+  a = wave_sum(a);
+  b = wave_sum(b);
+  if (lane_id() == 0) {
+    const float m = a / (float)row_len;
+    const float var = b / (float)row_len - m * m;
+    mean[row] = m;
+    rstd[row] = rsqrtf(fmaxf(var, 0.f) + eps);
+  }
+}
+
+void launch_gn_finalize_scratch(const float *scratch, long cols, float *mean,
+                                float *rstd, long row_len, int rows,
+                                float eps, hipStream_t stream) {
+  const int wpb = 256 / WAVE;
+  hipLaunchKernelGGL(gn_finalize_scratch_kernel,
+                     dim3((rows + wpb - 1) / wpb), dim3(256), 0, stream,
+                     scratch, cols, mean, rstd, row_len, rows, eps);
+}
+
 // pass 3: y = act((x - mean) * rstd * gamma + beta), per channel slice.
 // Stats are folded inline from the reduce partials; the (blockIdx.x==0,
 // first channel of the group) block also writes mean/rstd for backward.

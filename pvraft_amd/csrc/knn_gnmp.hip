@@ -40,8 +40,8 @@
 
 // shared folding / finalize / pick infrastructure (edge_gnmp.hip,
 // group_norm.hip)
-void launch_gn_finalize(float *, float *, float *, long, int, float,
-                        hipStream_t);
+void launch_gn_finalize_scratch(const float *, long, float *, float *, long,
+                                int, float, hipStream_t);
 __global__ void egnmp_sum_partials_kernel(const float *__restrict__,
                                           float *__restrict__, long, int);
 void launch_gnmp_pick(const float *, const float *, const unsigned char *,
@@ -408,13 +408,9 @@ void launch_kg_fwd(const float *raw, const float *W, const float *cb,
   hipLaunchKernelGGL(kg_fwd_reduce_kernel<4>, grid, dim3(KG_THREADS), 0,
                      stream, raw, W, cb, scratch, vmax, vmin, amax, amin, N,
                      K, C, G);
-  const int n_out = B * G * 2;
-  const int wpb = KG_THREADS / WAVE;
-  hipLaunchKernelGGL(egnmp_sum_partials_kernel,
-                     dim3((n_out + wpb - 1) / wpb), dim3(KG_THREADS), 0,
-                     stream, scratch, ws, (long)nblk * chunks * B, n_out);
-  launch_gn_finalize(ws, mean, rstd, (long)(C / G) * K * N, B * G, eps,
-                     stream);
+  (void)ws;
+  launch_gn_finalize_scratch(scratch, (long)nblk * chunks * B, mean, rstd,
+                             (long)(C / G) * K * N, B * G, eps, stream);
   launch_gnmp_pick(vmax, vmin, amax, amin, mean, rstd, gamma, beta, y, am,
                    vsel, (long)B * N * C, (long)N * C, C, G, 2, 0.f,
                    slope_ptr, bf16, stream);
