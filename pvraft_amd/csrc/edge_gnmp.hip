@@ -29,19 +29,23 @@
 // kernel (trivial next to the 100 MB it replaces).
 //
 // Reductions are DETERMINISTIC: each block writes its partial sums to a
-// per-block scratch slot (no global atomics) and a wave-per-output sum
-// kernel folds the partials into the workspace the shared finalize /
-// extract kernels consume.
+// per-block scratch slot (no global atomics); the forward GN stats fold
+// straight from that scratch in the finalize kernel, the backward sums
+// through one wave-per-output fold.
 //
 // Backward data path is deterministic too: the gradient w.r.t. WgT at
 // point p is
 //   dWg[m, p] = sum_{edges e=(j,n): idx[n,j]=p} dx1[m, j, n]   (incoming)
 //             - sum_j dx1[m, j, p]                             (centre)
 // where dx1 is the standard GroupNorm+act+maxpool backward element
-// (exact same formulas as gnmp_bwd_* in group_norm.hip), recomputed on
-// the fly from WgT / argmax / saved stats.  The incoming sum walks the
-// same inverse-adjacency CSR (order/offsets, edge id = j*N + n) the
-// round-1 CSR backward used; every (p, m) output is written exactly once.
+// (exact same formulas as gnmp_bwd_* in group_norm.hip).  The CENTRE sum
+// closes over j analytically (only j = argmax carries dy, recovered from
+// the saved pre-GN extreme; sum_j xhat_j derives from the per-point
+// gather sum the forward reduce stores) -- no K-loop in the apply pass.
+// The INCOMING sum walks the inverse-adjacency CSR via the (order_n,
+// order_j) side arrays -- source point and neighbour slot per ordered
+// edge, no id decomposition, dy a scalar load only on the argmax hit;
+// every (p, m) output is written exactly once.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include "common.h"
