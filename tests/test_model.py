@@ -201,3 +201,46 @@ def test_cast_cache_invalidated_by_inplace_update():
     refresh_casts()
     assert torch.equal(_cast_cached(w, torch.bfloat16), w.to(torch.bfloat16))
     clear_step_cache()
+
+def test_gn_defer_targets_active_with_grad_disabled():
+    """Regression: Function.backward runs with grad mode DISABLED; the
+    deferred-gradient gate must key on the explicit _DEFER flag, not
+    is_grad_enabled() -- the old gate silently rerouted every GN weight
+    grad through AccumulateGrad, whose side-stream-pinned nodes race
+    under hipGraph replay (graphed training stopped learning)."""
+    import torch.nn as nn
+
+    from pvraft_amd.model import pointwise
+    from pvraft_amd.ops import _gn_defer_targets
+
+    w = nn.Parameter(torch.randn(8))
+    b = nn.Parameter(torch.randn(8))
+    pointwise.wgrad_defer_begin()
+    try:
+        with torch.no_grad():  # the grad mode backward actually runs in
+            tgt = _gn_defer_targets(w, b, None, act=1)
+        assert tgt is not None
+        assert tgt[0].shape == (8,) and tgt[1].shape == (8,)
+    finally:
+        pointwise.wgrad_defer_end()
+        w.grad = None
+        b.grad = None
+
+
+def test_cast_registry_does_not_grow_across_steps():
+    """Regression: per-forward weight stacks and fresh views must REPLACE
+    their registry entries (stable keys), not accumulate one entry per
+    step (unbounded eager-mode growth)."""
+    from pvraft_amd.model.pointwise import _CASTS, _cast_cached, clear_step_cache
+
+    clear_step_cache()
+    base = torch.randn(4, 8, 1)
+    for _ in range(5):
+        v = base.squeeze(-1)          # fresh view object each "step"
+        _cast_cached(v, torch.bfloat16)
+        stack = torch.cat([base[:, :4, 0], base[:, 4:, 0]], dim=1).contiguous()
+        stack._cast_key = (id(base), "stack")
+        _cast_cached(stack, torch.bfloat16)
+    assert len(_CASTS) == 2, len(_CASTS)
+    clear_step_cache()
+
